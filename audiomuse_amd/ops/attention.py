@@ -1,0 +1,58 @@
+"""Windowed attention dispatch.
+
+The HTSAT encoder's hot op: multi-head attention over 8x8 = 64-token
+windows with a relative-position bias and (for shifted windows) an
+additive group mask.
+
+GPU path: fused HIP kernel (ops/csrc/attention.hip) — one wavefront per
+(window, head): the 64-token window maps 1:1 onto the 64-lane wave, QK^T
+and PV are mfma_f32_16x16x32_bf16 tiles (head_dim 32 = one K-step), and
+softmax+bias+mask stay in registers. Falls back to torch SDPA where the
+kernel is unavailable (CPU) or shapes don't fit.
+
+Reference behavior being replaced: ONNX Runtime attention inside the
+DCLAP student (/root/reference/tasks/clap_analyzer.py:478-500).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from audiomuse_amd.ops import _ext
+
+
+def _sdpa_reference(q, k, v, bias, mask, scale):
+    nW, h, T, d = q.shape
+    if mask is None:
+        am = bias.to(q.dtype)                              # (h, T, T)
+        return F.scaled_dot_product_attention(q, k, v, attn_mask=am, scale=scale)
+    nw = mask.shape[0]
+    B = nW // nw
+    am = (bias.unsqueeze(0) + mask.unsqueeze(1)).to(q.dtype)   # (nw, h, T, T)
+    q4 = q.reshape(B, nw * h, T, d)
+    k4 = k.reshape(B, nw * h, T, d)
+    v4 = v.reshape(B, nw * h, T, d)
+    out = F.scaled_dot_product_attention(q4, k4, v4,
+                                         attn_mask=am.view(nw * h, T, T),
+                                         scale=scale)
+    return out.view(nW, h, T, d)
+
+
+def window_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     bias: torch.Tensor, mask: torch.Tensor | None,
+                     scale: float) -> torch.Tensor:
+    """q, k, v: (nW, heads, T, head_dim); bias: (heads, T, T);
+    mask: (windows_per_image, T, T) additive or None. Returns (nW, h, T, d)."""
+    if (q.is_cuda and not torch.is_grad_enabled()
+            and q.dtype == torch.bfloat16
+            and q.shape[-1] == 32 and q.shape[-2] == 64):
+        ext = _ext.native_or_none()
+        if ext is not None and hasattr(ext, "window_attn_fwd"):
+            nw = 0 if mask is None else mask.shape[0]
+            m = mask if mask is not None else q.new_zeros(0)
+            return ext.window_attn_fwd(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                bias.to(torch.float32).contiguous(),
+                m.to(torch.float32).contiguous(), nw, scale)
+    return _sdpa_reference(q, k, v, bias, mask, scale)
