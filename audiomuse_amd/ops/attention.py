@@ -36,7 +36,7 @@ def _sdpa_reference(q, k, v, bias, mask, scale):
     out = F.scaled_dot_product_attention(q4, k4, v4,
                                          attn_mask=am.view(nw * h, T, T),
                                          scale=scale)
-    return out.view(nW, h, T, d)
+    return out.reshape(nW, h, T, d)
 
 
 def window_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
