@@ -33,6 +33,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from audiomuse_amd.ops.attention import window_attention
+from audiomuse_amd.ops.norms import FusedLayerNorm
 
 
 @dataclass
@@ -104,9 +105,9 @@ class SwinBlock(nn.Module):
         super().__init__()
         self.window = window
         self.shift = shift
-        self.norm1 = nn.LayerNorm(dim)
+        self.norm1 = FusedLayerNorm(dim)
         self.attn = WindowAttention(dim, heads, window)
-        self.norm2 = nn.LayerNorm(dim)
+        self.norm2 = FusedLayerNorm(dim)
         hidden = int(dim * mlp_ratio)
         self.mlp = nn.Sequential(nn.Linear(dim, hidden), nn.GELU(), nn.Linear(hidden, dim))
 
@@ -131,7 +132,7 @@ class PatchMerging(nn.Module):
 
     def __init__(self, dim: int):
         super().__init__()
-        self.norm = nn.LayerNorm(4 * dim)
+        self.norm = FusedLayerNorm(4 * dim)
         self.reduction = nn.Linear(4 * dim, 2 * dim, bias=False)
 
     def forward(self, x: torch.Tensor, H: int, W: int) -> torch.Tensor:
@@ -161,7 +162,12 @@ class HTSATEncoder(nn.Module):
     def __init__(self, cfg: HTSATConfig | None = None):
         super().__init__()
         self.cfg = cfg = cfg or HTSATConfig()
-        self.patch_embed = nn.Conv2d(1, cfg.embed_dim, cfg.patch_size, cfg.patch_size)
+        # Patch embed: stride == kernel, so the conv is exactly a reshape +
+        # GEMM. MIOpen falls back to naive_conv for bf16 NCHW 4x4/4 (measured:
+        # 11.6% of step time, profiles/r01_bench_baseline.md) — the reshape
+        # path runs on hipBLASLt instead.
+        p = cfg.patch_size
+        self.patch_proj = nn.Linear(p * p, cfg.embed_dim, bias=True)
         self.pos_drop = nn.Identity()
 
         self.stages = nn.ModuleList()
@@ -184,7 +190,7 @@ class HTSATEncoder(nn.Module):
                 self.mergers.append(PatchMerging(dim))
                 dim *= 2
                 H, W = H // 2, W // 2
-        self.norm = nn.LayerNorm(dim)
+        self.norm = FusedLayerNorm(dim)
         self.head = nn.Linear(dim, cfg.out_dim)
         self._mask_cache: dict = {}
 
@@ -204,9 +210,11 @@ class HTSATEncoder(nn.Module):
             mel = F.pad(mel, (0, cfg.n_frames - T))
         elif T > cfg.n_frames:
             mel = mel[..., : cfg.n_frames]
-        x = self.patch_embed(mel.unsqueeze(1))            # (B, C, H, W)
-        H, W = x.shape[2], x.shape[3]
-        x = x.flatten(2).transpose(1, 2)                  # (B, H*W, C)
+        p = cfg.patch_size
+        H, W = mel.shape[1] // p, mel.shape[2] // p
+        # (B, H*p, W*p) -> (B, H, W, p*p) -> GEMM to embed_dim
+        patches = mel.view(B, H, p, W, p).permute(0, 1, 3, 2, 4).reshape(B, H * W, p * p)
+        x = self.patch_proj(patches)                      # (B, H*W, C)
 
         for si, blocks in enumerate(self.stages):
             win = self.stage_windows[si]

@@ -9,7 +9,16 @@ void launch_mel_fwd(const float* audio, float* out, const float* window,
                     const float2* twiddle, const int* mel_rowptr,
                     const int* mel_bin, const float* mel_w, int B, int T,
                     int n_frames, int hop, int n_mels, int n_fft, int center,
-                    int log_mode, hipStream_t stream);
+                    int log_mode, int quant16, hipStream_t stream);
+void launch_ivf_scan(int dtype_code, int metric, const void* query,
+                     const float* qnorm, const void* data,
+                     const float* row_norm, const int* probe,
+                     const int* cell_off, const long long* cand_off,
+                     float* out_dist, int* out_row, int Q, int d, int nprobe,
+                     long long cap, hipStream_t stream);
+void launch_layernorm_bf16(const void* x, void* y, const void* w,
+                           const void* b, long long n_rows, int dim, float eps,
+                           hipStream_t stream);
 }
 
 #define AM_CHECK(x, msg) TORCH_CHECK(x, msg)
@@ -22,7 +31,7 @@ static torch::Tensor mel_fwd(torch::Tensor audio, torch::Tensor window,
                              torch::Tensor twiddle, torch::Tensor mel_rowptr,
                              torch::Tensor mel_bin, torch::Tensor mel_w,
                              int64_t hop, int64_t n_fft, bool center,
-                             int64_t log_mode) {
+                             int64_t log_mode, bool quant16) {
   AM_CHECK_GPU_F32_CONTIG(audio);
   AM_CHECK_GPU_F32_CONTIG(window);
   AM_CHECK_GPU_F32_CONTIG(mel_w);
@@ -52,15 +61,69 @@ static torch::Tensor mel_fwd(torch::Tensor audio, torch::Tensor window,
       reinterpret_cast<const float2*>(twiddle.data_ptr<float>()),
       mel_rowptr.data_ptr<int>(), mel_bin.data_ptr<int>(),
       mel_w.data_ptr<float>(), (int)B, (int)T, (int)n_frames, (int)hop,
-      (int)n_mels, (int)n_fft, center ? 1 : 0, (int)log_mode, stream.stream());
+      (int)n_mels, (int)n_fft, center ? 1 : 0, (int)log_mode, quant16 ? 1 : 0,
+      stream.stream());
   C10_HIP_CHECK(hipGetLastError());
   return out;
 }
 
+static void ivf_scan(int64_t dtype_code, int64_t metric, torch::Tensor query,
+                     torch::Tensor qnorm, torch::Tensor data,
+                     torch::Tensor row_norm, torch::Tensor probe,
+                     torch::Tensor cell_off, torch::Tensor cand_off,
+                     torch::Tensor out_dist, torch::Tensor out_row,
+                     int64_t dim_pad) {
+  AM_CHECK(query.is_cuda() && data.is_cuda(), "ivf_scan needs GPU tensors");
+  AM_CHECK(probe.scalar_type() == at::kInt && cell_off.scalar_type() == at::kInt,
+           "probe/cell_off must be int32");
+  AM_CHECK(cand_off.scalar_type() == at::kLong, "cand_off must be int64");
+  AM_CHECK(out_dist.is_contiguous() && out_row.is_contiguous(),
+           "outputs must be contiguous");
+  const int Q = probe.size(0);
+  const int nprobe = probe.size(1);
+  const long long cap = out_dist.size(1);
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_ivf_scan(
+      (int)dtype_code, (int)metric, query.data_ptr(), qnorm.data_ptr<float>(),
+      data.data_ptr(), row_norm.data_ptr<float>(), probe.data_ptr<int>(),
+      cell_off.data_ptr<int>(),
+      reinterpret_cast<const long long*>(cand_off.data_ptr<int64_t>()),
+      out_dist.data_ptr<float>(), out_row.data_ptr<int>(), Q, (int)dim_pad,
+      nprobe, cap, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+}
+
+static torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor w,
+                                    torch::Tensor b, double eps) {
+  AM_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous(),
+           "x must be contiguous bf16 on GPU");
+  const int dim = x.size(-1);
+  AM_CHECK(dim % 4 == 0 && dim <= 4096,
+           "dim must be a multiple of 4 and <= 4096");
+  AM_CHECK(w.is_contiguous() && b.is_contiguous() &&
+               w.scalar_type() == at::kBFloat16 &&
+               b.scalar_type() == at::kBFloat16 && w.numel() == dim &&
+               b.numel() == dim,
+           "w/b must be contiguous bf16 of size dim");
+  auto y = torch::empty_like(x);
+  const long long n_rows = x.numel() / dim;
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_layernorm_bf16(x.data_ptr(), y.data_ptr(), w.data_ptr(),
+                                   b.data_ptr(), n_rows, dim, (float)eps,
+                                   stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "AudioMuse-AMD native CDNA4 kernels";
+  m.def("layernorm_bf16", &layernorm_bf16,
+        "Fused LayerNorm forward, bf16 in/out, fp32 stats (x, w, b, eps)");
   m.def("mel_fwd", &mel_fwd,
         "Fused STFT+mel+log spectrogram (audio, window, twiddle, rowptr, "
         "bin, w, hop, n_fft, center, log_mode)");
+  m.def("ivf_scan", &ivf_scan,
+        "IVF probed-cell distance scan (dtype, metric, query, qnorm, data, "
+        "row_norm, probe, cell_off, cand_off, out_dist, out_row, dim_pad)");
   m.attr("gfx_arch") = "gfx950";
 }

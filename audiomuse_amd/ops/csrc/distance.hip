@@ -23,6 +23,7 @@
 // (torch.topk over the candidate buffer).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
 
 #include <cfloat>
 
@@ -37,6 +38,10 @@ __device__ __forceinline__ int wave_reduce_sum_i32(int v) {
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
   return v;
 }
+
+// torch builds with __HIP_NO_HALF_CONVERSIONS__: convert explicitly
+__device__ __forceinline__ float to_f32(float v) { return v; }
+__device__ __forceinline__ float to_f32(__half v) { return __half2float(v); }
 
 // ---------------------------------------------------------------------------
 // i8 angular scan. d must be a multiple of 4 (padded at build time).
@@ -123,7 +128,7 @@ __global__ __launch_bounds__(256) void ivf_scan_float(
     const T* row = data + (long long)r * d;
     float acc = 0.0f;
     for (int j = lane; j < d; j += 64) {
-      const float v = (float)row[j];
+      const float v = to_f32(row[j]);
       if (METRIC == 1) {
         const float diff = v - qf[j];
         acc += diff * diff;

@@ -31,7 +31,8 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     const int* __restrict__ mel_rowptr, // (n_mels+1) CSR over mel bins
     const int* __restrict__ mel_bin,    // (nnz) fft-bin indices
     const float* __restrict__ mel_w,    // (nnz) filter weights
-    int T, int n_frames, int hop, int n_mels, int center, int log_mode) {
+    int T, int n_frames, int hop, int n_mels, int center, int log_mode,
+    int quant16) {
   __shared__ float2 zbuf[NFFT];
   __shared__ float pw[NFFT / 2 + 1];
 
@@ -49,7 +50,14 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     if (g < 0) g = -g;                    // librosa reflect (no edge repeat)
     if (g >= T) g = 2 * (T - 1) - g;
     g = max(0, min(T - 1, g));            // safety for tiny T
-    float v = src[g] * window[i];
+    float v = src[g];
+    if (quant16) {
+      // fused int16 round-trip (clap_analyzer.py:453-455): clip +-1,
+      // numpy int16 cast truncates toward zero
+      v = fminf(1.0f, fmaxf(-1.0f, v));
+      v = (float)(int)(v * 32767.0f) / 32767.0f;
+    }
+    v *= window[i];
     int rev = __brev((unsigned)i) >> (32 - LOG2N);
     zbuf[rev] = make_float2(v, 0.0f);
   }
@@ -104,14 +112,14 @@ void launch_mel_fwd(const float* audio, float* out, const float* window,
                     const float2* twiddle, const int* mel_rowptr,
                     const int* mel_bin, const float* mel_w, int B, int T,
                     int n_frames, int hop, int n_mels, int n_fft, int center,
-                    int log_mode, hipStream_t stream) {
+                    int log_mode, int quant16, hipStream_t stream) {
   dim3 grid(n_frames, B);
   dim3 block(256);
 #define AM_MEL_CASE(N, L)                                                     \
   case N:                                                                     \
     hipLaunchKernelGGL((mel_fwd_kernel<N, L>), grid, block, 0, stream, audio, \
                        out, window, twiddle, mel_rowptr, mel_bin, mel_w, T,   \
-                       n_frames, hop, n_mels, center, log_mode);              \
+                       n_frames, hop, n_mels, center, log_mode, quant16);     \
     break;
   switch (n_fft) {
     AM_MEL_CASE(256, 8)

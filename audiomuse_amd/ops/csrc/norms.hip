@@ -1,0 +1,110 @@
+// Fused LayerNorm forward (bf16 in/out, fp32 stats) for gfx950.
+//
+// The HTSAT encoder spends ~18.5% of its step time in eager LayerNorm
+// (measured: profiles/r01_bench_baseline.md). This kernel does one HBM
+// read per element: each 64-lane wave owns one row, loads it vectorized
+// (short4 = 4 bf16 / 8 B per lane per iteration, Guideline 13), reduces
+// mean/var with wave shuffles, and writes the normalized row.
+//
+// NIT (per-lane short4 iterations) is a template parameter so the value
+// cache stays in registers — runtime-indexed local arrays spill to
+// scratch on hipcc (cdna_hip_programming.md, common-mistake #20).
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+namespace audiomuse {
+
+__device__ __forceinline__ float wrsum(float v) {
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return __shfl(v, 0, 64);
+}
+
+// one wave per row; block = 256 threads = 4 rows
+template <int NIT>
+__global__ __launch_bounds__(256) void layernorm_bf16_kernel(
+    const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
+    const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
+    long long n_rows, int dim, float eps) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long long row = (long long)blockIdx.x * 4 + wave;
+  if (row >= n_rows) return;
+
+  const __hip_bfloat16* xr = x + row * dim;
+  __hip_bfloat16* yr = y + row * dim;
+
+  float vals[NIT * 4];
+  float sum = 0.0f;
+#pragma unroll
+  for (int t = 0; t < NIT; ++t) {
+    const int i = lane * 4 + t * 256;
+    if (i < dim) {
+      const short4 p = *reinterpret_cast<const short4*>(xr + i);
+      const __hip_bfloat16* pb = reinterpret_cast<const __hip_bfloat16*>(&p);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float f = __bfloat162float(pb[j]);
+        vals[t * 4 + j] = f;
+        sum += f;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) vals[t * 4 + j] = 0.0f;
+    }
+  }
+  const float mean = wrsum(sum) / dim;
+  float var = 0.0f;
+#pragma unroll
+  for (int t = 0; t < NIT; ++t) {
+    const int i = lane * 4 + t * 256;
+    if (i < dim) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float d = vals[t * 4 + j] - mean;
+        var += d * d;
+      }
+    }
+  }
+  const float rstd = rsqrtf(wrsum(var) / dim + eps);
+
+#pragma unroll
+  for (int t = 0; t < NIT; ++t) {
+    const int i = lane * 4 + t * 256;
+    if (i >= dim) continue;
+    const short4 pw = *reinterpret_cast<const short4*>(w + i);
+    const short4 pbv = *reinterpret_cast<const short4*>(b + i);
+    const __hip_bfloat16* wb = reinterpret_cast<const __hip_bfloat16*>(&pw);
+    const __hip_bfloat16* bb = reinterpret_cast<const __hip_bfloat16*>(&pbv);
+    short4 out;
+    __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float f = (vals[t * 4 + j] - mean) * rstd;
+      ob[j] = __float2bfloat16(f * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
+    }
+    *reinterpret_cast<short4*>(yr + i) = out;
+  }
+}
+
+void launch_layernorm_bf16(const void* x, void* y, const void* w,
+                           const void* b, long long n_rows, int dim, float eps,
+                           hipStream_t stream) {
+  const long long blocks = (n_rows + 3) / 4;
+  const dim3 grid((unsigned)blocks);
+  const dim3 block(256);
+#define AM_LN_CASE(NIT)                                                   \
+  hipLaunchKernelGGL((layernorm_bf16_kernel<NIT>), grid, block, 0, stream, \
+                     (const __hip_bfloat16*)x, (__hip_bfloat16*)y,        \
+                     (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,  \
+                     n_rows, dim, eps)
+  const int nit = (dim + 255) / 256;
+  if (nit <= 1) AM_LN_CASE(1);
+  else if (nit <= 2) AM_LN_CASE(2);
+  else if (nit <= 4) AM_LN_CASE(4);
+  else if (nit <= 8) AM_LN_CASE(8);
+  else AM_LN_CASE(16);
+#undef AM_LN_CASE
+}
+
+}  // namespace audiomuse

@@ -62,7 +62,8 @@ def _reference(cfg: MelConfig, device: torch.device) -> MelFrontend:
 
 
 def mel_spectrogram(audio: torch.Tensor, cfg: MelConfig,
-                    force_reference: bool = False) -> torch.Tensor:
+                    force_reference: bool = False,
+                    quantize_int16: bool = False) -> torch.Tensor:
     """Log-mel spectrogram. audio (B, T) or (T,) fp32 -> (B, n_mels, frames).
 
     GPU inputs run the fused HIP kernel (ops/csrc/mel.hip); CPU inputs (or
@@ -83,7 +84,10 @@ def mel_spectrogram(audio: torch.Tensor, cfg: MelConfig,
             out = ext.mel_fwd(audio, plan["window"], plan["twiddle"],
                               plan["rowptr"], plan["bin"], plan["w"],
                               cfg.hop, cfg.n_fft, cfg.center,
-                              _LOG_MODE[cfg.log_mode])
+                              _LOG_MODE[cfg.log_mode], quantize_int16)
             return out[0] if single else out
+    if quantize_int16:
+        from audiomuse_amd.ops.dsp import int16_roundtrip
+        audio = int16_roundtrip(audio)
     out = _reference(cfg, audio.device)(audio)
     return out[0] if single else out

@@ -67,8 +67,8 @@ def main() -> None:
 
     def step() -> torch.Tensor:
         with torch.inference_mode():
-            a = dsp.int16_roundtrip(audio)
-            mel = hip_ops.mel_spectrogram(a, mel_cfg)          # HIP kernel, fp32
+            # int16 round-trip fused into the mel kernel
+            mel = hip_ops.mel_spectrogram(audio, mel_cfg, quantize_int16=True)
             emb = model(mel.to(torch.bfloat16))                 # bf16 encoder
             emb = emb.float()
             return emb / (emb.norm(dim=1, keepdim=True) + 1e-9)
