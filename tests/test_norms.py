@@ -1,0 +1,51 @@
+import pytest
+import torch
+import torch.nn.functional as F
+
+from audiomuse_amd.ops.norms import FusedLayerNorm
+
+
+def test_fused_ln_cpu_matches_torch():
+    ln = FusedLayerNorm(128)
+    torch.nn.init.normal_(ln.weight)
+    torch.nn.init.normal_(ln.bias)
+    x = torch.randn(4, 10, 128)
+    torch.testing.assert_close(ln(x), F.layer_norm(x, (128,), ln.weight, ln.bias, ln.eps))
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dim", [128, 256, 512, 1024])
+def test_fused_ln_gpu_matches_reference(dim):
+    torch.manual_seed(0)
+    ln = FusedLayerNorm(dim).to("cuda", torch.bfloat16)
+    with torch.no_grad():
+        ln.weight.normal_()
+        ln.bias.normal_()
+    x = torch.randn(3, 97, dim, device="cuda", dtype=torch.bfloat16)
+    with torch.inference_mode():
+        got = ln(x)
+    expect = F.layer_norm(x.float(), (dim,), ln.weight.float(), ln.bias.float(),
+                          ln.eps).to(torch.bfloat16)
+    # bf16 output: one-ulp differences expected
+    torch.testing.assert_close(got.float(), expect.float(), rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_fused_ln_odd_row_count():
+    ln = FusedLayerNorm(256).to("cuda", torch.bfloat16)
+    x = torch.randn(1, 5, 256, device="cuda", dtype=torch.bfloat16)
+    with torch.inference_mode():
+        got = ln(x)
+    assert got.shape == x.shape and torch.isfinite(got.float()).all()
+
+
+@pytest.mark.gpu
+def test_mel_fused_quant16_matches_reference():
+    from audiomuse_amd.ops import dsp, hip_ops
+
+    cfg = dsp.clap_mel_config()
+    audio = (torch.randn(2, 96000, device="cuda") * 0.4).clamp(-1, 1)
+    fused = hip_ops.mel_spectrogram(audio, cfg, quantize_int16=True)
+    ref = hip_ops.mel_spectrogram(dsp.int16_roundtrip(audio), cfg,
+                                  force_reference=True)
+    torch.testing.assert_close(fused, ref, rtol=1e-3, atol=2e-3)
