@@ -1,0 +1,167 @@
+"""Queue-semantics tests (modeled on the reference's
+test_taskqueue_abandoned.py / test_queue_locks.py pyramid: several
+connections/threads against one DB, simulated worker death)."""
+
+import threading
+import time
+
+import pytest
+
+from audiomuse_amd.db import connect
+from audiomuse_amd.db.schema import init_db
+from audiomuse_amd.taskqueue import (FAILURE, PENDING, REVOKED, RUNNING,
+                                     SUCCESS, cancel_task_recursive, enqueue,
+                                     task_row)
+from audiomuse_amd.taskqueue import sql as qsql
+from audiomuse_amd.taskqueue.worker import TaskContext, Worker, task_handler
+
+
+@pytest.fixture
+def db(tmp_db_url):
+    conn = connect(tmp_db_url)
+    init_db(conn)
+    yield conn, tmp_db_url
+    conn.close()
+
+
+def test_enqueue_claim_finish(db):
+    conn, url = db
+    tid = enqueue(conn, "noop", {"x": 1}, queue="high")
+    assert task_row(conn, tid)["status"] == PENDING
+    row = qsql.claim(conn, "w1")
+    assert row["task_id"] == tid and row["status"] == RUNNING
+    assert row["attempts"] == 1
+    assert qsql.finish(conn, tid, "w1", SUCCESS, {"ok": True})
+    assert task_row(conn, tid)["status"] == SUCCESS
+
+
+def test_high_queue_priority_order(db):
+    conn, _ = db
+    t_def = enqueue(conn, "noop", queue="default")
+    t_high = enqueue(conn, "noop", queue="high")
+    # claim order scans queues jointly; priority + created_at break ties,
+    # but the reference runs separate high/default workers — emulate:
+    row = qsql.claim(conn, "w1", queues=("high",))
+    assert row["task_id"] == t_high
+    row = qsql.claim(conn, "w1", queues=("high", "default"))
+    assert row["task_id"] == t_def
+
+
+def test_concurrent_claim_exclusive(db):
+    _, url = db
+    conn0 = connect(url)
+    ids = [enqueue(conn0, "noop") for _ in range(20)]
+    claimed = []
+    lock = threading.Lock()
+
+    def worker(wid):
+        conn = connect(url)
+        while True:
+            row = qsql.claim(conn, wid)
+            if row is None:
+                break
+            with lock:
+                claimed.append(row["task_id"])
+        conn.close()
+
+    threads = [threading.Thread(target=worker, args=(f"w{i}",)) for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert sorted(claimed) == sorted(ids)
+    assert len(set(claimed)) == 20  # no double claims
+
+
+def test_lease_expiry_reclaim_and_requeue(db):
+    conn, _ = db
+    tid = enqueue(conn, "noop", max_attempts=2)
+    qsql.claim(conn, "dead-worker", lease_seconds=0.05)
+    time.sleep(0.1)
+    assert qsql.reclaim_orphans(conn) == 1
+    assert task_row(conn, tid)["status"] == PENDING
+    # second claim exhausts attempts; next reclaim fails it for good
+    qsql.claim(conn, "dead-worker-2", lease_seconds=0.05)
+    time.sleep(0.1)
+    assert qsql.reclaim_orphans(conn) == 1
+    assert task_row(conn, tid)["status"] == FAILURE
+
+
+def test_heartbeat_extends_lease(db):
+    conn, _ = db
+    tid = enqueue(conn, "noop")
+    qsql.claim(conn, "w1", lease_seconds=0.2)
+    time.sleep(0.1)
+    assert qsql.heartbeat(conn, tid, "w1", lease_seconds=10.0)
+    time.sleep(0.15)
+    assert qsql.reclaim_orphans(conn) == 0  # lease extended, not an orphan
+    assert task_row(conn, tid)["status"] == RUNNING
+
+
+def test_recursive_cancel(db):
+    conn, _ = db
+    parent = enqueue(conn, "parent")
+    kids = [enqueue(conn, "child", parent_task_id=parent) for _ in range(3)]
+    grand = enqueue(conn, "grandchild", parent_task_id=kids[0])
+    n = cancel_task_recursive(conn, parent)
+    assert n == 5
+    for tid in [parent, *kids, grand]:
+        assert task_row(conn, tid)["status"] == REVOKED
+    assert qsql.is_cancelled(conn, grand)
+
+
+def test_cancelled_via_ancestor(db):
+    conn, _ = db
+    parent = enqueue(conn, "parent")
+    child = enqueue(conn, "child", parent_task_id=parent)
+    # revoke only the parent directly (not recursive)
+    cancel_task_recursive(conn, parent)
+    assert qsql.is_cancelled(conn, child)
+
+
+def test_worker_runs_registered_handler(db):
+    conn, url = db
+    ran = []
+
+    @task_handler("test_job_x")
+    def handler(ctx: TaskContext, payload):
+        ran.append(payload["v"])
+        ctx.report(50.0, "halfway")
+        return {"doubled": payload["v"] * 2}
+
+    tid = enqueue(conn, "test_job_x", {"v": 21})
+    w = Worker(db_url=url, max_jobs=1)
+    w.run_forever(idle_timeout=0.5)
+    row = task_row(conn, tid)
+    assert row["status"] == SUCCESS
+    assert ran == [21]
+    assert '"doubled": 42' in row["result"]
+
+
+def test_worker_failure_records_traceback(db):
+    conn, url = db
+
+    @task_handler("test_job_boom")
+    def handler(ctx, payload):
+        raise ValueError("boom")
+
+    tid = enqueue(conn, "test_job_boom")
+    Worker(db_url=url, max_jobs=1).run_forever(idle_timeout=0.5)
+    row = task_row(conn, tid)
+    assert row["status"] == FAILURE
+    assert "boom" in row["result"]
+
+
+def test_worker_unknown_type_fails(db):
+    conn, url = db
+    tid = enqueue(conn, "never_registered")
+    Worker(db_url=url, max_jobs=1).run_forever(idle_timeout=0.5)
+    assert task_row(conn, tid)["status"] == FAILURE
+
+
+def test_pending_children_backpressure(db):
+    conn, _ = db
+    parent = enqueue(conn, "parent")
+    for _ in range(4):
+        enqueue(conn, "child", parent_task_id=parent)
+    assert qsql.pending_children(conn, parent) == 4
