@@ -88,13 +88,18 @@ class WindowAttention(nn.Module):
         self.register_buffer("rel_index", idx, persistent=False)
         nn.init.trunc_normal_(self.rel_bias, std=0.02)
 
+    def full_bias(self) -> torch.Tensor:
+        """(heads, T, T) f32 relative-position bias table."""
+        T = self.window * self.window
+        bias = self.rel_bias[self.rel_index.view(-1)].view(T, T, self.heads)
+        return bias.permute(2, 0, 1).float().contiguous()
+
     def forward(self, x: torch.Tensor, mask: torch.Tensor | None) -> torch.Tensor:
         """x: (nW, T, C) with T = window*window; mask: (groups, T, T) or None."""
         nW, T, C = x.shape
         qkv = self.qkv(x).reshape(nW, T, 3, self.heads, C // self.heads)
         q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)   # (nW, h, T, d)
-        bias = self.rel_bias[self.rel_index.view(-1)].view(T, T, self.heads)
-        bias = bias.permute(2, 0, 1).contiguous()        # (h, T, T)
+        bias = self.full_bias().to(x.dtype)              # (h, T, T)
         out = window_attention(q, k, v, bias, mask, self.scale)
         out = out.transpose(1, 2).reshape(nW, T, C)
         return self.proj(out)
@@ -111,9 +116,30 @@ class SwinBlock(nn.Module):
         hidden = int(dim * mlp_ratio)
         self.mlp = nn.Sequential(nn.Linear(dim, hidden), nn.GELU(), nn.Linear(hidden, dim))
 
+    def _fused_attn_available(self, x: torch.Tensor) -> bool:
+        if not (x.is_cuda and x.dtype == torch.bfloat16
+                and not torch.is_grad_enabled() and self.window == 8
+                and self.attn.heads % 4 == 0
+                and self.attn.dim // self.attn.heads == 32):
+            return False
+        from audiomuse_amd.ops import _ext
+        ext = _ext.native_or_none()
+        return ext is not None and hasattr(ext, "window_attn_fwd")
+
     def forward(self, x: torch.Tensor, H: int, W: int,
                 mask: torch.Tensor | None) -> torch.Tensor:
         B, L, C = x.shape
+        if self._fused_attn_available(x):
+            # fused kernel folds roll + partition + attention + reverse
+            from audiomuse_amd.ops import _ext
+            ext = _ext.require()
+            xn = self.norm1(x)
+            qkv = self.attn.qkv(xn)                       # (B, L, 3C)
+            out = ext.window_attn_fwd(
+                qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
+                self.attn.heads, self.shift, self.attn.scale)
+            x = x + self.attn.proj(out.view(B, L, C))
+            return x + self.mlp(self.norm2(x))
         shortcut = x
         x = self.norm1(x).view(B, H, W, C)
         if self.shift:

@@ -1,0 +1,279 @@
+// Fused shifted-window attention for the HTSAT encoder (gfx950, CDNA4).
+//
+// Replaces, in one launch: roll(-shift) -> window_partition -> per-head
+// QK^T -> +rel_bias -> +shift mask -> softmax -> PV -> window_reverse ->
+// roll(+shift)  (the eager chain measured at ~25% of encoder step time:
+// profiles/r01_bench_clap.md).
+//
+// Geometry (fixed by the model design, models/htsat.py):
+//   window 8x8 = 64 tokens  == one 64-lane wavefront's MFMA row space
+//   head_dim 32             == one K-step of mfma_f32_16x16x32_bf16
+// Each workgroup = 4 waves handles one window; wave w computes heads
+// w, w+4, ... . S = QK^T is a 4x4 grid of 16x16 MFMA tiles (K=32, one
+// instruction each); softmax runs in registers (rows live across 16
+// lanes -> 4-step shfl_xor reduction); P is staged through a padded LDS
+// tile; PV is a 4x2 grid of 16x16 tiles with K=64 (two instructions).
+// Q and K fragments load straight from global memory (16 B per lane,
+// L2-resident across heads); V is staged transposed in LDS.
+//
+// The shifted-window mask is computed inline from wrap bits: after a
+// roll by -shift, two tokens of a window may attend iff both wrapped or
+// neither wrapped in each axis (equivalent to the reference's 9-region
+// mask construction, models/htsat.py:_shift_mask).
+//
+// Fragment layouts (A/B: row/col = lane&15, k = 8*(lane>>4)+j;
+// C/D: col = lane&15, row = (lane>>4)*4 + reg) are verified on hardware
+// by tests/test_attention.py::test_mfma_probe_layout via the mfma_probe
+// binding below.
+
+#include <hip/hip_runtime.h>
+
+namespace audiomuse {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ float fmax4(const f32x4 v) {
+  return fmaxf(fmaxf(v[0], v[1]), fmaxf(v[2], v[3]));
+}
+
+// ---------------------------------------------------------------------------
+// Layout probe: D = A(16x32) @ B(32x16) with the layout above; the GPU
+// test compares against torch.matmul to pin the mapping (guide G9:
+// asymmetric random inputs, transpose-detecting).
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe_kernel(const __bf16* __restrict__ A,
+                                  const __bf16* __restrict__ B,
+                                  float* __restrict__ D) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int r = lane & 15;
+    const int k = 8 * (lane >> 4) + j;
+    a[j] = *(const __bf16*)(A + r * 32 + k);   // A row-major (16,32)
+    b[j] = *(const __bf16*)(B + k * 16 + r);   // B row-major (32,16), col=r
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = (lane >> 4) * 4 + reg;
+    const int col = lane & 15;
+    D[row * 16 + col] = c[reg];
+  }
+}
+
+void launch_mfma_probe(const void* A, const void* B, float* D,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const __bf16*)A, (const __bf16*)B, D);
+}
+
+// ---------------------------------------------------------------------------
+// Fused window attention
+// ---------------------------------------------------------------------------
+
+// per-wave LDS: VT[32][64+8] bf16 rows padded to 144 B; P[64][64+8] bf16.
+constexpr int VT_STRIDE = 72;   // bf16 elements per VT row (64 + 8 pad)
+constexpr int P_STRIDE = 72;    // bf16 elements per P row
+constexpr int WAVE_LDS_HALF = 32 * VT_STRIDE + 64 * P_STRIDE;  // bf16 elems
+
+__global__ __launch_bounds__(256) void window_attn_kernel(
+    const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
+    __bf16* __restrict__ out,        // (B, H, W, C)
+    const float* __restrict__ bias,          // (heads, 64, 64)
+    int Bn, int H, int W, int C, int heads, int shift, float scale) {
+  extern __shared__ __bf16 lds[];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  const int nWw = W >> 3;
+  const int nWh = H >> 3;
+  const int win = blockIdx.x;
+  const int b = win / (nWh * nWw);
+  const int wrem = win - b * (nWh * nWw);
+  const int wh = wrem / nWw;
+  const int ww = wrem - wh * nWw;
+
+  __bf16* VT = lds + wave * WAVE_LDS_HALF;
+  __bf16* P = VT + 32 * VT_STRIDE;
+
+  // token t (0..63) -> source coords + wrap bits (shifted windows)
+  auto src_of = [&](int t, int& si, int& sj, int& wrap) {
+    const int ri = t >> 3, ci = t & 7;
+    int gi = wh * 8 + ri + shift;
+    int gj = ww * 8 + ci + shift;
+    const int wr = gi >= H;
+    const int wc = gj >= W;
+    si = wr ? gi - H : gi;
+    sj = wc ? gj - W : gj;
+    wrap = (wr << 1) | wc;
+  };
+
+  // wrap bits for every token, in registers (same for all heads)
+  int my_si, my_sj, my_wrap;
+  src_of(lane, my_si, my_sj, my_wrap);
+  const long long my_base = (((long long)b * H + my_si) * W + my_sj) * 3 * C;
+
+  // wrap bits of all 64 tokens as a packed pair of 64-bit masks via ballot
+  const unsigned long long wrap_r_mask = __ballot(my_wrap & 2);
+  const unsigned long long wrap_c_mask = __ballot(my_wrap & 1);
+
+  for (int h = wave; h < heads; h += 4) {
+    // ---- stage V transposed: VT[d][t] = V[t][d] ----
+    {
+      const __bf16* vptr = qkv + my_base + 2 * C + h * 32;
+      bf16x8 vv[4];
+#pragma unroll
+      for (int g = 0; g < 4; ++g) vv[g] = *(const bf16x8*)(vptr + g * 8);
+#pragma unroll
+      for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          VT[(g * 8 + j) * VT_STRIDE + lane] = (__bf16)vv[g][j];
+    }
+
+    // ---- Q (A-frags) and K (B-frags) straight from global ----
+    bf16x8 qf[4], kf[4];
+    const int kk = 8 * (lane >> 4);   // k-offset of this lane's fragment
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr) {
+      const int t = tr * 16 + (lane & 15);
+      int si, sj, wr_;
+      src_of(t, si, sj, wr_);
+      const long long base = (((long long)b * H + si) * W + sj) * 3 * C + h * 32;
+      qf[tr] = *(const bf16x8*)(qkv + base + kk);          // Q slice
+      kf[tr] = *(const bf16x8*)(qkv + base + C + kk);      // K slice
+    }
+
+    // ---- S = QK^T: 4x4 tiles of 16x16, one MFMA each ----
+    f32x4 s[4][4];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        s[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[tr], kf[tc],
+                                                            acc, 0, 0, 0);
+      }
+
+    // ---- scale + bias + shift mask, then row softmax in registers ----
+    // C-frag: col = lane&15, row = (lane>>4)*4 + reg
+    const int col_in_tile = lane & 15;
+    const int row_grp = (lane >> 4) * 4;
+    float rmax[4][4];  // [tr][reg]
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+        const int rwrap = (((wrap_r_mask >> row) & 1ull) << 1) |
+                          ((wrap_c_mask >> row) & 1ull);
+        float m = -1e30f;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const int col = tc * 16 + col_in_tile;
+          const int cwrap = (((wrap_r_mask >> col) & 1ull) << 1) |
+                            ((wrap_c_mask >> col) & 1ull);
+          float v = s[tr][tc][reg] * scale + bias[(h * 64 + row) * 64 + col];
+          if (shift && rwrap != cwrap) v = -1e30f;
+          s[tr][tc][reg] = v;
+          m = fmaxf(m, v);
+        }
+        rmax[tr][reg] = m;
+      }
+    }
+    // reduce max across the 16 lanes holding each row
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float m = rmax[tr][reg];
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1)
+          m = fmaxf(m, __shfl_xor(m, d, 64));
+        rmax[tr][reg] = m;
+      }
+    float rsum[4][4];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float sum = 0.f;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const float e = __expf(s[tr][tc][reg] - rmax[tr][reg]);
+          s[tr][tc][reg] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1) sum += __shfl_xor(sum, d, 64);
+        rsum[tr][reg] = sum;
+      }
+
+    // ---- write P = softmax(S) to LDS (bf16) ----
+    __syncthreads();  // VT writes (and previous round's P reads) settled
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const int col = tc * 16 + col_in_tile;
+          P[row * P_STRIDE + col] =
+              (__bf16)(s[tr][tc][reg] / (rsum[tr][reg] + 1e-20f));
+        }
+      }
+    __syncthreads();
+
+    // ---- O = P @ V: 4x2 tiles of 16x16, K = 64 (2 MFMA each) ----
+    f32x4 o[4][2];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 2; ++tc) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          // A = P tile: row = tr*16 + (lane&15), k = ks*32 + kk + j
+          bf16x8 pa = *(const bf16x8*)(P + (tr * 16 + (lane & 15)) * P_STRIDE +
+                                       ks * 32 + kk);
+          // B = V tile: col(dim) = tc*16 + (lane&15), k(token) = ks*32+kk+j
+          bf16x8 vb = *(const bf16x8*)(VT + (tc * 16 + (lane & 15)) * VT_STRIDE +
+                                       ks * 32 + kk);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc, 0, 0, 0);
+        }
+        o[tr][tc] = acc;
+      }
+
+    // ---- scatter O back to (B, H, W, C) with the inverse roll ----
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int t = tr * 16 + row_grp + reg;
+        int si, sj, w_;
+        src_of(t, si, sj, w_);
+        __bf16* op =
+            out + (((long long)b * H + si) * W + sj) * C + h * 32;
+#pragma unroll
+        for (int tc = 0; tc < 2; ++tc)
+          op[tc * 16 + col_in_tile] = (__bf16)o[tr][tc][reg];
+      }
+    __syncthreads();  // P/VT reuse next round
+  }
+}
+
+void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
+                        int H, int W, int C, int heads, int shift, float scale,
+                        hipStream_t stream) {
+  const int n_windows = Bn * (H >> 3) * (W >> 3);
+  const size_t lds_bytes = 4 * WAVE_LDS_HALF * sizeof(__bf16);
+  hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows), dim3(256), lds_bytes,
+                     stream, (const __bf16*)qkv, (__bf16*)out,
+                     bias, Bn, H, W, C, heads, shift, scale);
+}
+
+}  // namespace audiomuse

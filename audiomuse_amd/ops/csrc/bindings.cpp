@@ -19,6 +19,11 @@ void launch_ivf_scan(int dtype_code, int metric, const void* query,
 void launch_layernorm_bf16(const void* x, void* y, const void* w,
                            const void* b, long long n_rows, int dim, float eps,
                            hipStream_t stream);
+void launch_mfma_probe(const void* A, const void* B, float* D,
+                       hipStream_t stream);
+void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
+                        int H, int W, int C, int heads, int shift, float scale,
+                        hipStream_t stream);
 }
 
 #define AM_CHECK(x, msg) TORCH_CHECK(x, msg)
@@ -115,8 +120,52 @@ static torch::Tensor layernorm_bf16(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+static torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  AM_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+               A.is_contiguous() && A.size(0) == 16 && A.size(1) == 32,
+           "A must be (16,32) bf16 GPU");
+  AM_CHECK(B.is_cuda() && B.scalar_type() == at::kBFloat16 &&
+               B.is_contiguous() && B.size(0) == 32 && B.size(1) == 16,
+           "B must be (32,16) bf16 GPU");
+  auto D = torch::empty({16, 16}, A.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_mfma_probe(A.data_ptr(), B.data_ptr(),
+                               D.data_ptr<float>(), stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return D;
+}
+
+static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
+                                     int64_t heads, int64_t shift,
+                                     double scale) {
+  AM_CHECK(qkv.is_cuda() && qkv.scalar_type() == at::kBFloat16 &&
+               qkv.is_contiguous() && qkv.dim() == 4,
+           "qkv must be (B, H, W, 3C) bf16 contiguous GPU");
+  const int64_t Bn = qkv.size(0), H = qkv.size(1), W = qkv.size(2);
+  const int64_t C = qkv.size(3) / 3;
+  AM_CHECK(qkv.size(3) == 3 * C && C == heads * 32,
+           "C must be heads*32 and last dim 3C");
+  AM_CHECK(H % 8 == 0 && W % 8 == 0, "H, W must be multiples of 8");
+  AM_CHECK(heads % 4 == 0, "heads must be a multiple of 4");
+  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kFloat &&
+               bias.is_contiguous() && bias.numel() == heads * 64 * 64,
+           "bias must be (heads, 64, 64) f32 contiguous");
+  auto out = torch::empty({Bn, H, W, C}, qkv.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_window_attn(qkv.data_ptr(), out.data_ptr(),
+                                bias.data_ptr<float>(), (int)Bn, (int)H,
+                                (int)W, (int)C, (int)heads, (int)shift,
+                                (float)scale, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "AudioMuse-AMD native CDNA4 kernels";
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("window_attn_fwd", &window_attn_fwd,
+        "Fused shifted-window attention (qkv BHW3C bf16, bias, heads, "
+        "shift, scale) -> (B,H,W,C)");
   m.def("layernorm_bf16", &layernorm_bf16,
         "Fused LayerNorm forward, bf16 in/out, fp32 stats (x, w, b, eps)");
   m.def("mel_fwd", &mel_fwd,
