@@ -1,0 +1,139 @@
+"""Tests: basic features (tempo/energy/key), MusiCNN, simhash identity."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from audiomuse_amd import config as C
+from audiomuse_amd.engines import simhash
+from audiomuse_amd.models.musicnn import (MusiCNNConfig, MusiCNNEmbedding,
+                                          MusiCNNPrediction, aggregate_track)
+from audiomuse_amd.ops import features
+
+
+def _tone(freq, sr, seconds=3.0, amp=0.5):
+    t = torch.arange(int(sr * seconds)) / sr
+    return amp * torch.sin(2 * math.pi * freq * t)
+
+
+def test_energy_silence_and_full_scale():
+    assert features.estimate_energy(torch.zeros(16000)) == 0.0
+    loud = features.estimate_energy(_tone(440, 16000, amp=0.9))
+    quiet = features.estimate_energy(_tone(440, 16000, amp=0.01))
+    assert 0.0 <= quiet < loud <= 1.0
+
+
+def test_tempo_click_track():
+    sr = 16000
+    bpm = 120.0
+    n = int(sr * 10)
+    audio = torch.zeros(n)
+    step = int(sr * 60 / bpm)
+    for s in range(0, n, step):
+        audio[s : s + 200] = torch.randn(min(200, n - s)) * 0.8
+    tempo = features.estimate_tempo(audio, sr)
+    # accept the octave family around 120
+    assert any(abs(tempo - b) < 6.0 for b in (60.0, 120.0, 180.0)), tempo
+
+
+def test_tempo_folding_range():
+    sr = 16000
+    audio = torch.randn(sr * 5) * 0.1
+    tempo = features.estimate_tempo(audio, sr)
+    assert tempo == 0.0 or features.TEMPO_MIN_BPM <= tempo <= features.TEMPO_MAX_BPM
+
+
+def test_key_detection_major_triad():
+    sr = 22050
+    # C major triad: C4, E4, G4
+    audio = _tone(261.63, sr) + _tone(329.63, sr) + _tone(392.0, sr)
+    key, scale = features.estimate_key_scale(audio, sr)
+    assert key in ("C", "E", "G", "A")  # triad tones / relative minor family
+    # empty input default
+    assert features.estimate_key_scale(torch.zeros(0), sr) == ("C", "major")
+
+
+def test_musicnn_shapes_and_aggregation():
+    torch.manual_seed(0)
+    emb_model = MusiCNNEmbedding()
+    pred_model = MusiCNNPrediction()
+    patches = torch.randn(3, 187, 96)
+    with torch.no_grad():
+        emb = emb_model(patches)
+        logits = pred_model(emb)
+    assert emb.shape == (3, 200) and logits.shape == (3, 50)
+    track_emb, moods = aggregate_track(emb, logits)
+    assert track_emb.shape == (200,)
+    assert set(moods) == set(C.MOOD_LABELS)
+    assert all(0.0 < v < 1.0 for v in moods.values())
+    # reference aggregation formula: sigmoid(mean(sigmoid(logits)))
+    expect = torch.sigmoid(torch.sigmoid(logits).mean(dim=0))
+    np.testing.assert_allclose(
+        [moods[l] for l in C.MOOD_LABELS], expect.numpy(), rtol=1e-5)
+
+
+def test_signature_bits_above_mean():
+    v = np.zeros(200, dtype=np.float32)
+    v[:10] = 5.0  # only first 10 dims above mean
+    sig = simhash.embedding_signature(v)
+    assert sig is not None
+    assert sig == int("1" * 10 + "0" * 190, 2)
+
+
+def test_signature_invalid_inputs():
+    assert simhash.embedding_signature(None) is None
+    assert simhash.embedding_signature(np.zeros(100)) is None     # wrong dim
+    assert simhash.embedding_signature(np.ones(200)) is None      # constant
+    bad = np.ones(200); bad[0] = np.nan
+    assert simhash.embedding_signature(bad) is None
+
+
+def test_canonical_id_format_and_mint_collision():
+    sig = 12345
+    cid = simhash.canonical_id_str(sig)
+    assert cid.startswith("fp_4") and len(cid) == simhash.CANONICAL_ID_LEN
+    assert simhash.is_signature_id(cid)
+    assert simhash.signature_from_id(cid) == sig
+    # collision steps to next value
+    nxt = simhash.mint_canonical_id(sig, {cid})
+    assert simhash.signature_from_id(nxt) == sig + 1
+
+
+def test_signature_index_hamming_lookup():
+    rng = np.random.default_rng(0)
+    idx = simhash.SignatureIndex()
+    base = rng.standard_normal(200).astype(np.float32)
+    sig = simhash.embedding_signature(base)
+    idx.add("a", sig, duration=200.0)
+    idx.add("b", sig ^ (1 << 50) ^ (1 << 100), duration=200.0)   # hamming 2
+    far = simhash.embedding_signature(rng.standard_normal(200).astype(np.float32))
+    idx.add("c", far, duration=200.0)
+    hits = idx.lookup(sig)
+    assert [h[0] for h in hits[:2]] == ["a", "b"]
+    assert hits[0][1] == 0 and hits[1][1] == 2
+    # duration gate filters
+    idx2 = simhash.SignatureIndex()
+    idx2.add("x", sig, duration=100.0)
+    assert idx.lookup(sig, duration=200.0)[0][0] == "a"
+    assert idx2.lookup(sig, duration=300.0) == []
+
+
+def test_catalog_resolver_match_and_mint():
+    rng = np.random.default_rng(1)
+    r = simhash.CatalogResolver()
+    emb = rng.standard_normal(200).astype(np.float32)
+    id1, matched = r.resolve(emb, 180.0, "srv", "t1")
+    assert not matched and simhash.is_signature_id(id1)
+    # same recording, tiny noise -> matches
+    id2, matched = r.resolve(emb + rng.standard_normal(200).astype(np.float32) * 1e-4,
+                             180.0, "srv", "t2")
+    assert matched and id2 == id1
+    # different recording -> new id
+    id3, matched = r.resolve(rng.standard_normal(200).astype(np.float32),
+                             180.0, "srv", "t3")
+    assert not matched and id3 != id1
+    # unusable embedding -> fp_0 id
+    id4, matched = r.resolve(None, 0.0, "srv", "t4")
+    assert id4.startswith("fp_0") and not matched
