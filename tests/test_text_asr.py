@@ -1,0 +1,153 @@
+"""Text encoders, Whisper decode loop, VAD, lyrics pipeline (tiny configs)."""
+
+import numpy as np
+import pytest
+import torch
+
+from audiomuse_amd.engines.lyrics import (LyricsPipeline, detect_language,
+                                          quality_gate, score_axes)
+from audiomuse_amd.models.text import (HashTokenizer, TextEmbedder,
+                                       TextEncoderConfig)
+from audiomuse_amd.models.vad import (SileroStyleVAD, speech_probabilities,
+                                      speech_ratio, speech_segments)
+from audiomuse_amd.models.whisper import (TOK_EOT, WhisperConfig, WhisperModel,
+                                          beam_decode, greedy_decode)
+
+
+def tiny_text_cfg(**kw):
+    d = dict(vocab_size=512, dim=32, layers=2, heads=2, max_len=32,
+             out_dim=16, pool="cls")
+    d.update(kw)
+    return TextEncoderConfig(**d)
+
+
+def test_tokenizer_deterministic_and_bounded():
+    tok = HashTokenizer(vocab_size=1000, max_len=10)
+    a = tok.encode("Hello, world! Hello")
+    b = tok.encode("Hello, world! Hello")
+    assert a == b
+    assert a[0] == 1 and a[-1] == 2           # CLS ... SEP
+    assert all(0 <= t < 1000 for t in a)
+    long = tok.encode("word " * 100)
+    assert len(long) <= 10
+
+
+def test_text_embedder_normalized_and_distinct():
+    emb = TextEmbedder(tiny_text_cfg(), seed=0)
+    out = emb.embed(["a happy upbeat dance song", "sad slow piano ballad"])
+    assert out.shape == (2, 16)
+    torch.testing.assert_close(out.norm(dim=1), torch.ones(2), atol=1e-4,
+                               rtol=0)
+    assert float((out[0] - out[1]).norm()) > 1e-3
+
+
+def tiny_whisper():
+    return WhisperModel(WhisperConfig(n_mels=8, n_frames=64, dim=32,
+                                      enc_layers=1, dec_layers=1, heads=2,
+                                      vocab_size=128, max_tokens=24))
+
+
+def test_whisper_greedy_decode_terminates():
+    torch.manual_seed(0)
+    m = tiny_whisper().eval()
+    mel = torch.randn(8, 64)
+    seq = greedy_decode(m, mel, max_tokens=10)
+    assert isinstance(seq, list) and len(seq) <= 10
+    assert TOK_EOT not in seq
+
+
+def test_whisper_decode_cache_consistent_with_full_pass():
+    """Incremental KV-cache logits == full-sequence forward logits."""
+    torch.manual_seed(1)
+    m = tiny_whisper().eval()
+    mel = torch.randn(8, 64)
+    with torch.inference_mode():
+        enc = m.encode(mel.unsqueeze(0))
+        ckv = m.cross_kvs(enc)
+        toks = torch.tensor([[1, 10, 30, 40]])
+        # full pass
+        caches_a = m.make_caches(1, enc.device, enc.dtype)
+        full = m.decode_step(toks, 0, caches_a, ckv)
+        # incremental
+        caches_b = m.make_caches(1, enc.device, enc.dtype)
+        outs = []
+        for i in range(toks.shape[1]):
+            outs.append(m.decode_step(toks[:, i : i + 1], i, caches_b, ckv))
+        inc = torch.cat(outs, dim=1)
+    torch.testing.assert_close(full, inc, rtol=1e-4, atol=1e-4)
+
+
+def test_whisper_beam_decode_runs():
+    torch.manual_seed(2)
+    m = tiny_whisper().eval()
+    seq = beam_decode(m, torch.randn(8, 64), beam=2, max_tokens=8)
+    assert isinstance(seq, list) and len(seq) <= 8
+
+
+def test_vad_windowing_and_segments():
+    torch.manual_seed(3)
+    vad = SileroStyleVAD().eval()
+    audio = torch.randn(16000)        # 1 s -> 31 windows
+    probs = speech_probabilities(vad, audio)
+    assert probs.shape[0] == 31
+    assert ((probs >= 0) & (probs <= 1)).all()
+    # synthetic prob pattern -> segments
+    p = torch.tensor([0.1, 0.9, 0.9, 0.9, 0.1, 0.9, 0.9, 0.1])
+    segs = speech_segments(p, threshold=0.5, min_windows=3)
+    assert len(segs) == 1
+    assert abs(segs[0][0] - 1 * 512 / 16000) < 1e-6
+    assert speech_ratio(p) == pytest.approx(5 / 8)
+
+
+def test_language_detection_and_quality_gate():
+    assert detect_language("the quick fox and you have that with this") == "en"
+    assert detect_language("los ninos que cantan por las calles con una") == "es"
+    assert detect_language("xyz qrs") == "unknown"
+    assert quality_gate("many different words appear in this long enough text ok")
+    assert not quality_gate("la la la la la la la la la la")
+    assert not quality_gate("too short")
+
+
+def test_score_axes_softmax_range():
+    rng = np.random.default_rng(0)
+    emb = rng.standard_normal(16).astype(np.float32)
+    axes = {"love": rng.standard_normal((2, 16)).astype(np.float32),
+            "party": rng.standard_normal((2, 16)).astype(np.float32)}
+    scores = score_axes(emb, axes, temperature=0.1)
+    assert set(scores) == {"love", "party"}
+    assert all(0.0 <= v <= 1.0 for v in scores.values())
+
+
+def test_lyrics_pipeline_provided_and_instrumental():
+    emb = TextEmbedder(tiny_text_cfg(out_dim=0, pool="cls"), seed=1)
+    pipe = LyricsPipeline(emb, vad=None, asr_fn=None,
+                          axis_labels=["love", "party", "sadness"])
+    res = pipe.analyze(provided_lyrics="you and me dancing all night long baby")
+    assert res.source == "provided" and not res.instrumental
+    assert res.embedding is not None and res.embedding.shape == (32,)
+    assert set(res.axis_scores) == {"love", "party", "sadness"}
+    # no lyrics, no asr -> instrumental sentinel
+    res2 = pipe.analyze(audio=torch.randn(16000))
+    assert res2.instrumental and res2.source == "instrumental"
+
+
+def test_lyrics_pipeline_vad_gates_asr():
+    emb = TextEmbedder(tiny_text_cfg(out_dim=0), seed=2)
+    vad = SileroStyleVAD().eval()
+    calls = []
+
+    def asr(audio):
+        calls.append(1)
+        return "we sing the words of a long and meaningful chorus tonight"
+
+    pipe = LyricsPipeline(emb, vad=vad, asr_fn=asr, vad_speech_threshold=1.1,
+                          axis_labels=["love"])
+    res = pipe.analyze(audio=torch.randn(16000) * 0.01)
+    # threshold 1.1 is unreachable -> VAD gate fires, ASR never called
+    assert res.instrumental and not calls
+
+    pipe2 = LyricsPipeline(emb, vad=vad, asr_fn=asr, vad_speech_threshold=0.0,
+                           axis_labels=["love"])
+    res2 = pipe2.analyze(audio=torch.randn(16000) * 0.01)
+    assert calls and res2.source == "asr"
+    assert len(res2.text.split()) <= 300
