@@ -1,0 +1,191 @@
+"""Per-track analysis runtime: decode -> DSP -> encoders -> features.
+
+Reference: /root/reference/tasks/analysis/song.py (analyze_track :478)
++ clap_analyzer.analyze_audio_file (:432) + the staged per-track
+pipeline in tasks/analysis/album.py:290. The ONNX sessions are replaced
+by our torch/HIP models; on a GPU the whole album batches through the
+encoders at once instead of the reference's
+chunked per-track loops.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from audiomuse_amd import config as C
+from audiomuse_amd.models.htsat import (HTSATConfig, HTSATEncoder,
+                                        clap_track_embedding)
+from audiomuse_amd.models.musicnn import (MusiCNNEmbedding, MusiCNNPrediction,
+                                          aggregate_track)
+from audiomuse_amd.models.text import TextEmbedder, clap_text_config
+from audiomuse_amd.ops import dsp, features, hip_ops
+from audiomuse_amd.ops.audio_io import load_audio
+
+
+@dataclass
+class TrackAnalysis:
+    tempo: float = 0.0
+    energy: float = 0.0
+    key: str = "C"
+    scale: str = "major"
+    duration: float = 0.0
+    embedding: Optional[np.ndarray] = None          # 200-d MusiCNN
+    moods: Dict[str, float] = field(default_factory=dict)
+    clap_embedding: Optional[np.ndarray] = None     # 512-d
+    other_features: Dict[str, float] = field(default_factory=dict)
+
+
+class AnalysisRuntime:
+    """Holds the models; one instance per worker process/GPU rank."""
+
+    def __init__(self, device: str = "cpu", seed: int = 0,
+                 enable_clap: Optional[bool] = None):
+        self.device = torch.device(device)
+        self.dtype = (torch.bfloat16 if self.device.type == "cuda"
+                      else torch.float32)
+        self.enable_clap = C.CLAP_ENABLED if enable_clap is None else enable_clap
+        torch.manual_seed(seed)
+        self.musicnn_emb = MusiCNNEmbedding().to(self.device).eval()
+        self.musicnn_pred = MusiCNNPrediction().to(self.device).eval()
+        self.htsat = (HTSATEncoder(HTSATConfig())
+                      .to(self.device, self.dtype).eval()
+                      if self.enable_clap else None)
+        self._clap_text: Optional[TextEmbedder] = None
+        self._other_label_embs: Optional[np.ndarray] = None
+
+    # -- label text embeddings (clap_analyzer.py:580: cached .npz) -------
+
+    def other_feature_label_embeddings(self) -> np.ndarray:
+        if self._other_label_embs is None:
+            if self._clap_text is None:
+                self._clap_text = TextEmbedder(clap_text_config(),
+                                               device=str(self.device),
+                                               dtype=self.dtype)
+            texts = [f"this song is {lbl}" for lbl in C.OTHER_FEATURE_LABELS]
+            self._other_label_embs = self._clap_text.embed(texts).cpu().numpy()
+        return self._other_label_embs
+
+    # -- single-track analysis ------------------------------------------
+
+    @torch.inference_mode()
+    def analyze_track(self, audio_source, sr: Optional[int] = None
+                      ) -> Optional[TrackAnalysis]:
+        """audio_source: wav bytes/path, or float tensor (with sr)."""
+        if isinstance(audio_source, torch.Tensor):
+            audio, in_sr = audio_source, sr or 44100
+        else:
+            audio, in_sr = load_audio(audio_source)
+            if audio is None:
+                return None
+        out = TrackAnalysis(duration=audio.shape[-1] / in_sr)
+
+        from audiomuse_amd.ops.audio_io import resample
+
+        a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE).to(self.device)
+        out.tempo, out.energy, out.key, out.scale = \
+            features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+
+        # MusiCNN patches (song.py:240-256): 187-frame log-mel windows
+        mel = hip_ops.mel_spectrogram(a16, dsp.musicnn_mel_config())
+        frames = mel.shape[-1]
+        P = C.MUSICNN_PATCH_FRAMES
+        if frames >= P:
+            patches = torch.stack([mel[:, i : i + P]
+                                   for i in range(0, frames - P + 1, P)])
+            patches = patches.transpose(1, 2)          # (N, 187, 96)
+            emb = self.musicnn_emb(patches.float())
+            logits = self.musicnn_pred(emb)
+            track_emb, moods = aggregate_track(emb, logits)
+            out.embedding = track_emb.cpu().numpy().astype(np.float32)
+            out.moods = moods
+
+        if self.htsat is not None:
+            a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE).to(self.device)
+            out.clap_embedding = self._clap_embed(a48)
+            out.other_features = self._score_other_features(out.clap_embedding)
+        return out
+
+    def _clap_embed(self, a48: torch.Tensor) -> np.ndarray:
+        """Segments -> fused mel -> HTSAT -> mean + L2
+        (clap_analyzer.py:432-511; int16 round-trip fused in kernel)."""
+        segs = dsp.segment_audio(a48, C.CLAP_SEGMENT_SAMPLES,
+                                 C.CLAP_SEGMENT_HOP_SAMPLES)
+        mel = hip_ops.mel_spectrogram(segs, dsp.clap_mel_config(),
+                                      quantize_int16=True)
+        emb = self.htsat(mel.to(self.dtype)).float()
+        return clap_track_embedding(emb).cpu().numpy().astype(np.float32)
+
+    def _score_other_features(self, clap_emb: np.ndarray) -> Dict[str, float]:
+        """cosine vs the 6 cached label embeddings (clap_analyzer.py:641)."""
+        labels = self.other_feature_label_embeddings()
+        v = clap_emb / (np.linalg.norm(clap_emb) + 1e-9)
+        sims = labels @ v
+        return {lbl: float(s) for lbl, s in zip(C.OTHER_FEATURE_LABELS, sims)}
+
+    # -- batched album analysis (GPU path) --------------------------------
+
+    @torch.inference_mode()
+    def analyze_album_batch(self, wav_blobs: List[bytes]
+                            ) -> List[Optional[TrackAnalysis]]:
+        """All tracks of an album; CLAP segments of every track batch
+        into one encoder pass."""
+        results: List[Optional[TrackAnalysis]] = []
+        seg_batches: List[torch.Tensor] = []
+        seg_owner: List[int] = []
+        from audiomuse_amd.ops.audio_io import resample
+
+        for i, blob in enumerate(wav_blobs):
+            audio, in_sr = load_audio(blob)
+            if audio is None:
+                results.append(None)
+                continue
+            res = self.analyze_track_base(audio, in_sr)
+            results.append(res)
+            if self.htsat is not None:
+                a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE).to(self.device)
+                segs = dsp.segment_audio(a48, C.CLAP_SEGMENT_SAMPLES,
+                                         C.CLAP_SEGMENT_HOP_SAMPLES)
+                seg_batches.append(segs)
+                seg_owner.extend([i] * segs.shape[0])
+        if self.htsat is not None and seg_batches:
+            all_segs = torch.cat(seg_batches, dim=0)
+            mel = hip_ops.mel_spectrogram(all_segs, dsp.clap_mel_config(),
+                                          quantize_int16=True)
+            embs = self.htsat(mel.to(self.dtype)).float()
+            owner = torch.tensor(seg_owner)
+            for i, res in enumerate(results):
+                if res is None:
+                    continue
+                mine = embs[owner == i]
+                if mine.shape[0] == 0:
+                    continue
+                res.clap_embedding = clap_track_embedding(mine).cpu().numpy()
+                res.other_features = self._score_other_features(res.clap_embedding)
+        return results
+
+    @torch.inference_mode()
+    def analyze_track_base(self, audio: torch.Tensor, in_sr: int
+                           ) -> TrackAnalysis:
+        """Features + MusiCNN only (no CLAP) — used by the batched path."""
+        from audiomuse_amd.ops.audio_io import resample
+
+        out = TrackAnalysis(duration=audio.shape[-1] / in_sr)
+        a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE).to(self.device)
+        out.tempo, out.energy, out.key, out.scale = \
+            features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+        mel = hip_ops.mel_spectrogram(a16, dsp.musicnn_mel_config())
+        P = C.MUSICNN_PATCH_FRAMES
+        if mel.shape[-1] >= P:
+            patches = torch.stack([mel[:, i : i + P]
+                                   for i in range(0, mel.shape[-1] - P + 1, P)])
+            patches = patches.transpose(1, 2)
+            emb = self.musicnn_emb(patches.float())
+            logits = self.musicnn_pred(emb)
+            track_emb, moods = aggregate_track(emb, logits)
+            out.embedding = track_emb.cpu().numpy().astype(np.float32)
+            out.moods = moods
+        return out

@@ -1,0 +1,171 @@
+"""Analysis task handlers (queue entry points).
+
+Reference call stack: run_analysis_task
+(/root/reference/tasks/analysis/main.py:777) -> per-server phases,
+work map vs track_server_map, child analyze_album_task dispatch with
+back-pressure <= MAX_QUEUED_ANALYSIS_JOBS, drain/monitor with
+cooperative cancel, periodic + final rebuild_all_indexes_task.
+analyze_album_task (album.py:393) stages per track: download ->
+analyze (MusiCNN + features + CLAP batched on GPU) -> identity
+(simhash resolve) -> persist.
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from audiomuse_amd import config as C
+from audiomuse_amd.analysis.index import run_all_index_builds
+from audiomuse_amd.analysis.pipeline import AnalysisRuntime
+from audiomuse_amd.db import get_db, write_txn
+from audiomuse_amd.db.store import (save_clap_embedding,
+                                    save_track_analysis_and_embedding)
+from audiomuse_amd.engines.simhash import CatalogResolver
+from audiomuse_amd.mediaserver import make_provider
+from audiomuse_amd.taskqueue import enqueue
+from audiomuse_amd.taskqueue import sql as qsql
+from audiomuse_amd.taskqueue.worker import CancelledError, TaskContext, task_handler
+
+logger = logging.getLogger(__name__)
+
+_RUNTIME: Optional[AnalysisRuntime] = None
+
+
+def get_runtime() -> AnalysisRuntime:
+    """One model runtime per worker process (one process per GPU rank)."""
+    global _RUNTIME
+    if _RUNTIME is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+        _RUNTIME = AnalysisRuntime(device=device)
+    return _RUNTIME
+
+
+def _resolver_from_db(conn) -> CatalogResolver:
+    resolver = CatalogResolver()
+    rows = conn.execute(
+        """SELECT e.item_id, e.embedding, s.duration FROM embedding e
+           LEFT JOIN score s ON s.item_id = e.item_id""").fetchall()
+    for r in rows:
+        emb = np.frombuffer(r["embedding"], dtype=np.float32)
+        resolver.register_existing(r["item_id"], emb, r["duration"] or 0.0)
+    return resolver
+
+
+def _analyzed_provider_ids(conn, server_id: str) -> set:
+    return {r["provider_id"] for r in conn.execute(
+        "SELECT provider_id FROM track_server_map WHERE server_id=?",
+        (server_id,))}
+
+
+@task_handler("analyze_album")
+def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Child task: one album end-to-end (album.py:393)."""
+    conn = ctx.conn
+    server_id = payload["server_id"]
+    provider = make_provider(payload["server_type"],
+                             **payload.get("server_config", {}))
+    runtime = get_runtime()
+    resolver = _resolver_from_db(conn)
+    done = _analyzed_provider_ids(conn, server_id)
+
+    tracks = provider.get_tracks_from_album(payload["album_id"])
+    todo = [t for t in tracks if t.provider_id not in done]
+    blobs: List[Optional[bytes]] = []
+    for t in todo:
+        ctx.check_cancelled()
+        blobs.append(provider.download_track(t.provider_id))
+    valid = [(t, b) for t, b in zip(todo, blobs) if b is not None]
+    if not valid:
+        return {"analyzed": 0, "skipped": len(tracks) - len(todo)}
+
+    results = runtime.analyze_album_batch([b for _, b in valid])
+    n = 0
+    for (track, _blob), res in zip(valid, results):
+        ctx.check_cancelled()
+        if res is None:
+            continue
+        item_id, _matched = resolver.resolve(
+            res.embedding, res.duration, server_id, track.provider_id)
+        save_track_analysis_and_embedding(
+            conn, item_id, title=track.title, author=track.author,
+            album=track.album, tempo=res.tempo, key=res.key, scale=res.scale,
+            mood_vector=res.moods, other_features=res.other_features,
+            energy=res.energy, duration=res.duration, embedding=res.embedding)
+        if res.clap_embedding is not None:
+            save_clap_embedding(conn, item_id, res.clap_embedding)
+        with write_txn(conn):
+            conn.execute(
+                """INSERT INTO track_server_map
+                       (provider_id, server_id, item_id, title, author, album,
+                        file_path)
+                   VALUES (?,?,?,?,?,?,?)
+                   ON CONFLICT(provider_id, server_id)
+                   DO UPDATE SET item_id=excluded.item_id""",
+                (track.provider_id, server_id, item_id, track.title,
+                 track.author, track.album, track.file_path))
+        n += 1
+    return {"analyzed": n, "skipped": len(tracks) - len(todo)}
+
+
+@task_handler("run_analysis")
+def run_analysis_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Parent task (main.py:777): preflight, work map, child dispatch with
+    back-pressure, drain, final index rebuild."""
+    conn = ctx.conn
+    server_id = payload.get("server_id", "default")
+    provider = make_provider(payload["server_type"],
+                             **payload.get("server_config", {}))
+    if not provider.test_connection():        # main.py:294 preflight
+        raise RuntimeError(f"media server {server_id!r} unreachable")
+
+    albums = provider.get_recent_albums(limit=payload.get("album_limit", 0))
+    ctx.report(5.0, f"{len(albums)} albums to scan")
+
+    child_ids: List[str] = []
+    dispatched = 0
+    for album in albums:
+        ctx.check_cancelled()
+        # back-pressure (main.py: <= MAX_QUEUED_ANALYSIS_JOBS live children)
+        while qsql.pending_children(conn, ctx.task_id) >= C.MAX_QUEUED_ANALYSIS_JOBS:
+            ctx.check_cancelled()
+            time.sleep(C.QUEUE_POLL_SECONDS)
+        child_ids.append(enqueue(
+            conn, "analyze_album",
+            {"server_type": payload["server_type"],
+             "server_config": payload.get("server_config", {}),
+             "server_id": server_id, "album_id": album.provider_id},
+            parent_task_id=ctx.task_id))
+        dispatched += 1
+        ctx.report(5.0 + 60.0 * dispatched / max(len(albums), 1),
+                   f"dispatched {dispatched}/{len(albums)}")
+
+    # drain loop (main.py stage 4)
+    deadline = time.time() + payload.get("drain_timeout", 3600.0)
+    while time.time() < deadline:
+        ctx.check_cancelled()
+        left = qsql.pending_children(conn, ctx.task_id)
+        if left == 0:
+            break
+        ctx.report(65.0 + 25.0 * (1 - left / max(dispatched, 1)),
+                   f"{left} album jobs outstanding")
+        time.sleep(C.QUEUE_POLL_SECONDS)
+
+    ctx.report(92.0, "rebuilding indexes")
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    built = run_all_index_builds(conn, device=device)
+    ctx.report(100.0, "done")
+    return {"albums": dispatched, "indexes": built}
+
+
+@task_handler("rebuild_indexes")
+def rebuild_indexes_task(ctx: TaskContext, payload: Dict) -> Dict:
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    built = run_all_index_builds(
+        ctx.conn, device=device,
+        progress_cb=lambda name, n: ctx.report(50.0, f"{name}: {n}"))
+    return {"indexes": built}

@@ -1,0 +1,76 @@
+"""Media-server adapters.
+
+Reference: /root/reference/tasks/mediaserver/ — a uniform 19-call
+surface (`_PUBLIC_SERVER_API`, __init__.py:56-62) dispatched to provider
+modules (navidrome/jellyfin/emby/lyrion/plex), a registry with
+provider<->canonical id translation (registry.py), and thread-local
+server binding (context.py).
+
+This build defines the same surface as a Provider ABC with:
+- `synthetic`: a deterministic in-process provider generating WAV tracks
+  (ops.audio_io.synthetic_track) — the test/bench source in an image
+  with no network;
+- `subsonic`: a Subsonic-API HTTP provider (Navidrome speaks Subsonic)
+  as the real-network reference implementation; the remaining providers
+  (jellyfin/emby/lyrion/plex) are HTTP adapters over the same ABC and
+  register identically.
+"""
+
+from __future__ import annotations
+
+import threading
+from typing import Dict, List, Optional
+
+from audiomuse_amd.mediaserver.base import Album, Provider, Track
+
+_PROVIDERS: Dict[str, type] = {}
+_LOCAL = threading.local()
+
+
+def register_provider(name: str):
+    def deco(cls):
+        _PROVIDERS[name] = cls
+        return cls
+    return deco
+
+
+def provider_types() -> List[str]:
+    return sorted(_PROVIDERS)
+
+
+def make_provider(server_type: str, **kwargs) -> Provider:
+    cls = _PROVIDERS.get(server_type)
+    if cls is None:
+        raise ValueError(f"unsupported media server type {server_type!r} "
+                         f"(supported: {', '.join(provider_types())})")
+    return cls(**kwargs)
+
+
+class BoundServer:
+    """Thread-local active-server binding (reference: context.py:35 +
+    BoundServer __init__.py:392)."""
+
+    def __init__(self, provider: Provider, server_id: str):
+        self.provider = provider
+        self.server_id = server_id
+
+    def __enter__(self):
+        stack = getattr(_LOCAL, "stack", None)
+        if stack is None:
+            stack = _LOCAL.stack = []
+        stack.append(self)
+        return self.provider
+
+    def __exit__(self, *exc):
+        _LOCAL.stack.pop()
+        return False
+
+
+def active_server() -> Optional[BoundServer]:
+    stack = getattr(_LOCAL, "stack", None)
+    return stack[-1] if stack else None
+
+
+# register built-ins
+from audiomuse_amd.mediaserver import synthetic  # noqa: E402,F401
+from audiomuse_amd.mediaserver import subsonic  # noqa: E402,F401
