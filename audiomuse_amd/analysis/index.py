@@ -180,7 +180,7 @@ def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
         coords = proj.cpu().numpy()
     buf = io.BytesIO()
     torch.save({"item_ids": ids,
-                "coords": coords.astype(np.float32)}, buf)
+                "coords": torch.from_numpy(coords.astype(np.float32))}, buf)
     store_index_blob(conn, SONG_MAP, buf.getvalue(), meta={"n": len(ids)})
     return len(ids)
 

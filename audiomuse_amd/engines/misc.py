@@ -79,7 +79,8 @@ def sonic_fingerprint(vectors: np.ndarray, played_at: Sequence[float],
         return None
     w = recency_weights(played_at, now=now)
     if w.sum() <= 0:
-        return None
+        # all plays far older than the half-life window: uniform fallback
+        w = np.ones(vectors.shape[0], dtype=np.float32)
     v = (vectors * w[:, None]).sum(axis=0) / w.sum()
     norm = float(np.linalg.norm(v))
     return v / norm if norm > 0 else None

@@ -1,0 +1,97 @@
+"""Auth + setup blueprint (reference: app_auth.py + app_setup.py routes)."""
+
+from __future__ import annotations
+
+from flask import Blueprint, current_app, jsonify, make_response, request
+
+from audiomuse_amd.db import write_txn
+from audiomuse_amd.web.auth import (SESSION_COOKIE, check_setup_needed,
+                                    hash_password, make_session_token,
+                                    require_auth, verify_password)
+
+bp = Blueprint("auth", __name__)
+
+
+def _state():
+    return current_app.extensions["audiomuse"]
+
+
+@bp.get("/api/setup/status")
+def setup_status():
+    return jsonify({"setup_needed": check_setup_needed(_state().conn())})
+
+
+@bp.post("/api/setup/admin")
+def setup_admin():
+    """First-boot admin creation (reference: setup wizard step)."""
+    conn = _state().conn()
+    if not check_setup_needed(conn):
+        return jsonify({"error": "already configured"}), 409
+    body = request.get_json(force=True, silent=True) or {}
+    username = (body.get("username") or "").strip()
+    password = body.get("password") or ""
+    if not username or len(password) < 8:
+        return jsonify({"error": "username and password (>=8 chars) required"}), 400
+    with write_txn(conn):
+        conn.execute(
+            "INSERT INTO audiomuse_users (username, password_hash, role) "
+            "VALUES (?,?, 'admin')", (username, hash_password(password)))
+    return jsonify({"created": username})
+
+
+@bp.post("/api/login")
+def login():
+    conn = _state().conn()
+    body = request.get_json(force=True, silent=True) or {}
+    row = conn.execute(
+        "SELECT username, password_hash, role FROM audiomuse_users "
+        "WHERE username=?", (body.get("username", ""),)).fetchone()
+    if row is None or not verify_password(body.get("password", ""),
+                                          row["password_hash"]):
+        return jsonify({"error": "invalid credentials"}), 401
+    token = make_session_token(row["username"], row["role"])
+    resp = make_response(jsonify({"ok": True, "user": row["username"]}))
+    resp.set_cookie(SESSION_COOKIE, token, httponly=True, samesite="Lax")
+    return resp
+
+
+@bp.post("/api/logout")
+def logout():
+    resp = make_response(jsonify({"ok": True}))
+    resp.delete_cookie(SESSION_COOKIE)
+    return resp
+
+
+@bp.get("/api/me")
+@require_auth
+def me():
+    from flask import g
+    return jsonify({"user": g.user["u"], "role": g.user["r"]})
+
+
+@bp.get("/api/config")
+@require_auth
+def get_config():
+    """Config snapshot + persisted overrides (reference: setup manager)."""
+    from audiomuse_amd import config as C
+    from audiomuse_amd.db.store import get_app_config
+
+    overrides = get_app_config(_state().conn())
+    safe = {k: getattr(C, k) for k in dir(C)
+            if k.isupper() and isinstance(getattr(C, k), (int, float, str, bool))
+            and "SECRET" not in k and "TOKEN" not in k and "PASSWORD" not in k}
+    return jsonify({"config": safe, "overrides": overrides})
+
+
+@bp.post("/api/config")
+@require_auth
+def set_config():
+    from audiomuse_amd.db.store import set_app_config
+
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    for k, v in body.items():
+        if not isinstance(k, str) or not k.isupper():
+            return jsonify({"error": f"bad key {k!r}"}), 400
+        set_app_config(conn, k, str(v))
+    return jsonify({"saved": len(body)})

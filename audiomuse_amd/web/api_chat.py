@@ -1,0 +1,228 @@
+"""Instant playlist (AI chat) blueprint.
+
+Reference: /root/reference/app_chat.py + tasks/ai/ (planner.py:1231
+plan_and_execute_once): regex hint pre-extraction, ONE tool-calling LLM
+request producing <= 4 tool calls (seed_search / text_match /
+knowledge_lookup / search_database), hallucination stripping, tiered
+re-rank with a relax loop, optional ordering.
+
+Provider clients speak the OpenAI-compatible chat/completions HTTP API
+(works for openai/mistral/self-hosted; gemini adapter maps the same
+call shape). With AI_PROVIDER=none the planner is a deterministic
+heuristic over the extracted hints, so the endpoint works offline —
+the tool layer and re-rank are identical either way.
+"""
+
+from __future__ import annotations
+
+import json
+import re
+from typing import Dict, List, Optional
+
+import torch
+from flask import Blueprint, current_app, jsonify, request
+
+from audiomuse_amd import config as C
+from audiomuse_amd.analysis import index as idx
+from audiomuse_amd.web.auth import require_auth
+
+bp = Blueprint("chat", __name__)
+
+_MOOD_WORDS = set(w.lower() for w in C.MOOD_LABELS)
+_HINT_COUNT = re.compile(r"\b(\d{1,3})\s*(?:songs|tracks)\b", re.I)
+_HINT_ARTIST = re.compile(r"\b(?:by|like|similar to)\s+([A-Z][\w&' ]{2,40})")
+
+
+def _state():
+    return current_app.extensions["audiomuse"]
+
+
+def extract_hints(prompt: str) -> Dict:
+    """Regex pre-extraction (planner.extract_hints :218)."""
+    hints: Dict = {"n": 20, "moods": [], "artists": [], "text": prompt}
+    m = _HINT_COUNT.search(prompt)
+    if m:
+        hints["n"] = max(1, min(int(m.group(1)), 100))
+    words = set(re.findall(r"[a-z']+", prompt.lower()))
+    hints["moods"] = sorted(words & _MOOD_WORDS)
+    hints["artists"] = [a.strip() for a in _HINT_ARTIST.findall(prompt)]
+    return hints
+
+
+# -- tools (tasks/ai/tool_impl.py equivalents) ------------------------------
+
+def tool_seed_search(args: Dict) -> List[Dict]:
+    """Similar tracks from seed titles/artists (tool_impl seed_search)."""
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return []
+    conn = _state().conn()
+    seeds = []
+    for name in args.get("seeds", []):
+        row = conn.execute(
+            "SELECT item_id FROM score WHERE LOWER(title) LIKE ? "
+            "OR LOWER(author) LIKE ? LIMIT 1",
+            (f"%{name.lower()}%", f"%{name.lower()}%")).fetchone()
+        if row:
+            seeds.append(row["item_id"])
+    out: List[Dict] = []
+    for sid in seeds[:4]:
+        out.extend(eng.find_similar_by_id(sid, int(args.get("n", 20))))
+    return out
+
+
+def tool_text_match(args: Dict) -> List[Dict]:
+    """CLAP text search (tool_impl text_match)."""
+    eng = _state().engine(idx.CLAP_INDEX)
+    if eng is None:
+        return []
+    from audiomuse_amd.web.api_queries import _make_clap_text_embedder
+
+    emb = current_app.extensions.setdefault("clap_text",
+                                            _make_clap_text_embedder())
+    vec = emb.embed([args.get("query", "")])[0]
+    return eng.find_similar_by_vector(vec, int(args.get("n", 20)))
+
+
+def tool_search_database(args: Dict) -> List[Dict]:
+    """Read-only metadata filters (tool_impl search_database via the
+    low-privilege role, mcp_helper)."""
+    conn = _state().conn()
+    clauses, params = [], []
+    if args.get("mood"):
+        clauses.append("mood_vector LIKE ?")
+        params.append(f"%\"{args['mood']}\"%")
+    if args.get("artist"):
+        clauses.append("LOWER(author) LIKE ?")
+        params.append(f"%{args['artist'].lower()}%")
+    if args.get("min_tempo"):
+        clauses.append("tempo >= ?")
+        params.append(float(args["min_tempo"]))
+    if args.get("max_tempo"):
+        clauses.append("tempo <= ?")
+        params.append(float(args["max_tempo"]))
+    where = (" WHERE " + " AND ".join(clauses)) if clauses else ""
+    rows = conn.execute(
+        f"SELECT item_id FROM score{where} LIMIT ?",
+        (*params, int(args.get("n", 50)))).fetchall()
+    return [{"item_id": r["item_id"], "distance": 0.5} for r in rows]
+
+
+def tool_knowledge_lookup(args: Dict) -> List[Dict]:
+    """Without network knowledge, fall back to text match."""
+    return tool_text_match({"query": args.get("query", ""),
+                            "n": args.get("n", 20)})
+
+
+TOOLS = {"seed_search": tool_seed_search, "text_match": tool_text_match,
+         "search_database": tool_search_database,
+         "knowledge_lookup": tool_knowledge_lookup}
+
+
+def validate_and_normalize_plan(plan: List[Dict]) -> List[Dict]:
+    """Strip hallucinated tools, dedupe, cap at AI_MAX_TOOL_CALLS
+    (planner.validate_and_normalize_plan :946)."""
+    seen = set()
+    out = []
+    for call in plan:
+        name = call.get("tool")
+        if name not in TOOLS:
+            continue
+        key = json.dumps(call, sort_keys=True)
+        if key in seen:
+            continue
+        seen.add(key)
+        out.append(call)
+        if len(out) >= C.AI_MAX_TOOL_CALLS:
+            break
+    return out
+
+
+def heuristic_plan(hints: Dict) -> List[Dict]:
+    """Deterministic offline planner (AI_PROVIDER=none)."""
+    plan: List[Dict] = []
+    if hints["artists"]:
+        plan.append({"tool": "seed_search",
+                     "args": {"seeds": hints["artists"], "n": hints["n"]}})
+    for mood in hints["moods"][:2]:
+        plan.append({"tool": "search_database",
+                     "args": {"mood": mood, "n": hints["n"] * 2}})
+    plan.append({"tool": "text_match",
+                 "args": {"query": hints["text"], "n": hints["n"]}})
+    return plan
+
+
+def llm_plan(prompt: str, hints: Dict) -> Optional[List[Dict]]:
+    """One tool-calling request to an OpenAI-compatible endpoint
+    (providers/openai.py equivalent). None on any failure -> heuristic."""
+    if C.AI_PROVIDER in ("", "none"):
+        return None
+    try:
+        import os
+
+        import requests
+
+        base = os.environ.get("AI_BASE_URL", "https://api.openai.com/v1")
+        key = os.environ.get("AI_API_KEY", "")
+        tools_desc = [{"type": "function", "function": {
+            "name": name, "parameters": {"type": "object"}}}
+            for name in TOOLS]
+        r = requests.post(
+            f"{base}/chat/completions",
+            headers={"Authorization": f"Bearer {key}"},
+            json={"model": C.AI_MODEL_NAME or "gpt-4o-mini",
+                  "messages": [
+                      {"role": "system",
+                       "content": "Plan music-library tool calls for the "
+                                  "user's playlist request. Use only the "
+                                  "provided tools."},
+                      {"role": "user", "content": prompt}],
+                  "tools": tools_desc},
+            timeout=30)
+        r.raise_for_status()
+        msg = r.json()["choices"][0]["message"]
+        plan = []
+        for tc in msg.get("tool_calls", []):
+            fn = tc.get("function", {})
+            plan.append({"tool": fn.get("name"),
+                         "args": json.loads(fn.get("arguments") or "{}")})
+        return plan or None
+    except Exception:
+        return None
+
+
+def rerank(results_per_tool: List[List[Dict]], n: int) -> List[str]:
+    """Tiered re-rank with intersection boost + relax loop
+    (tasks/ai/rerank.py)."""
+    scores: Dict[str, float] = {}
+    hits: Dict[str, int] = {}
+    for results in results_per_tool:
+        for r in results:
+            iid = r["item_id"]
+            scores[iid] = min(scores.get(iid, 10.0), r.get("distance", 1.0))
+            hits[iid] = hits.get(iid, 0) + 1
+    ranked = sorted(scores,
+                    key=lambda i: (-hits[i], scores[i]))  # intersections first
+    return ranked[:n]
+
+
+@bp.post("/chat/api/chatPlaylist")
+@require_auth
+def chat_playlist():
+    """reference: app_chat.py:144,264"""
+    body = request.get_json(force=True, silent=True) or {}
+    prompt = body.get("prompt", "")
+    if not prompt:
+        return jsonify({"error": "prompt required"}), 400
+    hints = extract_hints(prompt)
+    plan = llm_plan(prompt, hints) or heuristic_plan(hints)
+    plan = validate_and_normalize_plan(plan)
+    results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
+    ids = rerank(results, hints["n"])
+    state = _state()
+    tracks = []
+    for i in ids:
+        meta = state.meta_fn(i) or {}
+        tracks.append({"item_id": i, "title": meta.get("title"),
+                       "author": meta.get("author")})
+    return jsonify({"plan": plan, "tracks": tracks, "hints": hints})

@@ -1,0 +1,275 @@
+"""Similarity-feature blueprints.
+
+Reference endpoints (SURVEY.md §2.1 query blueprints): /api/similar_tracks
+(app_ivf.py:313), song path (app_path.py), alchemy (app_alchemy.py),
+music map (app_map.py), artist similarity (app_artist_similarity.py),
+CLAP text search (app_clap_search.py), lyrics search (app_lyrics.py),
+SemGrove (app_sem_grove.py), sonic fingerprint
+(app_sonic_fingerprint.py), hyperbolic explorer (app_hyperbolic.py).
+"""
+
+from __future__ import annotations
+
+import json
+from typing import List, Optional
+
+import numpy as np
+import torch
+from flask import Blueprint, current_app, jsonify, request
+
+from audiomuse_amd import config as C
+from audiomuse_amd.analysis import index as idx
+from audiomuse_amd.engines.alchemy import alchemy_query
+from audiomuse_amd.engines.misc import order_playlist, sonic_fingerprint
+from audiomuse_amd.engines.path import find_path
+from audiomuse_amd.web.auth import require_auth
+
+bp = Blueprint("queries", __name__)
+
+
+def _state():
+    return current_app.extensions["audiomuse"]
+
+
+def _with_meta(results: List[dict]) -> List[dict]:
+    state = _state()
+    out = []
+    for r in results:
+        meta = state.meta_fn(r["item_id"]) or {}
+        out.append({**r, "title": meta.get("title"),
+                    "author": meta.get("author")})
+    return out
+
+
+@bp.get("/api/similar_tracks")
+@require_auth
+def similar_tracks():
+    """reference: app_ivf.py:313"""
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    item_id = request.args.get("item_id", "")
+    n = int(request.args.get("n", 10))
+    radius = request.args.get("radius_similarity", "0") in ("1", "true")
+    mood = request.args.get("mood_filter") or None
+    cap = request.args.get("max_per_artist")
+    res = eng.find_similar_by_id(
+        item_id, n, radius=radius, mood_filter=mood,
+        eliminate_duplicates=request.args.get("eliminate_duplicates", "1")
+        in ("1", "true"),
+        max_per_artist=int(cap) if cap else None)
+    if not res and eng.vector_for_id(item_id) is None:
+        return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
+    return jsonify(_with_meta(res))
+
+
+@bp.get("/api/search_tracks")
+@require_auth
+def search_tracks():
+    """Unified metadata search (ivf_manager unified track search)."""
+    q = (request.args.get("q") or "").strip().lower()
+    if not q:
+        return jsonify([])
+    conn = _state().conn()
+    rows = conn.execute(
+        """SELECT item_id, title, author, album FROM score
+           WHERE LOWER(title) LIKE ? OR LOWER(author) LIKE ?
+           LIMIT ?""",
+        (f"%{q}%", f"%{q}%", int(request.args.get("n", 25)))).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.get("/api/path")
+@require_auth
+def song_path():
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    a = request.args.get("start", "")
+    b = request.args.get("end", "")
+    length = int(request.args.get("length", 12))
+    res = find_path(eng, a, b, length=length,
+                    max_per_artist=C.MAX_SONGS_PER_ARTIST or None)
+    if not res:
+        return jsonify({"error": "unknown endpoints"}), 404
+    return jsonify(_with_meta(res))
+
+
+@bp.post("/api/alchemy")
+@require_auth
+def alchemy():
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    body = request.get_json(force=True, silent=True) or {}
+    add_ids = body.get("add", [])
+    sub_ids = body.get("subtract", [])
+
+    def _vecs(ids):
+        out = []
+        for i in ids:
+            v = eng.vector_for_id(i)
+            if v is not None:
+                out.append(v.cpu().numpy())
+        return out
+
+    res = alchemy_query(
+        eng, _vecs(add_ids), _vecs(sub_ids),
+        n=int(body.get("n", 25)),
+        subtract_radius=float(body.get("subtract_radius", 0.0)),
+        temperature=float(body.get("temperature", 0.0)),
+        exclude=tuple(add_ids), seed=body.get("seed"))
+    return jsonify(_with_meta(res))
+
+
+@bp.get("/api/artist_similarity")
+@require_auth
+def artist_similarity():
+    sim = _state().engine(idx.ARTIST_INDEX)
+    if sim is None:
+        return jsonify({"error": "artist index not built"}), 503
+    artist = request.args.get("artist", "")
+    res = sim.find_similar_artists(artist, n=int(request.args.get("n", 10)))
+    if not res and artist not in sim.models:
+        return jsonify({"error": f"unknown artist {artist!r}"}), 404
+    return jsonify([{"artist": a, "distance": d} for a, d in res])
+
+
+@bp.get("/api/clap_search")
+@require_auth
+def clap_text_search():
+    """Text -> CLAP space -> IVF (app_clap_search.py; clap_text_search.py:171)."""
+    eng = _state().engine(idx.CLAP_INDEX)
+    if eng is None:
+        return jsonify({"error": "clap index not built"}), 503
+    q = request.args.get("q", "")
+    if not q:
+        return jsonify([])
+    emb = current_app.extensions.setdefault(
+        "clap_text", _make_clap_text_embedder())
+    vec = emb.embed([q])[0]
+    res = eng.find_similar_by_vector(vec, int(request.args.get("n", 20)))
+    return jsonify(_with_meta(res))
+
+
+def _make_clap_text_embedder():
+    from audiomuse_amd.models.text import TextEmbedder, clap_text_config
+    return TextEmbedder(clap_text_config(), device=_state().device)
+
+
+@bp.get("/api/lyrics_search")
+@require_auth
+def lyrics_search():
+    """Text -> GTE space -> lyrics IVF (app_lyrics.py)."""
+    eng = _state().engine(idx.LYRICS_INDEX)
+    if eng is None:
+        return jsonify({"error": "lyrics index not built"}), 503
+    q = request.args.get("q", "")
+    if not q:
+        return jsonify([])
+    emb = current_app.extensions.setdefault("gte_text", _make_gte_embedder())
+    vec = emb.embed([q])[0]
+    res = eng.find_similar_by_vector(vec, int(request.args.get("n", 20)))
+    return jsonify(_with_meta(res))
+
+
+def _make_gte_embedder():
+    from audiomuse_amd.models.text import TextEmbedder, gte_config
+    return TextEmbedder(gte_config(), device=_state().device)
+
+
+@bp.get("/api/semgrove")
+@require_auth
+def semgrove():
+    """Seed-song search in the fused space (app_sem_grove.py)."""
+    eng = _state().engine(idx.SEMGROVE_INDEX)
+    if eng is None:
+        return jsonify({"error": "semgrove index not built"}), 503
+    item_id = request.args.get("item_id", "")
+    res = eng.find_similar_by_id(
+        item_id, int(request.args.get("n", 15)),
+        radius=request.args.get("radius", "0") in ("1", "true"))
+    if not res and eng.vector_for_id(item_id) is None:
+        return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
+    return jsonify(_with_meta(res))
+
+
+@bp.get("/api/sonic_fingerprint")
+@require_auth
+def sonic_fp():
+    """Taste vector from listen history -> expansion (app_sonic_fingerprint)."""
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    body_ids = request.args.getlist("item_id")
+    played = request.args.getlist("played_at", type=float)
+    if not body_ids:
+        return jsonify({"error": "item_id params required"}), 400
+    vecs, times = [], []
+    for i, pid in enumerate(body_ids):
+        v = eng.vector_for_id(pid)
+        if v is not None:
+            vecs.append(v.cpu().numpy())
+            import time as _t
+            times.append(played[i] if i < len(played) else _t.time())
+    if not vecs:
+        return jsonify({"error": "no known tracks"}), 404
+    fp = sonic_fingerprint(np.stack(vecs), times)
+    res = eng.find_similar_by_vector(torch.from_numpy(fp),
+                                     int(request.args.get("n", 20)),
+                                     exclude=tuple(body_ids))
+    return jsonify(_with_meta(res))
+
+
+@bp.get("/api/hyperbolic_similar")
+@require_auth
+def hyperbolic_similar():
+    """Poincare-space neighbors (app_hyperbolic.py:450)."""
+    from audiomuse_amd.engines.hyperbolic import HyperbolicSpace
+
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    item_id = request.args.get("item_id", "")
+    pos = eng.pos.get(item_id)
+    if pos is None:
+        return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
+    space = current_app.extensions.get("hyperbolic")
+    if space is None or current_app.extensions.get("hyperbolic_n") != eng.index.n:
+        space = HyperbolicSpace(eng.index.vectors_f32)
+        current_app.extensions["hyperbolic"] = space
+        current_app.extensions["hyperbolic_n"] = eng.index.n
+    d, indices = space.similar(pos, int(request.args.get("n", 10)))
+    res = [{"item_id": eng.item_ids[int(i)], "distance": float(dd)}
+           for dd, i in zip(d, indices)]
+    return jsonify(_with_meta(res))
+
+
+@bp.get("/api/map")
+@require_auth
+def music_map():
+    """2-D map coordinates (app_map.py buckets)."""
+    data = _state().engine(idx.SONG_MAP)
+    if data is None:
+        return jsonify({"error": "map not built"}), 503
+    coords = data["coords"]
+    ids = data["item_ids"]
+    limit = int(request.args.get("n", 0)) or len(ids)
+    return jsonify([
+        {"item_id": ids[i], "x": float(coords[i][0]), "y": float(coords[i][1])}
+        for i in range(min(limit, len(ids)))])
+
+
+@bp.post("/api/order_playlist")
+@require_auth
+def order():
+    body = request.get_json(force=True, silent=True) or {}
+    ids = body.get("item_ids", [])
+    state = _state()
+    tracks = []
+    for i in ids:
+        meta = state.meta_fn(i) or {}
+        tracks.append({"item_id": i, **{k: meta.get(k) for k in
+                                        ("tempo", "energy", "key", "scale")}})
+    ordered = order_playlist(tracks, energy_arc=bool(body.get("energy_arc")))
+    return jsonify([t["item_id"] for t in ordered])

@@ -1,0 +1,208 @@
+"""Task-control blueprints.
+
+Reference: task control routes (app.py:415-768), queue-state blueprints
+(app_analysis.py / app_clustering.py / app_cron.py), admission gate
+(app_helper.admit_and_enqueue_main_task :95), recursive cancel (:478).
+"""
+
+from __future__ import annotations
+
+import json
+
+from flask import Blueprint, current_app, jsonify, request
+
+from audiomuse_amd import config as C
+from audiomuse_amd.taskqueue import (PENDING, RUNNING, cancel_task_recursive,
+                                     enqueue, task_row)
+from audiomuse_amd.taskqueue import sql as qsql
+from audiomuse_amd.web.auth import require_auth
+
+bp = Blueprint("tasks", __name__)
+
+MAIN_TASK_TYPES = ("run_analysis", "run_clustering")
+
+
+def _state():
+    return current_app.extensions["audiomuse"]
+
+
+def _admit_and_enqueue(conn, task_type: str, payload: dict) -> tuple:
+    """Admission gate: one live main task of a type at a time
+    (app_helper.py:95-131)."""
+    live = conn.execute(
+        "SELECT task_id FROM task_status WHERE task_type=? AND status IN (?,?)",
+        (task_type, PENDING, RUNNING)).fetchone()
+    if live is not None:
+        return None, live["task_id"]
+    return enqueue(conn, task_type, payload, queue="high"), None
+
+
+@bp.post("/api/analysis/start")
+@require_auth
+def analysis_start():
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    tid, existing = _admit_and_enqueue(conn, "run_analysis", {
+        "server_type": body.get("server_type", "synthetic"),
+        "server_config": body.get("server_config", {}),
+        "server_id": body.get("server_id", "default"),
+        "album_limit": int(body.get("album_limit", 0)),
+    })
+    if tid is None:
+        return jsonify({"error": "analysis already running",
+                        "task_id": existing}), 409
+    return jsonify({"task_id": tid}), 202
+
+
+@bp.post("/api/clustering/start")
+@require_auth
+def clustering_start():
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    tid, existing = _admit_and_enqueue(conn, "run_clustering", {
+        "algorithm": body.get("algorithm", C.CLUSTER_ALGORITHM),
+        "runs": int(body.get("runs", C.CLUSTERING_RUNS)),
+        "server_id": body.get("server_id", "default"),
+    })
+    if tid is None:
+        return jsonify({"error": "clustering already running",
+                        "task_id": existing}), 409
+    return jsonify({"task_id": tid}), 202
+
+
+@bp.post("/api/index/rebuild")
+@require_auth
+def index_rebuild():
+    conn = _state().conn()
+    tid = enqueue(conn, "rebuild_indexes", {}, queue="high")
+    return jsonify({"task_id": tid}), 202
+
+
+@bp.get("/api/task/<task_id>")
+@require_auth
+def task_status(task_id: str):
+    row = task_row(_state().conn(), task_id)
+    if row is None:
+        return jsonify({"error": "unknown task"}), 404
+    d = dict(row)
+    for k in ("payload", "result"):
+        try:
+            d[k] = json.loads(d[k]) if d[k] else None
+        except Exception:
+            pass
+    return jsonify(d)
+
+
+@bp.post("/api/task/<task_id>/cancel")
+@require_auth
+def task_cancel(task_id: str):
+    n = cancel_task_recursive(_state().conn(), task_id)
+    return jsonify({"cancelled": n})
+
+
+@bp.get("/api/active_tasks")
+@require_auth
+def active_tasks():
+    """reference: app.py:768"""
+    rows = _state().conn().execute(
+        """SELECT task_id, task_type, status, progress, details, queue,
+               parent_task_id, created_at FROM task_status
+           WHERE status IN (?, ?) ORDER BY created_at DESC LIMIT 200""",
+        (PENDING, RUNNING)).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.get("/api/queue/stats")
+@require_auth
+def queue_stats():
+    return jsonify(qsql.counts_by_status(_state().conn()))
+
+
+# -- music server registry (reference: app_music_servers.py) ---------------
+
+@bp.get("/api/servers")
+@require_auth
+def list_servers():
+    rows = _state().conn().execute(
+        "SELECT server_id, server_type, base_url, username, enabled "
+        "FROM music_servers").fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.post("/api/servers")
+@require_auth
+def add_server():
+    from audiomuse_amd.db import write_txn
+    from audiomuse_amd.mediaserver import provider_types
+
+    body = request.get_json(force=True, silent=True) or {}
+    stype = body.get("server_type", "")
+    if stype not in provider_types():
+        return jsonify({"error": f"unsupported type {stype!r}",
+                        "supported": provider_types()}), 400
+    conn = _state().conn()
+    with write_txn(conn):
+        conn.execute(
+            """INSERT INTO music_servers (server_id, server_type, base_url,
+                   username, credential, config)
+               VALUES (?,?,?,?,?,?)
+               ON CONFLICT(server_id) DO UPDATE SET
+                   server_type=excluded.server_type,
+                   base_url=excluded.base_url, username=excluded.username,
+                   credential=excluded.credential, config=excluded.config""",
+            (body.get("server_id", "default"), stype,
+             body.get("base_url", ""), body.get("username", ""),
+             body.get("credential", ""),
+             json.dumps(body.get("config", {}))))
+    return jsonify({"ok": True})
+
+
+@bp.delete("/api/servers/<server_id>")
+@require_auth
+def delete_server(server_id: str):
+    from audiomuse_amd.db import write_txn
+
+    conn = _state().conn()
+    with write_txn(conn):
+        cur = conn.execute("DELETE FROM music_servers WHERE server_id=?",
+                           (server_id,))
+    return jsonify({"deleted": cur.rowcount})
+
+
+# -- cron (reference: app_cron.py minute-claimed scheduler) -----------------
+
+@bp.get("/api/cron")
+@require_auth
+def cron_list():
+    rows = _state().conn().execute(
+        "SELECT id, name, schedule, task_type, payload, enabled FROM cron"
+    ).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.post("/api/cron")
+@require_auth
+def cron_add():
+    from audiomuse_amd.db import write_txn
+
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    with write_txn(conn):
+        cur = conn.execute(
+            "INSERT INTO cron (name, schedule, task_type, payload, enabled) "
+            "VALUES (?,?,?,?,1)",
+            (body.get("name", ""), body.get("schedule", "0 3 * * *"),
+             body.get("task_type", "rebuild_indexes"),
+             json.dumps(body.get("payload", {}))))
+    return jsonify({"id": cur.lastrowid})
+
+
+@bp.delete("/api/cron/<int:cron_id>")
+@require_auth
+def cron_delete(cron_id: int):
+    from audiomuse_amd.db import write_txn
+
+    conn = _state().conn()
+    with write_txn(conn):
+        cur = conn.execute("DELETE FROM cron WHERE id=?", (cron_id,))
+    return jsonify({"deleted": cur.rowcount})

@@ -1,0 +1,148 @@
+"""Flask application factory.
+
+Reference: /root/reference/app.py (1409 LoC) + 26 app_*.py blueprints —
+blueprint wiring, health, task status/cancel, startup threads
+(index-reload listener, cron loop). Here: create_app() wires the
+blueprints over the SQLite storage; the index cache reloads engines
+when ivf_dir.updated_at changes (the LISTEN/NOTIFY analog on SQLite).
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+import time
+from typing import Dict, Optional
+
+import numpy as np
+from flask import Flask, jsonify
+
+from audiomuse_amd import config as C
+from audiomuse_amd.analysis import index as idx
+from audiomuse_amd.db import connect
+from audiomuse_amd.db.schema import init_db
+
+
+class AppState:
+    """Per-app storage + cached engines (reference: the module-level index
+    caches reloaded by listen_for_index_reloads, app.py:971-1076)."""
+
+    def __init__(self, db_url: str, device: str = "cpu"):
+        self.db_url = db_url
+        self.device = device
+        self._local = threading.local()
+        self._engines: Dict[str, object] = {}
+        self._stamps: Dict[str, float] = {}
+        self._lock = threading.Lock()
+
+    def conn(self):
+        conn = getattr(self._local, "conn", None)
+        if conn is None:
+            conn = self._local.conn = connect(self.db_url)
+            init_db(conn)
+        return conn
+
+    def meta_fn(self, item_id: str) -> Optional[dict]:
+        row = self.conn().execute(
+            "SELECT title, author, mood_vector, other_features, tempo, energy,"
+            " key, scale, duration FROM score WHERE item_id=?",
+            (item_id,)).fetchone()
+        if row is None:
+            return None
+        d = dict(row)
+        for k in ("mood_vector", "other_features"):
+            try:
+                d[k] = json.loads(d[k]) if d[k] else {}
+            except Exception:
+                d[k] = {}
+        return d
+
+    def _stamp(self, name: str) -> Optional[float]:
+        row = self.conn().execute(
+            "SELECT updated_at FROM ivf_dir WHERE index_name=?",
+            (name,)).fetchone()
+        return float(row["updated_at"]) if row else None
+
+    def engine(self, name: str):
+        """Cached engine for an index blob; reloads when the stored stamp
+        changes."""
+        stamp = self._stamp(name)
+        if stamp is None:
+            return None
+        with self._lock:
+            if self._stamps.get(name) == stamp and name in self._engines:
+                return self._engines[name]
+        if name == idx.ARTIST_INDEX:
+            eng = idx.load_artist_similarity(self.conn())
+        elif name == idx.SONG_MAP:
+            got = None
+            from audiomuse_amd.db.store import load_index_blob
+            got = load_index_blob(self.conn(), name)
+            if got is None:
+                return None
+            import io
+            import torch
+            eng = torch.load(io.BytesIO(got[0]), map_location="cpu",
+                             weights_only=False)  # our own trusted blob
+        else:
+            eng = idx.load_ivf_engine(self.conn(), name, device=self.device,
+                                      meta_fn=self.meta_fn)
+        with self._lock:
+            self._engines[name] = eng
+            self._stamps[name] = stamp
+        return eng
+
+    def invalidate(self) -> None:
+        with self._lock:
+            self._engines.clear()
+            self._stamps.clear()
+
+
+def create_app(db_url: Optional[str] = None, device: str = "cpu",
+               auth_disabled: bool = False) -> Flask:
+    app = Flask("audiomuse_amd")
+    state = AppState(db_url or C.DATABASE_URL, device=device)
+    app.extensions["audiomuse"] = state
+    app.config["AUTH_DISABLED"] = auth_disabled
+
+    from audiomuse_amd.web.api_auth import bp as auth_bp
+    from audiomuse_amd.web.api_queries import bp as queries_bp
+    from audiomuse_amd.web.api_tasks import bp as tasks_bp
+    from audiomuse_amd.web.api_chat import bp as chat_bp
+
+    app.register_blueprint(auth_bp)
+    app.register_blueprint(queries_bp)
+    app.register_blueprint(tasks_bp)
+    app.register_blueprint(chat_bp)
+
+    from audiomuse_amd.web.auth import seed_admin_from_env
+
+    with app.app_context():
+        seed_admin_from_env(state.conn())
+
+    @app.get("/health")
+    def health():  # reference: app.py:227
+        try:
+            state.conn().execute("SELECT 1")
+            return jsonify({"status": "ok"})
+        except Exception as exc:  # noqa: BLE001
+            return jsonify({"status": "error", "detail": str(exc)}), 500
+
+    @app.get("/api/index_profile")
+    def index_profile():  # reference: _log_startup_index_profile app.py:1244
+        conn = state.conn()
+        rows = conn.execute(
+            "SELECT index_name, meta, n_parts, updated_at FROM ivf_dir"
+        ).fetchall()
+        out = []
+        for r in rows:
+            size = conn.execute(
+                "SELECT COALESCE(SUM(LENGTH(blob)), 0) AS b FROM ivf_cell "
+                "WHERE index_name=?", (r["index_name"],)).fetchone()
+            out.append({"name": r["index_name"],
+                        "meta": json.loads(r["meta"]),
+                        "bytes": int(size["b"]),
+                        "updated_at": r["updated_at"]})
+        return jsonify(out)
+
+    return app

@@ -1,0 +1,205 @@
+"""Web API tests: auth barrier, task control, query endpoints over a
+seeded catalogue (Flask test client)."""
+
+import json
+
+import numpy as np
+import pytest
+
+from audiomuse_amd.analysis.index import run_all_index_builds
+from audiomuse_amd.db import connect
+from audiomuse_amd.db.schema import init_db
+from audiomuse_amd.db.store import (save_clap_embedding,
+                                    save_lyrics_embedding,
+                                    save_track_analysis_and_embedding)
+from audiomuse_amd.web.app import create_app
+
+
+def _seed(conn, n=40):
+    rng = np.random.default_rng(0)
+    ids = []
+    for i in range(n):
+        iid = f"fp_4{'%050x' % i}"
+        ids.append(iid)
+        save_track_analysis_and_embedding(
+            conn, iid, title=f"Song {i}", author=f"Artist {i % 6}",
+            album=f"Album {i % 8}", tempo=90 + i, energy=(i % 10) / 10,
+            key="C", scale="major", duration=180.0,
+            mood_vector={"rock": (i % 3) / 2, "jazz": ((i + 1) % 3) / 2},
+            other_features={"happy": 0.5},
+            embedding=rng.standard_normal(200).astype(np.float32))
+        save_clap_embedding(conn, iid,
+                            rng.standard_normal(512).astype(np.float32))
+        if i % 2 == 0:
+            save_lyrics_embedding(conn, iid,
+                                  rng.standard_normal(768).astype(np.float32),
+                                  axis_scores={"love": 0.4})
+    run_all_index_builds(conn)
+    return ids
+
+
+@pytest.fixture(scope="module")
+def client_ids(tmp_path_factory):
+    url = "sqlite:///" + str(tmp_path_factory.mktemp("web") / "web.db")
+    conn = connect(url)
+    init_db(conn)
+    ids = _seed(conn)
+    app = create_app(url, auth_disabled=True)
+    app.testing = True
+    with app.test_client() as client:
+        yield client, ids
+    conn.close()
+
+
+def test_health(client_ids):
+    client, _ = client_ids
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json["status"] == "ok"
+
+
+def test_similar_tracks(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/similar_tracks?item_id={ids[0]}&n=5")
+    assert r.status_code == 200
+    body = r.json
+    assert len(body) == 5
+    assert all(b["item_id"] != ids[0] for b in body)
+    assert all("title" in b and "author" in b for b in body)
+    # unknown id -> 404
+    assert client.get("/api/similar_tracks?item_id=nope").status_code == 404
+
+
+def test_search_tracks(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/search_tracks?q=song 1")
+    assert r.status_code == 200 and len(r.json) >= 1
+
+
+def test_path(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/path?start={ids[0]}&end={ids[30]}&length=6")
+    assert r.status_code == 200
+    body = r.json
+    assert body[0]["item_id"] == ids[0] and body[-1]["item_id"] == ids[30]
+
+
+def test_alchemy(client_ids):
+    client, ids = client_ids
+    r = client.post("/api/alchemy", json={"add": [ids[1]], "n": 6})
+    assert r.status_code == 200 and len(r.json) >= 1
+
+
+def test_artist_similarity(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/artist_similarity?artist=Artist 1&n=3")
+    assert r.status_code == 200
+    assert len(r.json) == 3
+    assert client.get("/api/artist_similarity?artist=Unknown").status_code == 404
+
+
+def test_text_searches(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/clap_search?q=upbeat dance music&n=4")
+    assert r.status_code == 200 and len(r.json) == 4
+    r = client.get("/api/lyrics_search?q=love and heartbreak&n=4")
+    assert r.status_code == 200 and len(r.json) == 4
+
+
+def test_semgrove_and_hyperbolic(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/semgrove?item_id={ids[0]}&n=4")
+    assert r.status_code == 200 and len(r.json) >= 1
+    r = client.get(f"/api/hyperbolic_similar?item_id={ids[0]}&n=4")
+    assert r.status_code == 200 and len(r.json) == 4
+
+
+def test_sonic_fingerprint_and_map(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/sonic_fingerprint?item_id={ids[0]}&item_id={ids[1]}&n=5")
+    assert r.status_code == 200 and len(r.json) == 5
+    r = client.get("/api/map?n=10")
+    assert r.status_code == 200 and len(r.json) == 10
+    assert {"item_id", "x", "y"} <= set(r.json[0])
+
+
+def test_order_playlist_endpoint(client_ids):
+    client, ids = client_ids
+    r = client.post("/api/order_playlist", json={"item_ids": ids[:5]})
+    assert r.status_code == 200 and sorted(r.json) == sorted(ids[:5])
+
+
+def test_task_control_flow(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/analysis/start", json={
+        "server_type": "synthetic",
+        "server_config": {"n_albums": 1, "tracks_per_album": 1}})
+    assert r.status_code == 202
+    tid = r.json["task_id"]
+    # admission gate: second start conflicts
+    r2 = client.post("/api/analysis/start", json={"server_type": "synthetic"})
+    assert r2.status_code == 409
+    # status + active list
+    assert client.get(f"/api/task/{tid}").json["status"] == "PENDING"
+    active = client.get("/api/active_tasks").json
+    assert any(t["task_id"] == tid for t in active)
+    # cancel recursively
+    r3 = client.post(f"/api/task/{tid}/cancel")
+    assert r3.json["cancelled"] == 1
+    assert client.get(f"/api/task/{tid}").json["status"] == "REVOKED"
+
+
+def test_chat_playlist_offline_planner(client_ids):
+    client, _ = client_ids
+    r = client.post("/chat/api/chatPlaylist",
+                    json={"prompt": "15 songs of happy rock by Artist 1"})
+    assert r.status_code == 200
+    body = r.json
+    assert body["hints"]["n"] == 15
+    assert "rock" in body["hints"]["moods"]
+    assert body["plan"] and all(c["tool"] in
+                                ("seed_search", "text_match",
+                                 "search_database", "knowledge_lookup")
+                                for c in body["plan"])
+    assert body["tracks"] and len(body["tracks"]) <= 15
+
+
+def test_server_registry_crud(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/servers", json={"server_id": "s1",
+                                          "server_type": "synthetic"})
+    assert r.status_code == 200
+    assert any(s["server_id"] == "s1" for s in client.get("/api/servers").json)
+    assert client.post("/api/servers",
+                       json={"server_type": "bogus"}).status_code == 400
+    assert client.delete("/api/servers/s1").json["deleted"] == 1
+
+
+def test_auth_barrier():
+    """Without auth_disabled: setup barrier, then login flow."""
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        url = f"sqlite:///{td}/auth.db"
+        conn = connect(url)
+        init_db(conn)
+        app = create_app(url, auth_disabled=False)
+        app.testing = True
+        with app.test_client() as client:
+            # setup needed -> 403 on protected endpoints
+            assert client.get("/api/active_tasks").status_code == 403
+            assert client.get("/api/setup/status").json["setup_needed"]
+            r = client.post("/api/setup/admin",
+                            json={"username": "admin",
+                                  "password": "longenough1"})
+            assert r.status_code == 200
+            # wrong password
+            assert client.post("/api/login",
+                               json={"username": "admin",
+                                     "password": "bad"}).status_code == 401
+            # login sets cookie; protected endpoint works
+            r = client.post("/api/login", json={"username": "admin",
+                                                "password": "longenough1"})
+            assert r.status_code == 200
+            assert client.get("/api/active_tasks").status_code == 200
+            assert client.get("/api/me").json["user"] == "admin"
+        conn.close()
