@@ -1,0 +1,112 @@
+"""2-D map projection: first-party UMAP-style layout on torch.
+
+Reference capability: /root/reference/alchemy_projections.py:78
+(_project_with_umap; umap-learn with PCA fallback) feeding the Music Map
+(app_map.py) and alchemy projections. umap-learn is not in this image;
+this is our own implementation of the same construction — exact kNN
+graph (chunked GEMM distances, GPU-capable), smooth-kNN fuzzy weights,
+PCA init, and the standard UMAP attract/repel SGD with negative
+sampling — vectorized over all edges per epoch on the device.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from audiomuse_amd.cluster.algorithms import _pairwise_sq, pca_fit_transform
+
+# curve parameters for min_dist=0.1 (standard UMAP fit)
+_A, _B = 1.577, 0.895
+
+
+def knn_graph(x: torch.Tensor, k: int, chunk: int = 4096
+              ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(dists (n,k), idx (n,k)) excluding self; euclidean."""
+    n = x.shape[0]
+    dists = torch.empty(n, k, device=x.device)
+    idx = torch.empty(n, k, dtype=torch.long, device=x.device)
+    for s in range(0, n, chunk):
+        d = _pairwise_sq(x[s : s + chunk], x)
+        d[torch.arange(d.shape[0]), torch.arange(s, s + d.shape[0])] = float("inf")
+        top = torch.topk(d, k, dim=1, largest=False)
+        dists[s : s + chunk] = top.values.clamp(min=0).sqrt()
+        idx[s : s + chunk] = top.indices
+    return dists, idx
+
+
+def smooth_knn_weights(dists: torch.Tensor, n_iter: int = 32) -> torch.Tensor:
+    """Per-point sigma via bisection so sum_j exp(-(d_j - rho)/sigma) =
+    log2(k) (UMAP smooth_knn_dist)."""
+    k = dists.shape[1]
+    target = math.log2(k)
+    rho = dists[:, 0:1]
+    lo = torch.full_like(rho, 1e-6)
+    hi = torch.full_like(rho, 1e3)
+    for _ in range(n_iter):
+        mid = (lo + hi) / 2
+        val = torch.exp(-(dists - rho).clamp(min=0) / mid).sum(dim=1, keepdim=True)
+        hi = torch.where(val > target, mid, hi)
+        lo = torch.where(val <= target, mid, lo)
+    sigma = (lo + hi) / 2
+    return torch.exp(-(dists - rho).clamp(min=0) / sigma)
+
+
+def umap_project(x: torch.Tensor, n_neighbors: int = 15, epochs: int = 200,
+                 lr: float = 1.0, neg_samples: int = 5,
+                 seed: int = 0) -> torch.Tensor:
+    """(n, d) -> (n, 2) layout."""
+    x = x.float()
+    n = x.shape[0]
+    if n <= 3:
+        return torch.zeros(n, 2, device=x.device)
+    k = min(n_neighbors, n - 1)
+    dists, idx = knn_graph(x, k)
+    w = smooth_knn_weights(dists)
+
+    # symmetrize: treat (i -> idx[i,j]) directed weights; w_sym = a+b-ab
+    rows = torch.arange(n, device=x.device).unsqueeze(1).expand(-1, k).reshape(-1)
+    cols = idx.reshape(-1)
+    vals = w.reshape(-1)
+    key = rows * n + cols
+    rkey = cols * n + rows
+    lookup = {}
+    keyl = key.tolist()
+    vall = vals.tolist()
+    for kk, vv in zip(keyl, vall):
+        lookup[kk] = vv
+    rvals = torch.tensor([lookup.get(int(rk), 0.0) for rk in rkey.tolist()],
+                         device=x.device)
+    wsym = vals + rvals - vals * rvals
+
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    init, _, _ = pca_fit_transform(x, 2)
+    emb = (init / (init.std() + 1e-9)).contiguous() * 10.0
+    emb += torch.randn(emb.shape, generator=g).to(x.device) * 0.01
+
+    edge_i, edge_j, edge_w = rows, cols, wsym / wsym.max().clamp(min=1e-12)
+    for epoch in range(epochs):
+        alpha = lr * (1.0 - epoch / epochs)
+        keep = torch.rand(edge_w.shape[0], generator=g).to(x.device) <= edge_w
+        ei, ej = edge_i[keep], edge_j[keep]
+        if ei.numel() == 0:
+            continue
+        # attraction
+        delta = emb[ei] - emb[ej]
+        d2 = delta.square().sum(dim=1, keepdim=True)
+        grad_coef = (-2.0 * _A * _B * d2.clamp(min=1e-12) ** (_B - 1)) / \
+                    (1.0 + _A * d2 ** _B)
+        grad = (grad_coef * delta).clamp(-4.0, 4.0)
+        emb.index_add_(0, ei, grad * alpha)
+        emb.index_add_(0, ej, -grad * alpha)
+        # repulsion: negative samples
+        for _ in range(neg_samples):
+            nj = torch.randint(0, n, (ei.shape[0],), generator=g).to(x.device)
+            delta = emb[ei] - emb[nj]
+            d2 = delta.square().sum(dim=1, keepdim=True)
+            grad_coef = (2.0 * _B) / ((0.001 + d2) * (1.0 + _A * d2 ** _B))
+            grad = (grad_coef * delta).clamp(-4.0, 4.0)
+            emb.index_add_(0, ei, grad * alpha)
+    return emb
