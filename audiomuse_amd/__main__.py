@@ -1,0 +1,68 @@
+"""CLI entry points: `python -m audiomuse_amd <command>`.
+
+Deployment shape (reference: deployment/docker-entrypoint.sh
+SERVICE_TYPE=flask|worker; supervisord runs gunicorn + queue workers):
+- web    : Flask app (dev server here; any WSGI server in production)
+- worker : queue worker loop (one process per GPU rank)
+- analyze: enqueue a full analysis run against a configured server
+- bench  : shortcut to the flagship benchmark
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser(prog="audiomuse_amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    web = sub.add_parser("web")
+    web.add_argument("--host", default="0.0.0.0")
+    web.add_argument("--port", type=int, default=8000)
+    web.add_argument("--db", default=None)
+    web.add_argument("--no-auth", action="store_true")
+
+    wk = sub.add_parser("worker")
+    wk.add_argument("--db", default=None)
+    wk.add_argument("--queues", default="high,default")
+    wk.add_argument("--max-jobs", type=int, default=None)
+
+    an = sub.add_parser("analyze")
+    an.add_argument("--db", default=None)
+    an.add_argument("--server-type", default="synthetic")
+    an.add_argument("--server-id", default="default")
+
+    sub.add_parser("bench")
+
+    args, rest = ap.parse_known_args()
+
+    if args.cmd == "web":
+        from audiomuse_amd.web.app import create_app
+
+        app = create_app(args.db, auth_disabled=args.no_auth)
+        app.run(host=args.host, port=args.port)
+    elif args.cmd == "worker":
+        from audiomuse_amd.taskqueue.worker import Worker
+
+        Worker(db_url=args.db, queues=tuple(args.queues.split(",")),
+               max_jobs=args.max_jobs).run_forever()
+    elif args.cmd == "analyze":
+        from audiomuse_amd.db import get_db
+        from audiomuse_amd.taskqueue import enqueue
+
+        conn = get_db(args.db)
+        tid = enqueue(conn, "run_analysis",
+                      {"server_type": args.server_type,
+                       "server_id": args.server_id}, queue="high")
+        print(f"enqueued analysis task {tid}")
+    elif args.cmd == "bench":
+        sys.argv = [sys.argv[0]] + rest
+        import runpy
+
+        runpy.run_path("bench.py", run_name="__main__")
+
+
+if __name__ == "__main__":
+    main()

@@ -43,6 +43,12 @@ def get_handler(task_type: str) -> Optional[Callable]:
     return _REGISTRY.get(task_type)
 
 
+def import_builtin_handlers() -> None:
+    """Register the built-in task modules (idempotent)."""
+    import audiomuse_amd.analysis.tasks  # noqa: F401
+    import audiomuse_amd.cluster.tasks  # noqa: F401
+
+
 class CancelledError(RuntimeError):
     pass
 
@@ -134,6 +140,10 @@ class Worker:
         conn = connect(self.db_url)
         from audiomuse_amd.db.schema import init_db
         init_db(conn)
+        try:
+            import_builtin_handlers()
+        except Exception:
+            logger.exception("builtin handler import failed")
         idle_since = time.time()
         last_maintenance = 0.0
         try:
