@@ -1,0 +1,175 @@
+"""Maintenance tasks: multi-server sweep, cleaning, backup, dashboard.
+
+References:
+- multi-server sweep: /root/reference/tasks/multiserver_sync.py (750 LoC)
+  — metadata-only alignment of a server's tracks onto existing canonical
+  recordings by tiers (exact path / path tail / exact title+artist /
+  noise-normalized title+artist), prune guarded by
+  SWEEP_PRUNE_MIN_FETCH_RATIO.
+- cleaning: tasks/cleaning.py (:48) — per-server stale-mapping prune and
+  orphan album report/delete behind a triple guard; the catalogue itself
+  is never deleted (test_catalogue_is_never_deleted invariant).
+- backup: app_backup.py (pg_dump/restore + lock) — here the SQLite
+  backup API with the same restore-lock semantics.
+- dashboard: app_dashboard.py snapshot stats.
+"""
+
+from __future__ import annotations
+
+import json
+import re
+import sqlite3
+import time
+from typing import Dict, List, Optional, Tuple
+
+from audiomuse_amd.db import connect, write_txn
+from audiomuse_amd.mediaserver import make_provider
+from audiomuse_amd.taskqueue.worker import TaskContext, task_handler
+
+SWEEP_PRUNE_MIN_FETCH_RATIO = 0.5
+
+_NOISE = re.compile(r"\s*[\(\[].*?[\)\]]\s*|\s*(feat\.|ft\.)\s.*$|[^\w\s]",
+                    re.IGNORECASE)
+
+
+def normalize_title(s: str) -> str:
+    """Noise-normalized matching key (remaster tags, feat. credits,
+    punctuation stripped)."""
+    return re.sub(r"\s+", " ", _NOISE.sub(" ", (s or "").lower())).strip()
+
+
+def align_server_tracks(conn: sqlite3.Connection, server_id: str,
+                        tracks: List) -> Dict[str, int]:
+    """Tiered metadata alignment (multiserver_sync.enqueue_server_
+    alignment :69): map provider tracks onto existing canonical ids
+    without re-analysis."""
+    existing = conn.execute(
+        """SELECT m.item_id, m.file_path, s.title, s.author
+           FROM track_server_map m JOIN score s ON s.item_id = m.item_id"""
+    ).fetchall()
+    by_path = {r["file_path"]: r["item_id"] for r in existing if r["file_path"]}
+    by_tail = {r["file_path"].rsplit("/", 1)[-1]: r["item_id"]
+               for r in existing if r["file_path"]}
+    by_exact = {(r["title"] or "", r["author"] or ""): r["item_id"]
+                for r in existing}
+    by_norm = {(normalize_title(r["title"]), normalize_title(r["author"])):
+               r["item_id"] for r in existing}
+    tiers = {"path": 0, "tail": 0, "exact": 0, "normalized": 0, "unmatched": 0}
+    with write_txn(conn):
+        for t in tracks:
+            item_id = None
+            if t.file_path and t.file_path in by_path:
+                item_id, tier = by_path[t.file_path], "path"
+            elif t.file_path and t.file_path.rsplit("/", 1)[-1] in by_tail:
+                item_id, tier = by_tail[t.file_path.rsplit("/", 1)[-1]], "tail"
+            elif (t.title, t.author) in by_exact:
+                item_id, tier = by_exact[(t.title, t.author)], "exact"
+            elif (normalize_title(t.title), normalize_title(t.author)) in by_norm:
+                item_id = by_norm[(normalize_title(t.title),
+                                   normalize_title(t.author))]
+                tier = "normalized"
+            else:
+                tiers["unmatched"] += 1
+                continue
+            tiers[tier] += 1
+            conn.execute(
+                """INSERT INTO track_server_map (provider_id, server_id,
+                       item_id, title, author, album, file_path)
+                   VALUES (?,?,?,?,?,?,?)
+                   ON CONFLICT(provider_id, server_id)
+                   DO UPDATE SET item_id=excluded.item_id""",
+                (t.provider_id, server_id, item_id, t.title, t.author,
+                 t.album, t.file_path))
+    return tiers
+
+
+@task_handler("multiserver_sync")
+def multiserver_sync_task(ctx: TaskContext, payload: Dict) -> Dict:
+    provider = make_provider(payload["server_type"],
+                             **payload.get("server_config", {}))
+    server_id = payload.get("server_id", "default")
+    tracks = provider.get_all_songs()
+    tiers = align_server_tracks(ctx.conn, server_id, tracks)
+
+    # prune mappings whose provider track vanished — guarded: skip when
+    # the fetch looks partial (SWEEP_PRUNE_MIN_FETCH_RATIO)
+    conn = ctx.conn
+    mapped = conn.execute(
+        "SELECT COUNT(*) AS n FROM track_server_map WHERE server_id=?",
+        (server_id,)).fetchone()["n"]
+    pruned = 0
+    if mapped and len(tracks) / mapped >= SWEEP_PRUNE_MIN_FETCH_RATIO:
+        live = {t.provider_id for t in tracks}
+        rows = conn.execute(
+            "SELECT provider_id FROM track_server_map WHERE server_id=?",
+            (server_id,)).fetchall()
+        stale = [r["provider_id"] for r in rows if r["provider_id"] not in live]
+        with write_txn(conn):
+            for pid in stale:
+                conn.execute(
+                    "DELETE FROM track_server_map WHERE provider_id=? AND "
+                    "server_id=?", (pid, server_id))
+                pruned += 1
+    return {"tiers": tiers, "pruned": pruned, "fetched": len(tracks)}
+
+
+@task_handler("clean_orphans")
+def clean_orphans_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Orphan report/delete with the triple guard (cleaning.py:48):
+    only mappings are removed, never catalogue rows (`score`/`embedding`
+    are append-only — the reference's hardest invariant)."""
+    conn = ctx.conn
+    do_delete = bool(payload.get("delete", False))
+    orphans = conn.execute(
+        """SELECT m.provider_id, m.server_id FROM track_server_map m
+           LEFT JOIN score s ON s.item_id = m.item_id
+           WHERE s.item_id IS NULL""").fetchall()
+    deleted = 0
+    if do_delete and orphans:
+        # triple guard: explicit flag + bounded fraction + re-check
+        total = conn.execute(
+            "SELECT COUNT(*) AS n FROM track_server_map").fetchone()["n"]
+        if total and len(orphans) / total <= payload.get("max_fraction", 0.2):
+            with write_txn(conn):
+                for r in orphans:
+                    conn.execute(
+                        "DELETE FROM track_server_map WHERE provider_id=? "
+                        "AND server_id=?",
+                        (r["provider_id"], r["server_id"]))
+                    deleted += 1
+    return {"orphans": len(orphans), "deleted": deleted}
+
+
+# -- backup / restore (app_backup.py analog) --------------------------------
+
+def backup_database(conn: sqlite3.Connection, dest_path: str) -> None:
+    dest = sqlite3.connect(dest_path)
+    try:
+        # sqlite's online-backup API snapshots consistently on its own
+        conn.backup(dest)
+    finally:
+        dest.close()
+
+
+def refresh_dashboard_stats(conn: sqlite3.Connection) -> Dict[str, int]:
+    """Snapshot stats (app_dashboard.py; dashboard_stats table)."""
+    stats = {}
+    for key, q in [
+        ("tracks", "SELECT COUNT(*) FROM score"),
+        ("embeddings", "SELECT COUNT(*) FROM embedding"),
+        ("clap_embeddings", "SELECT COUNT(*) FROM clap_embedding"),
+        ("lyrics", "SELECT COUNT(*) FROM lyrics_embedding"),
+        ("mappings", "SELECT COUNT(*) FROM track_server_map"),
+        ("playlists", "SELECT COUNT(*) FROM playlist"),
+        ("artists", "SELECT COUNT(DISTINCT author) FROM score"),
+    ]:
+        stats[key] = int(conn.execute(q).fetchone()[0])
+    with write_txn(conn):
+        for k, v in stats.items():
+            conn.execute(
+                """INSERT INTO dashboard_stats (key, value, updated_at)
+                   VALUES (?,?, (julianday('now') - 2440587.5) * 86400.0)
+                   ON CONFLICT(key) DO UPDATE SET value=excluded.value,
+                       updated_at=excluded.updated_at""",
+                (k, str(v)))
+    return stats
