@@ -119,7 +119,7 @@ class SwinBlock(nn.Module):
     def _fused_attn_available(self, x: torch.Tensor) -> bool:
         if not (x.is_cuda and x.dtype == torch.bfloat16
                 and not torch.is_grad_enabled() and self.window == 8
-                and self.attn.heads % 4 == 0
+                and self.attn.heads % 2 == 0
                 and self.attn.dim // self.attn.heads == 32):
             return False
         from audiomuse_amd.ops import _ext

@@ -75,11 +75,15 @@ void launch_mfma_probe(const void* A, const void* B, float* D,
 // ---------------------------------------------------------------------------
 
 // per-wave LDS: VT[32][64+8] bf16 rows padded to 144 B; P[64][64+8] bf16.
+// 2 waves per block: 27.6 KiB LDS -> 5 blocks/CU (10 waves/CU) vs 2
+// blocks at 4 waves (8/CU) — this kernel is gather-latency bound, so
+// occupancy is the lever (measured: profiles/r01_bench_clap.md).
+constexpr int ATTN_WAVES = 2;
 constexpr int VT_STRIDE = 72;   // bf16 elements per VT row (64 + 8 pad)
 constexpr int P_STRIDE = 72;    // bf16 elements per P row
 constexpr int WAVE_LDS_HALF = 32 * VT_STRIDE + 64 * P_STRIDE;  // bf16 elems
 
-__global__ __launch_bounds__(256) void window_attn_kernel(
+__global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
     const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
     __bf16* __restrict__ out,        // (B, H, W, C)
     const float* __restrict__ bias,          // (heads, 64, 64)
@@ -120,7 +124,7 @@ __global__ __launch_bounds__(256) void window_attn_kernel(
   const unsigned long long wrap_r_mask = __ballot(my_wrap & 2);
   const unsigned long long wrap_c_mask = __ballot(my_wrap & 1);
 
-  for (int h = wave; h < heads; h += 4) {
+  for (int h = wave; h < heads; h += ATTN_WAVES) {
     // ---- stage V transposed: VT[d][t] = V[t][d] ----
     {
       const __bf16* vptr = qkv + my_base + 2 * C + h * 32;
@@ -270,8 +274,9 @@ void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream) {
   const int n_windows = Bn * (H >> 3) * (W >> 3);
-  const size_t lds_bytes = 4 * WAVE_LDS_HALF * sizeof(__bf16);
-  hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows), dim3(256), lds_bytes,
+  const size_t lds_bytes = ATTN_WAVES * WAVE_LDS_HALF * sizeof(__bf16);
+  hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows),
+                     dim3(64 * ATTN_WAVES), lds_bytes,
                      stream, (const __bf16*)qkv, (__bf16*)out,
                      bias, Bn, H, W, C, heads, shift, scale);
 }

@@ -146,7 +146,7 @@ static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
   AM_CHECK(qkv.size(3) == 3 * C && C == heads * 32,
            "C must be heads*32 and last dim 3C");
   AM_CHECK(H % 8 == 0 && W % 8 == 0, "H, W must be multiples of 8");
-  AM_CHECK(heads % 4 == 0, "heads must be a multiple of 4");
+  AM_CHECK(heads % 2 == 0, "heads must be a multiple of 2");
   AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kFloat &&
                bias.is_contiguous() && bias.numel() == heads * 64 * 64,
            "bias must be (heads, 64, 64) f32 contiguous");

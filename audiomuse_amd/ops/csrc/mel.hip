@@ -3,20 +3,19 @@
 // Replaces the reference's librosa.feature.melspectrogram call sites
 // (/root/reference/tasks/clap_analyzer.py:394-430 CLAP shape 48k/2048/480/128;
 //  /root/reference/tasks/analysis/song.py:240-256 MusiCNN 16k/512/256/96)
-// with one kernel launch per batch: each workgroup computes one frame's
-// windowed 2^k-point FFT entirely in LDS (radix-2 DIT, twiddle table in
-// global/L2), the power spectrum, the sparse (CSR) slaney mel projection,
-// and the log compression, writing only the (n_mels) outputs to HBM.
+// with one kernel launch per batch: each workgroup computes TWO frames'
+// windowed 2^k-point FFT in LDS via the real-pair trick (frames packed as
+// re/im of one complex FFT, split by conjugate symmetry), the power
+// spectra, the sparse (CSR) slaney mel projection, and the log
+// compression, writing only the (n_mels) outputs per frame to HBM.
 //
 // Design notes (see /opt/skills/guides/cdna_hip_programming.md):
-// - block = 256 threads (4 waves); LDS = NFFT float2 + (NFFT/2+1) float
-//   (20 KiB at NFFT=2048) -> 8 blocks/CU, wave-capacity bound, good TLP.
-// - grid = n_frames x B  (for B=256 CLAP segments: 256k workgroups >> 256 CUs).
-// - All data stays in LDS between phases; HBM traffic is n_fft reads +
-//   n_mels writes per frame (the power spectrum is never materialized).
-// - fp32 throughout: the front-end feeds catalogue identity (simhash), so
-//   numeric fidelity vs the librosa reference matters more than speed here;
-//   the FFT work is ~0.2 GFLOP per 10 s clip, far from the bottleneck.
+// - block = 256 threads (4 waves); LDS = NFFT float2 + 2*(NFFT/2+1) float
+//   (24.4 KiB at NFFT=2048) -> LDS allows 6 blocks/CU (wave-capacity 8).
+// - grid = ceil(n_frames/2) x B: B=256 CLAP segments -> 128k workgroups.
+// - fp32 throughout: the front-end feeds catalogue identity (simhash);
+//   numeric fidelity vs the librosa reference matters.
+// - optional fused int16 round-trip on load (clap_analyzer.py:453-455).
 
 #include <hip/hip_runtime.h>
 
@@ -34,32 +33,40 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     int T, int n_frames, int hop, int n_mels, int center, int log_mode,
     int quant16) {
   __shared__ float2 zbuf[NFFT];
-  __shared__ float pw[NFFT / 2 + 1];
+  __shared__ float pw[2][NFFT / 2 + 1];
 
-  const int f = blockIdx.x;
+  const int f0 = blockIdx.x * 2;       // this block's frame pair
   const int b = blockIdx.y;
   const int tid = threadIdx.x;
-  if (f >= n_frames) return;
+  if (f0 >= n_frames) return;
+  const bool has_f1 = (f0 + 1) < n_frames;
 
   const float* src = audio + (long long)b * T;
-  const int start = f * hop - (center ? NFFT / 2 : 0);
+  const int start0 = f0 * hop - (center ? NFFT / 2 : 0);
 
-  // Phase 1: load + reflect-pad + window + bit-reverse scatter into LDS.
+  // Phase 1: load both frames (reflect pad + optional int16 round-trip +
+  // window), packed re/im, bit-reverse scatter into LDS.
   for (int i = tid; i < NFFT; i += blockDim.x) {
-    int g = start + i;
-    if (g < 0) g = -g;                    // librosa reflect (no edge repeat)
-    if (g >= T) g = 2 * (T - 1) - g;
-    g = max(0, min(T - 1, g));            // safety for tiny T
-    float v = src[g];
-    if (quant16) {
-      // fused int16 round-trip (clap_analyzer.py:453-455): clip +-1,
-      // numpy int16 cast truncates toward zero
-      v = fminf(1.0f, fmaxf(-1.0f, v));
-      v = (float)(int)(v * 32767.0f) / 32767.0f;
+    float v[2];
+#pragma unroll
+    for (int fr = 0; fr < 2; ++fr) {
+      int g = start0 + fr * hop + i;
+      if (fr == 1 && !has_f1) {
+        v[1] = 0.0f;
+        continue;
+      }
+      if (g < 0) g = -g;                  // librosa reflect (no edge repeat)
+      if (g >= T) g = 2 * (T - 1) - g;
+      g = max(0, min(T - 1, g));          // safety for tiny T
+      float x = src[g];
+      if (quant16) {
+        x = fminf(1.0f, fmaxf(-1.0f, x));
+        x = (float)(int)(x * 32767.0f) / 32767.0f;  // numpy int16 trunc
+      }
+      v[fr] = x * window[i];
     }
-    v *= window[i];
-    int rev = __brev((unsigned)i) >> (32 - LOG2N);
-    zbuf[rev] = make_float2(v, 0.0f);
+    const int rev = __brev((unsigned)i) >> (32 - LOG2N);
+    zbuf[rev] = make_float2(v[0], v[1]);
   }
   __syncthreads();
 
@@ -83,19 +90,29 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     __syncthreads();
   }
 
-  // Phase 3: power spectrum (one-sided).
+  // Phase 3: conjugate-split the packed pair and take power spectra.
+  // frame0[k] = (Z[k] + conj(Z[N-k])) / 2 ; frame1[k] = (Z[k] - conj(Z[N-k])) / 2i
   for (int k = tid; k <= NFFT / 2; k += blockDim.x) {
-    const float2 z = zbuf[k];
-    pw[k] = z.x * z.x + z.y * z.y;
+    const int nk = (NFFT - k) & (NFFT - 1);
+    const float2 zk = zbuf[k];
+    const float2 zn = zbuf[nk];
+    const float ar = 0.5f * (zk.x + zn.x);
+    const float ai = 0.5f * (zk.y - zn.y);
+    const float br = 0.5f * (zk.y + zn.y);
+    const float bi = 0.5f * (zn.x - zk.x);
+    pw[0][k] = ar * ar + ai * ai;
+    pw[1][k] = br * br + bi * bi;
   }
   __syncthreads();
 
-  // Phase 4: sparse mel projection + log, direct to HBM.
-  float* dst = out + ((long long)b * n_mels) * n_frames + f;
-  for (int m = tid; m < n_mels; m += blockDim.x) {
+  // Phase 4: sparse mel projection + log for both frames, direct to HBM.
+  for (int m = tid; m < 2 * n_mels; m += blockDim.x) {
+    const int fr = m >= n_mels ? 1 : 0;
+    if (fr == 1 && !has_f1) continue;
+    const int mm = m - fr * n_mels;
     float acc = 0.0f;
-    const int p0 = mel_rowptr[m], p1 = mel_rowptr[m + 1];
-    for (int p = p0; p < p1; ++p) acc += pw[mel_bin[p]] * mel_w[p];
+    const int p0 = mel_rowptr[mm], p1 = mel_rowptr[mm + 1];
+    for (int p = p0; p < p1; ++p) acc += pw[fr][mel_bin[p]] * mel_w[p];
     float y;
     if (log_mode == 0) {                       // librosa power_to_db, ref=1
       y = 10.0f * log10f(fmaxf(acc, 1e-10f));
@@ -104,7 +121,7 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     } else {                                   // raw power mel
       y = acc;
     }
-    dst[(long long)m * n_frames] = y;
+    out[((long long)b * n_mels + mm) * n_frames + (f0 + fr)] = y;
   }
 }
 
@@ -113,7 +130,7 @@ void launch_mel_fwd(const float* audio, float* out, const float* window,
                     const int* mel_bin, const float* mel_w, int B, int T,
                     int n_frames, int hop, int n_mels, int n_fft, int center,
                     int log_mode, int quant16, hipStream_t stream) {
-  dim3 grid(n_frames, B);
+  dim3 grid((n_frames + 1) / 2, B);
   dim3 block(256);
 #define AM_MEL_CASE(N, L)                                                     \
   case N:                                                                     \
