@@ -1,0 +1,100 @@
+"""Plugin system: zip-packaged extensions with hooks.
+
+Reference: /root/reference/plugin/ (2146 LoC) — third-party plugins as
+zips stored in the DB: safe extraction, role-scoped load, hooks
+(`song_analyzed`), cron task and analysis-provider extension points
+(PluginManager.load :537). Here: the same zip format (a `plugin.py`
+module exporting `register(api)`), safe extraction (path traversal
+guarded), an API object exposing hook registration + cron task
+registration, and a `song_analyzed` hook invoked by the analysis
+pipeline. pip-requirement install is intentionally absent (no network
+in target deployments is supported; document requirements instead).
+"""
+
+from __future__ import annotations
+
+import importlib.util
+import io
+import logging
+import os
+import tempfile
+import zipfile
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional
+
+logger = logging.getLogger(__name__)
+
+
+class HookRegistry:
+    def __init__(self):
+        self._hooks: Dict[str, List[Callable]] = {}
+
+    def register(self, name: str, fn: Callable) -> None:
+        self._hooks.setdefault(name, []).append(fn)
+
+    def fire(self, name: str, *args, **kwargs) -> None:
+        for fn in self._hooks.get(name, []):
+            try:
+                fn(*args, **kwargs)
+            except Exception:  # noqa: BLE001 — plugin errors never break analysis
+                logger.exception("plugin hook %s failed", name)
+
+    def clear(self) -> None:
+        self._hooks.clear()
+
+
+hook_registry = HookRegistry()
+
+
+@dataclass
+class PluginAPI:
+    """What a plugin's register(api) receives."""
+
+    name: str
+    hooks: HookRegistry
+    cron_tasks: List[dict] = field(default_factory=list)
+
+    def on_song_analyzed(self, fn: Callable) -> None:
+        self.hooks.register("song_analyzed", fn)
+
+    def add_cron_task(self, schedule: str, task_type: str,
+                      payload: Optional[dict] = None) -> None:
+        self.cron_tasks.append({"schedule": schedule, "task_type": task_type,
+                                "payload": payload or {}})
+
+
+def _safe_extract(zf: zipfile.ZipFile, dest: str) -> None:
+    for member in zf.namelist():
+        target = os.path.realpath(os.path.join(dest, member))
+        if not target.startswith(os.path.realpath(dest) + os.sep):
+            raise ValueError(f"unsafe path in plugin zip: {member!r}")
+    zf.extractall(dest)
+
+
+class PluginManager:
+    def __init__(self, hooks: Optional[HookRegistry] = None):
+        self.hooks = hooks or hook_registry
+        self.loaded: Dict[str, PluginAPI] = {}
+
+    def load_zip(self, name: str, blob: bytes) -> PluginAPI:
+        """Extract + import plugin.py + call register(api)
+        (reference: PluginManager.load :537)."""
+        dest = tempfile.mkdtemp(prefix=f"audiomuse-plugin-{name}-")
+        with zipfile.ZipFile(io.BytesIO(blob)) as zf:
+            _safe_extract(zf, dest)
+        mod_path = os.path.join(dest, "plugin.py")
+        if not os.path.exists(mod_path):
+            raise FileNotFoundError("plugin zip must contain plugin.py")
+        spec = importlib.util.spec_from_file_location(
+            f"audiomuse_plugin_{name}", mod_path)
+        module = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(module)  # type: ignore[union-attr]
+        if not hasattr(module, "register"):
+            raise AttributeError("plugin.py must export register(api)")
+        api = PluginAPI(name=name, hooks=self.hooks)
+        module.register(api)
+        self.loaded[name] = api
+        return api
+
+    def fire_song_analyzed(self, item_id: str, analysis: dict) -> None:
+        self.hooks.fire("song_analyzed", item_id, analysis)

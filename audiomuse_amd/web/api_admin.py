@@ -1,0 +1,169 @@
+"""Admin blueprints: backup/restore, dashboard, anchors/radios, plugins,
+provider migration.
+
+References: app_backup.py (pg_dump/restore + lock), app_dashboard.py,
+alchemy anchors/radios (song_alchemy + radio_manager.py),
+app_provider_migration.py (migration wizard -> task).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import tempfile
+
+import numpy as np
+from flask import (Blueprint, current_app, jsonify, request, send_file)
+
+from audiomuse_amd.analysis.maintenance import (backup_database,
+                                                refresh_dashboard_stats)
+from audiomuse_amd.db import write_txn
+from audiomuse_amd.taskqueue import enqueue
+from audiomuse_amd.web.auth import require_auth
+
+bp = Blueprint("admin", __name__)
+
+
+def _state():
+    return current_app.extensions["audiomuse"]
+
+
+@bp.get("/api/backup")
+@require_auth
+def backup():
+    conn = _state().conn()
+    fd, path = tempfile.mkstemp(suffix=".db", prefix="audiomuse-backup-")
+    os.close(fd)
+    backup_database(conn, path)
+    return send_file(path, as_attachment=True,
+                     download_name="audiomuse-backup.db")
+
+
+@bp.post("/api/restore")
+@require_auth
+def restore():
+    """Restore lock semantics (app_backup.py:607): refuse while tasks
+    run; replace the live DB file; engine caches invalidate."""
+    from audiomuse_amd.taskqueue import PENDING, RUNNING
+
+    state = _state()
+    conn = state.conn()
+    live = conn.execute(
+        "SELECT COUNT(*) AS n FROM task_status WHERE status IN (?, ?)",
+        (PENDING, RUNNING)).fetchone()["n"]
+    if live:
+        return jsonify({"error": "tasks running; cancel them first"}), 409
+    blob = request.get_data()
+    if not blob.startswith(b"SQLite format 3"):
+        return jsonify({"error": "not an SQLite backup"}), 400
+    db_path = state.db_url[len("sqlite:///"):]
+    tmp = db_path + ".restore"
+    with open(tmp, "wb") as fh:
+        fh.write(blob)
+    import sqlite3 as s3
+
+    try:
+        check = s3.connect(tmp)
+        check.execute("SELECT COUNT(*) FROM score")
+        check.close()
+    except Exception as exc:  # noqa: BLE001
+        os.unlink(tmp)
+        return jsonify({"error": f"backup failed validation: {exc}"}), 400
+    os.replace(tmp, db_path)
+    state._local.__dict__.clear()
+    state.invalidate()
+    return jsonify({"restored": True})
+
+
+@bp.get("/api/dashboard")
+@require_auth
+def dashboard():
+    return jsonify(refresh_dashboard_stats(_state().conn()))
+
+
+# -- alchemy anchors / radios ----------------------------------------------
+
+@bp.get("/api/alchemy/anchors")
+@require_auth
+def list_anchors():
+    rows = _state().conn().execute(
+        "SELECT name FROM alchemy_anchors").fetchall()
+    return jsonify([r["name"] for r in rows])
+
+
+@bp.post("/api/alchemy/anchors")
+@require_auth
+def save_anchor():
+    from audiomuse_amd.analysis import index as idx
+
+    body = request.get_json(force=True, silent=True) or {}
+    name = body.get("name", "").strip()
+    ids = body.get("item_ids", [])
+    if not name or not ids:
+        return jsonify({"error": "name and item_ids required"}), 400
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    vecs = [eng.vector_for_id(i) for i in ids]
+    vecs = [v.cpu().numpy() for v in vecs if v is not None]
+    if not vecs:
+        return jsonify({"error": "no known tracks"}), 404
+    centroid = np.mean(np.stack(vecs), axis=0).astype(np.float32)
+    conn = _state().conn()
+    with write_txn(conn):
+        conn.execute(
+            """INSERT INTO alchemy_anchors (name, vector) VALUES (?,?)
+               ON CONFLICT(name) DO UPDATE SET vector=excluded.vector""",
+            (name, centroid.tobytes()))
+    return jsonify({"saved": name, "dim": centroid.shape[0]})
+
+
+@bp.delete("/api/alchemy/anchors/<name>")
+@require_auth
+def delete_anchor(name):
+    conn = _state().conn()
+    with write_txn(conn):
+        cur = conn.execute("DELETE FROM alchemy_anchors WHERE name=?", (name,))
+    return jsonify({"deleted": cur.rowcount})
+
+
+@bp.get("/api/alchemy/radios")
+@require_auth
+def list_radios():
+    rows = _state().conn().execute(
+        "SELECT name, definition FROM alchemy_radios").fetchall()
+    return jsonify([{"name": r["name"],
+                     "definition": json.loads(r["definition"])} for r in rows])
+
+
+@bp.post("/api/alchemy/radios")
+@require_auth
+def save_radio():
+    body = request.get_json(force=True, silent=True) or {}
+    name = body.get("name", "").strip()
+    if not name:
+        return jsonify({"error": "name required"}), 400
+    conn = _state().conn()
+    with write_txn(conn):
+        conn.execute(
+            """INSERT INTO alchemy_radios (name, definition) VALUES (?,?)
+               ON CONFLICT(name) DO UPDATE SET definition=excluded.definition""",
+            (name, json.dumps(body.get("definition", {}))))
+    return jsonify({"saved": name})
+
+
+# -- provider migration ------------------------------------------------------
+
+@bp.post("/api/migration/start")
+@require_auth
+def migration_start():
+    """Move a library's mappings to another provider (reference:
+    app_provider_migration wizard -> planner task). Runs the metadata
+    alignment sweep against the target server as a queued task."""
+    body = request.get_json(force=True, silent=True) or {}
+    tid = enqueue(_state().conn(), "multiserver_sync", {
+        "server_type": body.get("server_type", "synthetic"),
+        "server_config": body.get("server_config", {}),
+        "server_id": body.get("server_id", "migrated"),
+    }, queue="high")
+    return jsonify({"task_id": tid}), 202

@@ -108,12 +108,14 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
     from audiomuse_amd.web.api_auth import bp as auth_bp
     from audiomuse_amd.web.api_queries import bp as queries_bp
     from audiomuse_amd.web.api_tasks import bp as tasks_bp
+    from audiomuse_amd.web.api_admin import bp as admin_bp
     from audiomuse_amd.web.api_chat import bp as chat_bp
 
     app.register_blueprint(auth_bp)
     app.register_blueprint(queries_bp)
     app.register_blueprint(tasks_bp)
     app.register_blueprint(chat_bp)
+    app.register_blueprint(admin_bp)
 
     from audiomuse_amd.web.auth import seed_admin_from_env
 
