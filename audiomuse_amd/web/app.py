@@ -121,6 +121,15 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
 
     with app.app_context():
         seed_admin_from_env(state.conn())
+        # inline boot migrations (reference boot sequence)
+        try:
+            from audiomuse_amd.analysis.canonicalize import run_startup_migrations
+
+            run_startup_migrations(state.conn())
+        except Exception:  # noqa: BLE001 — boot must not die on migration
+            import logging
+
+            logging.getLogger(__name__).exception("startup migration failed")
 
     @app.get("/health")
     def health():  # reference: app.py:227
