@@ -74,3 +74,6 @@ def active_server() -> Optional[BoundServer]:
 # register built-ins
 from audiomuse_amd.mediaserver import synthetic  # noqa: E402,F401
 from audiomuse_amd.mediaserver import subsonic  # noqa: E402,F401
+from audiomuse_amd.mediaserver import jellyfin  # noqa: E402,F401
+from audiomuse_amd.mediaserver import plex  # noqa: E402,F401
+from audiomuse_amd.mediaserver import lyrion  # noqa: E402,F401
