@@ -56,6 +56,22 @@ class AnalysisRuntime:
                       if self.enable_clap else None)
         self._clap_text: Optional[TextEmbedder] = None
         self._other_label_embs: Optional[np.ndarray] = None
+        self._lyrics = None
+
+    def lyrics_pipeline(self):
+        """Lazy lyrics pipeline (GTE embedder + VAD; ASR off by default —
+        random-init Whisper transcripts never pass the quality gate, so
+        enabling it only burns compute until trained weights exist)."""
+        if self._lyrics is None:
+            from audiomuse_amd.engines.lyrics import LyricsPipeline
+            from audiomuse_amd.models.text import TextEmbedder, gte_config
+            from audiomuse_amd.models.vad import SileroStyleVAD
+
+            embedder = TextEmbedder(gte_config(), device=str(self.device),
+                                    dtype=self.dtype)
+            vad = SileroStyleVAD().to(self.device).eval()
+            self._lyrics = LyricsPipeline(embedder, vad=vad, asr_fn=None)
+        return self._lyrics
 
     # -- label text embeddings (clap_analyzer.py:580: cached .npz) -------
 

@@ -24,6 +24,7 @@ from audiomuse_amd.analysis.index import run_all_index_builds
 from audiomuse_amd.analysis.pipeline import AnalysisRuntime
 from audiomuse_amd.db import get_db, write_txn
 from audiomuse_amd.db.store import (save_clap_embedding,
+                                    save_lyrics_embedding,
                                     save_track_analysis_and_embedding)
 from audiomuse_amd.engines.simhash import CatalogResolver
 from audiomuse_amd.mediaserver import make_provider
@@ -98,6 +99,20 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
             energy=res.energy, duration=res.duration, embedding=res.embedding)
         if res.clap_embedding is not None:
             save_clap_embedding(conn, item_id, res.clap_embedding)
+        if C.LYRICS_ENABLED:
+            # stage 8 (album.py:276): provided lyrics win; ASR optional
+            lyr = runtime.lyrics_pipeline().analyze(
+                provided_lyrics=provider.get_lyrics(track.provider_id))
+            save_lyrics_embedding(conn, item_id, lyr.embedding,
+                                  axis_scores=lyr.axis_scores,
+                                  lyrics_text=lyr.text, language=lyr.language,
+                                  instrumental=lyr.instrumental)
+        # plugin hook (reference: song.py:101)
+        from audiomuse_amd.plugin import hook_registry
+
+        hook_registry.fire("song_analyzed", item_id,
+                           {"tempo": res.tempo, "energy": res.energy,
+                            "moods": res.moods})
         with write_txn(conn):
             conn.execute(
                 """INSERT INTO track_server_map
