@@ -206,6 +206,42 @@ def rerank(results_per_tool: List[List[Dict]], n: int) -> List[str]:
     return ranked[:n]
 
 
+@bp.post("/chat/api/chatPlaylistStream")
+@require_auth
+def chat_playlist_stream():
+    """SSE variant (reference: app_chat.py:292): streams plan, then each
+    tool's results, then the final playlist."""
+    import flask
+
+    body = request.get_json(force=True, silent=True) or {}
+    prompt = body.get("prompt", "")
+    if not prompt:
+        return jsonify({"error": "prompt required"}), 400
+    hints = extract_hints(prompt)
+    plan = llm_plan(prompt, hints) or heuristic_plan(hints)
+    plan = validate_and_normalize_plan(plan)
+    state = _state()
+
+    def generate():
+        yield f"event: plan\ndata: {json.dumps(plan)}\n\n"
+        results = []
+        for call in plan:
+            res = TOOLS[call["tool"]](call.get("args", {}))
+            results.append(res)
+            yield (f"event: tool\ndata: "
+                   f"{json.dumps({'tool': call['tool'], 'n': len(res)})}\n\n")
+        ids = rerank(results, hints["n"])
+        tracks = []
+        for i in ids:
+            meta = state.meta_fn(i) or {}
+            tracks.append({"item_id": i, "title": meta.get("title"),
+                           "author": meta.get("author")})
+        yield f"event: playlist\ndata: {json.dumps(tracks)}\n\n"
+
+    return flask.Response(flask.stream_with_context(generate()),
+                          mimetype="text/event-stream")
+
+
 @bp.post("/chat/api/chatPlaylist")
 @require_auth
 def chat_playlist():

@@ -178,6 +178,45 @@ def _make_gte_embedder():
     return TextEmbedder(gte_config(), device=_state().device)
 
 
+# category-weighted suggested queries (reference: tasks/query.json used by
+# clap_text_search.py suggested-queries generator)
+_SUGGESTED_QUERIES = {
+    "mood": (3, ["uplifting summer anthems", "melancholic rainy day songs",
+                 "high energy workout tracks", "calm focus music"]),
+    "genre": (3, ["classic soul grooves", "90s alternative rock",
+                  "smooth jazz evenings", "underground hip-hop"]),
+    "instrument": (2, ["acoustic guitar ballads", "piano-driven pieces",
+                       "heavy synth textures"]),
+    "scene": (2, ["late night driving", "sunday morning coffee",
+                  "beach party at sunset"]),
+}
+
+
+@bp.get("/api/clap_search/suggestions")
+@require_auth
+def clap_search_suggestions():
+    import random
+
+    n = int(request.args.get("n", 6))
+    seed = request.args.get("seed")
+    rng = random.Random(int(seed) if seed else None)
+    pool = []
+    for _cat, (weight, queries) in _SUGGESTED_QUERIES.items():
+        pool.extend((weight, q) for q in queries)
+    picks = []
+    while pool and len(picks) < n:
+        total = sum(w for w, _ in pool)
+        r = rng.uniform(0, total)
+        acc = 0.0
+        for i, (w, q) in enumerate(pool):
+            acc += w
+            if r <= acc:
+                picks.append(q)
+                pool.pop(i)
+                break
+    return jsonify(picks)
+
+
 @bp.get("/api/semgrove")
 @require_auth
 def semgrove():
