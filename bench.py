@@ -34,6 +34,11 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=C.CLAP_GPU_BATCH)
+    ap.add_argument("--mode", choices=["clap", "knn", "train"], default="clap",
+                    help="clap: flagship analysis throughput (driver "
+                         "contract); knn: 1M-resident search latency/qps "
+                         "(BASELINE config 3); train: student_clap DDP "
+                         "distillation (BASELINE config 5)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -53,6 +58,11 @@ def main() -> None:
     from audiomuse_amd.ops import _ext
 
     _ext.require()
+
+    if args.mode == "knn":
+        return bench_knn(args, world, rank, device, dist)
+    if args.mode == "train":
+        return bench_train(args, world, rank, device, dist)
 
     batch = args.batch
     mel_cfg = dsp.clap_mel_config()
@@ -118,6 +128,78 @@ def main() -> None:
         }))
     if dist is not None:
         dist.destroy_process_group()
+
+
+def _finish(dist, device, rank, elapsed, payload):
+    if dist is not None:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    if rank == 0:
+        print(json.dumps(payload(elapsed)))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+def bench_knn(args, world, rank, device, dist):
+    """BASELINE config 3: text+audio top-k search over 1M resident
+    embeddings (i8 IVF scan + f32 re-rank)."""
+    from audiomuse_amd.index.ivf import IVFIndex
+
+    torch.manual_seed(7)
+    n, d = 1_000_000, 512
+    x = torch.randn(n, d, device=device)
+    idx = IVFIndex.build(x, metric="angular", storage="i8", device=device,
+                         seed=0)
+    q = x[: max(args.batch, 1)] + torch.randn(max(args.batch, 1), d,
+                                              device=device) * 0.01
+    for _ in range(args.warmup):
+        idx.query(q, k=10)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        idx.query(q, k=10)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    qps = args.batch * args.steps * world / elapsed
+    _finish(dist, device, rank, elapsed, lambda e: {
+        "metric": "knn_queries_per_sec_1M_resident", "value": round(qps, 1),
+        "unit": "queries/s", "n_gpus": world, "steps": args.steps,
+        "warmup": args.warmup, "ms_per_step": round(e / args.steps * 1000, 3),
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "i8+f32", "data": "synthetic",
+        "config": {"model": "ivf_i8_1Mx512", "global_batch": args.batch * world,
+                   "seq_len": 512, "parallelism": f"dp{world}"}})
+
+
+def bench_train(args, world, rank, device, dist):
+    """BASELINE config 5: student_clap distillation, DP grad all-reduce."""
+    from audiomuse_amd.parallel.trainer import DistillConfig, DistillTrainer
+
+    trainer = DistillTrainer(DistillConfig(batch=min(args.batch, 64)),
+                             device=str(device))
+    for i in range(args.warmup):
+        trainer.step(i)
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        loss = trainer.step(1000 + i)
+    torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    sps = trainer.cfg.batch * args.steps * world / elapsed
+    _finish(dist, device, rank, elapsed, lambda e: {
+        "metric": "distill_samples_per_sec", "value": round(sps, 1),
+        "unit": "samples/s", "n_gpus": world, "steps": args.steps,
+        "warmup": args.warmup, "ms_per_step": round(e / args.steps * 1000, 3),
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "bf16", "data": "synthetic",
+        "config": {"model": "htsat_student49M_teacher", "loss": round(loss, 4),
+                   "global_batch": trainer.cfg.batch * world,
+                   "seq_len": 1001, "parallelism": f"ddp{world}"}})
 
 
 if __name__ == "__main__":
