@@ -63,6 +63,16 @@ def finish(conn: sqlite3.Connection, task_id: str, worker_id: str,
                WHERE task_id=? AND worker_id=? AND status=?""",
             (status, json.dumps(result or {}), error_code, task_id, worker_id,
              RUNNING))
+        if cur.rowcount == 1:
+            # task history trail (reference: record_task_history :223)
+            row = conn.execute(
+                "SELECT task_type FROM task_status WHERE task_id=?",
+                (task_id,)).fetchone()
+            conn.execute(
+                "INSERT INTO task_history (task_id, task_type, status, note) "
+                "VALUES (?,?,?,?)",
+                (task_id, row["task_type"] if row else "", status,
+                 (json.dumps(result)[:500] if result else None)))
     return cur.rowcount == 1
 
 

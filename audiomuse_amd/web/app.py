@@ -131,6 +131,30 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
 
             logging.getLogger(__name__).exception("startup migration failed")
 
+    # request-level logging (reference: note_request_start/end, app.py:194)
+    import logging as _logging
+    import time as _time
+
+    from flask import g as _g
+    from flask import request as _request
+
+    _req_log = _logging.getLogger("audiomuse.requests")
+
+    @app.before_request
+    def _note_request_start():
+        _g._t0 = _time.monotonic()
+
+    @app.after_request
+    def _note_request_end(resp):
+        try:
+            dt = (_time.monotonic() - getattr(_g, "_t0", _time.monotonic()))
+            if _request.path.startswith(("/api/", "/chat/")):
+                _req_log.info("%s %s -> %d in %.1f ms", _request.method,
+                              _request.path, resp.status_code, dt * 1000)
+        except Exception:
+            pass
+        return resp
+
     @app.get("/health")
     def health():  # reference: app.py:227
         try:
