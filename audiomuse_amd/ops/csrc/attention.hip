@@ -124,7 +124,10 @@ __global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
   const unsigned long long wrap_r_mask = __ballot(my_wrap & 2);
   const unsigned long long wrap_c_mask = __ballot(my_wrap & 1);
 
-  for (int h = wave; h < heads; h += ATTN_WAVES) {
+  // one head per wave; heads ride gridDim.y so late stages (few windows,
+  // many heads) still fill all 256 CUs
+  const int h = blockIdx.y * ATTN_WAVES + wave;
+  if (h < heads) {
     // ---- stage V transposed: VT[d][t] = V[t][d] ----
     {
       const __bf16* vptr = qkv + my_base + 2 * C + h * 32;
@@ -274,8 +277,9 @@ void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream) {
   const int n_windows = Bn * (H >> 3) * (W >> 3);
+  const int head_groups = (heads + ATTN_WAVES - 1) / ATTN_WAVES;
   const size_t lds_bytes = ATTN_WAVES * WAVE_LDS_HALF * sizeof(__bf16);
-  hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows),
+  hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows, head_groups),
                      dim3(64 * ATTN_WAVES), lds_bytes,
                      stream, (const __bf16*)qkv, (__bf16*)out,
                      bias, Bn, H, W, C, heads, shift, scale);
