@@ -32,8 +32,12 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
     const float* __restrict__ mel_w,    // (nnz) filter weights
     int T, int n_frames, int hop, int n_mels, int center, int log_mode,
     int quant16) {
+  // zbuf is XOR-swizzled: the bit-reversal scatter otherwise lands all 64
+  // lanes on one bank pair (measured 5.7M SQ_LDS_BANK_CONFLICT/dispatch) —
+  // folding bits 5-9 into the bank bits spreads lanes across all banks.
   __shared__ float2 zbuf[NFFT];
   __shared__ float pw[2][NFFT / 2 + 1];
+#define AM_ZS(i) ((i) ^ (((i) >> 5) & 31))
 
   const int f0 = blockIdx.x * 2;       // this block's frame pair
   const int b = blockIdx.y;
@@ -66,7 +70,7 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
       v[fr] = x * window[i];
     }
     const int rev = __brev((unsigned)i) >> (32 - LOG2N);
-    zbuf[rev] = make_float2(v[0], v[1]);
+    zbuf[AM_ZS(rev)] = make_float2(v[0], v[1]);
   }
   __syncthreads();
 
@@ -80,12 +84,12 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
       const int i0 = (grp << s) + j;
       const int i1 = i0 + half;
       const float2 w = twiddle[j * tw_step];
-      const float2 a = zbuf[i0];
-      const float2 c = zbuf[i1];
+      const float2 a = zbuf[AM_ZS(i0)];
+      const float2 c = zbuf[AM_ZS(i1)];
       const float tr = w.x * c.x - w.y * c.y;
       const float ti = w.x * c.y + w.y * c.x;
-      zbuf[i0] = make_float2(a.x + tr, a.y + ti);
-      zbuf[i1] = make_float2(a.x - tr, a.y - ti);
+      zbuf[AM_ZS(i0)] = make_float2(a.x + tr, a.y + ti);
+      zbuf[AM_ZS(i1)] = make_float2(a.x - tr, a.y - ti);
     }
     __syncthreads();
   }
@@ -94,8 +98,8 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
   // frame0[k] = (Z[k] + conj(Z[N-k])) / 2 ; frame1[k] = (Z[k] - conj(Z[N-k])) / 2i
   for (int k = tid; k <= NFFT / 2; k += blockDim.x) {
     const int nk = (NFFT - k) & (NFFT - 1);
-    const float2 zk = zbuf[k];
-    const float2 zn = zbuf[nk];
+    const float2 zk = zbuf[AM_ZS(k)];
+    const float2 zn = zbuf[AM_ZS(nk)];
     const float ar = 0.5f * (zk.x + zn.x);
     const float ai = 0.5f * (zk.y - zn.y);
     const float br = 0.5f * (zk.y + zn.y);

@@ -20,6 +20,59 @@ __device__ __forceinline__ float wrsum(float v) {
   return __shfl(v, 0, 64);
 }
 
+__device__ __forceinline__ float hrsum(float v) {
+  // reduce within each 32-lane half (2 rows share one wave)
+  for (int off = 16; off > 0; off >>= 1) v += __shfl_xor(v, off, 32);
+  return v;
+}
+
+// dim <= 128: two rows per wave (lanes 0-31 / 32-63), full-wave utilization
+// (measured: the one-row variant at dim 128 runs 80% VALUBusy on half-idle
+// waves — profiles/r01_bench_clap.md).
+__global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
+    const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
+    const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
+    long long n_rows, int dim, float eps) {
+  const int sl = threadIdx.x & 31;          // lane within the half
+  const long long row = (long long)blockIdx.x * 8 + (threadIdx.x >> 5);
+  if (row >= n_rows) return;
+
+  const __hip_bfloat16* xr = x + row * dim;
+  __hip_bfloat16* yr = y + row * dim;
+  float vals[4] = {0.f, 0.f, 0.f, 0.f};
+  float sum = 0.0f;
+  const int i = sl * 4;
+  if (i < dim) {
+    const short4 p = *reinterpret_cast<const short4*>(xr + i);
+    const __hip_bfloat16* pb = reinterpret_cast<const __hip_bfloat16*>(&p);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      vals[j] = __bfloat162float(pb[j]);
+      sum += vals[j];
+    }
+  }
+  const float mean = hrsum(sum) / dim;
+  float var = 0.0f;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float d = i < dim ? vals[j] - mean : 0.0f;
+    var += d * d;
+  }
+  const float rstd = rsqrtf(hrsum(var) / dim + eps);
+  if (i < dim) {
+    const short4 pw = *reinterpret_cast<const short4*>(w + i);
+    const short4 pbv = *reinterpret_cast<const short4*>(b + i);
+    const __hip_bfloat16* wb = reinterpret_cast<const __hip_bfloat16*>(&pw);
+    const __hip_bfloat16* bb = reinterpret_cast<const __hip_bfloat16*>(&pbv);
+    short4 out;
+    __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      ob[j] = __float2bfloat16(((vals[j] - mean) * rstd) * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
+    *reinterpret_cast<short4*>(yr + i) = out;
+  }
+}
+
 // one wave per row; block = 256 threads = 4 rows
 template <int NIT>
 __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
@@ -90,6 +143,14 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
 void launch_layernorm_bf16(const void* x, void* y, const void* w,
                            const void* b, long long n_rows, int dim, float eps,
                            hipStream_t stream) {
+  if (dim <= 128) {
+    const long long blocks2 = (n_rows + 7) / 8;
+    hipLaunchKernelGGL(layernorm_bf16_half_kernel, dim3((unsigned)blocks2),
+                       dim3(256), 0, stream, (const __hip_bfloat16*)x,
+                       (__hip_bfloat16*)y, (const __hip_bfloat16*)w,
+                       (const __hip_bfloat16*)b, n_rows, dim, eps);
+    return;
+  }
   const long long blocks = (n_rows + 3) / 4;
   const dim3 grid((unsigned)blocks);
   const dim3 block(256);
