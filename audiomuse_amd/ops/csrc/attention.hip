@@ -74,16 +74,17 @@ void launch_mfma_probe(const void* A, const void* B, float* D,
 // Fused window attention
 // ---------------------------------------------------------------------------
 
-// per-wave LDS: VT[32][64+8] bf16 rows padded to 144 B; P[64][64+8] bf16.
-// 2 waves per block: 27.6 KiB LDS -> 5 blocks/CU (10 waves/CU) vs 2
-// blocks at 4 waves (8/CU) — this kernel is gather-latency bound, so
-// occupancy is the lever (measured: profiles/r01_bench_clap.md).
-constexpr int ATTN_WAVES = 2;
-constexpr int VT_STRIDE = 72;   // bf16 elements per VT row (64 + 8 pad)
-constexpr int P_STRIDE = 72;    // bf16 elements per P row
-constexpr int WAVE_LDS_HALF = 32 * VT_STRIDE + 64 * P_STRIDE;  // bf16 elems
+// per-wave LDS: VT[32][64] + P[64][64] bf16, XOR-swizzled instead of
+// padded (swizzle: element col ^ ((row&7)<<3) keeps 16 B reads aligned
+// and spreads the 16-lane column reads across all banks). 12 KiB/wave;
+// 1-wave blocks -> ~13 blocks/CU, registers fit 4 waves/SIMD with the
+// launch-bounds hint (was: 181 regs -> 2 waves/SIMD, 11.9% occupancy
+// measured — this kernel is gather-latency bound, occupancy is the lever).
+constexpr int ATTN_WAVES = 1;
+constexpr int WAVE_LDS_HALF = 32 * 64 + 64 * 64;  // bf16 elems (VT + P)
+#define AM_SWZ(row, col) (((row) << 6) + ((col) ^ (((row) & 7) << 3)))
 
-__global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
+__global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
     const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
     __bf16* __restrict__ out,        // (B, H, W, C)
     const float* __restrict__ bias,          // (heads, 64, 64)
@@ -101,7 +102,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
   const int ww = wrem - wh * nWw;
 
   __bf16* VT = lds + wave * WAVE_LDS_HALF;
-  __bf16* P = VT + 32 * VT_STRIDE;
+  __bf16* P = VT + 32 * 64;
 
   // token t (0..63) -> source coords + wrap bits (shifted windows)
   auto src_of = [&](int t, int& si, int& sj, int& wrap) {
@@ -138,7 +139,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
       for (int g = 0; g < 4; ++g)
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          VT[(g * 8 + j) * VT_STRIDE + lane] = (__bf16)vv[g][j];
+          VT[AM_SWZ(g * 8 + j, lane)] = (__bf16)vv[g][j];
     }
 
     // ---- Q (A-frags) and K (B-frags) straight from global ----
@@ -229,7 +230,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
 #pragma unroll
         for (int tc = 0; tc < 4; ++tc) {
           const int col = tc * 16 + col_in_tile;
-          P[row * P_STRIDE + col] =
+          P[AM_SWZ(row, col)] =
               (__bf16)(s[tr][tc][reg] / (rsum[tr][reg] + 1e-20f));
         }
       }
@@ -245,11 +246,11 @@ __global__ __launch_bounds__(64 * ATTN_WAVES) void window_attn_kernel(
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
           // A = P tile: row = tr*16 + (lane&15), k = ks*32 + kk + j
-          bf16x8 pa = *(const bf16x8*)(P + (tr * 16 + (lane & 15)) * P_STRIDE +
-                                       ks * 32 + kk);
+          bf16x8 pa = *(const bf16x8*)(
+              P + AM_SWZ(tr * 16 + (lane & 15), ks * 32 + kk));
           // B = V tile: col(dim) = tc*16 + (lane&15), k(token) = ks*32+kk+j
-          bf16x8 vb = *(const bf16x8*)(VT + (tc * 16 + (lane & 15)) * VT_STRIDE +
-                                       ks * 32 + kk);
+          bf16x8 vb = *(const bf16x8*)(
+              VT + AM_SWZ(tc * 16 + (lane & 15), ks * 32 + kk));
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc, 0, 0, 0);
         }
         o[tr][tc] = acc;
