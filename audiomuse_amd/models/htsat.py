@@ -138,8 +138,14 @@ class SwinBlock(nn.Module):
             out = ext.window_attn_fwd(
                 qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
                 self.attn.heads, self.shift, self.attn.scale)
-            x = x + self.attn.proj(out.view(B, L, C))
-            return x + self.mlp(self.norm2(x))
+            proj = self.attn.proj(out.view(B, L, C))
+            # fused residual add + norm2 (one pass instead of add->LN)
+            x2, xn2 = ext.add_layernorm_bf16(
+                x.contiguous(), proj.contiguous(),
+                self.norm2.weight.to(torch.bfloat16).contiguous(),
+                self.norm2.bias.to(torch.bfloat16).contiguous(),
+                self.norm2.eps)
+            return x2 + self.mlp(xn2)
         shortcut = x
         x = self.norm1(x).view(B, H, W, C)
         if self.shift:

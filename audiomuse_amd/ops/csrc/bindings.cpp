@@ -2,6 +2,8 @@
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 
+#include <vector>
+
 #include <c10/hip/HIPStream.h>
 
 namespace audiomuse {
@@ -19,6 +21,10 @@ void launch_ivf_scan(int dtype_code, int metric, const void* query,
 void launch_layernorm_bf16(const void* x, void* y, const void* w,
                            const void* b, long long n_rows, int dim, float eps,
                            hipStream_t stream);
+void launch_add_layernorm_bf16(const void* x, const void* in2, void* sum_out,
+                               void* y, const void* w, const void* b,
+                               long long n_rows, int dim, float eps,
+                               hipStream_t stream);
 void launch_mfma_probe(const void* A, const void* B, float* D,
                        hipStream_t stream);
 void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
@@ -160,8 +166,33 @@ static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
   return out;
 }
 
+static std::vector<torch::Tensor> add_layernorm_bf16(torch::Tensor x,
+                                                     torch::Tensor other,
+                                                     torch::Tensor w,
+                                                     torch::Tensor b,
+                                                     double eps) {
+  AM_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous(),
+           "x must be contiguous bf16 on GPU");
+  AM_CHECK(other.is_contiguous() && other.sizes() == x.sizes() &&
+               other.scalar_type() == at::kBFloat16,
+           "other must match x");
+  const int dim = x.size(-1);
+  AM_CHECK(dim % 4 == 0 && dim <= 4096, "dim must be /4 and <= 4096");
+  auto sum = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  const long long n_rows = x.numel() / dim;
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_add_layernorm_bf16(
+      x.data_ptr(), other.data_ptr(), sum.data_ptr(), y.data_ptr(),
+      w.data_ptr(), b.data_ptr(), n_rows, dim, (float)eps, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return {sum, y};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "AudioMuse-AMD native CDNA4 kernels";
+  m.def("add_layernorm_bf16", &add_layernorm_bf16,
+        "Fused residual add + LayerNorm: returns (x+other, LN(x+other))");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("window_attn_fwd", &window_attn_fwd,
         "Fused shifted-window attention (qkv BHW3C bf16, bias, heads, "

@@ -29,9 +29,14 @@ __device__ __forceinline__ float hrsum(float v) {
 // dim <= 128: two rows per wave (lanes 0-31 / 32-63), full-wave utilization
 // (measured: the one-row variant at dim 128 runs 80% VALUBusy on half-idle
 // waves — profiles/r01_bench_clap.md).
+// ADD variant fuses the preceding residual add: in2 != nullptr adds it to
+// x, writes the sum to `sum`, and normalizes the sum (saves one full
+// tensor read between the eager add and the LN).
+template <bool ADD>
 __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
     const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
     const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
+    const __hip_bfloat16* __restrict__ in2, __hip_bfloat16* __restrict__ sum_out,
     long long n_rows, int dim, float eps) {
   const int sl = threadIdx.x & 31;          // lane within the half
   const long long row = (long long)blockIdx.x * 8 + (threadIdx.x >> 5);
@@ -45,10 +50,21 @@ __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
   if (i < dim) {
     const short4 p = *reinterpret_cast<const short4*>(xr + i);
     const __hip_bfloat16* pb = reinterpret_cast<const __hip_bfloat16*>(&p);
+    short4 q;
+    const __hip_bfloat16* qb = reinterpret_cast<const __hip_bfloat16*>(&q);
+    if (ADD) q = *reinterpret_cast<const short4*>(in2 + row * dim + i);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       vals[j] = __bfloat162float(pb[j]);
+      if (ADD) vals[j] += __bfloat162float(qb[j]);
       sum += vals[j];
+    }
+    if (ADD) {
+      short4 so;
+      __hip_bfloat16* sb = reinterpret_cast<__hip_bfloat16*>(&so);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) sb[j] = __float2bfloat16(vals[j]);
+      *reinterpret_cast<short4*>(sum_out + row * dim + i) = so;
     }
   }
   const float mean = hrsum(sum) / dim;
@@ -74,10 +90,11 @@ __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
 }
 
 // one wave per row; block = 256 threads = 4 rows
-template <int NIT>
+template <int NIT, bool ADD>
 __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
     const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
+    const __hip_bfloat16* __restrict__ in2, __hip_bfloat16* __restrict__ sum_out,
     long long n_rows, int dim, float eps) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -95,11 +112,22 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
     if (i < dim) {
       const short4 p = *reinterpret_cast<const short4*>(xr + i);
       const __hip_bfloat16* pb = reinterpret_cast<const __hip_bfloat16*>(&p);
+      short4 q;
+      const __hip_bfloat16* qb = reinterpret_cast<const __hip_bfloat16*>(&q);
+      if (ADD) q = *reinterpret_cast<const short4*>(in2 + row * dim + i);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        const float f = __bfloat162float(pb[j]);
+        float f = __bfloat162float(pb[j]);
+        if (ADD) f += __bfloat162float(qb[j]);
         vals[t * 4 + j] = f;
         sum += f;
+      }
+      if (ADD) {
+        short4 so;
+        __hip_bfloat16* sb = reinterpret_cast<__hip_bfloat16*>(&so);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) sb[j] = __float2bfloat16(vals[t * 4 + j]);
+        *reinterpret_cast<short4*>(sum_out + row * dim + i) = so;
       }
     } else {
 #pragma unroll
@@ -140,32 +168,65 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
   }
 }
 
-void launch_layernorm_bf16(const void* x, void* y, const void* w,
-                           const void* b, long long n_rows, int dim, float eps,
-                           hipStream_t stream) {
+void launch_layernorm_bf16_impl(const void* x, void* y, const void* w,
+                                const void* b, const void* in2, void* sum_out,
+                                long long n_rows, int dim, float eps,
+                                hipStream_t stream) {
+  const bool add = in2 != nullptr;
   if (dim <= 128) {
     const long long blocks2 = (n_rows + 7) / 8;
-    hipLaunchKernelGGL(layernorm_bf16_half_kernel, dim3((unsigned)blocks2),
-                       dim3(256), 0, stream, (const __hip_bfloat16*)x,
-                       (__hip_bfloat16*)y, (const __hip_bfloat16*)w,
-                       (const __hip_bfloat16*)b, n_rows, dim, eps);
+    if (add)
+      hipLaunchKernelGGL((layernorm_bf16_half_kernel<true>),
+                         dim3((unsigned)blocks2), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x, (__hip_bfloat16*)y,
+                         (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                         (const __hip_bfloat16*)in2, (__hip_bfloat16*)sum_out,
+                         n_rows, dim, eps);
+    else
+      hipLaunchKernelGGL((layernorm_bf16_half_kernel<false>),
+                         dim3((unsigned)blocks2), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x, (__hip_bfloat16*)y,
+                         (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                         nullptr, nullptr, n_rows, dim, eps);
     return;
   }
   const long long blocks = (n_rows + 3) / 4;
   const dim3 grid((unsigned)blocks);
   const dim3 block(256);
-#define AM_LN_CASE(NIT)                                                   \
-  hipLaunchKernelGGL((layernorm_bf16_kernel<NIT>), grid, block, 0, stream, \
-                     (const __hip_bfloat16*)x, (__hip_bfloat16*)y,        \
-                     (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,  \
+#define AM_LN_CASE(NIT, A)                                                     \
+  hipLaunchKernelGGL((layernorm_bf16_kernel<NIT, A>), grid, block, 0, stream,  \
+                     (const __hip_bfloat16*)x, (__hip_bfloat16*)y,             \
+                     (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,       \
+                     (const __hip_bfloat16*)in2, (__hip_bfloat16*)sum_out,     \
                      n_rows, dim, eps)
+#define AM_LN_DISPATCH(NIT)                                                    \
+  do {                                                                         \
+    if (add) AM_LN_CASE(NIT, true);                                            \
+    else AM_LN_CASE(NIT, false);                                               \
+  } while (0)
   const int nit = (dim + 255) / 256;
-  if (nit <= 1) AM_LN_CASE(1);
-  else if (nit <= 2) AM_LN_CASE(2);
-  else if (nit <= 4) AM_LN_CASE(4);
-  else if (nit <= 8) AM_LN_CASE(8);
-  else AM_LN_CASE(16);
+  if (nit <= 1) AM_LN_DISPATCH(1);
+  else if (nit <= 2) AM_LN_DISPATCH(2);
+  else if (nit <= 4) AM_LN_DISPATCH(4);
+  else if (nit <= 8) AM_LN_DISPATCH(8);
+  else AM_LN_DISPATCH(16);
+#undef AM_LN_DISPATCH
 #undef AM_LN_CASE
+}
+
+void launch_layernorm_bf16(const void* x, void* y, const void* w,
+                           const void* b, long long n_rows, int dim, float eps,
+                           hipStream_t stream) {
+  launch_layernorm_bf16_impl(x, y, w, b, nullptr, nullptr, n_rows, dim, eps,
+                             stream);
+}
+
+void launch_add_layernorm_bf16(const void* x, const void* in2, void* sum_out,
+                               void* y, const void* w, const void* b,
+                               long long n_rows, int dim, float eps,
+                               hipStream_t stream) {
+  launch_layernorm_bf16_impl(x, y, w, b, in2, sum_out, n_rows, dim, eps,
+                             stream);
 }
 
 }  // namespace audiomuse

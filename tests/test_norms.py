@@ -49,3 +49,22 @@ def test_mel_fused_quant16_matches_reference():
     ref = hip_ops.mel_spectrogram(dsp.int16_roundtrip(audio), cfg,
                                   force_reference=True)
     torch.testing.assert_close(fused, ref, rtol=1e-3, atol=2e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dim", [128, 512])
+def test_fused_add_ln_matches_reference(dim):
+    import audiomuse_amd._C as C
+
+    torch.manual_seed(0)
+    x = torch.randn(5, 33, dim, device="cuda", dtype=torch.bfloat16)
+    other = torch.randn_like(x)
+    w = torch.randn(dim, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(dim, device="cuda", dtype=torch.bfloat16)
+    s, y = C.add_layernorm_bf16(x.contiguous(), other.contiguous(),
+                                w.contiguous(), b.contiguous(), 1e-5)
+    expect_sum = (x.float() + other.float())
+    torch.testing.assert_close(s.float(), expect_sum.to(torch.bfloat16).float(),
+                               rtol=2e-2, atol=2e-2)
+    expect = F.layer_norm(expect_sum, (dim,), w.float(), b.float(), 1e-5)
+    torch.testing.assert_close(y.float(), expect, rtol=3e-2, atol=3e-2)
