@@ -107,7 +107,16 @@ def alchemy():
 
     def _vecs(ids):
         out = []
+        conn = _state().conn()
         for i in ids:
+            if isinstance(i, str) and i.startswith("anchor:"):
+                # saved anchors join the mix (reference: alchemy_anchors)
+                row = conn.execute(
+                    "SELECT vector FROM alchemy_anchors WHERE name=?",
+                    (i[len("anchor:"):],)).fetchone()
+                if row is not None:
+                    out.append(np.frombuffer(row["vector"], dtype=np.float32))
+                continue
             v = eng.vector_for_id(i)
             if v is not None:
                 out.append(v.cpu().numpy())
@@ -120,6 +129,21 @@ def alchemy():
         temperature=float(body.get("temperature", 0.0)),
         exclude=tuple(add_ids), seed=body.get("seed"))
     return jsonify(_with_meta(res))
+
+
+@bp.post("/api/alchemy/radios/<name>/play")
+@require_auth
+def play_radio(name):
+    """Run a saved radio definition through alchemy
+    (reference: radio_manager.py)."""
+    conn = _state().conn()
+    row = conn.execute("SELECT definition FROM alchemy_radios WHERE name=?",
+                       (name,)).fetchone()
+    if row is None:
+        return jsonify({"error": f"unknown radio {name!r}"}), 404
+    definition = json.loads(row["definition"])
+    with current_app.test_request_context(json=definition):
+        return alchemy()
 
 
 @bp.get("/api/artist_similarity")

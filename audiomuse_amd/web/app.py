@@ -155,6 +155,22 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             pass
         return resp
 
+    # cron scheduler thread (reference: app.py cron loop)
+    import threading as _threading
+
+    if not app.config.get("TESTING"):
+        _stop = _threading.Event()
+        app.extensions["cron_stop"] = _stop
+
+        def _cron_thread():
+            from audiomuse_amd.db import connect as _connect
+            from audiomuse_amd.utils.cron import cron_loop
+
+            conn = _connect(state.db_url)
+            cron_loop(conn, _stop)
+
+        _threading.Thread(target=_cron_thread, daemon=True).start()
+
     @app.get("/health")
     def health():  # reference: app.py:227
         try:

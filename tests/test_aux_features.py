@@ -160,3 +160,20 @@ def test_restore_rejects_garbage(admin_client):
     client, _ = admin_client
     r = client.post("/api/restore", data=b"not a database")
     assert r.status_code == 400
+
+
+def test_anchor_in_alchemy_mix(admin_client):
+    client, ids = admin_client
+    client.post("/api/alchemy/anchors", json={"name": "mix1",
+                                              "item_ids": ids[:4]})
+    r = client.post("/api/alchemy", json={"add": ["anchor:mix1"], "n": 5})
+    assert r.status_code == 200 and len(r.json) == 5
+
+
+def test_radio_play(admin_client):
+    client, ids = admin_client
+    client.post("/api/alchemy/radios",
+                json={"name": "r1", "definition": {"add": [ids[0]], "n": 4}})
+    r = client.post("/api/alchemy/radios/r1/play")
+    assert r.status_code == 200 and len(r.json) == 4
+    assert client.post("/api/alchemy/radios/nope/play").status_code == 404
