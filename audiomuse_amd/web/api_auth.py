@@ -69,6 +69,15 @@ def me():
     return jsonify({"user": g.user["u"], "role": g.user["r"]})
 
 
+@bp.get("/api/users")
+@require_auth
+def list_users():
+    """reference: app_users.py"""
+    rows = _state().conn().execute(
+        "SELECT username, role, created_at FROM audiomuse_users").fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
 @bp.get("/api/config")
 @require_auth
 def get_config():

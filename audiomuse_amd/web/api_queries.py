@@ -257,6 +257,69 @@ def semgrove():
     return jsonify(_with_meta(res))
 
 
+@bp.get("/api/lyrics_axes")
+@require_auth
+def lyrics_axes():
+    """Axis-score search (reference: lyrics-axes index) — rank tracks by
+    one of the 27 thematic axes."""
+    axis = request.args.get("axis", "")
+    if axis not in C.LYRICS_AXES:
+        return jsonify({"error": f"unknown axis {axis!r}",
+                        "axes": C.LYRICS_AXES}), 400
+    conn = _state().conn()
+    rows = conn.execute(
+        "SELECT item_id, axis_scores FROM lyrics_embedding "
+        "WHERE axis_scores IS NOT NULL").fetchall()
+    scored = []
+    for r in rows:
+        try:
+            score = json.loads(r["axis_scores"]).get(axis)
+        except Exception:
+            continue
+        if score is not None:
+            scored.append((r["item_id"], float(score)))
+    scored.sort(key=lambda t: -t[1])
+    n = int(request.args.get("n", 20))
+    return jsonify(_with_meta([
+        {"item_id": i, "distance": 1.0 - s} for i, s in scored[:n]]))
+
+
+@bp.get("/api/hyperbolic_tree")
+@require_auth
+def hyperbolic_tree():
+    """Mood-rooted explorer tree (reference: hyperbolic_manager genre/
+    mood tree cache :613): root -> predominant moods -> nearest tracks
+    in hyperbolic space."""
+    from audiomuse_amd.engines.hyperbolic import HyperbolicSpace
+
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    per_node = int(request.args.get("per_node", 8))
+    state = _state()
+    space = current_app.extensions.get("hyperbolic")
+    if space is None or current_app.extensions.get("hyperbolic_n") != eng.index.n:
+        space = HyperbolicSpace(eng.index.vectors_f32)
+        current_app.extensions["hyperbolic"] = space
+        current_app.extensions["hyperbolic_n"] = eng.index.n
+    by_mood = {}
+    for pos, item_id in enumerate(eng.item_ids):
+        meta = state.meta_fn(item_id) or {}
+        moods = meta.get("mood_vector") or {}
+        if not moods:
+            continue
+        top = max(moods, key=moods.get)
+        by_mood.setdefault(top, []).append((pos, item_id, moods[top]))
+    tree = []
+    for mood, members in sorted(by_mood.items(), key=lambda kv: -len(kv[1])):
+        members.sort(key=lambda t: -t[2])
+        kids = [{"item_id": iid,
+                 "radius": float(space.points[pos].norm())}
+                for pos, iid, _s in members[:per_node]]
+        tree.append({"mood": mood, "count": len(members), "children": kids})
+    return jsonify(tree)
+
+
 @bp.get("/api/sonic_fingerprint")
 @require_auth
 def sonic_fp():

@@ -203,3 +203,25 @@ def test_auth_barrier():
             assert client.get("/api/active_tasks").status_code == 200
             assert client.get("/api/me").json["user"] == "admin"
         conn.close()
+
+
+def test_lyrics_axes_endpoint(client_ids):
+    client, ids = client_ids
+    r = client.get("/api/lyrics_axes?axis=love&n=5")
+    assert r.status_code == 200 and len(r.json) == 5
+    assert client.get("/api/lyrics_axes?axis=bogus").status_code == 400
+
+
+def test_hyperbolic_tree_endpoint(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/hyperbolic_tree?per_node=3")
+    assert r.status_code == 200
+    tree = r.json
+    assert tree and all("mood" in n and "children" in n for n in tree)
+    assert all(len(n["children"]) <= 3 for n in tree)
+
+
+def test_users_endpoint(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/users")
+    assert r.status_code == 200
