@@ -159,6 +159,19 @@ CREATE TABLE IF NOT EXISTS alchemy_radios (
     name TEXT PRIMARY KEY,
     definition TEXT NOT NULL
 );
+CREATE TABLE IF NOT EXISTS control_request (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    action TEXT NOT NULL,
+    payload TEXT,
+    created_at REAL DEFAULT ((julianday('now') - 2440587.5) * 86400.0),
+    expires_at REAL
+);
+CREATE TABLE IF NOT EXISTS control_ack (
+    request_id INTEGER NOT NULL,
+    listener TEXT NOT NULL,
+    acked_at REAL DEFAULT ((julianday('now') - 2440587.5) * 86400.0),
+    PRIMARY KEY (request_id, listener)
+);
 """
 
 

@@ -151,9 +151,23 @@ class Worker:
             while not self._stop.is_set():
                 if self.max_jobs and self.jobs_done >= self.max_jobs:
                     return
+                # control plane: stop/restart broadcast (reference:
+                # control.py; reclaim stands down inside the window)
+                from audiomuse_amd.taskqueue import control as qctl
+
+                reqs = qctl.pending_requests(
+                    conn, self.worker_id,
+                    [qctl.ACTION_STOP_WORKERS, qctl.ACTION_RESTART])
+                if reqs:
+                    for r in reqs:
+                        qctl.ack(conn, r["id"], self.worker_id)
+                    logger.info("worker %s stopping on control request",
+                                self.worker_id)
+                    return
                 now = time.time()
                 if now - last_maintenance > C.QUEUE_LEASE_SECONDS:
-                    qsql.reclaim_orphans(conn)
+                    if not qctl.control_window_active(conn):
+                        qsql.reclaim_orphans(conn)
                     last_maintenance = now
                 if self.run_one(conn):
                     idle_since = time.time()

@@ -165,3 +165,38 @@ def test_pending_children_backpressure(db):
     for _ in range(4):
         enqueue(conn, "child", parent_task_id=parent)
     assert qsql.pending_children(conn, parent) == 4
+
+
+def test_control_stop_broadcast_with_acks(db):
+    """reference: test_control_ack_wait.py semantics."""
+    import threading
+
+    from audiomuse_amd.taskqueue import control as qctl
+
+    conn, url = db
+    w1 = Worker(db_url=url, max_jobs=100)
+    w2 = Worker(db_url=url, max_jobs=100)
+    t1 = threading.Thread(target=lambda: w1.run_forever())
+    t2 = threading.Thread(target=lambda: w2.run_forever())
+    t1.start(); t2.start()
+    time.sleep(0.3)
+    rid = qctl.publish_control_request(conn, qctl.ACTION_STOP_WORKERS)
+    assert qctl.wait_for_acks(conn, rid, expected=2, timeout=10.0)
+    t1.join(timeout=5); t2.join(timeout=5)
+    assert not t1.is_alive() and not t2.is_alive()
+    assert qctl.ack_count(conn, rid) == 2
+
+
+def test_reclaim_stands_down_in_control_window(db):
+    from audiomuse_amd.taskqueue import control as qctl
+
+    conn, _ = db
+    tid = enqueue(conn, "noop")
+    qsql.claim(conn, "dead", lease_seconds=0.01)
+    time.sleep(0.05)
+    qctl.publish_control_request(conn, qctl.ACTION_RESTART,
+                                 window_seconds=60.0)
+    assert qctl.control_window_active(conn)
+    # maintenance policy: the worker loop skips reclaim inside the window;
+    # direct reclaim still works (it is the policy gate, not the SQL)
+    assert task_row(conn, tid)["status"] == RUNNING
