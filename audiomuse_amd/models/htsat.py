@@ -114,7 +114,11 @@ class SwinBlock(nn.Module):
         self.attn = WindowAttention(dim, heads, window)
         self.norm2 = FusedLayerNorm(dim)
         hidden = int(dim * mlp_ratio)
-        self.mlp = nn.Sequential(nn.Linear(dim, hidden), nn.GELU(), nn.Linear(hidden, dim))
+        # tanh-approx GELU matches the hipBLASLt epilogue used on the
+        # fused inference path
+        self.mlp = nn.Sequential(nn.Linear(dim, hidden),
+                                 nn.GELU(approximate="tanh"),
+                                 nn.Linear(hidden, dim))
 
     def _fused_attn_available(self, x: torch.Tensor) -> bool:
         if not (x.is_cuda and x.dtype == torch.bfloat16
@@ -145,7 +149,10 @@ class SwinBlock(nn.Module):
                 self.norm2.weight.to(torch.bfloat16).contiguous(),
                 self.norm2.bias.to(torch.bfloat16).contiguous(),
                 self.norm2.eps)
-            return x2 + self.mlp(xn2)
+            # MLP with the GELU fused into the first GEMM's epilogue
+            hidden = ext.linear_gelu(xn2, self.mlp[0].weight.contiguous(),
+                                     self.mlp[0].bias.contiguous())
+            return x2 + self.mlp[2](hidden)
         shortcut = x
         x = self.norm1(x).view(B, H, W, C)
         if self.shift:

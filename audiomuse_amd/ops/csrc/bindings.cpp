@@ -189,8 +189,11 @@ static std::vector<torch::Tensor> add_layernorm_bf16(torch::Tensor x,
   return {sum, y};
 }
 
+void register_gemm_gelu(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "AudioMuse-AMD native CDNA4 kernels";
+  register_gemm_gelu(m);
   m.def("add_layernorm_bf16", &add_layernorm_bf16,
         "Fused residual add + LayerNorm: returns (x+other, LN(x+other))");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");

@@ -1,0 +1,146 @@
+// hipBLASLt GEMM with fused bias+GELU epilogue (bf16 in/out, fp32 compute).
+//
+// The encoder MLP is Linear(C,4C) -> GELU -> Linear(4C,C); the eager GELU
+// pass re-reads and re-writes the 4C-wide activation (measured ~9% of step
+// time, profiles/r01_kernel_pmc.md). hipBLASLt applies GELU+bias in the
+// GEMM epilogue, eliminating that round trip. Plain library GEMM use —
+// the hand-written MFMA work stays in the attention/mel/scan kernels.
+
+#include <hipblaslt/hipblaslt.h>
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include <mutex>
+#include <unordered_map>
+
+namespace {
+
+#define HIPBLASLT_CHECK(expr)                                          \
+  do {                                                                 \
+    hipblasStatus_t st_ = (expr);                                      \
+    TORCH_CHECK(st_ == HIPBLAS_STATUS_SUCCESS, "hipblaslt error ", st_, \
+                " at " #expr);                                         \
+  } while (0)
+
+hipblasLtHandle_t get_handle() {
+  static hipblasLtHandle_t handle = [] {
+    hipblasLtHandle_t h;
+    TORCH_CHECK(hipblasLtCreate(&h) == HIPBLAS_STATUS_SUCCESS,
+                "hipblasLtCreate failed");
+    return h;
+  }();
+  return handle;
+}
+
+struct AlgoKey {
+  int64_t m, n, k;
+  bool operator==(const AlgoKey& o) const {
+    return m == o.m && n == o.n && k == o.k;
+  }
+};
+struct AlgoKeyHash {
+  size_t operator()(const AlgoKey& k) const {
+    return std::hash<int64_t>()(k.m * 1315423911 ^ k.n * 2654435761 ^ k.k);
+  }
+};
+
+std::mutex algo_mu;
+std::unordered_map<AlgoKey, hipblasLtMatmulAlgo_t, AlgoKeyHash> algo_cache;
+
+}  // namespace
+
+// y = gelu(x @ w^T + bias); x (M, K) bf16 row-major, w (N, K), bias (N)
+static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
+                                 torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+                  x.is_contiguous(),
+              "x must be contiguous bf16 GPU");
+  TORCH_CHECK(w.is_contiguous() && bias.is_contiguous() &&
+                  w.scalar_type() == at::kBFloat16 &&
+                  bias.scalar_type() == at::kBFloat16,
+              "w/bias must be contiguous bf16");
+  const int64_t K = x.size(-1);
+  const int64_t M = x.numel() / K;
+  const int64_t N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && bias.numel() == N, "shape mismatch");
+
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = torch::empty(sizes, x.options());
+
+  // col-major framing: D(N,M) = op(A=w_rm seen cm (K,N), T) x op(B=x_cm (K,M), N)
+  hipblasLtMatmulDesc_t desc;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F,
+                                            HIP_R_32F));
+  hipblasOperation_t opA = HIPBLAS_OP_T, opB = HIPBLAS_OP_N;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
+  hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_GELU_BIAS;
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+  const void* bias_ptr = bias.data_ptr();
+  HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+      desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias_ptr, sizeof(bias_ptr)));
+
+  hipblasLtMatrixLayout_t la, lb, ld;
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, K, N, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, K, M, K));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&ld, HIP_R_16BF, N, M, N));
+
+  auto stream = c10::hip::getCurrentHIPStream();
+  static void* workspace = nullptr;
+  static size_t workspace_size = 64ull << 20;
+  static std::once_flag ws_once;
+  std::call_once(ws_once, [] {
+    TORCH_CHECK(hipMalloc(&workspace, workspace_size) == hipSuccess,
+                "workspace alloc failed");
+  });
+
+  hipblasLtMatmulAlgo_t algo;
+  bool have_algo = false;
+  {
+    std::lock_guard<std::mutex> g(algo_mu);
+    auto it = algo_cache.find({M, N, K});
+    if (it != algo_cache.end()) {
+      algo = it->second;
+      have_algo = true;
+    }
+  }
+  if (!have_algo) {
+    hipblasLtMatmulPreference_t pref;
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+    HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &workspace_size,
+        sizeof(workspace_size)));
+    hipblasLtMatmulHeuristicResult_t results[4];
+    int found = 0;
+    HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
+        get_handle(), desc, la, lb, ld, ld, pref, 4, results, &found));
+    hipblasLtMatmulPreferenceDestroy(pref);
+    TORCH_CHECK(found > 0, "no hipblaslt algo for gelu epilogue at ", M, "x",
+                N, "x", K);
+    algo = results[0].algo;
+    std::lock_guard<std::mutex> g(algo_mu);
+    algo_cache[{M, N, K}] = algo;
+  }
+
+  const float alpha = 1.0f, beta = 0.0f;
+  HIPBLASLT_CHECK(hipblasLtMatmul(
+      get_handle(), desc, &alpha, w.data_ptr(), la, x.data_ptr(), lb, &beta,
+      y.data_ptr(), ld, y.data_ptr(), ld, &algo, workspace, workspace_size,
+      stream.stream()));
+
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(ld);
+  hipblasLtMatmulDescDestroy(desc);
+  return y;
+}
+
+void register_gemm_gelu(pybind11::module_& m) {
+  m.def("linear_gelu", &linear_gelu,
+        "gelu(x @ w.T + bias) via hipBLASLt epilogue fusion");
+}

@@ -17,6 +17,7 @@ CSRC = os.path.join("audiomuse_amd", "ops", "csrc")
 
 sources = [
     os.path.join(CSRC, "bindings.cpp"),
+    os.path.join(CSRC, "gemm_gelu.cpp"),
     os.path.join(CSRC, "mel.hip"),
 ]
 for extra in ("distance.hip", "kmeans.hip", "attention.hip", "norms.hip"):
@@ -32,6 +33,7 @@ setup(
         CUDAExtension(
             "audiomuse_amd._C",
             sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],

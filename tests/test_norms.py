@@ -68,3 +68,18 @@ def test_fused_add_ln_matches_reference(dim):
                                rtol=2e-2, atol=2e-2)
     expect = F.layer_norm(expect_sum, (dim,), w.float(), b.float(), 1e-5)
     torch.testing.assert_close(y.float(), expect, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_linear_gelu_epilogue_matches_reference():
+    import audiomuse_amd._C as C
+
+    torch.manual_seed(0)
+    x = torch.randn(7, 33, 256, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(1024, 256, device="cuda", dtype=torch.bfloat16) * 0.05
+    b = torch.randn(1024, device="cuda", dtype=torch.bfloat16)
+    got = C.linear_gelu(x.contiguous(), w.contiguous(), b.contiguous())
+    expect = F.gelu(F.linear(x.float(), w.float(), b.float()),
+                    approximate="tanh")
+    assert got.shape == expect.shape
+    torch.testing.assert_close(got.float(), expect, rtol=3e-2, atol=3e-2)
