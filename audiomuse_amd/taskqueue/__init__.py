@@ -19,6 +19,7 @@ cancel) on the SQLite backend:
 
 from __future__ import annotations
 
+import hashlib
 import json
 import sqlite3
 import uuid
@@ -43,18 +44,51 @@ def enqueue(conn: sqlite3.Connection, task_type: str,
             payload: Optional[Dict[str, Any]] = None, *,
             queue: str = QUEUE_DEFAULT, parent_task_id: Optional[str] = None,
             priority: int = 0, task_id: Optional[str] = None,
-            max_attempts: Optional[int] = None) -> str:
-    """Insert a PENDING job (reference: taskqueue/__init__.py:159)."""
+            max_attempts: Optional[int] = None,
+            shared_blob: Optional[bytes] = None) -> str:
+    """Insert a PENDING job (reference: taskqueue/__init__.py:159).
+    Large artifacts ride the deduplicated shared_payload table via a
+    content token (reference: sql.py:652-705)."""
     task_id = task_id or uuid.uuid4().hex
+    token = None
     with write_txn(conn):
+        if shared_blob is not None:
+            token = hashlib.sha256(shared_blob).hexdigest()[:32]
+            conn.execute(
+                """INSERT INTO shared_payload (token, payload, refcount)
+                   VALUES (?,?,1)
+                   ON CONFLICT(token)
+                   DO UPDATE SET refcount = refcount + 1""",
+                (token, shared_blob))
         conn.execute(
             """INSERT INTO task_status (task_id, task_type, parent_task_id,
-                   queue, status, priority, payload, max_attempts)
-               VALUES (?,?,?,?,?,?,?,?)""",
+                   queue, status, priority, payload, shared_token,
+                   max_attempts)
+               VALUES (?,?,?,?,?,?,?,?,?)""",
             (task_id, task_type, parent_task_id, queue, PENDING, priority,
-             json.dumps(payload or {}),
+             json.dumps(payload or {}), token,
              max_attempts if max_attempts is not None else C.QUEUE_MAX_ATTEMPTS))
     return task_id
+
+
+def get_shared_blob(conn: sqlite3.Connection, task_id: str) -> Optional[bytes]:
+    row = conn.execute(
+        """SELECT p.payload FROM task_status t
+           JOIN shared_payload p ON p.token = t.shared_token
+           WHERE t.task_id = ?""", (task_id,)).fetchone()
+    return bytes(row["payload"]) if row else None
+
+
+def vacuum_shared_payloads(conn: sqlite3.Connection) -> int:
+    """Drop blobs no live (non-terminal) task references (reference:
+    maintenance.py blob VACUUM; test_taskqueue_blob_reclaim)."""
+    with write_txn(conn):
+        cur = conn.execute(
+            """DELETE FROM shared_payload WHERE token NOT IN (
+                   SELECT shared_token FROM task_status
+                   WHERE shared_token IS NOT NULL
+                     AND status IN (?, ?))""", (PENDING, RUNNING))
+    return cur.rowcount
 
 
 def task_row(conn: sqlite3.Connection, task_id: str) -> Optional[sqlite3.Row]:

@@ -200,3 +200,25 @@ def test_reclaim_stands_down_in_control_window(db):
     # maintenance policy: the worker loop skips reclaim inside the window;
     # direct reclaim still works (it is the policy gate, not the SQL)
     assert task_row(conn, tid)["status"] == RUNNING
+
+
+def test_shared_payload_dedupe_and_vacuum(db):
+    from audiomuse_amd.taskqueue import (get_shared_blob,
+                                         vacuum_shared_payloads)
+
+    conn, _ = db
+    blob = b"x" * 10000
+    t1 = enqueue(conn, "noop", shared_blob=blob)
+    t2 = enqueue(conn, "noop", shared_blob=blob)      # dedupes to 1 row
+    row = conn.execute("SELECT COUNT(*) AS n, MAX(refcount) AS rc "
+                       "FROM shared_payload").fetchone()
+    assert row["n"] == 1 and row["rc"] == 2
+    assert get_shared_blob(conn, t1) == blob
+    # live tasks keep the blob
+    assert vacuum_shared_payloads(conn) == 0
+    for t in (t1, t2):
+        qsql.claim(conn, "w")
+    qsql.finish(conn, t1, "w", SUCCESS)
+    qsql.finish(conn, t2, "w", SUCCESS)
+    assert vacuum_shared_payloads(conn) == 1
+    assert get_shared_blob(conn, t1) is None
