@@ -70,8 +70,39 @@ class AnalysisRuntime:
             embedder = TextEmbedder(gte_config(), device=str(self.device),
                                     dtype=self.dtype)
             vad = SileroStyleVAD().to(self.device).eval()
-            self._lyrics = LyricsPipeline(embedder, vad=vad, asr_fn=None)
+            asr_fn = self._make_asr_fn() if C.LYRICS_ASR_ENABLED else None
+            self._lyrics = LyricsPipeline(embedder, vad=vad, asr_fn=asr_fn)
         return self._lyrics
+
+    def _make_asr_fn(self):
+        """Whisper greedy/beam decode over 30 s chunks
+        (whisper_onnx.py:173-178 chunking; random-init weights produce
+        gate-rejected transcripts until trained weights are loaded)."""
+        from audiomuse_amd.models.whisper import (WhisperModel, beam_decode,
+                                                  greedy_decode)
+        from audiomuse_amd.ops.dsp import whisper_mel_config
+
+        model = WhisperModel().to(self.device, self.dtype).eval()
+        mel_cfg = whisper_mel_config()
+
+        def asr(audio16k: torch.Tensor) -> str:
+            chunk = C.WHISPER_CHUNK_SECONDS * C.WHISPER_SAMPLE_RATE
+            tokens = []
+            for s0 in range(0, audio16k.shape[-1], chunk):
+                seg = audio16k[s0 : s0 + chunk].to(self.device)
+                mel = hip_ops.mel_spectrogram(seg, mel_cfg,
+                                              force_reference=True)
+                mel = mel.to(self.dtype)
+                if C.LYRICS_ASR_BEAM_SIZE > 1:
+                    toks = beam_decode(model, mel,
+                                       beam=C.LYRICS_ASR_BEAM_SIZE)
+                else:
+                    toks = greedy_decode(model, mel)
+                tokens.extend(toks)
+            # token ids -> placeholder wordpieces (no trained vocab in-image)
+            return " ".join(f"tok{t}" for t in tokens)
+
+        return asr
 
     # -- label text embeddings (clap_analyzer.py:580: cached .npz) -------
 

@@ -81,6 +81,7 @@ CLAP_AUDIO_FMAX = _env_float("CLAP_AUDIO_FMAX", 14000.0)
 CLAP_EMBEDDING_DIMENSION = _env_int("CLAP_EMBEDDING_DIMENSION", 512)
 CLAP_ENABLED = _env_bool("CLAP_ENABLED", True)
 LYRICS_ENABLED = _env_bool("LYRICS_ENABLED", False)
+LYRICS_ASR_ENABLED = _env_bool("LYRICS_ASR_ENABLED", False)
 
 # MusiCNN mel (reference: song.py:240-256)
 MUSICNN_SAMPLE_RATE = 16000
