@@ -100,7 +100,12 @@ class AppState:
 
 def create_app(db_url: Optional[str] = None, device: str = "cpu",
                auth_disabled: bool = False) -> Flask:
-    app = Flask("audiomuse_amd")
+    import os as _os
+
+    app = Flask("audiomuse_amd",
+                static_folder=_os.path.join(_os.path.dirname(
+                    _os.path.abspath(__file__)), "static"),
+                static_url_path="/static")
     state = AppState(db_url or C.DATABASE_URL, device=device)
     app.extensions["audiomuse"] = state
     app.config["AUTH_DISABLED"] = auth_disabled
@@ -170,6 +175,10 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             cron_loop(conn, _stop)
 
         _threading.Thread(target=_cron_thread, daemon=True).start()
+
+    @app.get("/")
+    def ui_index():  # L7: minimal first-party UI over the API
+        return app.send_static_file("index.html")
 
     @app.get("/health")
     def health():  # reference: app.py:227

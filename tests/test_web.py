@@ -225,3 +225,10 @@ def test_users_endpoint(client_ids):
     client, _ = client_ids
     r = client.get("/api/users")
     assert r.status_code == 200
+
+
+def test_ui_served(client_ids):
+    client, _ = client_ids
+    r = client.get("/")
+    assert r.status_code == 200
+    assert b"AudioMuse-AMD" in r.data and b"/api/map" in r.data
