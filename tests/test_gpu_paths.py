@@ -81,3 +81,39 @@ def test_distill_trainer_gpu_step():
     t = DistillTrainer(DistillConfig(batch=8), device="cuda")
     losses = [t.step(i) for i in range(3)]
     assert all(np.isfinite(v) for v in losses)
+
+
+@pytest.mark.gpu
+def test_worker_analysis_end_to_end_on_gpu(tmp_path):
+    """Full queue-driven analysis on the GPU: synthetic provider ->
+    worker -> native mel/HTSAT -> catalogue + indexes (the deployment
+    path the driver's round-end native-code check exercises)."""
+    import audiomuse_amd.analysis.tasks as atasks
+    from audiomuse_amd.analysis.index import AUDIO_INDEX, load_ivf_engine
+    from audiomuse_amd.db import connect
+    from audiomuse_amd.db.schema import init_db
+    from audiomuse_amd.taskqueue import SUCCESS, enqueue, task_row
+    from audiomuse_amd.taskqueue.worker import Worker
+
+    url = f"sqlite:///{tmp_path}/gpu_e2e.db"
+    conn = connect(url)
+    init_db(conn)
+    atasks._RUNTIME = None
+    tid = enqueue(conn, "analyze_album", {
+        "server_type": "synthetic", "server_id": "gpu-srv",
+        "server_config": {"n_albums": 1, "tracks_per_album": 3,
+                          "seconds": 11.0, "sr": 44100},
+        "album_id": "a0"}, queue="high")
+    Worker(db_url=url, max_jobs=1).run_forever(idle_timeout=10.0)
+    row = task_row(conn, tid)
+    assert row["status"] == SUCCESS, row["result"]
+    n_clap = conn.execute("SELECT COUNT(*) FROM clap_embedding").fetchone()[0]
+    assert n_clap >= 1                      # CLAP ran on the GPU path
+    from audiomuse_amd.analysis.index import run_all_index_builds
+
+    built = run_all_index_builds(conn, device="cuda")
+    assert built["audio"] >= 1 and built["clap"] >= 1
+    eng = load_ivf_engine(conn, AUDIO_INDEX, device="cuda")
+    assert eng is not None
+    conn.close()
+    atasks._RUNTIME = None
