@@ -74,8 +74,47 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
   }
   __syncthreads();
 
-  // Phase 2: radix-2 DIT FFT, LOG2N stages, NFFT/2 butterflies each.
-  for (int s = 1; s <= LOG2N; ++s) {
+  // Phase 2: DIT FFT over bit-reversed data, two radix-2 stages merged
+  // into one radix-4 pass (one LDS round-trip + one barrier per pair —
+  // measured: the stage loop dominated the kernel's VALU+LDS time).
+  int s = 1;
+  for (; s + 1 <= LOG2N; s += 2) {
+    const int h = 1 << (s - 1);          // stage-s half
+    for (int q = tid; q < NFFT / 4; q += blockDim.x) {
+      const int grp = q >> (s - 1);
+      const int j = q & (h - 1);
+      const int i0 = (grp << (s + 1)) + j;
+      // twiddles: w1 = W(j, 2h), w2 = W(j, 4h), w3 = -i * w2
+      const float2 w1 = twiddle[j * (NFFT >> s)];
+      const float2 w2 = twiddle[j * (NFFT >> (s + 1))];
+      float2 x0 = zbuf[AM_ZS(i0)];
+      float2 x1 = zbuf[AM_ZS(i0 + h)];
+      float2 x2 = zbuf[AM_ZS(i0 + 2 * h)];
+      float2 x3 = zbuf[AM_ZS(i0 + 3 * h)];
+      // stage s: (x0,x1) and (x2,x3), both with twiddle w1
+      float tr = w1.x * x1.x - w1.y * x1.y;
+      float ti = w1.x * x1.y + w1.y * x1.x;
+      const float2 a0 = make_float2(x0.x + tr, x0.y + ti);
+      const float2 a1 = make_float2(x0.x - tr, x0.y - ti);
+      tr = w1.x * x3.x - w1.y * x3.y;
+      ti = w1.x * x3.y + w1.y * x3.x;
+      const float2 a2 = make_float2(x2.x + tr, x2.y + ti);
+      const float2 a3 = make_float2(x2.x - tr, x2.y - ti);
+      // stage s+1: (a0,a2) with w2; (a1,a3) with w3 = -i*w2
+      tr = w2.x * a2.x - w2.y * a2.y;
+      ti = w2.x * a2.y + w2.y * a2.x;
+      zbuf[AM_ZS(i0)] = make_float2(a0.x + tr, a0.y + ti);
+      zbuf[AM_ZS(i0 + 2 * h)] = make_float2(a0.x - tr, a0.y - ti);
+      const float w3x = w2.y, w3y = -w2.x;   // -i * w2
+      tr = w3x * a3.x - w3y * a3.y;
+      ti = w3x * a3.y + w3y * a3.x;
+      zbuf[AM_ZS(i0 + h)] = make_float2(a1.x + tr, a1.y + ti);
+      zbuf[AM_ZS(i0 + 3 * h)] = make_float2(a1.x - tr, a1.y - ti);
+    }
+    __syncthreads();
+  }
+  // leftover radix-2 stage when LOG2N is odd
+  for (; s <= LOG2N; ++s) {
     const int half = 1 << (s - 1);
     const int tw_step = NFFT >> s;
     for (int bf = tid; bf < NFFT / 2; bf += blockDim.x) {

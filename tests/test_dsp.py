@@ -86,8 +86,9 @@ def test_torch_reference_matches_numpy(cfg_fn):
 
 
 def _kernel_mirror_fft(x):
-    """Numpy mirror of the radix-2 DIT FFT in ops/csrc/mel.hip (same
-    bit-reversal, twiddle layout and butterfly indexing)."""
+    """Numpy mirror of the DIT FFT in ops/csrc/mel.hip: bit-reversal,
+    then merged radix-4 passes (two radix-2 stages per LDS round-trip)
+    with a leftover radix-2 stage for odd log2(N)."""
     n = len(x)
     log2n = n.bit_length() - 1
     assert 1 << log2n == n
@@ -96,17 +97,32 @@ def _kernel_mirror_fft(x):
     rev = np.array([int(f"{i:0{log2n}b}"[::-1], 2) for i in range(n)])
     z = np.zeros(n, dtype=np.complex128)
     z[rev] = x
-    for s in range(1, log2n + 1):
+    s = 1
+    while s + 1 <= log2n:
+        h = 1 << (s - 1)
+        for q in range(n // 4):
+            grp, jj = q >> (s - 1), q & (h - 1)
+            i0 = (grp << (s + 1)) + jj
+            w1 = tw[jj * (n >> s)]
+            w2 = tw[jj * (n >> (s + 1))]
+            x0, x1, x2, x3 = z[i0], z[i0 + h], z[i0 + 2 * h], z[i0 + 3 * h]
+            a0, a1 = x0 + w1 * x1, x0 - w1 * x1
+            a2, a3 = x2 + w1 * x3, x2 - w1 * x3
+            w3 = -1j * w2
+            z[i0], z[i0 + 2 * h] = a0 + w2 * a2, a0 - w2 * a2
+            z[i0 + h], z[i0 + 3 * h] = a1 + w3 * a3, a1 - w3 * a3
+        s += 2
+    while s <= log2n:
         half = 1 << (s - 1)
-        tw_step = n >> s
+        step = n >> s
         for bf in range(n // 2):
             grp, jj = bf >> (s - 1), bf & (half - 1)
             i0 = (grp << s) + jj
             i1 = i0 + half
-            w = tw[jj * tw_step]
+            w = tw[jj * step]
             a, c = z[i0], z[i1]
-            t = w * c
-            z[i0], z[i1] = a + t, a - t
+            z[i0], z[i1] = a + w * c, a - w * c
+        s += 1
     return z
 
 
