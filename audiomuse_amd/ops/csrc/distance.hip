@@ -29,13 +29,16 @@
 
 namespace audiomuse {
 
-__device__ __forceinline__ float wave_reduce_sum(float v) {
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+// 16-lane sub-group reductions: each wave scans 4 rows concurrently
+// (4x fewer shuffle steps per row than a full-wave reduce, full
+// coalescing preserved: 16 lanes x 4 B = one 64 B segment per row).
+__device__ __forceinline__ float sub_reduce_sum(float v) {
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 16);
   return v;
 }
 
-__device__ __forceinline__ int wave_reduce_sum_i32(int v) {
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+__device__ __forceinline__ int sub_reduce_sum_i32(int v) {
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 16);
   return v;
 }
 
@@ -65,6 +68,8 @@ __global__ __launch_bounds__(256) void ivf_scan_i8_angular(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
+  const int sub = lane >> 4;       // row slot within the wave (0..3)
+  const int sl = lane & 15;        // lane within the 16-lane sub-group
 
   extern __shared__ int qs[];  // d/4 ints
   for (int i = threadIdx.x; i < d4; i += blockDim.x)
@@ -76,13 +81,13 @@ __global__ __launch_bounds__(256) void ivf_scan_i8_angular(
   float* dq = out_dist + (long long)q * cap;
   int* rq = out_row + (long long)q * cap;
 
-  for (int r = r0 + wave; r < r1; r += nwaves) {
+  for (int r = r0 + wave * 4 + sub; r < r1; r += nwaves * 4) {
     const int* row = data + (long long)r * d4;
     int acc = 0;
-    for (int j = lane; j < d4; j += 64)
+    for (int j = sl; j < d4; j += 16)
       acc = __builtin_amdgcn_sdot4(qs[j], row[j], acc, false);
-    acc = wave_reduce_sum_i32(acc);
-    if (lane == 0) {
+    acc = sub_reduce_sum_i32(acc);
+    if (sl == 0) {
       const float denom = qn * row_norm[r] + 1e-12f;
       float cosv = (float)acc / denom;
       cosv = fminf(1.0f, fmaxf(-1.0f, cosv));
@@ -113,6 +118,8 @@ __global__ __launch_bounds__(256) void ivf_scan_float(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
+  const int sub = lane >> 4;
+  const int sl = lane & 15;
 
   extern __shared__ float qf[];  // d floats
   for (int i = threadIdx.x; i < d; i += blockDim.x)
@@ -124,10 +131,10 @@ __global__ __launch_bounds__(256) void ivf_scan_float(
   float* dq = out_dist + (long long)q * cap;
   int* rq = out_row + (long long)q * cap;
 
-  for (int r = r0 + wave; r < r1; r += nwaves) {
+  for (int r = r0 + wave * 4 + sub; r < r1; r += nwaves * 4) {
     const T* row = data + (long long)r * d;
     float acc = 0.0f;
-    for (int j = lane; j < d; j += 64) {
+    for (int j = sl; j < d; j += 16) {
       const float v = to_f32(row[j]);
       if (METRIC == 1) {
         const float diff = v - qf[j];
@@ -136,8 +143,8 @@ __global__ __launch_bounds__(256) void ivf_scan_float(
         acc += v * qf[j];
       }
     }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) {
+    acc = sub_reduce_sum(acc);
+    if (sl == 0) {
       float dist;
       if (METRIC == 0) {
         float cosv = acc / (qn * row_norm[r] + 1e-12f);
