@@ -222,3 +222,31 @@ def test_shared_payload_dedupe_and_vacuum(db):
     qsql.finish(conn, t2, "w", SUCCESS)
     assert vacuum_shared_payloads(conn) == 1
     assert get_shared_blob(conn, t1) is None
+
+
+@pytest.mark.slow
+def test_crashed_fork_worker_job_is_reclaimed(db, monkeypatch):
+    """Fault injection: a worker process dying mid-job
+    leaves the row RUNNING with a decaying lease; maintenance requeues it
+    (reference: advisory lock released by connection death)."""
+    import os
+
+    from audiomuse_amd import config as C
+    from audiomuse_amd.taskqueue.worker import run_in_subprocess
+
+    conn, url = db
+    monkeypatch.setattr(C, "QUEUE_LEASE_SECONDS", 0.3)
+    monkeypatch.setattr(C, "QUEUE_HEARTBEAT_SECONDS", 10.0)
+
+    @task_handler("crash_job")
+    def crash(ctx, payload):
+        os._exit(13)           # simulated hard crash (no cleanup)
+
+    tid = enqueue(conn, "crash_job")
+    rc = run_in_subprocess(url, ("high", "default"), max_jobs=1)
+    assert rc == 13
+    row = task_row(conn, tid)
+    assert row["status"] == RUNNING          # stuck: worker died holding it
+    time.sleep(0.4)                          # lease decays
+    assert qsql.reclaim_orphans(conn) == 1
+    assert task_row(conn, tid)["status"] == PENDING
