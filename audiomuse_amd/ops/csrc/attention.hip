@@ -173,6 +173,15 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
     float rmax[4][4];  // [tr][reg]
 #pragma unroll
     for (int tr = 0; tr < 4; ++tr) {
+      // issue this row-tile's 16 independent bias loads up front — the
+      // fmax chain otherwise serializes on one ~500-cycle L2 hit at a time
+      float bvals[4][4];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc)
+          bvals[reg][tc] = bias[(h * 64 + tr * 16 + row_grp + reg) * 64 +
+                                tc * 16 + col_in_tile];
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int row = tr * 16 + row_grp + reg;
@@ -184,7 +193,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
           const int col = tc * 16 + col_in_tile;
           const int cwrap = (((wrap_r_mask >> col) & 1ull) << 1) |
                             ((wrap_c_mask >> col) & 1ull);
-          float v = s[tr][tc][reg] * scale + bias[(h * 64 + row) * 64 + col];
+          float v = s[tr][tc][reg] * scale + bvals[reg][tc];
           if (shift && rwrap != cwrap) v = -1e30f;
           s[tr][tc][reg] = v;
           m = fmaxf(m, v);
