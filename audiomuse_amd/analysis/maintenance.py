@@ -140,6 +140,55 @@ def clean_orphans_task(ctx: TaskContext, payload: Dict) -> Dict:
     return {"orphans": len(orphans), "deleted": deleted}
 
 
+@task_handler("sonic_fingerprint")
+def sonic_fingerprint_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Cron-able fingerprint playlist (reference:
+    sonic_fingerprint_manager cron task :33): top-played -> recency
+    weights -> taste vector -> IVF expansion -> stored playlist."""
+    import json as _json
+
+    import numpy as np
+    import torch
+
+    from audiomuse_amd.analysis.index import AUDIO_INDEX, load_ivf_engine
+    from audiomuse_amd.engines.misc import sonic_fingerprint
+
+    conn = ctx.conn
+    provider = make_provider(payload.get("server_type", "synthetic"),
+                             **payload.get("server_config", {}))
+    server_id = payload.get("server_id", "default")
+    eng = load_ivf_engine(conn, AUDIO_INDEX)
+    if eng is None:
+        return {"error": "audio index not built"}
+    top = provider.get_top_played_songs(payload.get("top_n", 100))
+    mapped = {r["provider_id"]: r["item_id"] for r in conn.execute(
+        "SELECT provider_id, item_id FROM track_server_map WHERE server_id=?",
+        (server_id,))}
+    vecs, times = [], []
+    for t in top:
+        iid = mapped.get(t.provider_id)
+        v = eng.vector_for_id(iid) if iid else None
+        if v is None:
+            continue
+        vecs.append(v.cpu().numpy())
+        times.append(provider.get_last_played_time(t.provider_id) or 0.0)
+    if not vecs:
+        return {"tracks": 0}
+    fp = sonic_fingerprint(np.stack(vecs), times)
+    res = eng.find_similar_by_vector(torch.from_numpy(fp),
+                                     payload.get("n", 30))
+    name = payload.get("name", "Sonic Fingerprint_automatic")
+    with write_txn(conn):
+        conn.execute("DELETE FROM playlist WHERE name=?", (name,))
+        conn.execute(
+            "INSERT INTO playlist (name, server_id, item_ids, kind) "
+            "VALUES (?,?,?, 'fingerprint')",
+            (name, server_id, _json.dumps([r["item_id"] for r in res])))
+        conn.execute("INSERT INTO playlist_name_history (name) VALUES (?)",
+                     (name,))
+    return {"tracks": len(res)}
+
+
 # -- backup / restore (app_backup.py analog) --------------------------------
 
 def backup_database(conn: sqlite3.Connection, dest_path: str) -> None:

@@ -136,6 +136,8 @@ def run_clustering_task(ctx: TaskContext, payload: Dict) -> Dict:
             conn.execute(
                 "INSERT INTO playlist (name, item_ids, kind) VALUES (?,?,?)",
                 (name, json.dumps(ids), "automatic"))
+            conn.execute(
+                "INSERT INTO playlist_name_history (name) VALUES (?)", (name,))
     if payload.get("server_type"):
         try:
             from audiomuse_amd.mediaserver import make_provider

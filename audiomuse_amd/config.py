@@ -210,30 +210,38 @@ def set_db_override_provider(provider: Optional[Callable[[], Dict[str, str]]]) -
     _DB_OVERRIDE_PROVIDER = provider
 
 
-def refresh_config() -> None:
-    """Re-read env and re-apply DB overrides (reference: config.py:1389)."""
-    import importlib
+def apply_db_overrides(overrides: Dict[str, str]) -> int:
+    """Layer persisted overrides onto the module (reference:
+    config._apply_db_overrides :1395). Non-destructive: only keys that
+    already exist change; types coerce to the current value's type."""
     import sys
 
     module = sys.modules[__name__]
-    provider = _DB_OVERRIDE_PROVIDER
-    importlib.reload(module)
-    if provider is not None:
-        module.set_db_override_provider(provider)
-        overrides = provider() or {}
-        for key, raw in overrides.items():
-            if not hasattr(module, key):
-                continue
-            current = getattr(module, key)
-            try:
-                if isinstance(current, bool):
-                    value = str(raw).strip().lower() in ("1", "true", "yes", "on")
-                elif isinstance(current, int):
-                    value = int(raw)
-                elif isinstance(current, float):
-                    value = float(raw)
-                else:
-                    value = raw
-            except (TypeError, ValueError):
-                continue
-            setattr(module, key, value)
+    applied = 0
+    for key, raw in (overrides or {}).items():
+        if not key.isupper() or not hasattr(module, key):
+            continue
+        current = getattr(module, key)
+        try:
+            if isinstance(current, bool):
+                value = str(raw).strip().lower() in ("1", "true", "yes", "on")
+            elif isinstance(current, int):
+                value = int(raw)
+            elif isinstance(current, float):
+                value = float(raw)
+            elif isinstance(current, str):
+                value = str(raw)
+            else:
+                continue  # lists/complex values are code-owned
+        except (TypeError, ValueError):
+            continue
+        setattr(module, key, value)
+        applied += 1
+    return applied
+
+
+def refresh_config() -> None:
+    """Re-apply the registered DB override provider
+    (reference: config.refresh_config :1389)."""
+    if _DB_OVERRIDE_PROVIDER is not None:
+        apply_db_overrides(_DB_OVERRIDE_PROVIDER() or {})

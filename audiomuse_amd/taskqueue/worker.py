@@ -145,6 +145,14 @@ class Worker:
             import_builtin_handlers()
         except Exception:
             logger.exception("builtin handler import failed")
+        try:
+            # hydrate persisted config overrides (reference: worker.py:698)
+            from audiomuse_amd.db.store import get_app_config
+
+            C.set_db_override_provider(lambda: get_app_config(connect(self.db_url)))
+            C.refresh_config()
+        except Exception:
+            logger.exception("config hydrate failed")
         idle_since = time.time()
         last_maintenance = 0.0
         try:

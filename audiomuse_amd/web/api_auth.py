@@ -103,4 +103,7 @@ def set_config():
         if not isinstance(k, str) or not k.isupper():
             return jsonify({"error": f"bad key {k!r}"}), 400
         set_app_config(conn, k, str(v))
+    from audiomuse_amd import config as C
+
+    C.refresh_config()      # workers re-hydrate per job (worker loop)
     return jsonify({"saved": len(body)})

@@ -124,6 +124,13 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
 
     from audiomuse_amd.web.auth import seed_admin_from_env
 
+    # persisted config overrides (reference: config._apply_db_overrides)
+    from audiomuse_amd import config as _C
+    from audiomuse_amd.db.store import get_app_config as _gac
+
+    _C.set_db_override_provider(lambda: _gac(state.conn()))
+    _C.refresh_config()
+
     with app.app_context():
         seed_admin_from_env(state.conn())
         # inline boot migrations (reference boot sequence)
@@ -187,6 +194,19 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             return jsonify({"status": "ok"})
         except Exception as exc:  # noqa: BLE001
             return jsonify({"status": "error", "detail": str(exc)}), 500
+
+    @app.get("/api/spec")
+    def api_spec():  # reference: Swagger at /apidocs (flasgger)
+        out = []
+        for rule in app.url_map.iter_rules():
+            if rule.endpoint == "static":
+                continue
+            fn = app.view_functions[rule.endpoint]
+            out.append({"path": str(rule),
+                        "methods": sorted(m for m in rule.methods
+                                          if m not in ("HEAD", "OPTIONS")),
+                        "doc": (fn.__doc__ or "").strip().split("\n")[0]})
+        return jsonify(sorted(out, key=lambda r: r["path"]))
 
     @app.get("/api/index_profile")
     def index_profile():  # reference: _log_startup_index_profile app.py:1244

@@ -232,3 +232,28 @@ def test_ui_served(client_ids):
     r = client.get("/")
     assert r.status_code == 200
     assert b"AudioMuse-AMD" in r.data and b"/api/map" in r.data
+
+
+def test_config_override_round_trip(client_ids):
+    from audiomuse_amd import config as C
+
+    client, _ = client_ids
+    original = C.IVF_RERANK_OVERFETCH
+    try:
+        r = client.post("/api/config", json={"IVF_RERANK_OVERFETCH": "7"})
+        assert r.status_code == 200 and r.json["saved"] == 1
+        assert C.IVF_RERANK_OVERFETCH == 7
+        body = client.get("/api/config").json
+        assert body["overrides"]["IVF_RERANK_OVERFETCH"] == "7"
+        assert client.post("/api/config",
+                           json={"lowercase": "x"}).status_code == 400
+    finally:
+        C.IVF_RERANK_OVERFETCH = original
+
+
+def test_api_spec_lists_routes(client_ids):
+    client, _ = client_ids
+    spec = client.get("/api/spec").json
+    paths = {r["path"] for r in spec}
+    assert "/api/similar_tracks" in paths and "/chat/api/chatPlaylist" in paths
+    assert all("methods" in r for r in spec)
