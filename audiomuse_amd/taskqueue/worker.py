@@ -149,8 +149,7 @@ class Worker:
             # hydrate persisted config overrides (reference: worker.py:698)
             from audiomuse_amd.db.store import get_app_config
 
-            C.set_db_override_provider(lambda: get_app_config(connect(self.db_url)))
-            C.refresh_config()
+            C.apply_db_overrides(get_app_config(conn))
         except Exception:
             logger.exception("config hydrate failed")
         idle_since = time.time()

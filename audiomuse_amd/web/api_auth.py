@@ -104,6 +104,7 @@ def set_config():
             return jsonify({"error": f"bad key {k!r}"}), 400
         set_app_config(conn, k, str(v))
     from audiomuse_amd import config as C
+    from audiomuse_amd.db.store import get_app_config
 
-    C.refresh_config()      # workers re-hydrate per job (worker loop)
+    C.apply_db_overrides(get_app_config(conn))   # workers hydrate on loop
     return jsonify({"saved": len(body)})

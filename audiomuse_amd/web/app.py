@@ -128,8 +128,7 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
     from audiomuse_amd import config as _C
     from audiomuse_amd.db.store import get_app_config as _gac
 
-    _C.set_db_override_provider(lambda: _gac(state.conn()))
-    _C.refresh_config()
+    _C.apply_db_overrides(_gac(state.conn()))
 
     with app.app_context():
         seed_admin_from_env(state.conn())
