@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import logging
 import sqlite3
-from typing import Dict, List, Tuple
+from typing import Dict
 
 import numpy as np
 

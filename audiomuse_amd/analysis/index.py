@@ -11,10 +11,9 @@ reloads by watching ivf_dir.updated_at (the LISTEN/NOTIFY analog).
 from __future__ import annotations
 
 import io
-import json
 import logging
 import sqlite3
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch

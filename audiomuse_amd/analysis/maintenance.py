@@ -20,7 +20,7 @@ import json
 import re
 import sqlite3
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List
 
 from audiomuse_amd.db import connect, write_txn
 from audiomuse_amd.mediaserver import make_provider

@@ -22,7 +22,7 @@ import torch
 from audiomuse_amd import config as C
 from audiomuse_amd.analysis.index import run_all_index_builds
 from audiomuse_amd.analysis.pipeline import AnalysisRuntime
-from audiomuse_amd.db import get_db, write_txn
+from audiomuse_amd.db import write_txn
 from audiomuse_amd.db.store import (save_clap_embedding,
                                     save_lyrics_embedding,
                                     save_track_analysis_and_embedding)
@@ -30,7 +30,7 @@ from audiomuse_amd.engines.simhash import CatalogResolver
 from audiomuse_amd.mediaserver import make_provider
 from audiomuse_amd.taskqueue import enqueue
 from audiomuse_amd.taskqueue import sql as qsql
-from audiomuse_amd.taskqueue.worker import CancelledError, TaskContext, task_handler
+from audiomuse_amd.taskqueue.worker import TaskContext, task_handler
 
 logger = logging.getLogger(__name__)
 

@@ -16,10 +16,9 @@ reference's Postgres-mediated reduction).
 
 from __future__ import annotations
 
-import json
 import random
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Sequence
 
 import numpy as np
 import torch

@@ -12,7 +12,7 @@ sampling — vectorized over all edges per epoch on the device.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

@@ -21,7 +21,7 @@ from __future__ import annotations
 import threading
 from typing import Dict, List, Optional
 
-from audiomuse_amd.mediaserver.base import Album, Provider, Track
+from audiomuse_amd.mediaserver.base import Provider
 
 _PROVIDERS: Dict[str, type] = {}
 _LOCAL = threading.local()

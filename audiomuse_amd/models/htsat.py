@@ -24,8 +24,7 @@ fused HIP kernel on GPU and a torch reference on CPU.
 
 from __future__ import annotations
 
-import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Tuple
 
 import torch

@@ -16,7 +16,6 @@ fixed-shape (hipGraph-capturable later).
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import List, Optional, Tuple
 

@@ -19,7 +19,6 @@ from __future__ import annotations
 import torch
 import torch.nn.functional as F
 
-from audiomuse_amd.ops import _ext
 
 
 def _sdpa_reference(q, k, v, bias, mask, scale):

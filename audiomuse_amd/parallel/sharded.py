@@ -11,7 +11,7 @@ bucketing games.
 
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist

@@ -12,7 +12,7 @@ from __future__ import annotations
 import json
 import sqlite3
 import time
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from audiomuse_amd.db import write_txn
 from audiomuse_amd.taskqueue import enqueue

@@ -8,7 +8,7 @@ to the UI, tracebacks only to logs (error_manager.py:112-177).
 from __future__ import annotations
 
 import logging
-from typing import Optional, Tuple
+from typing import Tuple
 
 logger = logging.getLogger(__name__)
 

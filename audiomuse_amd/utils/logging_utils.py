@@ -12,7 +12,6 @@ import ipaddress
 import logging
 import re
 import sys
-from typing import Optional
 from urllib.parse import urlparse
 
 _CTRL = re.compile(r"[\x00-\x1f\x7f]")

@@ -12,7 +12,6 @@ from __future__ import annotations
 import ctypes
 import gc
 import os
-from typing import Optional
 
 import torch
 

@@ -19,7 +19,6 @@ import json
 import re
 from typing import Dict, List, Optional
 
-import torch
 from flask import Blueprint, current_app, jsonify, request
 
 from audiomuse_amd import config as C

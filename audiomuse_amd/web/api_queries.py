@@ -11,7 +11,7 @@ SemGrove (app_sem_grove.py), sonic fingerprint
 from __future__ import annotations
 
 import json
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 import torch

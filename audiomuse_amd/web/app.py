@@ -14,7 +14,6 @@ import threading
 import time
 from typing import Dict, Optional
 
-import numpy as np
 from flask import Flask, jsonify
 
 from audiomuse_amd import config as C
