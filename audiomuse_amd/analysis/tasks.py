@@ -46,7 +46,21 @@ def get_runtime() -> AnalysisRuntime:
     return _RUNTIME
 
 
+_RESOLVER: Optional[CatalogResolver] = None
+_RESOLVER_EXPECT_N = -1
+
+
 def _resolver_from_db(conn) -> CatalogResolver:
+    """Per-worker cached resolver. A fresh load is O(catalogue); reloading
+    per album would be O(N^2) across a library scan. The cache tracks the
+    ids it minted itself; if the DB grew beyond that, a sibling worker
+    wrote rows we have not seen -> rebuild. A short blind window between
+    siblings can mint duplicate canonical ids; the duplicate-repair
+    migration collapses those (reference tolerates the same race)."""
+    global _RESOLVER, _RESOLVER_EXPECT_N
+    n = conn.execute("SELECT COUNT(*) FROM embedding").fetchone()[0]
+    if _RESOLVER is not None and n <= _RESOLVER_EXPECT_N:
+        return _RESOLVER
     resolver = CatalogResolver()
     rows = conn.execute(
         """SELECT e.item_id, e.embedding, s.duration FROM embedding e
@@ -54,6 +68,8 @@ def _resolver_from_db(conn) -> CatalogResolver:
     for r in rows:
         emb = np.frombuffer(r["embedding"], dtype=np.float32)
         resolver.register_existing(r["item_id"], emb, r["duration"] or 0.0)
+    _RESOLVER = resolver
+    _RESOLVER_EXPECT_N = n
     return resolver
 
 
@@ -92,6 +108,9 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
             continue
         item_id, _matched = resolver.resolve(
             res.embedding, res.duration, server_id, track.provider_id)
+        global _RESOLVER_EXPECT_N
+        if not _matched:
+            _RESOLVER_EXPECT_N += 1     # we will insert one embedding row
         save_track_analysis_and_embedding(
             conn, item_id, title=track.title, author=track.author,
             album=track.album, tempo=res.tempo, key=res.key, scale=res.scale,
