@@ -257,3 +257,38 @@ def test_api_spec_lists_routes(client_ids):
     paths = {r["path"] for r in spec}
     assert "/api/similar_tracks" in paths and "/chat/api/chatPlaylist" in paths
     assert all("methods" in r for r in spec)
+
+
+def test_engine_cache_reloads_on_rebuild(tmp_path):
+    """The web engine cache must pick up a rebuilt index (reference:
+    listen_for_index_reloads, app.py:971)."""
+    import numpy as np
+
+    from audiomuse_amd.analysis.index import AUDIO_INDEX, build_audio_index
+    from audiomuse_amd.web.app import AppState
+
+    url = f"sqlite:///{tmp_path}/reload.db"
+    conn = connect(url)
+    init_db(conn)
+    rng = np.random.default_rng(0)
+    for i in range(10):
+        save_track_analysis_and_embedding(
+            conn, f"fp_4{'%050x' % i}", title=f"T{i}", author="A",
+            embedding=rng.standard_normal(200).astype(np.float32))
+    build_audio_index(conn)
+    state = AppState(url)
+    eng1 = state.engine(AUDIO_INDEX)
+    assert eng1 is not None and eng1.index.n == 10
+    assert state.engine(AUDIO_INDEX) is eng1          # cached
+    # grow + rebuild -> stamp changes -> new engine object
+    import time as _t
+
+    _t.sleep(0.01)
+    for i in range(10, 15):
+        save_track_analysis_and_embedding(
+            conn, f"fp_4{'%050x' % i}", title=f"T{i}", author="A",
+            embedding=rng.standard_normal(200).astype(np.float32))
+    build_audio_index(conn)
+    eng2 = state.engine(AUDIO_INDEX)
+    assert eng2 is not eng1 and eng2.index.n == 15
+    conn.close()
