@@ -292,3 +292,20 @@ def test_engine_cache_reloads_on_rebuild(tmp_path):
     eng2 = state.engine(AUDIO_INDEX)
     assert eng2 is not eng1 and eng2.index.n == 15
     conn.close()
+
+
+def test_chat_stream_sse(client_ids):
+    client, _ = client_ids
+    r = client.post("/chat/api/chatPlaylistStream",
+                    json={"prompt": "8 songs of jazz"})
+    assert r.status_code == 200
+    assert r.mimetype == "text/event-stream"
+    body = r.get_data(as_text=True)
+    assert "event: plan" in body and "event: playlist" in body
+
+
+def test_suggested_queries(client_ids):
+    client, _ = client_ids
+    a = client.get("/api/clap_search/suggestions?n=5&seed=3").json
+    b = client.get("/api/clap_search/suggestions?n=5&seed=3").json
+    assert a == b and len(a) == 5 and len(set(a)) == 5
