@@ -132,7 +132,8 @@ class AnalysisRuntime:
 
         from audiomuse_amd.ops.audio_io import resample
 
-        a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE).to(self.device)
+        audio = audio.to(self.device)          # resample runs on-device
+        a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE)
         out.tempo, out.energy, out.key, out.scale = \
             features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
 
@@ -151,7 +152,7 @@ class AnalysisRuntime:
             out.moods = moods
 
         if self.htsat is not None:
-            a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE).to(self.device)
+            a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE)
             out.clap_embedding = self._clap_embed(a48)
             out.other_features = self._score_other_features(out.clap_embedding)
         return out
@@ -183,6 +184,8 @@ class AnalysisRuntime:
         results: List[Optional[TrackAnalysis]] = []
         seg_batches: List[torch.Tensor] = []
         seg_owner: List[int] = []
+        patch_batches: List[torch.Tensor] = []
+        patch_owner: List[int] = []
         from audiomuse_amd.ops.audio_io import resample
 
         for i, blob in enumerate(wav_blobs):
@@ -190,14 +193,43 @@ class AnalysisRuntime:
             if audio is None:
                 results.append(None)
                 continue
-            res = self.analyze_track_base(audio, in_sr)
+            audio = audio.to(self.device)      # resample + DSP on-device
+            res = TrackAnalysis(duration=audio.shape[-1] / in_sr)
+            a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE)
+            res.tempo, res.energy, res.key, res.scale = \
+                features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+            mel = hip_ops.mel_spectrogram(a16, dsp.musicnn_mel_config())
+            P = C.MUSICNN_PATCH_FRAMES
+            if mel.shape[-1] >= P:
+                patches = torch.stack(
+                    [mel[:, j : j + P]
+                     for j in range(0, mel.shape[-1] - P + 1, P)]
+                ).transpose(1, 2)              # (n, 187, 96)
+                patch_batches.append(patches)
+                patch_owner.extend([i] * patches.shape[0])
             results.append(res)
             if self.htsat is not None:
-                a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE).to(self.device)
+                a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE)
                 segs = dsp.segment_audio(a48, C.CLAP_SEGMENT_SAMPLES,
                                          C.CLAP_SEGMENT_HOP_SAMPLES)
                 seg_batches.append(segs)
                 seg_owner.extend([i] * segs.shape[0])
+
+        # one MusiCNN pass for the whole album
+        if patch_batches:
+            all_patches = torch.cat(patch_batches, dim=0).float()
+            emb = self.musicnn_emb(all_patches)
+            logits = self.musicnn_pred(emb)
+            owner = torch.tensor(patch_owner)
+            for i, res in enumerate(results):
+                if res is None:
+                    continue
+                mine = owner == i
+                if not bool(mine.any()):
+                    continue
+                track_emb, moods = aggregate_track(emb[mine], logits[mine])
+                res.embedding = track_emb.cpu().numpy().astype(np.float32)
+                res.moods = moods
         if self.htsat is not None and seg_batches:
             all_segs = torch.cat(seg_batches, dim=0)
             mel = hip_ops.mel_spectrogram(all_segs, dsp.clap_mel_config(),
