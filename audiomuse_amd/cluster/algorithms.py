@@ -99,27 +99,45 @@ def dbscan_fit(x: torch.Tensor, eps: float, min_samples: int,
         neighbor_rows.append(mask)
     adj = torch.cat(neighbor_rows, dim=0)      # (n, n) bool; subset-sized inputs
     core = counts >= min_samples
-    labels = torch.full((n,), -1, dtype=torch.long, device=device)
-    cluster = 0
-    adj_core = adj & core.unsqueeze(0)          # edges into core points
-    visited = torch.zeros(n, dtype=torch.bool, device=device)
-    for i in range(n):
-        if visited[i] or not core[i]:
-            continue
-        # BFS over core connectivity (vectorized frontier expansion)
-        frontier = torch.zeros(n, dtype=torch.bool, device=device)
-        frontier[i] = True
-        member = frontier.clone()
-        while frontier.any():
-            reach = adj[frontier].any(dim=0)
-            new = reach & ~member
-            member |= new
-            frontier = new & core
-        member_idx = member.nonzero(as_tuple=True)[0]
-        labels[member_idx] = cluster
-        visited |= member
-        cluster += 1
-    return labels
+    # connected components over core-core edges via min-label propagation
+    # with path halving — fully vectorized (the per-cluster BFS frontier
+    # loop was the CPU-parity bottleneck: profiles/r01_kernel_pmc.md)
+    INF = n
+    labels = torch.where(core, torch.arange(n, device=device),
+                         torch.full((n,), INF, device=device))
+    adj_cc = adj & core.unsqueeze(0) & core.unsqueeze(1)
+    big = torch.full((n, n), INF, device=device, dtype=torch.long)
+    for _ in range(n):  # converges in O(log diameter) with halving
+        nb = torch.where(adj_cc, labels.unsqueeze(0).expand(n, n), big)
+        new = torch.minimum(labels, nb.min(dim=1).values)
+        safe = torch.clamp(new, max=n - 1)
+        new = torch.minimum(new, torch.where(new < INF, new[safe], new))
+        if torch.equal(new, labels):
+            break
+        labels = new
+    # border points take a core neighbor's component; isolated stay noise
+    nb_core = torch.where(adj & core.unsqueeze(0),
+                          labels.unsqueeze(0).expand(n, n), big)
+    border_lab = nb_core.min(dim=1).values
+    labels = torch.where(core, labels,
+                         torch.where(border_lab < INF, border_lab,
+                                     torch.full_like(labels, INF)))
+    # compress to sklearn-style 0..k-1 in first-appearance order; noise -1
+    out = torch.full((n,), -1, dtype=torch.long, device=device)
+    valid = labels < INF
+    if bool(valid.any()):
+        roots = labels[valid]
+        uniq = []
+        seen = set()
+        for r in roots.tolist():
+            if r not in seen:
+                seen.add(r)
+                uniq.append(r)
+        remap = torch.full((n,), -1, dtype=torch.long, device=device)
+        remap[torch.tensor(uniq, device=device)] = torch.arange(
+            len(uniq), device=device)
+        out[valid] = remap[roots]
+    return out
 
 
 def pca_fit_transform(x: torch.Tensor, n_components: int
