@@ -142,12 +142,28 @@ def dbscan_fit(x: torch.Tensor, eps: float, min_samples: int,
 
 def pca_fit_transform(x: torch.Tensor, n_components: int
                       ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """(projected, components, mean) via torch.linalg SVD (rocSOLVER)."""
+    """(projected, components, mean). Small inputs use exact SVD
+    (rocSOLVER); large ones use randomized subspace iteration (all-GEMM,
+    Halko et al.) — the dense SVD was the GPU PCA bottleneck."""
     x = x.float()
     mean = x.mean(dim=0)
     xc = x - mean
     q = min(n_components, min(xc.shape) - 1)
-    U, S, Vh = torch.linalg.svd(xc, full_matrices=False)
+    n, d = xc.shape
+    if n * d <= 1_000_000 or q >= d // 2:
+        U, S, Vh = torch.linalg.svd(xc, full_matrices=False)
+        comps = Vh[:q]
+        return xc @ comps.T, comps, mean
+    # randomized range finder: oversample + 2 power iterations
+    g = torch.Generator(device="cpu").manual_seed(0)
+    p = min(d, q + 8)
+    omega = torch.randn(d, p, generator=g).to(xc.device)
+    y = xc @ omega
+    for _ in range(2):
+        y = xc @ (xc.T @ y)
+        y, _ = torch.linalg.qr(y)
+    b = y.T @ xc                                  # (p, d)
+    _, _, Vh = torch.linalg.svd(b, full_matrices=False)
     comps = Vh[:q]
     return xc @ comps.T, comps, mean
 
