@@ -17,9 +17,11 @@ transformer), with every shape chosen for CDNA4:
 
 ~50M params (~100 MB bf16): same class as the reference student.
 
-The hot path (windowed attention) runs through
-audiomuse_amd.ops.attention.window_attention, which dispatches to the
-fused HIP kernel on GPU and a torch reference on CPU.
+Inference on GPU takes the fused path in SwinBlock.forward: the
+window-attention HIP kernel (roll+partition+attention+reverse in one
+launch), fused residual-add+LayerNorm, and the hipBLASLt GELU-epilogue
+MLP GEMM. CPU and training fall back to the eager torch reference
+(ops/attention.py), which the kernels are numerics-tested against.
 """
 
 from __future__ import annotations

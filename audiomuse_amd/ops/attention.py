@@ -1,14 +1,13 @@
-"""Windowed attention dispatch.
+"""Windowed attention: torch SDPA reference path.
 
-The HTSAT encoder's hot op: multi-head attention over 8x8 = 64-token
+The HTSAT encoder's hot op — multi-head attention over 8x8 = 64-token
 windows with a relative-position bias and (for shifted windows) an
-additive group mask.
-
-GPU path: fused HIP kernel (ops/csrc/attention.hip) — one wavefront per
-(window, head): the 64-token window maps 1:1 onto the 64-lane wave, QK^T
-and PV are mfma_f32_16x16x32_bf16 tiles (head_dim 32 = one K-step), and
-softmax+bias+mask stay in registers. Falls back to torch SDPA where the
-kernel is unavailable (CPU) or shapes don't fit.
+additive group mask. The fused HIP kernel (ops/csrc/attention.hip: one
+wavefront per (window, head), mfma_f32_16x16x32_bf16 QK^T/PV, softmax +
+bias + inline shift mask in registers) operates on the full
+(B, H, W, 3C) image layout and dispatches one level up
+(models/htsat.py SwinBlock.forward). This per-window entry point is the
+numerics reference (CPU + training path) the kernel is tested against.
 
 Reference behavior being replaced: ONNX Runtime attention inside the
 DCLAP student (/root/reference/tasks/clap_analyzer.py:478-500).
