@@ -190,6 +190,9 @@ RADIUS_INSTRUMENTATION = _env_bool("RADIUS_INSTRUMENTATION", False)
 # --------------------------------------------------------------------------
 GPU_DTYPE = _env("GPU_DTYPE", "bf16")
 CLAP_GPU_BATCH = _env_int("CLAP_GPU_BATCH", 256)
+# opt-in fp8 e4m3 serving path for encoder GEMMs (ops/fp8.py); the
+# headline bench stays bf16 regardless of this flag unless --fp8 is passed
+CLAP_FP8_SERVING = _env_bool("AUDIOMUSE_FP8_SERVING", False)
 HIP_REQUIRE_NATIVE = _env_bool("HIP_REQUIRE_NATIVE", True)  # fail loudly on GPU without .so
 RCCL_BUCKET_CAP_MB = _env_int("RCCL_BUCKET_CAP_MB", 64)
 

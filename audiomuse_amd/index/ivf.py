@@ -138,6 +138,97 @@ class IVFIndex:
     def _rebuild_id_map(self) -> None:
         self.id_to_row = {int(v): i for i, v in enumerate(self.ids.tolist())}
 
+    # -- incremental update ------------------------------------------------
+    # The reference rebuilds its IVF wholesale on every index task
+    # (ivf_manager.py rebuild path); with the whole index HBM-resident a
+    # packed-layout splice is cheap, so new/removed tracks can be folded in
+    # without re-training the coarse quantizer. The quantizer is only as
+    # stale as the distribution drift — callers still schedule full
+    # rebuilds periodically (analysis/index.py), matching reference
+    # behavior, but between rebuilds queries see fresh rows.
+
+    def _row_assignments(self) -> torch.Tensor:
+        """Recover each packed row's cell from cell_off: (N,) int64."""
+        counts = (self.cell_off[1:] - self.cell_off[:-1]).long()
+        return torch.repeat_interleave(
+            torch.arange(self.nlist, device=self.device), counts)
+
+    def _repack(self, assign: torch.Tensor, unit: torch.Tensor,
+                raw_f32: Optional[torch.Tensor], ids: torch.Tensor) -> None:
+        """Rebuild packed arrays from per-row assignments (stable order)."""
+        order = torch.argsort(assign, stable=True)
+        counts = torch.bincount(assign, minlength=self.nlist)
+        cell_off = torch.zeros(self.nlist + 1, dtype=torch.int32,
+                               device=self.device)
+        cell_off[1:] = torch.cumsum(counts, dim=0).to(torch.int32)
+        enc = encode_vectors(unit[order], self.storage)
+        n = enc.shape[0]
+        if self.dim_pad != self.dim:
+            pad = torch.zeros(n, self.dim_pad - self.dim, dtype=enc.dtype,
+                              device=self.device)
+            enc = torch.cat([enc, pad], dim=1)
+        self.data = enc.contiguous()
+        self.row_norm = enc.float().norm(dim=1).contiguous()
+        self.cell_off = cell_off.contiguous()
+        self.ids = ids[order].contiguous()
+        self.vectors_f32 = (None if raw_f32 is None
+                            else raw_f32[order].contiguous())
+        self._rebuild_id_map()
+
+    def add(self, vectors: torch.Tensor, ids: torch.Tensor) -> None:
+        """Fold new vectors into the packed cells without re-training the
+        coarse quantizer. An id already present is replaced (upsert)."""
+        if self.centroids is None:
+            raise RuntimeError("add() requires a built index")
+        vectors = torch.as_tensor(vectors, dtype=torch.float32).to(self.device)
+        if vectors.dim() == 1:
+            vectors = vectors.unsqueeze(0)
+        new_ids = torch.as_tensor(ids, dtype=torch.int64).to(self.device)
+        dup = [self.id_to_row[i] for i in new_ids.tolist() if i in self.id_to_row]
+        if dup:
+            self._drop_rows(torch.tensor(dup, dtype=torch.int64,
+                                         device=self.device))
+        if self.metric == "angular":
+            unit_new = vectors / vectors.norm(dim=1, keepdim=True).clamp(min=1e-12)
+        else:
+            unit_new = vectors
+        assign_new = assign_to_centroids(unit_new, self.centroids)
+
+        old_unit = self._decode_unit()
+        assign = torch.cat([self._row_assignments(), assign_new.long()])
+        unit = torch.cat([old_unit, unit_new])
+        raw = (torch.cat([self.vectors_f32, vectors])
+               if self.vectors_f32 is not None else None)
+        self._repack(assign, unit, raw, torch.cat([self.ids, new_ids]))
+
+    def remove(self, ids: torch.Tensor) -> int:
+        """Drop rows by id; returns how many were present and removed."""
+        rows = [self.id_to_row[i] for i in
+                torch.as_tensor(ids, dtype=torch.int64).tolist()
+                if i in self.id_to_row]
+        if rows:
+            self._drop_rows(torch.tensor(sorted(rows), dtype=torch.int64,
+                                         device=self.device))
+        return len(rows)
+
+    def _drop_rows(self, rows: torch.Tensor) -> None:
+        keep = torch.ones(self.n, dtype=torch.bool, device=self.device)
+        keep[rows] = False
+        assign = self._row_assignments()[keep]
+        unit = self._decode_unit()[keep]
+        raw = self.vectors_f32[keep] if self.vectors_f32 is not None else None
+        self._repack(assign, unit, raw, self.ids[keep])
+
+    def _decode_unit(self) -> torch.Tensor:
+        """Packed rows back to the encoded-domain f32 unit vectors."""
+        if self.vectors_f32 is not None:
+            v = self.vectors_f32
+            if self.metric == "angular":
+                return v / v.norm(dim=1, keepdim=True).clamp(min=1e-12)
+            return v.float()
+        v = self.data[:, : self.dim].float()
+        return v / 127.0 if self.storage == "i8" else v
+
     @property
     def n(self) -> int:
         return 0 if self.ids is None else int(self.ids.shape[0])

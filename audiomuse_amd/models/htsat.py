@@ -136,20 +136,37 @@ class SwinBlock(nn.Module):
         B, L, C = x.shape
         if self._fused_attn_available(x):
             # fused kernel folds roll + partition + attention + reverse
-            from audiomuse_amd.ops import _ext
+            from audiomuse_amd.ops import _ext, fp8
             ext = _ext.require()
+            use_fp8 = fp8.serving_enabled() and fp8.available(x.device)
             xn = self.norm1(x)
-            qkv = self.attn.qkv(xn)                       # (B, L, 3C)
+            if use_fp8:
+                qkv = fp8.scaled_linear(xn, self.attn.qkv.weight,
+                                        self.attn.qkv.bias)
+            else:
+                qkv = self.attn.qkv(xn)                   # (B, L, 3C)
             out = ext.window_attn_fwd(
                 qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
                 self.attn.heads, self.shift, self.attn.scale)
-            proj = self.attn.proj(out.view(B, L, C))
+            out = out.view(B, L, C)
+            proj = (fp8.scaled_linear(out, self.attn.proj.weight,
+                                      self.attn.proj.bias)
+                    if use_fp8 else self.attn.proj(out))
             # fused residual add + norm2 (one pass instead of add->LN)
             x2, xn2 = ext.add_layernorm_bf16(
                 x.contiguous(), proj.contiguous(),
                 self.norm2.weight.to(torch.bfloat16).contiguous(),
                 self.norm2.bias.to(torch.bfloat16).contiguous(),
                 self.norm2.eps)
+            if use_fp8:
+                # fp8 GEMM + eager tanh-GELU (scaled_mm has no epilogue);
+                # the fp8 rate + halved operand traffic beats the fused
+                # bf16 epilogue at these shapes
+                hidden = F.gelu(fp8.scaled_linear(xn2, self.mlp[0].weight,
+                                                  self.mlp[0].bias),
+                                approximate="tanh")
+                return x2 + fp8.scaled_linear(hidden, self.mlp[2].weight,
+                                              self.mlp[2].bias)
             # MLP with the GELU fused into the first GEMM's epilogue
             hidden = ext.linear_gelu(xn2, self.mlp[0].weight.contiguous(),
                                      self.mlp[0].bias.contiguous())

@@ -1,0 +1,83 @@
+"""Opt-in fp8 (OCP e4m3) serving path for the CLAP encoder's GEMMs.
+
+gfx950's MFMA runs fp8 at 2x the bf16 dense rate (and the fp8 operands
+halve the HBM traffic of the weight/activation reads, which is what
+actually matters for these memory-bound shapes — see
+profiles/r01_kernel_pmc.md). `torch._scaled_mm` maps to hipBLASLt fp8
+GEMMs on ROCm and was validated on MI355X (scripts/fp8_probe.py:
+637 TF fp8 vs 457 TF bf16 at the encoder's dominant shape).
+
+This is OFF by default: the BASELINE headline metric is bf16 per the
+precision contract, and fp8 is an explicitly-labelled serving mode
+(`AUDIOMUSE_FP8_SERVING=1`, or `bench.py --fp8` which reports
+`"dtype": "fp8_e4m3"` so the number is never mistaken for the bf16
+headline). Quantization is dynamic per-tensor e4m3 (weights cached
+after first quantization; activations scaled by abs-max per call).
+
+The reference has no quantized serving path (its ONNX students run
+fp32 CUDA/DML — /root/reference/tasks/clap_analyzer.py); this is an
+MI355X-native extra, not a parity item.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from audiomuse_amd import config as C
+
+_E4M3 = torch.float8_e4m3fn
+_wcache: Dict[Tuple[int, int], Tuple[torch.Tensor, torch.Tensor]] = {}
+
+
+def serving_enabled() -> bool:
+    return bool(getattr(C, "CLAP_FP8_SERVING", False))
+
+
+def available(device: Optional[torch.device] = None) -> bool:
+    """fp8 scaled-mm usable here (needs a gfx950-class GPU + torch op)."""
+    if not hasattr(torch, "_scaled_mm"):
+        return False
+    if device is not None and device.type != "cuda":
+        return False
+    return torch.cuda.is_available()
+
+
+def quantize_weight(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(N, K) bf16/f32 -> (wq (N, K) e4m3 contiguous, scale 0-d f32).
+
+    Cached by (data_ptr, _version): serving weights are frozen, so the
+    quantization runs once per weight tensor.
+    """
+    key = (w.data_ptr(), w._version)
+    hit = _wcache.get(key)
+    if hit is not None:
+        return hit
+    fmax = torch.finfo(_E4M3).max
+    s = (w.detach().abs().amax().float() / fmax).clamp(min=1e-12)
+    wq = (w.detach().float() / s).clamp(-fmax, fmax).to(_E4M3).contiguous()
+    if len(_wcache) > 256:   # bounded: model swaps must not leak memory
+        _wcache.clear()
+    _wcache[key] = (wq, s)
+    return wq, s
+
+
+def scaled_linear(x: torch.Tensor, weight: torch.Tensor,
+                  bias: Optional[torch.Tensor]) -> torch.Tensor:
+    """Linear in fp8: x (..., K) bf16 @ weight (N, K) -> (..., N) bf16.
+
+    Dynamic per-tensor activation scale (abs-max), cached per-tensor
+    weight scale. `_scaled_mm` needs A row-major and B column-major,
+    which `wq.t()` of a contiguous (N, K) quantized weight provides.
+    """
+    fmax = torch.finfo(_E4M3).max
+    lead = x.shape[:-1]
+    x2 = x.reshape(-1, x.shape[-1])
+    xs = (x2.abs().amax().float() / fmax).clamp(min=1e-12)
+    xq = (x2.float() / xs).clamp(-fmax, fmax).to(_E4M3)
+    wq, ws = quantize_weight(weight)
+    out = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                           bias=None if bias is None else bias.to(torch.bfloat16),
+                           out_dtype=torch.bfloat16)
+    return out.reshape(*lead, -1)

@@ -39,7 +39,13 @@ def main() -> None:
                          "contract); knn: 1M-resident search latency/qps "
                          "(BASELINE config 3); train: student_clap DDP "
                          "distillation (BASELINE config 5)")
+    ap.add_argument("--fp8", action="store_true",
+                    help="opt-in fp8 e4m3 serving mode for encoder GEMMs "
+                         "(ops/fp8.py). NOT the headline: the JSON reports "
+                         'dtype "fp8_e4m3" so it is never mistaken for bf16')
     args = ap.parse_args()
+    if args.fp8:
+        C.CLAP_FP8_SERVING = True
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -143,7 +149,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "fp8_e4m3" if args.fp8 else "bf16",
             "data": "synthetic",
             "config": {
                 "model": "htsat_clap_student_49M",

@@ -147,3 +147,68 @@ def test_gpu_query_end_to_end_large():
     dist, ids = idx.query(q, k=10)
     assert (ids[:, 0].cpu() == torch.arange(16)).all()
     assert torch.isfinite(dist).all()
+
+
+# -- incremental update (MI355X-native extra: packed-layout splice instead
+# -- of the reference's wholesale rebuild) ---------------------------------
+
+def test_incremental_add_found_by_query():
+    x = _corpus(1500, 48)
+    idx = IVFIndex.build(x, metric="angular", storage="i8", seed=0)
+    extra = _corpus(40, 48, seed=9) + 0.05
+    new_ids = torch.arange(10_000, 10_040)
+    idx.add(extra, new_ids)
+    assert idx.n == 1540
+    # every added vector finds itself as nearest neighbor
+    _, ids = idx.query(extra, k=1, nprobe=idx.nlist)
+    assert torch.equal(ids.view(-1), new_ids)
+    # packed invariants hold
+    assert int(idx.cell_off[-1]) == idx.n
+    assert idx.data.shape[0] == idx.ids.shape[0] == idx.vectors_f32.shape[0]
+
+
+def test_incremental_add_matches_fresh_build_results():
+    """After add(), queries return the same neighbors as a scan over the
+    union corpus (full-probe, so coarse quantization cannot differ)."""
+    x = _corpus(800, 32)
+    extra = _corpus(60, 32, seed=5)
+    idx = IVFIndex.build(x, metric="angular", storage="f32", seed=0)
+    idx.add(extra, torch.arange(800, 860))
+    q = _corpus(10, 32, seed=7)
+    d, ids = idx.query(q, k=5, nprobe=idx.nlist)
+    both = torch.cat([x, extra])
+    want = _brute_force_ids(both, q, 5, "angular")
+    assert (ids == want).float().mean() > 0.95
+
+
+def test_incremental_upsert_replaces_vector():
+    x = _corpus(500, 32)
+    idx = IVFIndex.build(x, metric="angular", storage="f16", seed=0)
+    moved = torch.randn(1, 32) * 2
+    idx.add(moved, torch.tensor([123]))
+    assert idx.n == 500  # replaced, not appended
+    v = idx.vector_for_id(123)
+    assert torch.allclose(v, moved[0], atol=1e-3)
+
+
+def test_incremental_remove():
+    x = _corpus(600, 32)
+    idx = IVFIndex.build(x, metric="euclidean", storage="f32", seed=0)
+    gone = torch.arange(0, 50)
+    assert idx.remove(gone) == 50
+    assert idx.n == 550
+    assert idx.remove(gone) == 0  # already gone
+    _, ids = idx.query(x[:50], k=3, nprobe=idx.nlist)
+    assert not (ids.unsqueeze(-1) == gone.view(1, 1, -1)).any()
+    # survivors still found
+    _, ids2 = idx.query(x[100:110], k=1, nprobe=idx.nlist)
+    assert torch.equal(ids2.view(-1), torch.arange(100, 110))
+
+
+def test_incremental_add_without_f32_rerank_store():
+    x = _corpus(400, 36)
+    idx = IVFIndex.build(x, metric="angular", storage="i8", keep_f32=False)
+    idx.add(_corpus(10, 36, seed=3), torch.arange(400, 410))
+    assert idx.n == 410 and idx.vectors_f32 is None
+    d, ids = idx.query(x[:5], k=1, nprobe=idx.nlist, rerank=False)
+    assert torch.equal(ids.view(-1), torch.arange(5))
