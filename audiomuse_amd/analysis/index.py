@@ -184,6 +184,73 @@ def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
     return len(ids)
 
 
+_REFRESHABLE = {AUDIO_INDEX: "embedding", CLAP_INDEX: "clap_embedding",
+                LYRICS_INDEX: "lyrics_embedding"}
+
+
+def refresh_ivf_index(conn: sqlite3.Connection, name: str,
+                      device: str = "cpu",
+                      max_drift: float = 0.25) -> Dict[str, int]:
+    """Incremental IVF refresh: splice new/removed tracks into the stored
+    packed index (IVFIndex.add/remove) instead of rebuilding.
+
+    The reference rebuilds wholesale on every index task; with the index
+    HBM-resident the splice is cheap, so between full rebuilds queries can
+    see fresh tracks immediately. Falls back to a full rebuild when more
+    than `max_drift` of the corpus changed (the coarse quantizer is only
+    trained on the old distribution) or when no index exists yet.
+
+    Returns {"added": a, "removed": r, "total": n, "rebuilt": 0|1}.
+    """
+    column = _REFRESHABLE[name]
+    ids, mat = load_all_embeddings(conn, column)
+    got = load_index_blob(conn, name)
+    if got is None or not ids:
+        n = {AUDIO_INDEX: build_audio_index, CLAP_INDEX: build_clap_index,
+             LYRICS_INDEX: build_lyrics_index}[name](conn, device)
+        return {"added": n, "removed": 0, "total": n, "rebuilt": 1}
+
+    payload = torch.load(io.BytesIO(got[0]), map_location="cpu",
+                         weights_only=True)
+    index = IVFIndex.deserialize(payload["index"], device=device)
+    old_ids: List[str] = payload["item_ids"]
+    old_pos = {s: i for i, s in enumerate(old_ids)}
+    new_pos = {s: i for i, s in enumerate(ids)}
+    added = [s for s in ids if s not in old_pos]
+    removed = [s for s in old_ids if s not in new_pos]
+    changed = len(added) + len(removed)
+    if changed == 0:
+        return {"added": 0, "removed": 0, "total": len(ids), "rebuilt": 0}
+    if changed > max_drift * max(len(ids), 1):
+        n = {AUDIO_INDEX: build_audio_index, CLAP_INDEX: build_clap_index,
+             LYRICS_INDEX: build_lyrics_index}[name](conn, device)
+        return {"added": len(added), "removed": len(removed), "total": n,
+                "rebuilt": 1}
+
+    # The engine's item_ids list maps packed int rows -> string ids; keep
+    # the integer key stable per string id across the splice.
+    if removed:
+        index.remove(torch.tensor([old_pos[s] for s in removed]))
+    item_ids = list(old_ids)
+    if added:
+        removed_set = set(removed)
+        free = [i for i, s in enumerate(item_ids) if s in removed_set]
+        rows = []
+        for s in added:
+            if free:
+                slot = free.pop()
+                item_ids[slot] = s
+            else:
+                slot = len(item_ids)
+                item_ids.append(s)
+            rows.append(slot)
+        vecs = torch.from_numpy(mat[[new_pos[s] for s in added]])
+        index.add(vecs, torch.tensor(rows, dtype=torch.int64))
+    _store_ivf(conn, name, index, item_ids)
+    return {"added": len(added), "removed": len(removed),
+            "total": index.n, "rebuilt": 0}
+
+
 def run_all_index_builds(conn: sqlite3.Connection, device: str = "cpu",
                          progress_cb=None) -> Dict[str, int]:
     """The ordered build set (_run_all_index_builds, index.py:47).

@@ -203,3 +203,24 @@ def rebuild_indexes_task(ctx: TaskContext, payload: Dict) -> Dict:
         ctx.conn, device=device,
         progress_cb=lambda name, n: ctx.report(50.0, f"{name}: {n}"))
     return {"indexes": built}
+
+
+@task_handler("refresh_indexes")
+def refresh_indexes_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Incremental IVF refresh (index.refresh_ivf_index): splice
+    new/removed tracks into the stored packed indexes between full
+    rebuilds. MI355X-native extra over the reference's wholesale
+    rebuild path."""
+    from audiomuse_amd.analysis import index as idx_mod
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    out: Dict[str, Dict] = {}
+    for name in idx_mod._REFRESHABLE:
+        try:
+            out[name] = idx_mod.refresh_ivf_index(ctx.conn, name,
+                                                  device=device)
+        except Exception:
+            logger.exception("incremental refresh %s failed", name)
+            out[name] = {"error": 1}
+        ctx.report(50.0, f"{name}: {out[name]}")
+    return {"indexes": out}

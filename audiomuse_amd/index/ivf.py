@@ -148,17 +148,20 @@ class IVFIndex:
     # behavior, but between rebuilds queries see fresh rows.
 
     def _row_assignments(self) -> torch.Tensor:
-        """Recover each packed row's cell from cell_off: (N,) int64."""
+        """Recover each packed row's cell from cell_off: (N,) int64.
+        Cell count comes from cell_off, which may exceed nlist when the
+        build clamped k-means k to the training-sample size."""
         counts = (self.cell_off[1:] - self.cell_off[:-1]).long()
         return torch.repeat_interleave(
-            torch.arange(self.nlist, device=self.device), counts)
+            torch.arange(counts.shape[0], device=self.device), counts)
 
     def _repack(self, assign: torch.Tensor, unit: torch.Tensor,
                 raw_f32: Optional[torch.Tensor], ids: torch.Tensor) -> None:
         """Rebuild packed arrays from per-row assignments (stable order)."""
+        ncells = int(self.cell_off.shape[0]) - 1
         order = torch.argsort(assign, stable=True)
-        counts = torch.bincount(assign, minlength=self.nlist)
-        cell_off = torch.zeros(self.nlist + 1, dtype=torch.int32,
+        counts = torch.bincount(assign, minlength=ncells)
+        cell_off = torch.zeros(ncells + 1, dtype=torch.int32,
                                device=self.device)
         cell_off[1:] = torch.cumsum(counts, dim=0).to(torch.int32)
         enc = encode_vectors(unit[order], self.storage)

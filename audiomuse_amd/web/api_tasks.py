@@ -78,6 +78,17 @@ def index_rebuild():
     return jsonify({"task_id": tid}), 202
 
 
+@bp.post("/api/index/refresh")
+@require_auth
+def index_refresh():
+    """Incremental IVF splice (analysis.index.refresh_ivf_index):
+    new/removed tracks folded into the stored packed indexes without a
+    full rebuild — cheap enough to run after every analysis batch."""
+    conn = _state().conn()
+    tid = enqueue(conn, "refresh_indexes", {}, queue="high")
+    return jsonify({"task_id": tid}), 202
+
+
 @bp.get("/api/task/<task_id>")
 @require_auth
 def task_status(task_id: str):

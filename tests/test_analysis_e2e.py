@@ -129,3 +129,69 @@ def test_index_builds_on_synthetic_embeddings(db):
     assert built["semgrove"] == 15
     assert built["artist"] == 5
     assert built["song_map"] == 30
+
+
+def test_incremental_index_refresh(db):
+    conn, _url = db
+    import torch
+
+    from audiomuse_amd.analysis.index import (AUDIO_INDEX, build_audio_index,
+                                              load_ivf_engine,
+                                              refresh_ivf_index)
+    from audiomuse_amd.db import write_txn
+    from audiomuse_amd.db.store import save_track_analysis_and_embedding
+
+    rng = np.random.default_rng(1)
+    vecs = {}
+    for i in range(40):
+        vecs[f"t{i}"] = rng.standard_normal(64).astype(np.float32)
+        save_track_analysis_and_embedding(conn, f"t{i}", title=f"T{i}",
+                                          embedding=vecs[f"t{i}"])
+    assert build_audio_index(conn) == 40
+
+    # no-op refresh
+    out = refresh_ivf_index(conn, AUDIO_INDEX)
+    assert out == {"added": 0, "removed": 0, "total": 40, "rebuilt": 0}
+
+    # +5 tracks, -3 tracks -> splice, not rebuild
+    for i in range(40, 45):
+        vecs[f"t{i}"] = rng.standard_normal(64).astype(np.float32)
+        save_track_analysis_and_embedding(conn, f"t{i}", title=f"T{i}",
+                                          embedding=vecs[f"t{i}"])
+    with write_txn(conn):
+        conn.execute("DELETE FROM embedding WHERE item_id IN ('t0','t1','t2')")
+    out = refresh_ivf_index(conn, AUDIO_INDEX)
+    assert out["added"] == 5 and out["removed"] == 3 and out["rebuilt"] == 0
+    assert out["total"] == 42
+
+    eng = load_ivf_engine(conn, AUDIO_INDEX)
+    assert "t42" in eng.pos and "t0" not in [
+        eng.item_ids[int(r)] for r in eng.index.ids.tolist()]
+    # each spliced-in vector finds itself
+    for iid in ("t40", "t44"):
+        _, rows = eng.index.query(torch.from_numpy(vecs[iid]), k=1,
+                                  nprobe=eng.index.nlist)
+        assert eng.item_ids[int(rows[0])] == iid
+
+    # large drift -> full rebuild
+    with write_txn(conn):
+        conn.execute("DELETE FROM embedding WHERE item_id IN ("
+                     + ",".join(f"'t{i}'" for i in range(3, 25)) + ")")
+    out = refresh_ivf_index(conn, AUDIO_INDEX)
+    assert out["rebuilt"] == 1 and out["total"] == 20
+
+
+def test_refresh_indexes_task_runs(db):
+    conn, url = db
+    from audiomuse_amd.db.store import save_track_analysis_and_embedding
+
+    rng = np.random.default_rng(2)
+    for i in range(10):
+        save_track_analysis_and_embedding(
+            conn, f"r{i}", title=f"R{i}",
+            embedding=rng.standard_normal(32).astype(np.float32))
+    tid = enqueue(conn, "refresh_indexes", {})
+    Worker(db_url=url, max_jobs=1).run_forever(idle_timeout=3.0)
+    row = task_row(conn, tid)
+    assert row["status"] == SUCCESS
+    assert '"rebuilt": 1' in row["result"]  # first run builds audio index

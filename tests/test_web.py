@@ -309,3 +309,9 @@ def test_suggested_queries(client_ids):
     a = client.get("/api/clap_search/suggestions?n=5&seed=3").json
     b = client.get("/api/clap_search/suggestions?n=5&seed=3").json
     assert a == b and len(a) == 5 and len(set(a)) == 5
+
+
+def test_index_refresh_endpoint(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/index/refresh")
+    assert r.status_code == 202 and "task_id" in r.json
