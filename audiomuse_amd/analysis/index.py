@@ -32,9 +32,11 @@ logger = logging.getLogger(__name__)
 AUDIO_INDEX = "audio_ivf"
 CLAP_INDEX = "clap_ivf"
 LYRICS_INDEX = "lyrics_ivf"
+LYRICS_AXES_INDEX = "lyrics_axes_ivf"
 SEMGROVE_INDEX = "semgrove_ivf"
 ARTIST_INDEX = "artist_models"
 SONG_MAP = "song_map_projection"
+ARTIST_MAP = "artist_map_projection"
 
 
 def _store_ivf(conn: sqlite3.Connection, name: str, index: IVFIndex,
@@ -87,6 +89,37 @@ def build_lyrics_index(conn: sqlite3.Connection, device: str = "cpu") -> int:
     idx = IVFIndex.build(torch.from_numpy(mat), metric="angular",
                          device=device)
     _store_ivf(conn, LYRICS_INDEX, idx, ids)
+    return len(ids)
+
+
+def build_lyrics_axes_index(conn: sqlite3.Connection,
+                            device: str = "cpu") -> int:
+    """27-axis thematic-profile index (reference: the lyrics-axes build in
+    _run_all_index_builds, index.py:47, backed by lyrics_manager's axis
+    store): each track's 27 axis scores (engines/lyrics.score_axes) form a
+    vector; angular IVF over those enables 'similar thematic profile'
+    queries (/api/lyrics_axes_similar) beyond the single-axis ranking."""
+    import json as _json
+
+    rows = conn.execute(
+        "SELECT item_id, axis_scores FROM lyrics_embedding "
+        "WHERE axis_scores IS NOT NULL AND axis_scores != '{}'").fetchall()
+    ids, vecs = [], []
+    for r in rows:
+        try:
+            scores = _json.loads(r["axis_scores"])
+        except Exception:
+            continue
+        if not scores:
+            continue
+        ids.append(r["item_id"])
+        vecs.append([float(scores.get(a, 0.0)) for a in C.LYRICS_AXES])
+    if not ids:
+        return 0
+    mat = torch.tensor(vecs, dtype=torch.float32)
+    # 27-d profiles: f32 storage (i8 would quantize away the soft scores)
+    idx = IVFIndex.build(mat, metric="angular", storage="f32", device=device)
+    _store_ivf(conn, LYRICS_AXES_INDEX, idx, ids)
     return len(ids)
 
 
@@ -184,6 +217,43 @@ def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
     return len(ids)
 
 
+def build_artist_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
+    """2-D artist map (reference: the artist-map build in
+    _run_all_index_builds, index.py:47): one point per artist at the mean
+    of their track embeddings, projected with the same UMAP-style layout
+    as the song map (PCA fallback)."""
+    rows = conn.execute(
+        """SELECT s.author, e.embedding FROM score s
+           JOIN embedding e ON e.item_id = s.item_id
+           WHERE s.author IS NOT NULL AND s.author != ''""").fetchall()
+    per_artist: Dict[str, list] = {}
+    for r in rows:
+        per_artist.setdefault(r["author"], []).append(
+            np.frombuffer(r["embedding"], dtype=np.float32))
+    if not per_artist:
+        return 0
+    names = sorted(per_artist)
+    cents = torch.from_numpy(
+        np.stack([np.stack(per_artist[a]).mean(axis=0) for a in names]))
+    if len(names) < 4:   # too few points for any layout: place on a line
+        coords = np.stack([np.arange(len(names), dtype=np.float32),
+                           np.zeros(len(names), dtype=np.float32)], axis=1)
+    else:
+        try:
+            from audiomuse_amd.engines.projection import umap_project
+
+            coords = umap_project(cents.to(device), seed=0).cpu().numpy()
+        except Exception:
+            proj, _, _ = pca_fit_transform(cents.to(device), 2)
+            coords = proj.cpu().numpy()
+    buf = io.BytesIO()
+    torch.save({"item_ids": names,
+                "coords": torch.from_numpy(coords.astype(np.float32)),
+                "n_tracks": [len(per_artist[a]) for a in names]}, buf)
+    store_index_blob(conn, ARTIST_MAP, buf.getvalue(), meta={"n": len(names)})
+    return len(names)
+
+
 _REFRESHABLE = {AUDIO_INDEX: "embedding", CLAP_INDEX: "clap_embedding",
                 LYRICS_INDEX: "lyrics_embedding"}
 
@@ -259,9 +329,11 @@ def run_all_index_builds(conn: sqlite3.Connection, device: str = "cpu",
     results["audio"] = build_audio_index(conn, device)   # fatal on raise
     for name, fn in [("clap", build_clap_index),
                      ("lyrics", build_lyrics_index),
+                     ("lyrics_axes", build_lyrics_axes_index),
                      ("semgrove", build_semgrove_index),
                      ("artist", lambda c, **k: build_artist_index(c)),
-                     ("song_map", build_song_map)]:
+                     ("song_map", build_song_map),
+                     ("artist_map", build_artist_map)]:
         try:
             results[name] = fn(conn, device=device) if name != "artist" \
                 else build_artist_index(conn)

@@ -17,6 +17,20 @@ after first quantization; activations scaled by abs-max per call).
 The reference has no quantized serving path (its ONNX students run
 fp32 CUDA/DML — /root/reference/tasks/clap_analyzer.py); this is an
 MI355X-native extra, not a parity item.
+
+MEASURED STATUS (scripts/fp8_shapes.py on MI355X, 2026-09): the fp8
+GEMM alone beats bf16 at most encoder shapes (1.4-1.9x) but the
+unfused dynamic quantization costs ~2/3 of a full bf16 GEMM per input
+(it re-reads the very activation bytes fp8 saves), and hipBLASLt's fp8
+algo at the stage-1 qkv shape (M=2.1M, K=128, N=384) is 2.1x SLOWER
+than bf16. Net: this mode currently trades ~2.7x throughput for the
+accuracy headroom experiment. Making fp8 pay requires (a) quantization
+fused into the producing kernels (LayerNorm/attention emit e4m3 +
+running amax, delayed scaling) and (b) an algo-searched fp8 GEMM for
+the K=128 shapes — both round-2 items. gfx950 note: non-scaled fp8
+MFMA runs at the bf16 rate; fp8's entire win at these memory-bound
+shapes is the halved operand traffic, which is exactly what an
+unfused quantize pass spends back.
 """
 
 from __future__ import annotations

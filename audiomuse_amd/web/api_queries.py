@@ -284,6 +284,22 @@ def lyrics_axes():
         {"item_id": i, "distance": 1.0 - s} for i, s in scored[:n]]))
 
 
+@bp.get("/api/lyrics_axes_similar")
+@require_auth
+def lyrics_axes_similar():
+    """Similar thematic profile: nearest neighbors in the 27-axis space
+    (reference: lyrics-axes index queries)."""
+    item_id = request.args.get("item_id", "")
+    eng = _state().engine(idx.LYRICS_AXES_INDEX)
+    if eng is None:
+        return jsonify({"error": "lyrics-axes index not built"}), 503
+    n = int(request.args.get("n", 20))
+    if eng.vector_for_id(item_id) is None:
+        return jsonify({"error": f"no axis profile for {item_id!r}"}), 404
+    res = eng.find_similar_by_id(item_id, n=n)
+    return jsonify(_with_meta(res))
+
+
 @bp.get("/api/hyperbolic_tree")
 @require_auth
 def hyperbolic_tree():
@@ -374,15 +390,19 @@ def hyperbolic_similar():
 @bp.get("/api/map")
 @require_auth
 def music_map():
-    """2-D map coordinates (app_map.py buckets)."""
-    data = _state().engine(idx.SONG_MAP)
+    """2-D map coordinates (app_map.py buckets). kind=song (default)
+    or kind=artist (reference: artist-map build)."""
+    kind = request.args.get("kind", "song")
+    data = _state().engine(
+        idx.ARTIST_MAP if kind == "artist" else idx.SONG_MAP)
     if data is None:
         return jsonify({"error": "map not built"}), 503
     coords = data["coords"]
     ids = data["item_ids"]
     limit = int(request.args.get("n", 0)) or len(ids)
+    key = "artist" if kind == "artist" else "item_id"
     return jsonify([
-        {"item_id": ids[i], "x": float(coords[i][0]), "y": float(coords[i][1])}
+        {key: ids[i], "x": float(coords[i][0]), "y": float(coords[i][1])}
         for i in range(min(limit, len(ids)))])
 
 

@@ -73,7 +73,7 @@ class AppState:
                 return self._engines[name]
         if name == idx.ARTIST_INDEX:
             eng = idx.load_artist_similarity(self.conn())
-        elif name == idx.SONG_MAP:
+        elif name in (idx.SONG_MAP, idx.ARTIST_MAP):
             got = None
             from audiomuse_amd.db.store import load_index_blob
             got = load_index_blob(self.conn(), name)

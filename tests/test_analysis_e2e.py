@@ -126,9 +126,11 @@ def test_index_builds_on_synthetic_embeddings(db):
     assert built["audio"] == 30
     assert built["clap"] == 30
     assert built["lyrics"] == 15
+    assert built["lyrics_axes"] == 15
     assert built["semgrove"] == 15
     assert built["artist"] == 5
     assert built["song_map"] == 30
+    assert built["artist_map"] == 5
 
 
 def test_incremental_index_refresh(db):

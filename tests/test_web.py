@@ -315,3 +315,23 @@ def test_index_refresh_endpoint(client_ids):
     client, _ = client_ids
     r = client.post("/api/index/refresh")
     assert r.status_code == 202 and "task_id" in r.json
+
+
+def test_artist_map(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/map?kind=artist")
+    assert r.status_code == 200
+    body = r.json
+    assert len(body) == 6  # 6 artists seeded
+    assert all("artist" in p and "x" in p and "y" in p for p in body)
+
+
+def test_lyrics_axes_similar(client_ids):
+    client, ids = client_ids
+    # even-index tracks have axis profiles
+    r = client.get(f"/api/lyrics_axes_similar?item_id={ids[0]}&n=5")
+    assert r.status_code == 200 and len(r.json) >= 1
+    assert all(b["item_id"] != ids[0] for b in r.json)
+    # odd-index track has no profile -> 404
+    assert client.get(
+        f"/api/lyrics_axes_similar?item_id={ids[1]}").status_code == 404
