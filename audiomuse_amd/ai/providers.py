@@ -1,0 +1,137 @@
+"""LLM provider adapters for the instant-playlist planner.
+
+Reference: /root/reference/tasks/ai/providers/ — openai.py (tool-calling
+chat completions), gemini.py (generateContent + functionDeclarations),
+mistral.py (OpenAI-compatible chat with Mistral auth/model defaults).
+Each adapter turns one planning prompt + tool schema into a vendor
+request and parses the returned tool calls into the planner's
+[{"tool": name, "args": {...}}] form. All network failures return None
+(the planner falls back to the deterministic heuristic plan).
+
+Outbound URLs pass the SSRF guard (utils/logging_utils.py:
+validate_outbound_url) before any request is made — same policy as the
+reference's ssrf_guard.py.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+from typing import Dict, List, Optional
+
+from audiomuse_amd import config as C
+from audiomuse_amd.utils.logging_utils import validate_outbound_url
+
+logger = logging.getLogger(__name__)
+
+_SYSTEM = ("Plan music-library tool calls for the user's playlist "
+           "request. Use only the provided tools.")
+
+_DEFAULT_BASE = {
+    "openai": "https://api.openai.com/v1",
+    "mistral": "https://api.mistral.ai/v1",
+    "gemini": "https://generativelanguage.googleapis.com/v1beta",
+}
+_DEFAULT_MODEL = {
+    "openai": "gpt-4o-mini",
+    "mistral": "mistral-small-latest",
+    "gemini": "gemini-2.0-flash",
+}
+
+
+def _base_and_key(provider: str) -> tuple:
+    base = os.environ.get("AI_BASE_URL", _DEFAULT_BASE[provider]).rstrip("/")
+    key = os.environ.get("AI_API_KEY", "")
+    validate_outbound_url(base)
+    return base, key
+
+
+def build_openai_request(provider: str, prompt: str,
+                         tools: Dict[str, Dict]) -> Dict:
+    """OpenAI-compatible chat/completions body (openai.py + mistral.py:
+    Mistral's chat API is OpenAI-wire-compatible)."""
+    return {
+        "model": C.AI_MODEL_NAME or _DEFAULT_MODEL[provider],
+        "messages": [{"role": "system", "content": _SYSTEM},
+                     {"role": "user", "content": prompt}],
+        "tools": [{"type": "function",
+                   "function": {"name": name,
+                                "parameters": schema or {"type": "object"}}}
+                  for name, schema in tools.items()],
+        "tool_choice": "auto",
+    }
+
+
+def parse_openai_response(body: Dict) -> Optional[List[Dict]]:
+    msg = body["choices"][0]["message"]
+    plan = []
+    for tc in msg.get("tool_calls") or []:
+        fn = tc.get("function", {})
+        try:
+            args = json.loads(fn.get("arguments") or "{}")
+        except Exception:
+            args = {}
+        plan.append({"tool": fn.get("name"), "args": args})
+    return plan or None
+
+
+def build_gemini_request(prompt: str, tools: Dict[str, Dict]) -> Dict:
+    """generateContent body with functionDeclarations (gemini.py)."""
+    return {
+        "system_instruction": {"parts": [{"text": _SYSTEM}]},
+        "contents": [{"role": "user", "parts": [{"text": prompt}]}],
+        "tools": [{"functionDeclarations": [
+            {"name": name,
+             "parameters": schema or {"type": "object", "properties": {}}}
+            for name, schema in tools.items()]}],
+        "tool_config": {"function_calling_config": {"mode": "AUTO"}},
+    }
+
+
+def parse_gemini_response(body: Dict) -> Optional[List[Dict]]:
+    plan = []
+    for cand in body.get("candidates") or []:
+        for part in cand.get("content", {}).get("parts", []):
+            fc = part.get("functionCall")
+            if fc:
+                plan.append({"tool": fc.get("name"),
+                             "args": dict(fc.get("args") or {})})
+    return plan or None
+
+
+def plan_with_llm(prompt: str, tools: Dict[str, Dict],
+                  provider: Optional[str] = None,
+                  post=None) -> Optional[List[Dict]]:
+    """One tool-calling request to the configured provider; None on any
+    failure or when AI_PROVIDER is none. `post` injects the HTTP call in
+    tests (defaults to requests.post)."""
+    provider = (provider or C.AI_PROVIDER or "none").lower()
+    if provider in ("", "none"):
+        return None
+    if provider not in _DEFAULT_BASE:
+        logger.warning("unknown AI provider %r", provider)
+        return None
+    try:
+        if post is None:
+            import requests
+
+            post = requests.post
+        base, key = _base_and_key(provider)
+        if provider == "gemini":
+            model = C.AI_MODEL_NAME or _DEFAULT_MODEL[provider]
+            r = post(f"{base}/models/{model}:generateContent",
+                     headers={"x-goog-api-key": key},
+                     json=build_gemini_request(prompt, tools), timeout=30)
+            r.raise_for_status()
+            return parse_gemini_response(r.json())
+        r = post(f"{base}/chat/completions",
+                 headers={"Authorization": f"Bearer {key}"},
+                 json=build_openai_request(provider, prompt, tools),
+                 timeout=30)
+        r.raise_for_status()
+        return parse_openai_response(r.json())
+    except Exception:
+        logger.debug("LLM plan failed; falling back to heuristic",
+                     exc_info=True)
+        return None

@@ -152,42 +152,12 @@ def heuristic_plan(hints: Dict) -> List[Dict]:
 
 
 def llm_plan(prompt: str, hints: Dict) -> Optional[List[Dict]]:
-    """One tool-calling request to an OpenAI-compatible endpoint
-    (providers/openai.py equivalent). None on any failure -> heuristic."""
-    if C.AI_PROVIDER in ("", "none"):
-        return None
-    try:
-        import os
+    """One tool-calling request to the configured vendor adapter
+    (ai/providers.py: openai / mistral / gemini). None on any failure
+    -> heuristic plan."""
+    from audiomuse_amd.ai import plan_with_llm
 
-        import requests
-
-        base = os.environ.get("AI_BASE_URL", "https://api.openai.com/v1")
-        key = os.environ.get("AI_API_KEY", "")
-        tools_desc = [{"type": "function", "function": {
-            "name": name, "parameters": {"type": "object"}}}
-            for name in TOOLS]
-        r = requests.post(
-            f"{base}/chat/completions",
-            headers={"Authorization": f"Bearer {key}"},
-            json={"model": C.AI_MODEL_NAME or "gpt-4o-mini",
-                  "messages": [
-                      {"role": "system",
-                       "content": "Plan music-library tool calls for the "
-                                  "user's playlist request. Use only the "
-                                  "provided tools."},
-                      {"role": "user", "content": prompt}],
-                  "tools": tools_desc},
-            timeout=30)
-        r.raise_for_status()
-        msg = r.json()["choices"][0]["message"]
-        plan = []
-        for tc in msg.get("tool_calls", []):
-            fn = tc.get("function", {})
-            plan.append({"tool": fn.get("name"),
-                         "args": json.loads(fn.get("arguments") or "{}")})
-        return plan or None
-    except Exception:
-        return None
+    return plan_with_llm(prompt, {name: None for name in TOOLS})
 
 
 def rerank(results_per_tool: List[List[Dict]], n: int) -> List[str]:
