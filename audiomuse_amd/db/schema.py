@@ -145,6 +145,12 @@ CREATE TABLE IF NOT EXISTS app_config (
     key TEXT PRIMARY KEY,
     value TEXT
 );
+CREATE TABLE IF NOT EXISTS plugin (
+    name TEXT PRIMARY KEY,
+    blob BLOB NOT NULL,
+    enabled INTEGER DEFAULT 1,
+    uploaded_at REAL DEFAULT ((julianday('now') - 2440587.5) * 86400.0)
+);
 CREATE TABLE IF NOT EXISTS dashboard_stats (
     key TEXT PRIMARY KEY,
     value TEXT,

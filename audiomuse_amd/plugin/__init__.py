@@ -1,1 +1,6 @@
-from audiomuse_amd.plugin.manager import PluginManager, hook_registry  # noqa: F401
+from audiomuse_amd.plugin.manager import (PluginManager,  # noqa: F401
+                                          hook_registry)
+
+# process-wide manager: web app and workers load DB-stored plugins into
+# this instance at boot (manager.load_from_db)
+plugin_manager = PluginManager()

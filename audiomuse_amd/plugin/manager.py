@@ -62,6 +62,19 @@ class PluginAPI:
         self.cron_tasks.append({"schedule": schedule, "task_type": task_type,
                                 "payload": payload or {}})
 
+    def add_task_handler(self, task_type: str, fn: Callable) -> None:
+        """Analysis-provider extension point (reference: plugin cron task
+        + analysis-provider extension points, PluginManager.load :537):
+        the plugin contributes a queue task type, runnable by workers and
+        schedulable via add_cron_task. Namespaced to the plugin so a
+        plugin cannot shadow built-in handlers."""
+        from audiomuse_amd.taskqueue.worker import _REGISTRY
+
+        full = f"plugin.{self.name}.{task_type}"
+        _REGISTRY[full] = fn
+        self.task_types = getattr(self, "task_types", [])
+        self.task_types.append(full)
+
 
 def _safe_extract(zf: zipfile.ZipFile, dest: str) -> None:
     for member in zf.namelist():
@@ -98,3 +111,55 @@ class PluginManager:
 
     def fire_song_analyzed(self, item_id: str, analysis: dict) -> None:
         self.hooks.fire("song_analyzed", item_id, analysis)
+
+    def load_from_db(self, conn) -> int:
+        """Load every enabled plugin stored in the `plugin` table (the
+        reference keeps plugin zips in the DB and loads them at web and
+        worker boot), then sync their cron tasks. Individual plugin
+        failures are logged and skipped. Returns plugins loaded."""
+        rows = conn.execute(
+            "SELECT name, blob FROM plugin WHERE enabled = 1").fetchall()
+        n = 0
+        for r in rows:
+            try:
+                self.load_zip(r["name"], r["blob"])
+                n += 1
+            except Exception:  # noqa: BLE001 — one bad plugin never blocks boot
+                logger.exception("plugin %s failed to load", r["name"])
+        self.sync_cron(conn)
+        return n
+
+    def sync_cron(self, conn) -> int:
+        """Upsert every loaded plugin's cron tasks into the cron table
+        (name 'plugin:<plugin>:<i>'); prunes rows of plugins no longer
+        loaded. Returns the number of active plugin cron rows."""
+        import json as _json
+
+        from audiomuse_amd.db import write_txn
+
+        want = {}
+        for api in self.loaded.values():
+            for i, ct in enumerate(api.cron_tasks):
+                want[f"plugin:{api.name}:{i}"] = ct
+        with write_txn(conn):
+            rows = conn.execute(
+                "SELECT id, name FROM cron WHERE name LIKE 'plugin:%'"
+            ).fetchall()
+            for r in rows:
+                if r["name"] not in want:
+                    conn.execute("DELETE FROM cron WHERE id = ?", (r["id"],))
+            have = {r["name"] for r in rows}
+            for name, ct in want.items():
+                if name in have:
+                    conn.execute(
+                        "UPDATE cron SET schedule=?, task_type=?, payload=? "
+                        "WHERE name=?",
+                        (ct["schedule"], ct["task_type"],
+                         _json.dumps(ct["payload"]), name))
+                else:
+                    conn.execute(
+                        "INSERT INTO cron (name, schedule, task_type, "
+                        "payload, enabled) VALUES (?,?,?,?,1)",
+                        (name, ct["schedule"], ct["task_type"],
+                         _json.dumps(ct["payload"])))
+        return len(want)

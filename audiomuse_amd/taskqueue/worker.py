@@ -152,6 +152,13 @@ class Worker:
             C.apply_db_overrides(get_app_config(conn))
         except Exception:
             logger.exception("config hydrate failed")
+        try:
+            # DB-stored plugins contribute hooks + task handlers here too
+            from audiomuse_amd.plugin import plugin_manager
+
+            plugin_manager.load_from_db(conn)
+        except Exception:
+            logger.exception("plugin load failed")
         idle_since = time.time()
         last_maintenance = 0.0
         try:

@@ -167,3 +167,66 @@ def migration_start():
         "server_id": body.get("server_id", "migrated"),
     }, queue="high")
     return jsonify({"task_id": tid}), 202
+
+
+@bp.get("/api/plugins")
+@require_auth
+def list_plugins():
+    rows = _state().conn().execute(
+        "SELECT name, enabled, uploaded_at FROM plugin ORDER BY name"
+    ).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.post("/api/plugins")
+@require_auth
+def upload_plugin():
+    """Store + load a plugin zip (reference: plugin/blueprint.py upload).
+    Body: raw zip bytes with ?name=..., or JSON {name, zip_base64}."""
+    import base64
+
+    from audiomuse_amd.db import write_txn
+    from audiomuse_amd.plugin import plugin_manager
+
+    if request.is_json:
+        body = request.get_json(silent=True) or {}
+        name = body.get("name", "")
+        try:
+            blob = base64.b64decode(body.get("zip_base64", ""))
+        except Exception:
+            return jsonify({"error": "invalid zip_base64"}), 400
+    else:
+        name = request.args.get("name", "")
+        blob = request.get_data()
+    if not name or not name.replace("_", "").replace("-", "").isalnum():
+        return jsonify({"error": "plugin name must be alphanumeric"}), 400
+    if not blob:
+        return jsonify({"error": "empty plugin zip"}), 400
+    try:
+        plugin_manager.load_zip(name, blob)   # validate before persisting
+    except Exception as exc:  # noqa: BLE001 — surface the load error
+        return jsonify({"error": f"plugin failed to load: {exc}"}), 400
+    conn = _state().conn()
+    with write_txn(conn):
+        conn.execute(
+            "INSERT INTO plugin (name, blob, enabled) VALUES (?, ?, 1) "
+            "ON CONFLICT(name) DO UPDATE SET blob=excluded.blob, enabled=1",
+            (name, blob))
+    plugin_manager.sync_cron(conn)
+    return jsonify({"loaded": name}), 201
+
+
+@bp.delete("/api/plugins/<name>")
+@require_auth
+def delete_plugin(name):
+    from audiomuse_amd.db import write_txn
+    from audiomuse_amd.plugin import plugin_manager
+
+    conn = _state().conn()
+    with write_txn(conn):
+        cur = conn.execute("DELETE FROM plugin WHERE name = ?", (name,))
+    plugin_manager.loaded.pop(name, None)
+    plugin_manager.sync_cron(conn)
+    if cur.rowcount == 0:
+        return jsonify({"error": "unknown plugin"}), 404
+    return jsonify({"deleted": name})

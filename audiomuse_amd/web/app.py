@@ -140,6 +140,14 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             import logging
 
             logging.getLogger(__name__).exception("startup migration failed")
+        try:
+            from audiomuse_amd.plugin import plugin_manager
+
+            plugin_manager.load_from_db(state.conn())
+        except Exception:  # noqa: BLE001 — plugin failures never block boot
+            import logging
+
+            logging.getLogger(__name__).exception("plugin boot load failed")
 
     # request-level logging (reference: note_request_start/end, app.py:194)
     import logging as _logging
