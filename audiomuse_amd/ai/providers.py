@@ -17,7 +17,6 @@ from __future__ import annotations
 
 import json
 import logging
-import os
 from typing import Dict, List, Optional
 
 from audiomuse_amd import config as C
@@ -41,8 +40,8 @@ _DEFAULT_MODEL = {
 
 
 def _base_and_key(provider: str) -> tuple:
-    base = os.environ.get("AI_BASE_URL", _DEFAULT_BASE[provider]).rstrip("/")
-    key = os.environ.get("AI_API_KEY", "")
+    base = (C.AI_BASE_URL or _DEFAULT_BASE[provider]).rstrip("/")
+    key = C.AI_API_KEY
     validate_outbound_url(base)
     return base, key
 

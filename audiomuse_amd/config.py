@@ -199,6 +199,8 @@ RCCL_BUCKET_CAP_MB = _env_int("RCCL_BUCKET_CAP_MB", 64)
 # AI instant playlist (reference: tasks/ai/)
 AI_PROVIDER = _env("AI_PROVIDER", "none")  # none|openai|gemini|mistral
 AI_MODEL_NAME = _env("AI_MODEL_NAME", "")
+AI_BASE_URL = _env("AI_BASE_URL", "")      # override vendor default base
+AI_API_KEY = _env("AI_API_KEY", "")
 AI_MAX_TOOL_CALLS = _env_int("AI_MAX_TOOL_CALLS", 4)
 
 # Web

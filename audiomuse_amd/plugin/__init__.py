@@ -1,3 +1,5 @@
+"""Plugin package: zip-packaged extensions (see manager.py)."""
+
 from audiomuse_amd.plugin.manager import (PluginManager,  # noqa: F401
                                           hook_registry)
 

@@ -104,7 +104,7 @@ def test_network_failure_returns_none():
 
 
 def test_ssrf_guard_blocks_internal_base(monkeypatch):
-    monkeypatch.setenv("AI_BASE_URL", "http://169.254.169.254/latest")
+    monkeypatch.setattr(C, "AI_BASE_URL", "http://169.254.169.254/latest")
 
     def post(url, **kw):  # must never be reached
         raise AssertionError("request was sent to a blocked URL")
