@@ -24,6 +24,7 @@ from audiomuse_amd.models.musicnn import (MusiCNNEmbedding, MusiCNNPrediction,
 from audiomuse_amd.models.text import TextEmbedder, clap_text_config
 from audiomuse_amd.ops import dsp, features, hip_ops
 from audiomuse_amd.ops.audio_io import load_audio
+from audiomuse_amd.utils.resources import oom_retry
 
 
 @dataclass
@@ -218,8 +219,8 @@ class AnalysisRuntime:
         # one MusiCNN pass for the whole album
         if patch_batches:
             all_patches = torch.cat(patch_batches, dim=0).float()
-            emb = self.musicnn_emb(all_patches)
-            logits = self.musicnn_pred(emb)
+            emb = oom_retry(self.musicnn_emb, all_patches)
+            logits = oom_retry(self.musicnn_pred, emb)
             owner = torch.tensor(patch_owner)
             for i, res in enumerate(results):
                 if res is None:
@@ -234,7 +235,8 @@ class AnalysisRuntime:
             all_segs = torch.cat(seg_batches, dim=0)
             mel = hip_ops.mel_spectrogram(all_segs, dsp.clap_mel_config(),
                                           quantize_int16=True)
-            embs = self.htsat(mel.to(self.dtype)).float()
+            embs = oom_retry(lambda m: self.htsat(m.to(self.dtype)).float(),
+                             mel)
             owner = torch.tensor(seg_owner)
             for i, res in enumerate(results):
                 if res is None:

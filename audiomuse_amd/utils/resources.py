@@ -83,3 +83,24 @@ class ModelLifecycle:
     @property
     def loaded(self) -> bool:
         return self._model is not None
+
+
+def oom_retry(fn, x: torch.Tensor, min_chunk: int = 1):
+    """Run `fn(batch)` over `x`, halving the batch on HIP out-of-memory
+    until it fits (reference: memory_utils.handle_onnx_memory_error :282
+    — its ONNX OOM -> CPU-retry ladder; here the ladder is batch halving
+    on the same GPU, since 288 GB of HBM makes a CPU fallback strictly
+    worse). Results are concatenated along dim 0."""
+    n = x.shape[0]
+    chunk = n
+    while True:
+        try:
+            if chunk >= n:
+                return fn(x)
+            outs = [fn(x[i : i + chunk]) for i in range(0, n, chunk)]
+            return torch.cat(outs, dim=0)
+        except torch.cuda.OutOfMemoryError:
+            release_memory_to_os()
+            if chunk <= min_chunk:
+                raise
+            chunk = max(min_chunk, chunk // 2)

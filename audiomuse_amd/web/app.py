@@ -97,6 +97,26 @@ class AppState:
             self._stamps.clear()
 
 
+class ProxyPrefixMiddleware:
+    """Honor X-Forwarded-Prefix from a reverse proxy so url_for and
+    redirects carry the mount prefix (reference: proxy_prefix.py:31)."""
+
+    def __init__(self, wsgi_app):
+        self.wsgi_app = wsgi_app
+
+    def __call__(self, environ, start_response):
+        prefix = environ.get("HTTP_X_FORWARDED_PREFIX", "").rstrip("/")
+        if prefix and prefix.startswith("/"):
+            environ["SCRIPT_NAME"] = prefix
+            path = environ.get("PATH_INFO", "")
+            if path.startswith(prefix):
+                environ["PATH_INFO"] = path[len(prefix):] or "/"
+        scheme = environ.get("HTTP_X_FORWARDED_PROTO", "")
+        if scheme in ("http", "https"):
+            environ["wsgi.url_scheme"] = scheme
+        return self.wsgi_app(environ, start_response)
+
+
 def create_app(db_url: Optional[str] = None, device: str = "cpu",
                auth_disabled: bool = False) -> Flask:
     import os as _os
@@ -105,6 +125,7 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
                 static_folder=_os.path.join(_os.path.dirname(
                     _os.path.abspath(__file__)), "static"),
                 static_url_path="/static")
+    app.wsgi_app = ProxyPrefixMiddleware(app.wsgi_app)
     state = AppState(db_url or C.DATABASE_URL, device=device)
     app.extensions["audiomuse"] = state
     app.config["AUTH_DISABLED"] = auth_disabled

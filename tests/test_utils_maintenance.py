@@ -162,3 +162,28 @@ def test_catalogue_never_deleted_by_cleaning(db):
     # catalogue untouched
     assert db.execute("SELECT COUNT(*) FROM score").fetchone()[0] == 5
     assert db.execute("SELECT COUNT(*) FROM embedding").fetchone()[0] == 5
+
+
+def test_oom_retry_halves_batch():
+    import torch
+
+    from audiomuse_amd.utils.resources import oom_retry
+
+    calls = []
+
+    def fn(b):
+        calls.append(b.shape[0])
+        if b.shape[0] > 3:
+            raise torch.cuda.OutOfMemoryError("fake OOM")
+        return b * 2
+
+    x = torch.arange(10).reshape(10, 1).float()
+    assert torch.equal(oom_retry(fn, x), x * 2)
+    assert calls[0] == 10 and max(calls[2:]) <= 3  # halved until it fit
+
+    def always(b):
+        raise torch.cuda.OutOfMemoryError("cannot fit even one row")
+
+    import pytest as _pytest
+    with _pytest.raises(torch.cuda.OutOfMemoryError):
+        oom_retry(always, x)

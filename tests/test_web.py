@@ -335,3 +335,11 @@ def test_lyrics_axes_similar(client_ids):
     # odd-index track has no profile -> 404
     assert client.get(
         f"/api/lyrics_axes_similar?item_id={ids[1]}").status_code == 404
+
+
+def test_proxy_prefix_middleware(client_ids):
+    client, _ = client_ids
+    r = client.get("/am/health", headers={"X-Forwarded-Prefix": "/am"})
+    assert r.status_code == 200 and r.json["status"] == "ok"
+    # without the header the prefixed path does not exist
+    assert client.get("/am/health").status_code == 404
