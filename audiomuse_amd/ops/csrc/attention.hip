@@ -129,6 +129,20 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
   // many heads) still fill all 256 CUs
   const int h = blockIdx.y * ATTN_WAVES + wave;
   if (h < heads) {
+    // ---- Q (A-frags) and K (B-frags) straight from global, issued
+    // FIRST so the S MFMAs can start while the V staging drains ----
+    bf16x8 qf[4], kf[4];
+    const int kk = 8 * (lane >> 4);   // k-offset of this lane's fragment
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr) {
+      const int t = tr * 16 + (lane & 15);
+      int si, sj, wr_;
+      src_of(t, si, sj, wr_);
+      const long long base = (((long long)b * H + si) * W + sj) * 3 * C + h * 32;
+      qf[tr] = *(const bf16x8*)(qkv + base + kk);          // Q slice
+      kf[tr] = *(const bf16x8*)(qkv + base + C + kk);      // K slice
+    }
+
     // ---- stage V transposed: VT[d][t] = V[t][d] ----
     {
       const __bf16* vptr = qkv + my_base + 2 * C + h * 32;
@@ -140,19 +154,6 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           VT[AM_SWZ(g * 8 + j, lane)] = (__bf16)vv[g][j];
-    }
-
-    // ---- Q (A-frags) and K (B-frags) straight from global ----
-    bf16x8 qf[4], kf[4];
-    const int kk = 8 * (lane >> 4);   // k-offset of this lane's fragment
-#pragma unroll
-    for (int tr = 0; tr < 4; ++tr) {
-      const int t = tr * 16 + (lane & 15);
-      int si, sj, wr_;
-      src_of(t, si, sj, wr_);
-      const long long base = (((long long)b * H + si) * W + sj) * 3 * C + h * 32;
-      qf[tr] = *(const bf16x8*)(qkv + base + kk);          // Q slice
-      kf[tr] = *(const bf16x8*)(qkv + base + C + kk);      // K slice
     }
 
     // ---- S = QK^T: 4x4 tiles of 16x16, one MFMA each ----
@@ -236,11 +237,11 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int row = tr * 16 + row_grp + reg;
+        const float inv = __frcp_rn(rsum[tr][reg] + 1e-20f);
 #pragma unroll
         for (int tc = 0; tc < 4; ++tc) {
           const int col = tc * 16 + col_in_tile;
-          P[AM_SWZ(row, col)] =
-              (__bf16)(s[tr][tc][reg] / (rsum[tr][reg] + 1e-20f));
+          P[AM_SWZ(row, col)] = (__bf16)(s[tr][tc][reg] * inv);
         }
       }
     __syncthreads();
