@@ -167,10 +167,13 @@ class SwinBlock(nn.Module):
                                 approximate="tanh")
                 return x2 + fp8.scaled_linear(hidden, self.mlp[2].weight,
                                               self.mlp[2].bias)
-            # MLP with the GELU fused into the first GEMM's epilogue
+            # MLP with the GELU fused into the first GEMM's epilogue and
+            # the residual add folded into the second GEMM (beta=1)
             hidden = ext.linear_gelu(xn2, self.mlp[0].weight.contiguous(),
                                      self.mlp[0].bias.contiguous())
-            return x2 + self.mlp[2](hidden)
+            return ext.linear_bias_add(hidden,
+                                       self.mlp[2].weight.contiguous(),
+                                       self.mlp[2].bias.contiguous(), x2)
         shortcut = x
         x = self.norm1(x).view(B, H, W, C)
         if self.shift:

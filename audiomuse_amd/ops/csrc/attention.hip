@@ -157,6 +157,20 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
     }
 
     // ---- S = QK^T: 4x4 tiles of 16x16, one MFMA each ----
+    // Bias loads for the first row-tile are issued BEFORE the MFMAs so
+    // their ~L2 latency hides under the matrix work; each row-tile then
+    // prefetches the next tile's bias before processing its own
+    // (double-buffered software pipeline).
+    const int col_in_tile = lane & 15;
+    const int row_grp = (lane >> 4) * 4;
+    const float* bias_base = bias + (h * 64 + row_grp) * 64 + col_in_tile;
+    float bv[2][4][4];  // [buf][reg][tc]
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc)
+        bv[0][reg][tc] = bias_base[reg * 64 + tc * 16];
+
     f32x4 s[4][4];
 #pragma unroll
     for (int tr = 0; tr < 4; ++tr)
@@ -169,20 +183,17 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
 
     // ---- scale + bias + shift mask, then row softmax in registers ----
     // C-frag: col = lane&15, row = (lane>>4)*4 + reg
-    const int col_in_tile = lane & 15;
-    const int row_grp = (lane >> 4) * 4;
     float rmax[4][4];  // [tr][reg]
 #pragma unroll
     for (int tr = 0; tr < 4; ++tr) {
-      // issue this row-tile's 16 independent bias loads up front — the
-      // fmax chain otherwise serializes on one ~500-cycle L2 hit at a time
-      float bvals[4][4];
+      if (tr < 3) {
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg)
+        for (int reg = 0; reg < 4; ++reg)
 #pragma unroll
-        for (int tc = 0; tc < 4; ++tc)
-          bvals[reg][tc] = bias[(h * 64 + tr * 16 + row_grp + reg) * 64 +
-                                tc * 16 + col_in_tile];
+          for (int tc = 0; tc < 4; ++tc)
+            bv[(tr + 1) & 1][reg][tc] =
+                bias_base[((tr + 1) * 16 + reg) * 64 + tc * 16];
+      }
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int row = tr * 16 + row_grp + reg;
@@ -194,7 +205,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
           const int col = tc * 16 + col_in_tile;
           const int cwrap = (((wrap_r_mask >> col) & 1ull) << 1) |
                             ((wrap_c_mask >> col) & 1ull);
-          float v = s[tr][tc][reg] * scale + bvals[reg][tc];
+          float v = s[tr][tc][reg] * scale + bv[tr & 1][reg][tc];
           if (shift && rwrap != cwrap) v = -1e30f;
           s[tr][tc][reg] = v;
           m = fmaxf(m, v);

@@ -1,10 +1,14 @@
-// hipBLASLt GEMM with fused bias+GELU epilogue (bf16 in/out, fp32 compute).
+// hipBLASLt GEMMs with fused epilogues (bf16 in/out, fp32 compute).
 //
-// The encoder MLP is Linear(C,4C) -> GELU -> Linear(4C,C); the eager GELU
-// pass re-reads and re-writes the 4C-wide activation (measured ~9% of step
-// time, profiles/r01_kernel_pmc.md). hipBLASLt applies GELU+bias in the
-// GEMM epilogue, eliminating that round trip. Plain library GEMM use —
-// the hand-written MFMA work stays in the attention/mel/scan kernels.
+// linear_gelu:     y = gelu(x @ w^T + b)        (epilogue GELU_BIAS)
+// linear_bias_add: y = x @ w^T + b + residual   (epilogue BIAS, beta=1)
+//
+// The encoder MLP is Linear(C,4C) -> GELU -> Linear(4C,C) + residual; an
+// eager GELU re-reads and re-writes the 4C-wide activation (measured ~9%
+// of step time, profiles/r01_kernel_pmc.md) and the trailing residual
+// add re-reads both the GEMM output and the skip tensor. Both ride the
+// GEMM epilogue instead. Plain library GEMM use — the hand-written MFMA
+// work stays in the attention/mel/scan kernels.
 
 #include <hipblaslt/hipblaslt.h>
 #include <torch/extension.h>
@@ -35,24 +39,26 @@ hipblasLtHandle_t get_handle() {
 
 struct AlgoKey {
   int64_t m, n, k;
+  int epi;
   bool operator==(const AlgoKey& o) const {
-    return m == o.m && n == o.n && k == o.k;
+    return m == o.m && n == o.n && k == o.k && epi == o.epi;
   }
 };
 struct AlgoKeyHash {
   size_t operator()(const AlgoKey& k) const {
-    return std::hash<int64_t>()(k.m * 1315423911 ^ k.n * 2654435761 ^ k.k);
+    return std::hash<int64_t>()(k.m * 1315423911 ^ k.n * 2654435761 ^ k.k ^
+                                (int64_t)k.epi << 40);
   }
 };
 
 std::mutex algo_mu;
 std::unordered_map<AlgoKey, hipblasLtMatmulAlgo_t, AlgoKeyHash> algo_cache;
 
-}  // namespace
-
-// y = gelu(x @ w^T + bias); x (M, K) bf16 row-major, w (N, K), bias (N)
-static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
-                                 torch::Tensor bias) {
+// Shared driver: D(M,N) = epilogue(x(M,K) @ w(N,K)^T + bias) + beta*C.
+// Col-major framing: D(N,M)cm = op(A=w_rm seen cm (K,N), T) x op(B=x_cm
+// (K,M), N); C shares D's layout, so a row-major (M,N) residual is legal.
+torch::Tensor lt_linear(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                        hipblasLtEpilogue_t epi, const torch::Tensor* resid) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
                   x.is_contiguous(),
               "x must be contiguous bf16 GPU");
@@ -64,12 +70,17 @@ static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
   const int64_t M = x.numel() / K;
   const int64_t N = w.size(0);
   TORCH_CHECK(w.size(1) == K && bias.numel() == N, "shape mismatch");
+  if (resid != nullptr) {
+    TORCH_CHECK(resid->is_contiguous() &&
+                    resid->scalar_type() == at::kBFloat16 &&
+                    resid->numel() == M * N,
+                "residual must be contiguous bf16 of (M, N)");
+  }
 
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = torch::empty(sizes, x.options());
 
-  // col-major framing: D(N,M) = op(A=w_rm seen cm (K,N), T) x op(B=x_cm (K,M), N)
   hipblasLtMatmulDesc_t desc;
   HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F,
                                             HIP_R_32F));
@@ -78,7 +89,6 @@ static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
       desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
-  hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_GELU_BIAS;
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
   const void* bias_ptr = bias.data_ptr();
@@ -103,7 +113,7 @@ static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
   bool have_algo = false;
   {
     std::lock_guard<std::mutex> g(algo_mu);
-    auto it = algo_cache.find({M, N, K});
+    auto it = algo_cache.find({M, N, K, (int)epi});
     if (it != algo_cache.end()) {
       algo = it->second;
       have_algo = true;
@@ -120,17 +130,19 @@ static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
     HIPBLASLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
         get_handle(), desc, la, lb, ld, ld, pref, 4, results, &found));
     hipblasLtMatmulPreferenceDestroy(pref);
-    TORCH_CHECK(found > 0, "no hipblaslt algo for gelu epilogue at ", M, "x",
-                N, "x", K);
+    TORCH_CHECK(found > 0, "no hipblaslt algo for epilogue ", (int)epi,
+                " at ", M, "x", N, "x", K);
     algo = results[0].algo;
     std::lock_guard<std::mutex> g(algo_mu);
-    algo_cache[{M, N, K}] = algo;
+    algo_cache[{M, N, K, (int)epi}] = algo;
   }
 
-  const float alpha = 1.0f, beta = 0.0f;
+  const float alpha = 1.0f;
+  const float beta = resid != nullptr ? 1.0f : 0.0f;
+  const void* c_ptr = resid != nullptr ? resid->data_ptr() : y.data_ptr();
   HIPBLASLT_CHECK(hipblasLtMatmul(
       get_handle(), desc, &alpha, w.data_ptr(), la, x.data_ptr(), lb, &beta,
-      y.data_ptr(), ld, y.data_ptr(), ld, &algo, workspace, workspace_size,
+      c_ptr, ld, y.data_ptr(), ld, &algo, workspace, workspace_size,
       stream.stream()));
 
   hipblasLtMatrixLayoutDestroy(la);
@@ -140,7 +152,21 @@ static torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+torch::Tensor linear_gelu(torch::Tensor x, torch::Tensor w,
+                          torch::Tensor bias) {
+  return lt_linear(x, w, bias, HIPBLASLT_EPILOGUE_GELU_BIAS, nullptr);
+}
+
+torch::Tensor linear_bias_add(torch::Tensor x, torch::Tensor w,
+                              torch::Tensor bias, torch::Tensor resid) {
+  return lt_linear(x, w, bias, HIPBLASLT_EPILOGUE_BIAS, &resid);
+}
+
+}  // namespace
+
 void register_gemm_gelu(pybind11::module_& m) {
   m.def("linear_gelu", &linear_gelu,
         "gelu(x @ w.T + bias) via hipBLASLt epilogue fusion");
+  m.def("linear_bias_add", &linear_bias_add,
+        "x @ w.T + bias + residual via hipBLASLt beta=1");
 }
