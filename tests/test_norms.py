@@ -83,3 +83,18 @@ def test_linear_gelu_epilogue_matches_reference():
                     approximate="tanh")
     assert got.shape == expect.shape
     torch.testing.assert_close(got.float(), expect, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_linear_bias_add_matches_reference():
+    import audiomuse_amd._C as C
+
+    torch.manual_seed(1)
+    x = torch.randn(5, 17, 512, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(128, 512, device="cuda", dtype=torch.bfloat16) * 0.05
+    b = torch.randn(128, device="cuda", dtype=torch.bfloat16)
+    r = torch.randn(5, 17, 128, device="cuda", dtype=torch.bfloat16)
+    got = C.linear_bias_add(x.contiguous(), w.contiguous(), b.contiguous(),
+                            r.contiguous())
+    expect = F.linear(x.float(), w.float(), b.float()) + r.float()
+    torch.testing.assert_close(got.float(), expect, rtol=3e-2, atol=3e-2)
