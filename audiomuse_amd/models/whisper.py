@@ -80,6 +80,25 @@ class _SelfAttn(nn.Module):
             out = F.scaled_dot_product_attention(q, k, v, is_causal=self.causal)
         return self.proj(out.transpose(1, 2).reshape(B, L, D))
 
+    def forward_static(self, x: torch.Tensor, kv_cache, len_t: torch.Tensor,
+                       arange_T: torch.Tensor):
+        """Fixed-shape single-token step for hipGraph capture: the write
+        position and attention extent come from the device tensor `len_t`
+        (read at replay time), so every replay has identical shapes and
+        kernel arguments."""
+        B, L, D = x.shape                      # L == 1
+        h = self.heads
+        q, k, v = self.qkv(x).view(B, L, 3, h, D // h).permute(2, 0, 3, 1, 4).unbind(0)
+        ck, cv = kv_cache                      # (B, h, T_max, d)
+        idx = len_t.view(1)
+        ck.index_copy_(2, idx, k)
+        cv.index_copy_(2, idx, v)
+        # attend over the FULL preallocated cache; positions > len masked
+        amask = torch.where(arange_T.view(1, 1, 1, -1) <= len_t, 0.0,
+                            float("-inf")).to(q.dtype)
+        out = F.scaled_dot_product_attention(q, ck, cv, attn_mask=amask)
+        return self.proj(out.transpose(1, 2).reshape(B, L, D))
+
 
 class _CrossAttn(nn.Module):
     def __init__(self, dim: int, heads: int):
@@ -134,6 +153,12 @@ class DecoderBlock(nn.Module):
 
     def forward(self, x, self_cache, cache_len, cross_kv):
         x = x + self.self_attn(self.norm1(x), self_cache, cache_len)
+        x = x + self.cross(self.norm2(x), cross_kv)
+        return x + self.mlp(self.norm3(x))
+
+    def forward_static(self, x, self_cache, len_t, arange_T, cross_kv):
+        x = x + self.self_attn.forward_static(self.norm1(x), self_cache,
+                                              len_t, arange_T)
         x = x + self.cross(self.norm2(x), cross_kv)
         return x + self.mlp(self.norm3(x))
 
@@ -193,6 +218,62 @@ class WhisperModel(nn.Module):
     def cross_kvs(self, enc: torch.Tensor):
         return [blk.cross.precompute(enc) for blk in self.dec_blocks]
 
+    def decode_step_static(self, tok: torch.Tensor, len_t: torch.Tensor,
+                           caches, cross_kvs, arange_T) -> torch.Tensor:
+        """Fixed-shape (B, 1) decode step, hipGraph-capturable."""
+        x = self.tok_emb(tok) + self.dec_pos.index_select(0, len_t.view(1))[None]
+        for blk, cache, ckv in zip(self.dec_blocks, caches, cross_kvs):
+            x = blk.forward_static(x, cache, len_t, arange_T, ckv)
+        x = self.dec_norm(x)
+        return x @ self.tok_emb.weight.T
+
+
+class GraphedDecoder:
+    """hipGraph-captured single-token decode loop (greedy, B=1).
+
+    The eager decode_step launches ~150 tiny kernels per token (12 blocks
+    x GEMMs/LNs/SDPA at M=1) — pure launch latency. This captures ONE
+    fixed-shape step (decode_step_static) into a hipGraph; each token is
+    then tok/len update + one graph replay. The reference's ONNX decoder
+    re-runs a full session per token (whisper_onnx.py:332-528); this is
+    the MI355X-native answer to SURVEY hard part #1.
+    """
+
+    def __init__(self, model: WhisperModel, enc: torch.Tensor):
+        cfg = model.cfg
+        device, dtype = enc.device, enc.dtype
+        self.model = model
+        self.caches = model.make_caches(1, device, dtype)
+        self.ckv = model.cross_kvs(enc)
+        self.tok = torch.zeros(1, 1, dtype=torch.long, device=device)
+        self.len_t = torch.zeros((), dtype=torch.long, device=device)
+        self.arange_T = torch.arange(cfg.max_tokens, device=device)
+        # warm up the exact op sequence on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self.model.decode_step_static(self.tok, self.len_t,
+                                              self.caches, self.ckv,
+                                              self.arange_T)
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.logits = self.model.decode_step_static(
+                self.tok, self.len_t, self.caches, self.ckv, self.arange_T)
+        # the warmup/capture wrote junk at position 0; the caller's first
+        # eager prompt step overwrites it and the mask hides the rest
+        for ck, cv in self.caches:
+            ck.zero_()
+            cv.zero_()
+
+    def step(self, token: int, pos: int) -> torch.Tensor:
+        """Returns logits (1, 1, vocab) for `token` written at `pos`."""
+        self.tok.fill_(token)
+        self.len_t.fill_(pos)
+        self.graph.replay()
+        return self.logits
+
 
 def _block_repeats(logits: torch.Tensor, seq: List[int],
                    repetition_penalty: float, no_repeat_ngram: int) -> None:
@@ -226,13 +307,28 @@ def detect_language(model: WhisperModel, enc: torch.Tensor) -> int:
 def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
                   max_tokens: Optional[int] = None,
                   repetition_penalty: float = 1.2,
-                  no_repeat_ngram: int = 3) -> List[int]:
-    """Greedy KV-cache decode of one chunk. mel (n_mels, T)."""
+                  no_repeat_ngram: int = 3,
+                  use_graph: Optional[bool] = None) -> List[int]:
+    """Greedy KV-cache decode of one chunk. mel (n_mels, T).
+
+    On GPU the per-token step runs as one hipGraph replay
+    (GraphedDecoder) unless use_graph=False; CPU always runs eager."""
     cfg = model.cfg
     max_tokens = min(max_tokens or cfg.max_tokens - 4, cfg.max_tokens - 4)
     enc = model.encode(mel.unsqueeze(0))
-    caches = model.make_caches(1, enc.device, enc.dtype)
-    ckv = model.cross_kvs(enc)
+    if use_graph is None:
+        use_graph = enc.is_cuda
+    gdec: Optional[GraphedDecoder] = None
+    if use_graph and enc.is_cuda:
+        try:
+            gdec = GraphedDecoder(model, enc)
+        except Exception:  # noqa: BLE001 — capture failure falls back to eager
+            gdec = None
+    if gdec is not None:
+        caches, ckv = gdec.caches, gdec.ckv
+    else:
+        caches = model.make_caches(1, enc.device, enc.dtype)
+        ckv = model.cross_kvs(enc)
     lang = detect_language(model, enc)
     prompt = [TOK_SOT, TOK_LANG_BASE + lang]
     logits = model.decode_step(
@@ -246,8 +342,12 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
         if nxt == TOK_EOT:
             break
         seq.append(nxt)
-        out = model.decode_step(
-            torch.tensor([[nxt]], device=enc.device), cache_len, caches, ckv)
+        if gdec is not None:
+            out = gdec.step(nxt, cache_len)
+        else:
+            out = model.decode_step(
+                torch.tensor([[nxt]], device=enc.device), cache_len, caches,
+                ckv)
         cache_len += 1
         step_logits = out[0, -1].float()
     return seq

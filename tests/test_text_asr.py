@@ -151,3 +151,34 @@ def test_lyrics_pipeline_vad_gates_asr():
     res2 = pipe2.analyze(audio=torch.randn(16000) * 0.01)
     assert calls and res2.source == "asr"
     assert len(res2.text.split()) <= 300
+
+
+def test_whisper_static_step_matches_eager():
+    """decode_step_static (the hipGraph-capturable body) produces the
+    same logits as the eager decode_step at every position."""
+    torch.manual_seed(3)
+    m = tiny_whisper().eval()
+    mel = torch.randn(8, 64)
+    with torch.inference_mode():
+        enc = m.encode(mel.unsqueeze(0))
+        ckv = m.cross_kvs(enc)
+        toks = [1, 10, 31, 44, 17]
+        caches_a = m.make_caches(1, enc.device, enc.dtype)
+        caches_b = m.make_caches(1, enc.device, enc.dtype)
+        arange_T = torch.arange(m.cfg.max_tokens)
+        for i, t in enumerate(toks):
+            tt = torch.tensor([[t]])
+            ea = m.decode_step(tt, i, caches_a, ckv)
+            st = m.decode_step_static(tt, torch.tensor(i), caches_b, ckv,
+                                      arange_T)
+            torch.testing.assert_close(ea, st, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.gpu
+def test_whisper_graphed_decode_matches_eager_gpu():
+    torch.manual_seed(4)
+    m = tiny_whisper().to("cuda", torch.bfloat16).eval()
+    mel = torch.randn(8, 64, device="cuda", dtype=torch.bfloat16)
+    eager = greedy_decode(m, mel, max_tokens=12, use_graph=False)
+    graphed = greedy_decode(m, mel, max_tokens=12, use_graph=True)
+    assert graphed == eager
