@@ -141,6 +141,9 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
     app.register_blueprint(tasks_bp)
     app.register_blueprint(chat_bp)
     app.register_blueprint(admin_bp)
+    from audiomuse_amd.web.api_external import bp as external_bp
+
+    app.register_blueprint(external_bp)
 
     from audiomuse_amd.web.auth import seed_admin_from_env
 

@@ -343,3 +343,44 @@ def test_proxy_prefix_middleware(client_ids):
     assert r.status_code == 200 and r.json["status"] == "ok"
     # without the header the prefixed path does not exist
     assert client.get("/am/health").status_code == 404
+
+
+def test_external_api(client_ids):
+    client, ids = client_ids
+    # canonical id passes through
+    r = client.get(f"/external/get_score?id={ids[3]}")
+    assert r.status_code == 200 and r.json["title"] == "Song 3"
+    r = client.get(f"/external/get_embedding?id={ids[3]}")
+    assert r.status_code == 200 and len(r.json["embedding"]) == 200
+    assert isinstance(r.json["embedding"][0], float)
+    # unknown id -> 404; missing -> 400
+    assert client.get("/external/get_score?id=zzz").status_code == 404
+    assert client.get("/external/get_score").status_code == 400
+    # unified autocomplete
+    r = client.get("/external/search?q=Song 1")
+    assert r.status_code == 200 and len(r.json) >= 1
+    # legacy title/artist params
+    r = client.get("/external/search?title=Song&artist=Artist 2")
+    assert r.status_code == 200
+    assert all(b["author"] == "Artist 2" for b in r.json)
+
+
+def test_external_provider_id_resolution(client_ids):
+    client, ids = client_ids
+    from audiomuse_amd.db import write_txn
+
+    state = None
+    # map a provider id onto ids[5] through the app's own connection
+    import flask
+
+    app = client.application
+    conn = app.extensions["audiomuse"].conn()
+    with write_txn(conn):
+        conn.execute(
+            "INSERT OR REPLACE INTO track_server_map "
+            "(provider_id, server_id, item_id, title, author) "
+            "VALUES ('prov-42', 'srv1', ?, 'Song 5', 'Artist 5')", (ids[5],))
+    r = client.get("/external/get_score?id=prov-42&server=srv1")
+    assert r.status_code == 200 and r.json["item_id"] == ids[5]
+    r = client.get("/external/get_score?id=prov-42")  # default: any server
+    assert r.status_code == 200 and r.json["item_id"] == ids[5]
