@@ -121,15 +121,34 @@ class SwinBlock(nn.Module):
                                  nn.GELU(approximate="tanh"),
                                  nn.Linear(hidden, dim))
 
-    def _fused_attn_available(self, x: torch.Tensor) -> bool:
+    def _fused_available(self, x: torch.Tensor) -> bool:
+        """Fused LN/MLP path (any window size; inference on GPU)."""
         if not (x.is_cuda and x.dtype == torch.bfloat16
-                and not torch.is_grad_enabled() and self.window == 8
-                and self.attn.heads % 2 == 0
-                and self.attn.dim // self.attn.heads == 32):
+                and not torch.is_grad_enabled()):
             return False
         from audiomuse_amd.ops import _ext
         ext = _ext.native_or_none()
         return ext is not None and hasattr(ext, "window_attn_fwd")
+
+    def _fused_attn_available(self, x: torch.Tensor) -> bool:
+        return (self._fused_available(x) and self.window == 8
+                and self.attn.heads % 2 == 0
+                and self.attn.dim // self.attn.heads == 32)
+
+    def _eager_attn(self, x: torch.Tensor, H: int, W: int,
+                    mask: torch.Tensor | None) -> torch.Tensor:
+        """norm1 -> roll/partition/attention/reverse (the attention half
+        of the eager path; callers add the residual)."""
+        B, L, C = x.shape
+        x = self.norm1(x).view(B, H, W, C)
+        if self.shift:
+            x = torch.roll(x, shifts=(-self.shift, -self.shift), dims=(1, 2))
+        win = window_partition(x, self.window)
+        win = self.attn(win, mask if self.shift else None)
+        x = window_reverse(win, self.window, H, W)
+        if self.shift:
+            x = torch.roll(x, shifts=(self.shift, self.shift), dims=(1, 2))
+        return x.view(B, L, C)
 
     def forward(self, x: torch.Tensor, H: int, W: int,
                 mask: torch.Tensor | None) -> torch.Tensor:
@@ -174,16 +193,23 @@ class SwinBlock(nn.Module):
             return ext.linear_bias_add(hidden,
                                        self.mlp[2].weight.contiguous(),
                                        self.mlp[2].bias.contiguous(), x2)
-        shortcut = x
-        x = self.norm1(x).view(B, H, W, C)
-        if self.shift:
-            x = torch.roll(x, shifts=(-self.shift, -self.shift), dims=(1, 2))
-        win = window_partition(x, self.window)
-        win = self.attn(win, mask if self.shift else None)
-        x = window_reverse(win, self.window, H, W)
-        if self.shift:
-            x = torch.roll(x, shifts=(self.shift, self.shift), dims=(1, 2))
-        x = shortcut + x.view(B, L, C)
+        if self._fused_available(x):
+            # attention must run eager (window != 8), but the add+LN and
+            # MLP fusions still apply (stage 4's window-4 blocks)
+            from audiomuse_amd.ops import _ext
+            ext = _ext.require()
+            attn_out = self._eager_attn(x, H, W, mask)
+            x2, xn2 = ext.add_layernorm_bf16(
+                x.contiguous(), attn_out.contiguous(),
+                self.norm2.weight.to(torch.bfloat16).contiguous(),
+                self.norm2.bias.to(torch.bfloat16).contiguous(),
+                self.norm2.eps)
+            hidden = ext.linear_gelu(xn2, self.mlp[0].weight.contiguous(),
+                                     self.mlp[0].bias.contiguous())
+            return ext.linear_bias_add(hidden,
+                                       self.mlp[2].weight.contiguous(),
+                                       self.mlp[2].bias.contiguous(), x2)
+        x = x + self._eager_attn(x, H, W, mask)
         return x + self.mlp(self.norm2(x))
 
 
@@ -197,10 +223,13 @@ class PatchMerging(nn.Module):
 
     def forward(self, x: torch.Tensor, H: int, W: int) -> torch.Tensor:
         B, L, C = x.shape
-        x = x.view(B, H, W, C)
-        x = torch.cat([x[:, 0::2, 0::2], x[:, 1::2, 0::2],
-                       x[:, 0::2, 1::2], x[:, 1::2, 1::2]], dim=-1)
-        x = x.view(B, (H // 2) * (W // 2), 4 * C)
+        # one permute copy instead of 4 strided slices + cat (the cat
+        # measured 2.3x its roofline: CatArrayBatchedCopy, profiles/
+        # r01_final_profile.md). Channel-block order (0::2,0::2),
+        # (1::2,0::2), (0::2,1::2), (1::2,1::2) == (wpar, hpar) major.
+        x = x.view(B, H // 2, 2, W // 2, 2, C)
+        x = x.permute(0, 1, 3, 4, 2, 5).reshape(B, (H // 2) * (W // 2),
+                                                4 * C)
         return self.reduction(self.norm(x))
 
 
