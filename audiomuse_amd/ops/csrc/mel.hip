@@ -74,10 +74,65 @@ __global__ __launch_bounds__(256) void mel_fwd_kernel(
   }
   __syncthreads();
 
-  // Phase 2: DIT FFT over bit-reversed data, two radix-2 stages merged
-  // into one radix-4 pass (one LDS round-trip + one barrier per pair —
-  // measured: the stage loop dominated the kernel's VALU+LDS time).
+  // Phase 2: DIT FFT over bit-reversed data, THREE radix-2 stages merged
+  // into one radix-8 pass (one LDS round-trip + one barrier per triple —
+  // measured: the stage loop dominated the kernel's VALU+LDS time), with
+  // radix-4 / radix-2 leftovers for LOG2N % 3.
   int s = 1;
+  for (; s + 2 <= LOG2N; s += 3) {
+    const int h = 1 << (s - 1);          // stage-s half
+    for (int q = tid; q < NFFT / 8; q += blockDim.x) {
+      const int grp = q >> (s - 1);
+      const int j = q & (h - 1);
+      const int i0 = (grp << (s + 2)) + j;
+      // twiddles: w1 = W(j, 2h), w2 = W(j, 4h), w4 = W(j, 8h)
+      const float2 w1 = twiddle[j * (NFFT >> s)];
+      const float2 w2 = twiddle[j * (NFFT >> (s + 1))];
+      const float2 w4 = twiddle[j * (NFFT >> (s + 2))];
+      float2 x[8];
+#pragma unroll
+      for (int m = 0; m < 8; ++m) x[m] = zbuf[AM_ZS(i0 + m * h)];
+      // stage s: pairs (m, m+1), all with w1
+      float2 bb[8];
+#pragma unroll
+      for (int m = 0; m < 8; m += 2) {
+        const float tr = w1.x * x[m + 1].x - w1.y * x[m + 1].y;
+        const float ti = w1.x * x[m + 1].y + w1.y * x[m + 1].x;
+        bb[m] = make_float2(x[m].x + tr, x[m].y + ti);
+        bb[m + 1] = make_float2(x[m].x - tr, x[m].y - ti);
+      }
+      // stage s+1: (g, g+2) with w2, (g+1, g+3) with -i*w2
+      float2 dd[8];
+      const float2 w3 = make_float2(w2.y, -w2.x);    // -i * w2
+#pragma unroll
+      for (int g = 0; g < 8; g += 4) {
+        float tr = w2.x * bb[g + 2].x - w2.y * bb[g + 2].y;
+        float ti = w2.x * bb[g + 2].y + w2.y * bb[g + 2].x;
+        dd[g] = make_float2(bb[g].x + tr, bb[g].y + ti);
+        dd[g + 2] = make_float2(bb[g].x - tr, bb[g].y - ti);
+        tr = w3.x * bb[g + 3].x - w3.y * bb[g + 3].y;
+        ti = w3.x * bb[g + 3].y + w3.y * bb[g + 3].x;
+        dd[g + 1] = make_float2(bb[g + 1].x + tr, bb[g + 1].y + ti);
+        dd[g + 3] = make_float2(bb[g + 1].x - tr, bb[g + 1].y - ti);
+      }
+      // stage s+2: (m, m+4) with w4 * e^{-i pi m / 4}
+      const float R = 0.70710678118654752f;
+      float2 t4[4];
+      t4[0] = w4;
+      t4[1] = make_float2(R * (w4.x + w4.y), R * (w4.y - w4.x));
+      t4[2] = make_float2(w4.y, -w4.x);
+      t4[3] = make_float2(R * (w4.y - w4.x), -R * (w4.x + w4.y));
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        const float tr = t4[m].x * dd[m + 4].x - t4[m].y * dd[m + 4].y;
+        const float ti = t4[m].x * dd[m + 4].y + t4[m].y * dd[m + 4].x;
+        zbuf[AM_ZS(i0 + m * h)] = make_float2(dd[m].x + tr, dd[m].y + ti);
+        zbuf[AM_ZS(i0 + (m + 4) * h)] =
+            make_float2(dd[m].x - tr, dd[m].y - ti);
+      }
+    }
+    __syncthreads();
+  }
   for (; s + 1 <= LOG2N; s += 2) {
     const int h = 1 << (s - 1);          // stage-s half
     for (int q = tid; q < NFFT / 4; q += blockDim.x) {

@@ -87,8 +87,8 @@ def test_torch_reference_matches_numpy(cfg_fn):
 
 def _kernel_mirror_fft(x):
     """Numpy mirror of the DIT FFT in ops/csrc/mel.hip: bit-reversal,
-    then merged radix-4 passes (two radix-2 stages per LDS round-trip)
-    with a leftover radix-2 stage for odd log2(N)."""
+    then merged radix-8 passes (three radix-2 stages per LDS round-trip)
+    with radix-4 / radix-2 leftovers for log2(N) % 3."""
     n = len(x)
     log2n = n.bit_length() - 1
     assert 1 << log2n == n
@@ -98,6 +98,31 @@ def _kernel_mirror_fft(x):
     z = np.zeros(n, dtype=np.complex128)
     z[rev] = x
     s = 1
+    while s + 2 <= log2n:
+        h = 1 << (s - 1)
+        for q in range(n // 8):
+            grp, jj = q >> (s - 1), q & (h - 1)
+            i0 = (grp << (s + 2)) + jj
+            w1 = tw[jj * (n >> s)]
+            w2 = tw[jj * (n >> (s + 1))]
+            w4 = tw[jj * (n >> (s + 2))]
+            xs = [z[i0 + m * h] for m in range(8)]
+            b = [0.0] * 8
+            for m in range(0, 8, 2):
+                t = w1 * xs[m + 1]
+                b[m], b[m + 1] = xs[m] + t, xs[m] - t
+            d = [0.0] * 8
+            w3 = -1j * w2
+            for g in (0, 4):
+                t = w2 * b[g + 2]
+                d[g], d[g + 2] = b[g] + t, b[g] - t
+                t = w3 * b[g + 3]
+                d[g + 1], d[g + 3] = b[g + 1] + t, b[g + 1] - t
+            for m in range(4):
+                t = w4 * np.exp(-1j * np.pi * m / 4) * d[m + 4]
+                z[i0 + m * h] = d[m] + t
+                z[i0 + (m + 4) * h] = d[m] - t
+        s += 3
     while s + 1 <= log2n:
         h = 1 << (s - 1)
         for q in range(n // 4):
@@ -126,7 +151,7 @@ def _kernel_mirror_fft(x):
     return z
 
 
-@pytest.mark.parametrize("n", [256, 2048])
+@pytest.mark.parametrize("n", [256, 512, 1024, 2048, 4096])
 def test_kernel_fft_algorithm_matches_numpy(n):
     rng = np.random.default_rng(1)
     x = rng.standard_normal(n)

@@ -117,3 +117,38 @@ def test_worker_analysis_end_to_end_on_gpu(tmp_path):
     assert eng is not None
     conn.close()
     atasks._RUNTIME = None
+
+
+@pytest.mark.gpu
+def test_lyrics_pipeline_end_to_end_on_gpu(monkeypatch):
+    """Full lyrics chain on hardware: VAD gate -> chunked Whisper ASR
+    (hipGraph decode) -> quality gate -> GTE embed + 27 axis scores.
+    Random-init weights: provided lyrics exercise embed/axes; the ASR
+    leg is exercised for termination + instrumental sentinel shape."""
+    import numpy as np
+
+    from audiomuse_amd import config as C
+    from audiomuse_amd.analysis.pipeline import AnalysisRuntime
+
+    monkeypatch.setattr(C, "LYRICS_ENABLED", True)
+    monkeypatch.setattr(C, "LYRICS_ASR_ENABLED", True)
+    monkeypatch.setattr(C, "CLAP_ENABLED", False)
+    rt = AnalysisRuntime(device="cuda")
+    lp = rt.lyrics_pipeline
+    assert lp.asr_fn is not None and lp.vad is not None
+
+    # provided-lyrics leg: embedding + full axis coverage
+    res = lp.analyze(provided_lyrics="neon nights and engines burning "
+                                     "down the endless highway home")
+    assert res.source == "provided" and not res.instrumental
+    assert res.embedding.shape == (C.LYRICS_EMBEDDING_DIMENSION,)
+    assert set(res.axis_scores) == set(C.LYRICS_AXES)
+    assert all(np.isfinite(v) for v in res.axis_scores.values())
+
+    # audio leg: random audio through VAD + Whisper; random-init model
+    # yields either the instrumental sentinel or a gated/asr result
+    torch.manual_seed(0)
+    audio = (torch.randn(16000 * 4) * 0.2).clamp(-1, 1)
+    res2 = lp.analyze(audio=audio)
+    assert res2.source in ("instrumental", "asr", "none")
+    assert res2.embedding is not None
