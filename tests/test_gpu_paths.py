@@ -134,7 +134,7 @@ def test_lyrics_pipeline_end_to_end_on_gpu(monkeypatch):
     monkeypatch.setattr(C, "LYRICS_ASR_ENABLED", True)
     monkeypatch.setattr(C, "CLAP_ENABLED", False)
     rt = AnalysisRuntime(device="cuda")
-    lp = rt.lyrics_pipeline
+    lp = rt.lyrics_pipeline()
     assert lp.asr_fn is not None and lp.vad is not None
 
     # provided-lyrics leg: embedding + full axis coverage
