@@ -132,7 +132,9 @@ class LyricsPipeline:
             if audio.shape[-1] > C.LYRICS_MAX_AUDIO_SECONDS * 16000:
                 audio = audio[..., : C.LYRICS_MAX_AUDIO_SECONDS * 16000]
             if self.vad is not None:
-                probs = speech_probabilities(self.vad, audio)
+                vdev = next(self.vad.parameters()).device
+                probs = speech_probabilities(self.vad,
+                                             audio.to(vdev, torch.float32))
                 if speech_ratio(probs) < self.vad_speech_threshold:
                     return self._instrumental()
             transcript = self.asr_fn(audio) or ""
