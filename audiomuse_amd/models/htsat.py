@@ -122,8 +122,11 @@ class SwinBlock(nn.Module):
                                  nn.Linear(hidden, dim))
 
     def _fused_available(self, x: torch.Tensor) -> bool:
-        """Fused LN/MLP path (any window size; inference on GPU)."""
+        """Fused LN/MLP path (any window size; inference on GPU). Weights
+        must be bf16 too: an autocast teacher feeds bf16 activations
+        through fp32 parameters, which belongs on the eager path."""
         if not (x.is_cuda and x.dtype == torch.bfloat16
+                and self.mlp[0].weight.dtype == torch.bfloat16
                 and not torch.is_grad_enabled()):
             return False
         from audiomuse_amd.ops import _ext
