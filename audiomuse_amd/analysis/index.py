@@ -260,7 +260,7 @@ _REFRESHABLE = {AUDIO_INDEX: "embedding", CLAP_INDEX: "clap_embedding",
 
 def refresh_ivf_index(conn: sqlite3.Connection, name: str,
                       device: str = "cpu",
-                      max_drift: float = 0.25) -> Dict[str, int]:
+                      max_drift: Optional[float] = None) -> Dict[str, int]:
     """Incremental IVF refresh: splice new/removed tracks into the stored
     packed index (IVFIndex.add/remove) instead of rebuilding.
 
@@ -272,6 +272,8 @@ def refresh_ivf_index(conn: sqlite3.Connection, name: str,
 
     Returns {"added": a, "removed": r, "total": n, "rebuilt": 0|1}.
     """
+    if max_drift is None:
+        max_drift = C.IVF_REFRESH_MAX_DRIFT
     column = _REFRESHABLE[name]
     ids, mat = load_all_embeddings(conn, column)
     got = load_index_blob(conn, name)

@@ -138,6 +138,8 @@ IVF_NLIST_MAX = _env_int("IVF_NLIST_MAX", 8192)
 IVF_STORAGE_DTYPE = _env("IVF_STORAGE_DTYPE", "i8")  # i8 | f16 | f32
 IVF_TRAIN_POINTS_PER_CELL = _env_int("IVF_TRAIN_POINTS_PER_CELL", 256)
 IVF_RERANK_OVERFETCH = _env_int("IVF_RERANK_OVERFETCH", 4)
+# incremental refresh: full rebuild when more than this fraction changed
+IVF_REFRESH_MAX_DRIFT = _env_float("IVF_REFRESH_MAX_DRIFT", 0.25)
 IVF_KMEANS_ITERS = _env_int("IVF_KMEANS_ITERS", 25)
 IVF_MAX_PART_SIZE_MB = _env_int("IVF_MAX_PART_SIZE_MB", 32)
 
