@@ -387,23 +387,74 @@ def hyperbolic_similar():
     return jsonify(_with_meta(res))
 
 
+_MAP_PERCENTS = (100, 75, 50, 25)
+_map_cache: dict = {}
+
+
+def _map_bucket(kind: str, percent: int):
+    """Percent-sampled map bucket, JSON + gzip serialized once
+    (reference: app_map.build_map_cache :156 — deterministic samples at
+    100/75/50/25%, pre-gzipped, served with zero further work)."""
+    import gzip
+    import random
+
+    state = _state()
+    name = idx.ARTIST_MAP if kind == "artist" else idx.SONG_MAP
+    stamp = state._stamp(name)
+    key = (kind, percent, stamp)
+    hit = _map_cache.get(key)
+    if hit is not None:
+        return hit
+    data = state.engine(name)
+    if data is None:
+        return None
+    coords = data["coords"]
+    ids = data["item_ids"]
+    recs = []
+    id_key = "artist" if kind == "artist" else "item_id"
+    for i in range(len(ids)):
+        rec = {id_key: ids[i], "x": float(coords[i][0]),
+               "y": float(coords[i][1])}
+        if kind == "song":
+            meta = state.meta_fn(ids[i]) or {}
+            rec["title"] = meta.get("title", "")
+            rec["author"] = meta.get("author", "")
+            mv = meta.get("mood_vector") or {}
+            rec["mood"] = max(mv, key=mv.get) if mv else ""
+        recs.append(rec)
+    if percent < 100:
+        rng = random.Random(1234)           # deterministic sample
+        recs = rng.sample(recs, max(1, len(recs) * percent // 100))
+    raw = json.dumps(recs).encode()
+    entry = (raw, gzip.compress(raw))
+    if len(_map_cache) > 16:                # stamp change invalidates
+        _map_cache.clear()
+    _map_cache[key] = entry
+    return entry
+
+
 @bp.get("/api/map")
 @require_auth
 def music_map():
-    """2-D map coordinates (app_map.py buckets). kind=song (default)
-    or kind=artist (reference: artist-map build)."""
+    """2-D map coordinates (app_map.py buckets). kind=song (default) or
+    kind=artist; percent in {100, 75, 50, 25} selects the deterministic
+    sample; responses are served from a pre-gzipped cache."""
     kind = request.args.get("kind", "song")
-    data = _state().engine(
-        idx.ARTIST_MAP if kind == "artist" else idx.SONG_MAP)
-    if data is None:
+    percent = int(request.args.get("percent", 100))
+    if percent not in _MAP_PERCENTS:
+        return jsonify({"error": f"percent must be one of {_MAP_PERCENTS}"}), 400
+    entry = _map_bucket(kind, percent)
+    if entry is None:
         return jsonify({"error": "map not built"}), 503
-    coords = data["coords"]
-    ids = data["item_ids"]
-    limit = int(request.args.get("n", 0)) or len(ids)
-    key = "artist" if kind == "artist" else "item_id"
-    return jsonify([
-        {key: ids[i], "x": float(coords[i][0]), "y": float(coords[i][1])}
-        for i in range(min(limit, len(ids)))])
+    raw, gz = entry
+    from flask import Response
+
+    if "gzip" in (request.headers.get("Accept-Encoding") or ""):
+        return Response(gz, mimetype="application/json",
+                        headers={"Content-Encoding": "gzip",
+                                 "Cache-Control": "no-store"})
+    return Response(raw, mimetype="application/json",
+                    headers={"Cache-Control": "no-store"})
 
 
 @bp.post("/api/order_playlist")

@@ -117,8 +117,8 @@ def test_sonic_fingerprint_and_map(client_ids):
     client, ids = client_ids
     r = client.get(f"/api/sonic_fingerprint?item_id={ids[0]}&item_id={ids[1]}&n=5")
     assert r.status_code == 200 and len(r.json) == 5
-    r = client.get("/api/map?n=10")
-    assert r.status_code == 200 and len(r.json) == 10
+    r = client.get("/api/map?percent=25")
+    assert r.status_code == 200 and len(r.json) == 10  # 25% of 40
     assert {"item_id", "x", "y"} <= set(r.json[0])
 
 
@@ -384,3 +384,20 @@ def test_external_provider_id_resolution(client_ids):
     assert r.status_code == 200 and r.json["item_id"] == ids[5]
     r = client.get("/external/get_score?id=prov-42")  # default: any server
     assert r.status_code == 200 and r.json["item_id"] == ids[5]
+
+
+def test_map_percent_buckets_and_gzip(client_ids):
+    client, _ = client_ids
+    full = client.get("/api/map?percent=100").json
+    half = client.get("/api/map?percent=50").json
+    assert len(half) == len(full) // 2
+    assert {"item_id", "x", "y", "title", "author", "mood"} <= set(full[0])
+    # deterministic sample
+    again = client.get("/api/map?percent=50").json
+    assert again == half
+    assert client.get("/api/map?percent=33").status_code == 400
+    r = client.get("/api/map", headers={"Accept-Encoding": "gzip"})
+    assert r.headers.get("Content-Encoding") == "gzip"
+    import gzip as _gz
+    import json as _json
+    assert _json.loads(_gz.decompress(r.data)) == full
