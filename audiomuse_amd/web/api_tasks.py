@@ -129,6 +129,50 @@ def queue_stats():
     return jsonify(qsql.counts_by_status(_state().conn()))
 
 
+@bp.post("/api/create_playlist")
+@require_auth
+def create_playlist():
+    """Create a playlist ON the media server (reference: ALGORITHM.md
+    §6.2 playlist creation): canonical ids translate back to the
+    selected server's provider ids via track_server_map, tracks the
+    server does not have are dropped, and the response reports how many
+    were unavailable."""
+    from audiomuse_amd.mediaserver import make_provider
+
+    body = request.get_json(force=True, silent=True) or {}
+    name = body.get("name", "")
+    item_ids = body.get("item_ids", [])
+    if not name or not item_ids:
+        return jsonify({"error": "name and item_ids required"}), 400
+    conn = _state().conn()
+    server_id = body.get("server_id")
+    row = conn.execute(
+        "SELECT * FROM music_servers WHERE enabled = 1"
+        + (" AND server_id = ?" if server_id else "") + " LIMIT 1",
+        (server_id,) if server_id else ()).fetchone()
+    if row is None:
+        return jsonify({"error": "no configured media server"}), 404
+    provider_ids, missing = [], 0
+    for iid in item_ids:
+        m = conn.execute(
+            "SELECT provider_id FROM track_server_map WHERE item_id = ? "
+            "AND server_id = ?", (iid, row["server_id"])).fetchone()
+        if m is None:
+            missing += 1
+        else:
+            provider_ids.append(m["provider_id"])
+    if not provider_ids:
+        return jsonify({"error": "no tracks available on this server",
+                        "missing": missing}), 404
+    cfg = json.loads(row["config"] or "{}")
+    provider = make_provider(row["server_type"], base_url=row["base_url"],
+                             username=row["username"],
+                             credential=row["credential"], **cfg)
+    pid = provider.create_playlist(name, provider_ids)
+    return jsonify({"playlist_id": pid, "created": len(provider_ids),
+                    "missing": missing}), 201
+
+
 # -- music server registry (reference: app_music_servers.py) ---------------
 
 @bp.get("/api/servers")

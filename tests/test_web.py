@@ -401,3 +401,32 @@ def test_map_percent_buckets_and_gzip(client_ids):
     import gzip as _gz
     import json as _json
     assert _json.loads(_gz.decompress(r.data)) == full
+
+
+def test_create_playlist_on_server(client_ids):
+    client, ids = client_ids
+    from audiomuse_amd.db import write_txn
+
+    # no server registered yet -> 404
+    r = client.post("/api/create_playlist",
+                    json={"name": "Mix", "item_ids": ids[:3]})
+    assert r.status_code == 404
+    # register a synthetic server + map 2 of 3 ids onto it
+    assert client.post("/api/servers", json={
+        "server_id": "syn1", "server_type": "synthetic"}).status_code == 200
+    conn = client.application.extensions["audiomuse"].conn()
+    with write_txn(conn):
+        for k, iid in enumerate(ids[:2]):
+            conn.execute(
+                "INSERT OR REPLACE INTO track_server_map "
+                "(provider_id, server_id, item_id) VALUES (?, 'syn1', ?)",
+                (f"a0-t{k}", iid))
+    r = client.post("/api/create_playlist",
+                    json={"name": "Mix", "item_ids": ids[:3],
+                          "server_id": "syn1"})
+    assert r.status_code == 201, r.json
+    assert r.json["created"] == 2 and r.json["missing"] == 1
+    assert r.json["playlist_id"]
+    # missing args -> 400
+    assert client.post("/api/create_playlist",
+                       json={"name": "x"}).status_code == 400
