@@ -140,6 +140,9 @@ IVF_TRAIN_POINTS_PER_CELL = _env_int("IVF_TRAIN_POINTS_PER_CELL", 256)
 IVF_RERANK_OVERFETCH = _env_int("IVF_RERANK_OVERFETCH", 4)
 # incremental refresh: full rebuild when more than this fraction changed
 IVF_REFRESH_MAX_DRIFT = _env_float("IVF_REFRESH_MAX_DRIFT", 0.25)
+# cron queue-guard retry cadence (reference: ALGORITHM.md 16.2-16.3)
+CRON_RETRY_INTERVAL_MINUTES = _env_float("CRON_RETRY_INTERVAL_MINUTES", 5.0)
+CRON_RETRY_MAX_MINUTES = _env_float("CRON_RETRY_MAX_MINUTES", 120.0)
 IVF_KMEANS_ITERS = _env_int("IVF_KMEANS_ITERS", 25)
 IVF_MAX_PART_SIZE_MB = _env_int("IVF_MAX_PART_SIZE_MB", 32)
 

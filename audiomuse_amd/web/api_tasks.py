@@ -229,24 +229,42 @@ def delete_server(server_id: str):
 @bp.get("/api/cron")
 @require_auth
 def cron_list():
-    rows = _state().conn().execute(
+    """Rows plus retry-pending state (reference: ALGORITHM.md 16.2 step
+    6 — the Scheduled Tasks page shows a waiting schedule as waiting)."""
+    conn = _state().conn()
+    rows = conn.execute(
         "SELECT id, name, schedule, task_type, payload, enabled FROM cron"
     ).fetchall()
-    return jsonify([dict(r) for r in rows])
+    retries = {r["cron_id"]: r for r in conn.execute(
+        "SELECT cron_id, attempts, due_at FROM cron_retry").fetchall()}
+    out = []
+    for r in rows:
+        d = dict(r)
+        ret = retries.get(r["id"])
+        d["retry_pending"] = ret is not None
+        if ret is not None:
+            d["retry_attempts"] = ret["attempts"]
+            d["retry_due_at"] = ret["due_at"]
+        out.append(d)
+    return jsonify(out)
 
 
 @bp.post("/api/cron")
 @require_auth
 def cron_add():
     from audiomuse_amd.db import write_txn
+    from audiomuse_amd.utils.cron import validate_cron
 
     body = request.get_json(force=True, silent=True) or {}
+    schedule = body.get("schedule", "0 3 * * *")
+    if not validate_cron(schedule):
+        return jsonify({"error": f"invalid cron expression {schedule!r}"}), 400
     conn = _state().conn()
     with write_txn(conn):
         cur = conn.execute(
             "INSERT INTO cron (name, schedule, task_type, payload, enabled) "
             "VALUES (?,?,?,?,1)",
-            (body.get("name", ""), body.get("schedule", "0 3 * * *"),
+            (body.get("name", ""), schedule,
              body.get("task_type", "rebuild_indexes"),
              json.dumps(body.get("payload", {}))))
     return jsonify({"id": cur.lastrowid})

@@ -268,3 +268,61 @@ def test_plugin_upload_endpoint_and_worker_boot(tmp_path):
     finally:
         plugin_manager.loaded.clear()
         conn.close()
+
+
+def test_cron_validate():
+    from audiomuse_amd.utils.cron import validate_cron
+
+    assert validate_cron("0 3 * * *")
+    assert validate_cron("*/5 1-4 * * 0-5")
+    assert not validate_cron("99 * * * *")      # can never fire
+    assert not validate_cron("* * * *")          # wrong arity
+    assert not validate_cron("a b c d e")        # garbage
+
+
+def test_cron_queue_guard_and_retry(tmp_db_url, monkeypatch):
+    """A due guarded cron run is parked while another guarded task is
+    active, re-attempted, and surfaced as FAILURE when the retry window
+    expires (reference: ALGORITHM.md 16.2 steps 5-6)."""
+    import json as _json
+
+    from audiomuse_amd import config as C
+    from audiomuse_amd.db import connect, write_txn
+    from audiomuse_amd.db.schema import init_db
+    from audiomuse_amd.taskqueue import enqueue
+    from audiomuse_amd.utils.cron import run_due_cron_jobs
+
+    conn = connect(tmp_db_url)
+    init_db(conn)
+    monkeypatch.setattr(C, "CRON_RETRY_INTERVAL_MINUTES", 1.0)
+    monkeypatch.setattr(C, "CRON_RETRY_MAX_MINUTES", 2.0)  # 2 attempts
+    with write_txn(conn):
+        conn.execute(
+            "INSERT INTO cron (name, schedule, task_type, payload, enabled) "
+            "VALUES ('nightly', '* * * * *', 'run_analysis', '{}', 1)")
+    blocker = enqueue(conn, "run_clustering", {})   # active guarded task
+
+    t0 = 1_700_000_000.0
+    assert run_due_cron_jobs(conn, now=t0) == []    # parked, not enqueued
+    row = conn.execute("SELECT * FROM cron_retry").fetchone()
+    assert row is not None and row["attempts"] == 1
+
+    # retry due, still blocked -> window expires into a visible FAILURE
+    assert run_due_cron_jobs(conn, now=t0 + 61) == []
+    fail = conn.execute(
+        "SELECT * FROM task_status WHERE status='FAILURE' "
+        "AND task_type='run_analysis'").fetchone()
+    assert fail is not None
+    assert "cron retry window expired" in fail["details"]
+    assert conn.execute("SELECT COUNT(*) FROM cron_retry").fetchone()[0] == 0
+
+    # guard clears -> a due row enqueues normally
+    with write_txn(conn):
+        conn.execute("UPDATE task_status SET status='SUCCESS' "
+                     "WHERE task_id=?", (blocker,))
+    got = run_due_cron_jobs(conn, now=t0 + 120)
+    assert len(got) == 1
+    t = conn.execute("SELECT task_type, status FROM task_status "
+                     "WHERE task_id=?", (got[0],)).fetchone()
+    assert t["task_type"] == "run_analysis" and t["status"] == "PENDING"
+    conn.close()
