@@ -143,6 +143,8 @@ IVF_REFRESH_MAX_DRIFT = _env_float("IVF_REFRESH_MAX_DRIFT", 0.25)
 # cron queue-guard retry cadence (reference: ALGORITHM.md 16.2-16.3)
 CRON_RETRY_INTERVAL_MINUTES = _env_float("CRON_RETRY_INTERVAL_MINUTES", 5.0)
 CRON_RETRY_MAX_MINUTES = _env_float("CRON_RETRY_MAX_MINUTES", 120.0)
+# text-search model warm-up countdown (reference: clap_text_search.py:99)
+CLAP_TEXT_SEARCH_WARMUP_DURATION = _env_float("CLAP_TEXT_SEARCH_WARMUP_DURATION", 300.0)
 IVF_KMEANS_ITERS = _env_int("IVF_KMEANS_ITERS", 25)
 IVF_MAX_PART_SIZE_MB = _env_int("IVF_MAX_PART_SIZE_MB", 32)
 

@@ -73,6 +73,15 @@ class ModelLifecycle:
             self._model = self._factory()
         return self._model
 
+    def remaining(self) -> float:
+        """Seconds until idle unload; 0.0 when not loaded."""
+        if self._model is None:
+            return 0.0
+        return max(0.0, self._idle - (self._time() - self._last))
+
+    def loaded(self) -> bool:
+        return self._model is not None
+
     def maybe_unload(self) -> bool:
         if self._model is not None and self._time() - self._last > self._idle:
             self._model = None

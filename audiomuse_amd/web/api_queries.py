@@ -169,16 +169,46 @@ def clap_text_search():
     q = request.args.get("q", "")
     if not q:
         return jsonify([])
-    emb = current_app.extensions.setdefault(
-        "clap_text", _make_clap_text_embedder())
+    emb = _clap_text_lifecycle().get()      # load + reset the countdown
     vec = emb.embed([q])[0]
     res = eng.find_similar_by_vector(vec, int(request.args.get("n", 20)))
     return jsonify(_with_meta(res))
 
 
-def _make_clap_text_embedder():
-    from audiomuse_amd.models.text import TextEmbedder, clap_text_config
-    return TextEmbedder(clap_text_config(), device=_state().device)
+def _clap_text_lifecycle():
+    """Warm/idle lifecycle for the (large) text model: loaded on demand
+    in the web process only, unloaded after the warm-up countdown
+    expires; every search resets it (reference:
+    clap_text_search.warmup_text_search_model :99)."""
+    from audiomuse_amd.utils.resources import ModelLifecycle
+
+    lc = current_app.extensions.get("clap_text_lc")
+    if lc is None:
+        def factory():
+            from audiomuse_amd.models.text import TextEmbedder, clap_text_config
+            return TextEmbedder(clap_text_config(), device=_state().device)
+
+        lc = ModelLifecycle(factory,
+                            idle_seconds=C.CLAP_TEXT_SEARCH_WARMUP_DURATION)
+        current_app.extensions["clap_text_lc"] = lc
+    lc.maybe_unload()
+    return lc
+
+
+@bp.post("/api/clap/warmup")
+@require_auth
+def clap_warmup():
+    """Pre-load the text model and start the countdown."""
+    lc = _clap_text_lifecycle()
+    lc.get()
+    return jsonify({"loaded": True, "seconds": lc.remaining()})
+
+
+@bp.get("/api/clap/warmup/status")
+@require_auth
+def clap_warmup_status():
+    lc = _clap_text_lifecycle()
+    return jsonify({"loaded": lc.loaded(), "seconds": lc.remaining()})
 
 
 @bp.get("/api/lyrics_search")

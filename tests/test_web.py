@@ -430,3 +430,22 @@ def test_create_playlist_on_server(client_ids):
     # missing args -> 400
     assert client.post("/api/create_playlist",
                        json={"name": "x"}).status_code == 400
+
+
+def test_clap_warmup_lifecycle(client_ids, monkeypatch):
+    client, _ = client_ids
+    r = client.get("/api/clap/warmup/status")
+    assert r.status_code == 200
+    r = client.post("/api/clap/warmup")
+    assert r.status_code == 200 and r.json["loaded"]
+    assert r.json["seconds"] > 0
+    st = client.get("/api/clap/warmup/status").json
+    assert st["loaded"] and st["seconds"] > 0
+    # countdown expiry unloads the model
+    lc = client.application.extensions["clap_text_lc"]
+    monkeypatch.setattr(lc, "_last", lc._time() - 10_000)
+    st = client.get("/api/clap/warmup/status").json
+    assert not st["loaded"] and st["seconds"] == 0
+    # a search reloads it transparently
+    assert client.get("/api/clap_search?q=rainy night").status_code == 200
+    assert client.get("/api/clap/warmup/status").json["loaded"]
