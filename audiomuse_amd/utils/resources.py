@@ -79,9 +79,6 @@ class ModelLifecycle:
             return 0.0
         return max(0.0, self._idle - (self._time() - self._last))
 
-    def loaded(self) -> bool:
-        return self._model is not None
-
     def maybe_unload(self) -> bool:
         if self._model is not None and self._time() - self._last > self._idle:
             self._model = None

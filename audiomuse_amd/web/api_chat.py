@@ -75,10 +75,9 @@ def tool_text_match(args: Dict) -> List[Dict]:
     eng = _state().engine(idx.CLAP_INDEX)
     if eng is None:
         return []
-    from audiomuse_amd.web.api_queries import _make_clap_text_embedder
+    from audiomuse_amd.web.api_queries import _clap_text_lifecycle
 
-    emb = current_app.extensions.setdefault("clap_text",
-                                            _make_clap_text_embedder())
+    emb = _clap_text_lifecycle().get()
     vec = emb.embed([args.get("query", "")])[0]
     return eng.find_similar_by_vector(vec, int(args.get("n", 20)))
 

@@ -208,7 +208,7 @@ def clap_warmup():
 @require_auth
 def clap_warmup_status():
     lc = _clap_text_lifecycle()
-    return jsonify({"loaded": lc.loaded(), "seconds": lc.remaining()})
+    return jsonify({"loaded": lc.loaded, "seconds": lc.remaining()})
 
 
 @bp.get("/api/lyrics_search")

@@ -445,7 +445,7 @@ def test_clap_warmup_lifecycle(client_ids, monkeypatch):
     lc = client.application.extensions["clap_text_lc"]
     monkeypatch.setattr(lc, "_last", lc._time() - 10_000)
     st = client.get("/api/clap/warmup/status").json
-    assert not st["loaded"] and st["seconds"] == 0
+    assert not st["loaded"] and st["seconds"] == 0  # property
     # a search reloads it transparently
     assert client.get("/api/clap_search?q=rainy night").status_code == 200
     assert client.get("/api/clap/warmup/status").json["loaded"]
