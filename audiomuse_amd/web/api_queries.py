@@ -41,6 +41,26 @@ def _with_meta(results: List[dict]) -> List[dict]:
     return out
 
 
+def _server_scope(results: List[dict], n: int) -> List[dict]:
+    """Availability mask (reference: ALGORITHM.md 4.2): when the request
+    names a server, drop tracks that server does not have and attach the
+    server's own provider id so every returned track can actually play."""
+    server_id = request.args.get("server")
+    if not server_id:
+        return results[:n] if n else results
+    conn = _state().conn()
+    out = []
+    for r in results:
+        m = conn.execute(
+            "SELECT provider_id FROM track_server_map WHERE item_id = ? "
+            "AND server_id = ?", (r["item_id"], server_id)).fetchone()
+        if m is not None:
+            out.append({**r, "provider_id": m["provider_id"]})
+            if n and len(out) >= n:
+                break
+    return out
+
+
 @bp.get("/api/similar_tracks")
 @require_auth
 def similar_tracks():
@@ -53,14 +73,16 @@ def similar_tracks():
     radius = request.args.get("radius_similarity", "0") in ("1", "true")
     mood = request.args.get("mood_filter") or None
     cap = request.args.get("max_per_artist")
+    # over-fetch when a server scope will drop unmapped tracks
+    fetch = n * 2 if request.args.get("server") else n
     res = eng.find_similar_by_id(
-        item_id, n, radius=radius, mood_filter=mood,
+        item_id, fetch, radius=radius, mood_filter=mood,
         eliminate_duplicates=request.args.get("eliminate_duplicates", "1")
         in ("1", "true"),
         max_per_artist=int(cap) if cap else None)
     if not res and eng.vector_for_id(item_id) is None:
         return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
-    return jsonify(_with_meta(res))
+    return jsonify(_with_meta(_server_scope(res, n)))
 
 
 @bp.get("/api/search_tracks")
@@ -171,8 +193,10 @@ def clap_text_search():
         return jsonify([])
     emb = _clap_text_lifecycle().get()      # load + reset the countdown
     vec = emb.embed([q])[0]
-    res = eng.find_similar_by_vector(vec, int(request.args.get("n", 20)))
-    return jsonify(_with_meta(res))
+    n = int(request.args.get("n", 20))
+    fetch = n * 2 if request.args.get("server") else n
+    res = eng.find_similar_by_vector(vec, fetch)
+    return jsonify(_with_meta(_server_scope(res, n)))
 
 
 def _clap_text_lifecycle():

@@ -449,3 +449,27 @@ def test_clap_warmup_lifecycle(client_ids, monkeypatch):
     # a search reloads it transparently
     assert client.get("/api/clap_search?q=rainy night").status_code == 200
     assert client.get("/api/clap/warmup/status").json["loaded"]
+
+
+def test_server_availability_mask(client_ids):
+    """reference ALGORITHM.md 4.2: ?server= drops unmapped tracks and
+    attaches provider ids."""
+    client, ids = client_ids
+    from audiomuse_amd.db import write_txn
+
+    conn = client.application.extensions["audiomuse"].conn()
+    with write_txn(conn):
+        for k, iid in enumerate(ids[2:8]):
+            conn.execute(
+                "INSERT OR REPLACE INTO track_server_map "
+                "(provider_id, server_id, item_id) VALUES (?, 'navi', ?)",
+                (f"nv-{k}", iid))
+    r = client.get(f"/api/similar_tracks?item_id={ids[0]}&n=4&server=navi")
+    assert r.status_code == 200
+    assert 0 < len(r.json) <= 4
+    mapped = set(ids[2:8])
+    for b in r.json:
+        assert b["item_id"] in mapped and b["provider_id"].startswith("nv-")
+    # without the scope, unmapped tracks appear
+    r2 = client.get(f"/api/similar_tracks?item_id={ids[0]}&n=10")
+    assert any(b["item_id"] not in mapped for b in r2.json)
