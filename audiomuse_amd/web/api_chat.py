@@ -159,9 +159,15 @@ def llm_plan(prompt: str, hints: Dict) -> Optional[List[Dict]]:
     return plan_with_llm(prompt, {name: None for name in TOOLS})
 
 
-def rerank(results_per_tool: List[List[Dict]], n: int) -> List[str]:
-    """Tiered re-rank with intersection boost + relax loop
-    (tasks/ai/rerank.py)."""
+_FILLER_TITLE = re.compile(r"\b(intro|outro|skit|interlude|spoken)\b", re.I)
+
+
+def rerank(results_per_tool: List[List[Dict]], n: int,
+           exclude_artists: Optional[List[str]] = None) -> List[str]:
+    """Tiered re-rank (tasks/ai/rerank.py): intersections boost, primary
+    similarity orders within a tier, intro/skit/interlude titles are
+    pushed down, exclude_artists is the one hard cut."""
+    state = _state()
     scores: Dict[str, float] = {}
     hits: Dict[str, int] = {}
     for results in results_per_tool:
@@ -169,9 +175,31 @@ def rerank(results_per_tool: List[List[Dict]], n: int) -> List[str]:
             iid = r["item_id"]
             scores[iid] = min(scores.get(iid, 10.0), r.get("distance", 1.0))
             hits[iid] = hits.get(iid, 0) + 1
+    excl = {a.lower() for a in (exclude_artists or [])}
+    filler: Dict[str, bool] = {}
+    for iid in list(scores):
+        meta = state.meta_fn(iid) or {}
+        if excl and (meta.get("author") or "").lower() in excl:
+            scores.pop(iid)      # hard cut
+            continue
+        filler[iid] = bool(_FILLER_TITLE.search(meta.get("title") or ""))
     ranked = sorted(scores,
-                    key=lambda i: (-hits[i], scores[i]))  # intersections first
+                    key=lambda i: (filler[i], -hits[i], scores[i]))
     return ranked[:n]
+
+
+def run_plan(prompt: str, hints: Dict):
+    """Plan -> tools -> one replan on an empty pool (reference:
+    planner 'one replan' rule) -> (plan, per-tool results)."""
+    plan = llm_plan(prompt, hints) or heuristic_plan(hints)
+    plan = validate_and_normalize_plan(plan)
+    results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
+    if not any(results):
+        replan = validate_and_normalize_plan(heuristic_plan(hints))
+        if replan != plan:
+            plan = replan
+            results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
+    return plan, results
 
 
 @bp.post("/chat/api/chatPlaylistStream")
@@ -186,8 +214,7 @@ def chat_playlist_stream():
     if not prompt:
         return jsonify({"error": "prompt required"}), 400
     hints = extract_hints(prompt)
-    plan = llm_plan(prompt, hints) or heuristic_plan(hints)
-    plan = validate_and_normalize_plan(plan)
+    plan, _pre = run_plan(prompt, hints)
     state = _state()
 
     def generate():
@@ -219,14 +246,22 @@ def chat_playlist():
     if not prompt:
         return jsonify({"error": "prompt required"}), 400
     hints = extract_hints(prompt)
-    plan = llm_plan(prompt, hints) or heuristic_plan(hints)
-    plan = validate_and_normalize_plan(plan)
-    results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
-    ids = rerank(results, hints["n"])
+    plan, results = run_plan(prompt, hints)
+    ids = rerank(results, hints["n"],
+                 exclude_artists=body.get("exclude_artists"))
     state = _state()
     tracks = []
     for i in ids:
         meta = state.meta_fn(i) or {}
         tracks.append({"item_id": i, "title": meta.get("title"),
                        "author": meta.get("author")})
+    if body.get("order"):
+        rows = [{"item_id": t["item_id"],
+                 **{k: (state.meta_fn(t["item_id"]) or {}).get(k)
+                    for k in ("tempo", "energy", "key", "scale")}}
+                for t in tracks]
+        from audiomuse_amd.engines.misc import order_playlist
+        ordered = {r["item_id"]: i for i, r in enumerate(
+            order_playlist(rows, energy_arc=len(rows) >= 10))}
+        tracks.sort(key=lambda t: ordered.get(t["item_id"], 0))
     return jsonify({"plan": plan, "tracks": tracks, "hints": hints})
