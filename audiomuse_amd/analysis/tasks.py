@@ -92,10 +92,22 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
 
     tracks = provider.get_tracks_from_album(payload["album_id"])
     todo = [t for t in tracks if t.provider_id not in done]
-    blobs: List[Optional[bytes]] = []
-    for t in todo:
-        ctx.check_cancelled()
-        blobs.append(provider.download_track(t.provider_id))
+    ctx.check_cancelled()
+    # parallel downloads (reference downloads serially, album.py:290;
+    # provider IO and WAV decode release the GIL, so a small pool
+    # overlaps network/decode with the GPU batch of the previous album)
+    if len(todo) > 1:
+        from concurrent.futures import ThreadPoolExecutor
+
+        from audiomuse_amd.utils.resources import usable_cpu_count
+
+        workers = min(4, max(1, usable_cpu_count() - 1), len(todo))
+        with ThreadPoolExecutor(max_workers=workers) as pool:
+            blobs = list(pool.map(
+                lambda t: provider.download_track(t.provider_id), todo))
+    else:
+        blobs = [provider.download_track(t.provider_id) for t in todo]
+    ctx.check_cancelled()
     valid = [(t, b) for t, b in zip(todo, blobs) if b is not None]
     if not valid:
         return {"analyzed": 0, "skipped": len(tracks) - len(todo)}
