@@ -243,6 +243,7 @@ class GraphedDecoder:
         cfg = model.cfg
         device, dtype = enc.device, enc.dtype
         self.model = model
+        self.enc_shape = tuple(enc.shape)
         self.caches = model.make_caches(1, device, dtype)
         self.ckv = model.cross_kvs(enc)
         self.tok = torch.zeros(1, 1, dtype=torch.long, device=device)
@@ -267,12 +268,40 @@ class GraphedDecoder:
             ck.zero_()
             cv.zero_()
 
+    def reset(self, enc: torch.Tensor) -> None:
+        """Re-arm the captured graph for a NEW chunk: the fresh cross
+        K/V are copied into the captured buffers and the self caches are
+        zeroed. Capture cost (~2 s) is paid once per (model, enc shape);
+        every later chunk is a copy + replays."""
+        new = self.model.cross_kvs(enc)
+        for (kb, vb), (k, v) in zip(self.ckv, new):
+            kb.copy_(k)
+            vb.copy_(v)
+        for ck, cv in self.caches:
+            ck.zero_()
+            cv.zero_()
+
     def step(self, token: int, pos: int) -> torch.Tensor:
         """Returns logits (1, 1, vocab) for `token` written at `pos`."""
         self.tok.fill_(token)
         self.len_t.fill_(pos)
         self.graph.replay()
         return self.logits
+
+
+def _graphed_decoder_for(model: WhisperModel,
+                         enc: torch.Tensor) -> Optional[GraphedDecoder]:
+    """Per-model cached GraphedDecoder, re-armed per chunk."""
+    gdec = getattr(model, "_graphed_decoder", None)
+    if gdec is not None and gdec.enc_shape == tuple(enc.shape):
+        gdec.reset(enc)
+        return gdec
+    try:
+        gdec = GraphedDecoder(model, enc)
+    except Exception:  # noqa: BLE001 — capture failure falls back to eager
+        return None
+    model._graphed_decoder = gdec
+    return gdec
 
 
 def _block_repeats(logits: torch.Tensor, seq: List[int],
@@ -320,10 +349,7 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
         use_graph = enc.is_cuda
     gdec: Optional[GraphedDecoder] = None
     if use_graph and enc.is_cuda:
-        try:
-            gdec = GraphedDecoder(model, enc)
-        except Exception:  # noqa: BLE001 — capture failure falls back to eager
-            gdec = None
+        gdec = _graphed_decoder_for(model, enc)
     if gdec is not None:
         caches, ckv = gdec.caches, gdec.ckv
     else:

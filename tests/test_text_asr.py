@@ -182,3 +182,20 @@ def test_whisper_graphed_decode_matches_eager_gpu():
     eager = greedy_decode(m, mel, max_tokens=12, use_graph=False)
     graphed = greedy_decode(m, mel, max_tokens=12, use_graph=True)
     assert graphed == eager
+
+
+@pytest.mark.gpu
+def test_whisper_graphed_decoder_reset_across_chunks():
+    """The cached graph re-arms per chunk (reset path): chunk 2 decoded
+    after chunk 1 must match its own eager decode exactly."""
+    torch.manual_seed(5)
+    m = tiny_whisper().to("cuda", torch.bfloat16).eval()
+    mel1 = torch.randn(8, 64, device="cuda", dtype=torch.bfloat16)
+    mel2 = torch.randn(8, 64, device="cuda", dtype=torch.bfloat16) * 1.3
+    greedy_decode(m, mel1, max_tokens=8, use_graph=True)   # capture + use
+    first = getattr(m, "_graphed_decoder", None)
+    assert first is not None
+    g2 = greedy_decode(m, mel2, max_tokens=12, use_graph=True)
+    assert getattr(m, "_graphed_decoder") is first          # reused, no recapture
+    e2 = greedy_decode(m, mel2, max_tokens=12, use_graph=False)
+    assert g2 == e2
