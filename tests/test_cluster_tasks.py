@@ -62,3 +62,59 @@ def test_clustering_end_to_end(seeded):
         ids = json.loads(p["item_ids"])
         assert len(ids) >= 1
         assert p["name"].endswith("_automatic")
+
+
+def test_clustering_tolerates_failed_batches(seeded, monkeypatch):
+    """Stall-valve semantics (reference: clustering.py:1426-1449 +
+    absorb :1581): up to CLUSTERING_MAX_FAILED_BATCHES dead batches are
+    absorbed; one more fails the parent run."""
+    from audiomuse_amd import config as C
+    from audiomuse_amd.taskqueue import FAILURE
+    from audiomuse_amd.taskqueue.worker import _REGISTRY, import_builtin_handlers
+
+    conn, url = seeded
+    import_builtin_handlers()
+    real = _REGISTRY["run_clustering_batch"]
+
+    def flaky(ctx, payload):
+        # keyed on the batch seed so retries of a doomed batch also die
+        if payload["seed"] % 2 == 1:     # half the batches die
+            raise RuntimeError("injected batch crash")
+        return real(ctx, payload)
+
+    monkeypatch.setitem(_REGISTRY, "run_clustering_batch", flaky)
+    monkeypatch.setattr(C, "CLUSTERING_MAX_FAILED_BATCHES", 2)
+
+    def run_workers():
+        import threading
+
+        ws = [Worker(db_url=url, max_jobs=30) for _ in range(2)]
+        ts = [threading.Thread(target=lambda w=w: w.run_forever(
+            idle_timeout=10.0)) for w in ws]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=240)
+
+    tid = enqueue(conn, "run_clustering",
+                  {"runs": 8, "iterations_per_batch": 2,
+                   "algorithm": "kmeans", "drain_timeout": 120.0})
+    run_workers()
+    row = task_row(conn, tid)
+    assert row["status"] == SUCCESS, row["details"]
+    res = json.loads(row["result"])
+    assert res["failed_batches"] == 2 and res["playlists"] >= 1
+
+    # over the valve: 3 of 4 batches die -> parent FAILURE
+
+    def flakier(ctx, payload):
+        if payload["seed"] % 4 != 0:
+            raise RuntimeError("injected batch crash")
+        return real(ctx, payload)
+
+    monkeypatch.setitem(_REGISTRY, "run_clustering_batch", flakier)
+    tid2 = enqueue(conn, "run_clustering",
+                   {"runs": 8, "iterations_per_batch": 2,
+                    "algorithm": "kmeans", "drain_timeout": 120.0})
+    run_workers()
+    assert task_row(conn, tid2)["status"] == FAILURE
