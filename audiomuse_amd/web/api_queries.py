@@ -445,17 +445,19 @@ _MAP_PERCENTS = (100, 75, 50, 25)
 _map_cache: dict = {}
 
 
-def _map_bucket(kind: str, percent: int):
+def _map_bucket(kind: str, percent: int, server_id: str = ""):
     """Percent-sampled map bucket, JSON + gzip serialized once
     (reference: app_map.build_map_cache :156 — deterministic samples at
-    100/75/50/25%, pre-gzipped, served with zero further work)."""
+    100/75/50/25%, pre-gzipped, served with zero further work). A
+    per-server bucket (reference: the multi-server second cache layer)
+    holds that server's own provider ids and drops unmapped tracks."""
     import gzip
     import random
 
     state = _state()
     name = idx.ARTIST_MAP if kind == "artist" else idx.SONG_MAP
     stamp = state._stamp(name)
-    key = (kind, percent, stamp)
+    key = (kind, percent, server_id, stamp)
     hit = _map_cache.get(key)
     if hit is not None:
         return hit
@@ -464,12 +466,22 @@ def _map_bucket(kind: str, percent: int):
         return None
     coords = data["coords"]
     ids = data["item_ids"]
+    pid_of = {}
+    if server_id and kind == "song":
+        pid_of = {r["item_id"]: r["provider_id"] for r in state.conn().execute(
+            "SELECT item_id, provider_id FROM track_server_map "
+            "WHERE server_id = ?", (server_id,)).fetchall()}
     recs = []
     id_key = "artist" if kind == "artist" else "item_id"
     for i in range(len(ids)):
         rec = {id_key: ids[i], "x": float(coords[i][0]),
                "y": float(coords[i][1])}
         if kind == "song":
+            if server_id:
+                pid = pid_of.get(ids[i])
+                if pid is None:
+                    continue           # availability mask
+                rec["provider_id"] = pid
             meta = state.meta_fn(ids[i]) or {}
             rec["title"] = meta.get("title", "")
             rec["author"] = meta.get("author", "")
@@ -497,7 +509,7 @@ def music_map():
     percent = int(request.args.get("percent", 100))
     if percent not in _MAP_PERCENTS:
         return jsonify({"error": f"percent must be one of {_MAP_PERCENTS}"}), 400
-    entry = _map_bucket(kind, percent)
+    entry = _map_bucket(kind, percent, request.args.get("server", ""))
     if entry is None:
         return jsonify({"error": "map not built"}), 503
     raw, gz = entry

@@ -473,3 +473,13 @@ def test_server_availability_mask(client_ids):
     # without the scope, unmapped tracks appear
     r2 = client.get(f"/api/similar_tracks?item_id={ids[0]}&n=10")
     assert any(b["item_id"] not in mapped for b in r2.json)
+
+
+def test_map_per_server_bucket(client_ids):
+    client, ids = client_ids
+    # 'navi' was mapped onto ids[2:8] in the availability-mask test
+    pts = client.get("/api/map?server=navi").json
+    assert 0 < len(pts) <= 6
+    assert all("provider_id" in p for p in pts)
+    mapped = set(ids[2:8])
+    assert all(p["item_id"] in mapped for p in pts)
