@@ -114,6 +114,17 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
 
     results = runtime.analyze_album_batch([b for _, b in valid])
     n = 0
+    # one write transaction per album (write_txn is reentrant: the
+    # nested save_* helpers join it) — one fsync instead of ~3/track
+    with write_txn(conn):
+        n = _persist_album(ctx, conn, valid, results, resolver, server_id,
+                           runtime, provider)
+    return {"analyzed": n, "skipped": len(tracks) - len(todo)}
+
+
+def _persist_album(ctx, conn, valid, results, resolver, server_id,
+                   runtime, provider) -> int:
+    n = 0
     for (track, _blob), res in zip(valid, results):
         ctx.check_cancelled()
         if res is None:
@@ -155,7 +166,7 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
                 (track.provider_id, server_id, item_id, track.title,
                  track.author, track.album, track.file_path))
         n += 1
-    return {"analyzed": n, "skipped": len(tracks) - len(todo)}
+    return n
 
 
 @task_handler("run_analysis")

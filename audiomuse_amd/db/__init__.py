@@ -68,7 +68,13 @@ def get_db(url: Optional[str] = None) -> sqlite3.Connection:
 @contextmanager
 def write_txn(conn: sqlite3.Connection) -> Iterator[sqlite3.Connection]:
     """Exclusive write transaction — the SQLite analog of the reference's
-    row-locked claim sections (taskqueue/sql.py:415-430)."""
+    row-locked claim sections (taskqueue/sql.py:415-430). Reentrant: a
+    nested call joins the enclosing transaction (the outermost owns
+    commit/rollback), so callers can batch many save_* helpers into one
+    fsync."""
+    if conn.in_transaction:
+        yield conn
+        return
     conn.execute("BEGIN IMMEDIATE")
     try:
         yield conn

@@ -250,3 +250,27 @@ def test_crashed_fork_worker_job_is_reclaimed(db, monkeypatch):
     time.sleep(0.4)                          # lease decays
     assert qsql.reclaim_orphans(conn) == 1
     assert task_row(conn, tid)["status"] == PENDING
+
+
+def test_write_txn_reentrant(db):
+    conn, _ = db
+    from audiomuse_amd.db import write_txn
+
+    with write_txn(conn):
+        conn.execute("CREATE TABLE IF NOT EXISTS rt (v INTEGER)")
+        with write_txn(conn):          # joins, does not BEGIN/COMMIT
+            conn.execute("INSERT INTO rt VALUES (1)")
+        assert conn.in_transaction     # outer txn still open
+        conn.execute("INSERT INTO rt VALUES (2)")
+    assert not conn.in_transaction
+    assert conn.execute("SELECT COUNT(*) FROM rt").fetchone()[0] == 2
+
+    # outer rollback undoes the nested writes too
+    try:
+        with write_txn(conn):
+            with write_txn(conn):
+                conn.execute("INSERT INTO rt VALUES (3)")
+            raise RuntimeError("boom")
+    except RuntimeError:
+        pass
+    assert conn.execute("SELECT COUNT(*) FROM rt").fetchone()[0] == 2
