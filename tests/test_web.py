@@ -477,9 +477,17 @@ def test_server_availability_mask(client_ids):
 
 def test_map_per_server_bucket(client_ids):
     client, ids = client_ids
-    # 'navi' was mapped onto ids[2:8] in the availability-mask test
-    pts = client.get("/api/map?server=navi").json
+    from audiomuse_amd.db import write_txn
+
+    conn = client.application.extensions["audiomuse"].conn()
+    with write_txn(conn):
+        for k, iid in enumerate(ids[10:16]):
+            conn.execute(
+                "INSERT OR REPLACE INTO track_server_map "
+                "(provider_id, server_id, item_id) VALUES (?, 'mapsrv', ?)",
+                (f"ms-{k}", iid))
+    pts = client.get("/api/map?server=mapsrv").json
     assert 0 < len(pts) <= 6
     assert all("provider_id" in p for p in pts)
-    mapped = set(ids[2:8])
+    mapped = set(ids[10:16])
     assert all(p["item_id"] in mapped for p in pts)
