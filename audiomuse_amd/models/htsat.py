@@ -166,14 +166,20 @@ class SwinBlock(nn.Module):
                 qkv = fp8.scaled_linear(xn, self.attn.qkv.weight,
                                         self.attn.qkv.bias)
             else:
-                qkv = self.attn.qkv(xn)                   # (B, L, 3C)
+                # routed through the extension for the timed algo search
+                qkv = ext.linear_bias(xn.contiguous(),
+                                      self.attn.qkv.weight.contiguous(),
+                                      self.attn.qkv.bias.contiguous())
             out = ext.window_attn_fwd(
                 qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
                 self.attn.heads, self.shift, self.attn.scale)
             out = out.view(B, L, C)
             proj = (fp8.scaled_linear(out, self.attn.proj.weight,
                                       self.attn.proj.bias)
-                    if use_fp8 else self.attn.proj(out))
+                    if use_fp8 else
+                    ext.linear_bias(out.contiguous(),
+                                    self.attn.proj.weight.contiguous(),
+                                    self.attn.proj.bias.contiguous()))
             # fused residual add + norm2 (one pass instead of add->LN)
             x2, xn2 = ext.add_layernorm_bf16(
                 x.contiguous(), proj.contiguous(),

@@ -98,3 +98,16 @@ def test_linear_bias_add_matches_reference():
                             r.contiguous())
     expect = F.linear(x.float(), w.float(), b.float()) + r.float()
     torch.testing.assert_close(got.float(), expect, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_linear_bias_matches_reference():
+    import audiomuse_amd._C as C
+
+    torch.manual_seed(2)
+    x = torch.randn(3, 100, 128, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(384, 128, device="cuda", dtype=torch.bfloat16) * 0.05
+    b = torch.randn(384, device="cuda", dtype=torch.bfloat16)
+    got = C.linear_bias(x.contiguous(), w.contiguous(), b.contiguous())
+    expect = F.linear(x.float(), w.float(), b.float())
+    torch.testing.assert_close(got.float(), expect, rtol=3e-2, atol=3e-2)
