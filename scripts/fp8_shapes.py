@@ -38,6 +38,8 @@ def main():
         (131_072, 512, 1536), (131_072, 2048, 512),
         (32_768, 1024, 3072), (32_768, 4096, 1024),
     ]
+    import sys as _sys
+    _sys.path.insert(0, "/root/repo")
     import audiomuse_amd._C as ext
     print(f"{'M':>9} {'K':>5} {'N':>5} | {'bf16':>7} {'fp8mm':>7} "
           f"{'fp8ext':>7} {'q_bf16':>7} | fp8ext+q_bf16 vs bf16")
