@@ -161,11 +161,16 @@ class SwinBlock(nn.Module):
             from audiomuse_amd.ops import _ext, fp8
             ext = _ext.require()
             use_fp8 = fp8.serving_enabled() and fp8.available(x.device)
-            xn = self.norm1(x)
             if use_fp8:
-                qkv = fp8.scaled_linear(xn, self.attn.qkv.weight,
-                                        self.attn.qkv.bias)
+                # LN emits e4m3 directly (delayed scale); fp8 GEMM with
+                # searched algo — no standalone quantize pass
+                xq, xs = fp8.ln_fp8(self.norm1, x)
+                wq, ws = fp8.quantize_weight(self.attn.qkv.weight)
+                qkv = ext.linear_fp8(
+                    xq, wq, xs, ws,
+                    self.attn.qkv.bias.to(torch.bfloat16).contiguous())
             else:
+                xn = self.norm1(x)
                 # routed through the extension for the timed algo search
                 qkv = ext.linear_bias(xn.contiguous(),
                                       self.attn.qkv.weight.contiguous(),
@@ -174,27 +179,30 @@ class SwinBlock(nn.Module):
                 qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
                 self.attn.heads, self.shift, self.attn.scale)
             out = out.view(B, L, C)
-            proj = (fp8.scaled_linear(out, self.attn.proj.weight,
-                                      self.attn.proj.bias)
-                    if use_fp8 else
-                    ext.linear_bias(out.contiguous(),
-                                    self.attn.proj.weight.contiguous(),
-                                    self.attn.proj.bias.contiguous()))
+            # proj stays bf16 in fp8 mode too: its GEMM gain is smaller
+            # than any quantize cost at C x C shapes (fp8_shapes.py)
+            proj = ext.linear_bias(out.contiguous(),
+                                   self.attn.proj.weight.contiguous(),
+                                   self.attn.proj.bias.contiguous())
+            if use_fp8:
+                # fused residual add + LN emitting e4m3 for the mlp0 GEMM
+                x2, xn2q, xs2 = fp8.add_ln_fp8(self.norm2, x, proj)
+                w0q, w0s = fp8.quantize_weight(self.mlp[0].weight)
+                hidden = ext.linear_fp8(
+                    xn2q, w0q, xs2, w0s,
+                    self.mlp[0].bias.to(torch.bfloat16).contiguous(),
+                    gelu=True)
+                # mlp2 stays bf16 (quantizing the 4C-wide hidden costs
+                # more than the fp8 GEMM saves) with the residual folded
+                return ext.linear_bias_add(hidden,
+                                           self.mlp[2].weight.contiguous(),
+                                           self.mlp[2].bias.contiguous(), x2)
             # fused residual add + norm2 (one pass instead of add->LN)
             x2, xn2 = ext.add_layernorm_bf16(
                 x.contiguous(), proj.contiguous(),
                 self.norm2.weight.to(torch.bfloat16).contiguous(),
                 self.norm2.bias.to(torch.bfloat16).contiguous(),
                 self.norm2.eps)
-            if use_fp8:
-                # fp8 GEMM + eager tanh-GELU (scaled_mm has no epilogue);
-                # the fp8 rate + halved operand traffic beats the fused
-                # bf16 epilogue at these shapes
-                hidden = F.gelu(fp8.scaled_linear(xn2, self.mlp[0].weight,
-                                                  self.mlp[0].bias),
-                                approximate="tanh")
-                return x2 + fp8.scaled_linear(hidden, self.mlp[2].weight,
-                                              self.mlp[2].bias)
             # MLP with the GELU fused into the first GEMM's epilogue and
             # the residual add folded into the second GEMM (beta=1)
             hidden = ext.linear_gelu(xn2, self.mlp[0].weight.contiguous(),

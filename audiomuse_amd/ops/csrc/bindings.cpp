@@ -25,6 +25,11 @@ void launch_add_layernorm_bf16(const void* x, const void* in2, void* sum_out,
                                void* y, const void* w, const void* b,
                                long long n_rows, int dim, float eps,
                                hipStream_t stream);
+void launch_layernorm_bf16_fp8_impl(const void* x, void* y8, const void* w,
+                                    const void* b, const void* in2,
+                                    void* sum_out, const float* q_scale,
+                                    float* q_amax, long long n_rows, int dim,
+                                    float eps, hipStream_t stream);
 void launch_mfma_probe(const void* A, const void* B, float* D,
                        hipStream_t stream);
 void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
@@ -189,6 +194,50 @@ static std::vector<torch::Tensor> add_layernorm_bf16(torch::Tensor x,
   return {sum, y};
 }
 
+static torch::Tensor layernorm_bf16_fp8(torch::Tensor x, torch::Tensor w,
+                                        torch::Tensor b, double eps,
+                                        torch::Tensor scale,
+                                        torch::Tensor amax) {
+  AM_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous(),
+           "x must be contiguous bf16 on GPU");
+  AM_CHECK(scale.scalar_type() == at::kFloat && amax.scalar_type() == at::kFloat,
+           "scale/amax must be f32 device scalars");
+  const int dim = x.size(-1);
+  AM_CHECK(dim % 4 == 0 && dim <= 4096, "dim must be /4 and <= 4096");
+  auto y8 = torch::empty(x.sizes(),
+                         x.options().dtype(at::kFloat8_e4m3fn));
+  const long long n_rows = x.numel() / dim;
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_layernorm_bf16_fp8_impl(
+      x.data_ptr(), y8.data_ptr(), w.data_ptr(), b.data_ptr(), nullptr,
+      nullptr, scale.data_ptr<float>(), amax.data_ptr<float>(), n_rows, dim,
+      (float)eps, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return y8;
+}
+
+static std::vector<torch::Tensor> add_layernorm_bf16_fp8(
+    torch::Tensor x, torch::Tensor other, torch::Tensor w, torch::Tensor b,
+    double eps, torch::Tensor scale, torch::Tensor amax) {
+  AM_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous(),
+           "x must be contiguous bf16 on GPU");
+  AM_CHECK(other.is_contiguous() && other.sizes() == x.sizes() &&
+               other.scalar_type() == at::kBFloat16,
+           "other must match x");
+  const int dim = x.size(-1);
+  AM_CHECK(dim % 4 == 0 && dim <= 4096, "dim must be /4 and <= 4096");
+  auto sum = torch::empty_like(x);
+  auto y8 = torch::empty(x.sizes(), x.options().dtype(at::kFloat8_e4m3fn));
+  const long long n_rows = x.numel() / dim;
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_layernorm_bf16_fp8_impl(
+      x.data_ptr(), y8.data_ptr(), w.data_ptr(), b.data_ptr(),
+      other.data_ptr(), sum.data_ptr(), scale.data_ptr<float>(),
+      amax.data_ptr<float>(), n_rows, dim, (float)eps, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return {sum, y8};
+}
+
 void register_gemm_gelu(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -196,6 +245,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_gemm_gelu(m);
   m.def("add_layernorm_bf16", &add_layernorm_bf16,
         "Fused residual add + LayerNorm: returns (x+other, LN(x+other))");
+  m.def("layernorm_bf16_fp8", &layernorm_bf16_fp8,
+        "LayerNorm with fused e4m3 quantize (delayed scale + amax)");
+  m.def("add_layernorm_bf16_fp8", &add_layernorm_bf16_fp8,
+        "Residual add + LN with fused e4m3 quantize: (sum bf16, y fp8)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("window_attn_fwd", &window_attn_fwd,
         "Fused shifted-window attention (qkv BHW3C bf16, bias, heads, "

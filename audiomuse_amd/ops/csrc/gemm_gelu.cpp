@@ -209,7 +209,8 @@ torch::Tensor linear_bias(torch::Tensor x, torch::Tensor w,
 // searches instead.
 torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
                          torch::Tensor sa, torch::Tensor sb,
-                         c10::optional<torch::Tensor> bias) {
+                         c10::optional<torch::Tensor> bias,
+                         bool gelu) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat8_e4m3fn &&
                   x.is_contiguous(),
               "x must be contiguous fp8e4m3 GPU");
@@ -245,12 +246,17 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
       desc, HIPBLASLT_MATMUL_DESC_B_SCALE_POINTER, &b_scale,
       sizeof(b_scale)));
   if (bias.has_value()) {
-    hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_BIAS;
+    hipblasLtEpilogue_t epi =
+        gelu ? HIPBLASLT_EPILOGUE_GELU_BIAS : HIPBLASLT_EPILOGUE_BIAS;
     HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
         desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
     const void* bp = bias->data_ptr();
     HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
         desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bp, sizeof(bp)));
+  } else if (gelu) {
+    hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_GELU;
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
   }
 
   hipblasLtMatrixLayout_t la, lb, ld;
@@ -271,7 +277,7 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
   bool have_algo = false;
   {
     std::lock_guard<std::mutex> g(algo_mu);
-    auto it = algo_cache.find({M, N, K, 1000 + (bias.has_value() ? 1 : 0)});
+    auto it = algo_cache.find({M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0)});
     if (it != algo_cache.end()) {
       algo = it->second;
       have_algo = true;
@@ -321,7 +327,7 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
     (void)hipEventDestroy(ev1);
     algo = results[best].algo;
     std::lock_guard<std::mutex> g(algo_mu);
-    algo_cache[{M, N, K, 1000 + (bias.has_value() ? 1 : 0)}] = algo;
+    algo_cache[{M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0)}] = algo;
   }
 
   const float alpha = 1.0f, beta = 0.0f;
@@ -349,5 +355,6 @@ void register_gemm_gelu(pybind11::module_& m) {
   m.def("linear_fp8", &linear_fp8,
         "fp8 e4m3 GEMM with per-tensor scales + timed algo search",
         pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("sa"),
-        pybind11::arg("sb"), pybind11::arg("bias") = pybind11::none());
+        pybind11::arg("sb"), pybind11::arg("bias") = pybind11::none(),
+        pybind11::arg("gelu") = false);
 }

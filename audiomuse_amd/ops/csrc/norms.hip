@@ -11,9 +11,20 @@
 // scratch on hipcc (cdna_hip_programming.md, common-mistake #20).
 
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 
 namespace audiomuse {
+
+// fp8-output variants (template flag F8): the normalized row is written
+// as OCP e4m3 with a delayed per-tensor scale (read from device memory)
+// while the true amax accumulates via atomicMax for the NEXT step's
+// scale — this removes the standalone quantize pass that made unfused
+// fp8 serving slower than bf16 (profiles/r01_final_profile.md).
+__device__ __forceinline__ void am_atomic_fmax(float* addr, float v) {
+  // non-negative floats compare correctly as ints
+  atomicMax(reinterpret_cast<int*>(addr), __float_as_int(v));
+}
 
 __device__ __forceinline__ float wrsum(float v) {
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
@@ -32,12 +43,15 @@ __device__ __forceinline__ float hrsum(float v) {
 // ADD variant fuses the preceding residual add: in2 != nullptr adds it to
 // x, writes the sum to `sum`, and normalizes the sum (saves one full
 // tensor read between the eager add and the LN).
-template <bool ADD>
+template <bool ADD, bool F8 = false>
 __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
     const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
     const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
     const __hip_bfloat16* __restrict__ in2, __hip_bfloat16* __restrict__ sum_out,
-    long long n_rows, int dim, float eps) {
+    long long n_rows, int dim, float eps,
+    const float* __restrict__ q_scale = nullptr,
+    float* __restrict__ q_amax = nullptr,
+    unsigned char* __restrict__ y8 = nullptr) {
   const int sl = threadIdx.x & 31;          // lane within the half
   const long long row = (long long)blockIdx.x * 8 + (threadIdx.x >> 5);
   if (row >= n_rows) return;
@@ -80,22 +94,43 @@ __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
     const short4 pbv = *reinterpret_cast<const short4*>(b + i);
     const __hip_bfloat16* wb = reinterpret_cast<const __hip_bfloat16*>(&pw);
     const __hip_bfloat16* bb = reinterpret_cast<const __hip_bfloat16*>(&pbv);
-    short4 out;
-    __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+    if (F8) {
+      const float inv = 1.0f / q_scale[0];
+      float amax = 0.0f;
+      uchar4 o8;
+      unsigned char* o8b = reinterpret_cast<unsigned char*>(&o8);
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      ob[j] = __float2bfloat16(((vals[j] - mean) * rstd) * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
-    *reinterpret_cast<short4*>(yr + i) = out;
+      for (int j = 0; j < 4; ++j) {
+        const float f = ((vals[j] - mean) * rstd) * __bfloat162float(wb[j]) +
+                        __bfloat162float(bb[j]);
+        amax = fmaxf(amax, fabsf(f));
+        o8b[j] = __hip_cvt_float_to_fp8(f * inv, __HIP_SATFINITE, __HIP_E4M3);
+      }
+      *reinterpret_cast<uchar4*>(y8 + row * dim + i) = o8;
+      for (int off = 16; off > 0; off >>= 1)
+        amax = fmaxf(amax, __shfl_xor(amax, off, 32));
+      if (sl == 0) am_atomic_fmax(q_amax, amax);
+    } else {
+      short4 out;
+      __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        ob[j] = __float2bfloat16(((vals[j] - mean) * rstd) * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
+      *reinterpret_cast<short4*>(yr + i) = out;
+    }
   }
 }
 
 // one wave per row; block = 256 threads = 4 rows
-template <int NIT, bool ADD>
+template <int NIT, bool ADD, bool F8 = false>
 __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
     const __hip_bfloat16* __restrict__ w, const __hip_bfloat16* __restrict__ b,
     const __hip_bfloat16* __restrict__ in2, __hip_bfloat16* __restrict__ sum_out,
-    long long n_rows, int dim, float eps) {
+    long long n_rows, int dim, float eps,
+    const float* __restrict__ q_scale = nullptr,
+    float* __restrict__ q_amax = nullptr,
+    unsigned char* __restrict__ y8 = nullptr) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const long long row = (long long)blockIdx.x * 4 + wave;
@@ -149,6 +184,8 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
   }
   const float rstd = rsqrtf(wrsum(var) / dim + eps);
 
+  const float inv = F8 ? 1.0f / q_scale[0] : 0.0f;
+  float amax = 0.0f;
 #pragma unroll
   for (int t = 0; t < NIT; ++t) {
     const int i = lane * 4 + t * 256;
@@ -157,14 +194,32 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
     const short4 pbv = *reinterpret_cast<const short4*>(b + i);
     const __hip_bfloat16* wb = reinterpret_cast<const __hip_bfloat16*>(&pw);
     const __hip_bfloat16* bb = reinterpret_cast<const __hip_bfloat16*>(&pbv);
-    short4 out;
-    __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+    if (F8) {
+      uchar4 o8;
+      unsigned char* o8b = reinterpret_cast<unsigned char*>(&o8);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float f = (vals[t * 4 + j] - mean) * rstd;
-      ob[j] = __float2bfloat16(f * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
+      for (int j = 0; j < 4; ++j) {
+        const float f = (vals[t * 4 + j] - mean) * rstd *
+                        __bfloat162float(wb[j]) + __bfloat162float(bb[j]);
+        amax = fmaxf(amax, fabsf(f));
+        o8b[j] = __hip_cvt_float_to_fp8(f * inv, __HIP_SATFINITE, __HIP_E4M3);
+      }
+      *reinterpret_cast<uchar4*>(y8 + row * dim + i) = o8;
+    } else {
+      short4 out;
+      __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float f = (vals[t * 4 + j] - mean) * rstd;
+        ob[j] = __float2bfloat16(f * __bfloat162float(wb[j]) + __bfloat162float(bb[j]));
+      }
+      *reinterpret_cast<short4*>(yr + i) = out;
     }
-    *reinterpret_cast<short4*>(yr + i) = out;
+  }
+  if (F8) {
+    for (int off = 32; off > 0; off >>= 1)
+      amax = fmaxf(amax, __shfl_xor(amax, off, 64));
+    if (lane == 0) am_atomic_fmax(q_amax, amax);
   }
 }
 
@@ -212,6 +267,55 @@ void launch_layernorm_bf16_impl(const void* x, void* y, const void* w,
   else AM_LN_DISPATCH(16);
 #undef AM_LN_DISPATCH
 #undef AM_LN_CASE
+}
+
+void launch_layernorm_bf16_fp8_impl(const void* x, void* y8, const void* w,
+                                    const void* b, const void* in2,
+                                    void* sum_out, const float* q_scale,
+                                    float* q_amax, long long n_rows, int dim,
+                                    float eps, hipStream_t stream) {
+  const bool add = in2 != nullptr;
+  if (dim <= 128) {
+    const long long blocks2 = (n_rows + 7) / 8;
+    if (add)
+      hipLaunchKernelGGL((layernorm_bf16_half_kernel<true, true>),
+                         dim3((unsigned)blocks2), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x, nullptr,
+                         (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                         (const __hip_bfloat16*)in2, (__hip_bfloat16*)sum_out,
+                         n_rows, dim, eps, q_scale, q_amax,
+                         (unsigned char*)y8);
+    else
+      hipLaunchKernelGGL((layernorm_bf16_half_kernel<false, true>),
+                         dim3((unsigned)blocks2), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x, nullptr,
+                         (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                         nullptr, nullptr, n_rows, dim, eps, q_scale, q_amax,
+                         (unsigned char*)y8);
+    return;
+  }
+  const long long blocks = (n_rows + 3) / 4;
+  const dim3 grid((unsigned)blocks);
+  const dim3 block(256);
+#define AM_LN8_CASE(NIT, A)                                                    \
+  hipLaunchKernelGGL((layernorm_bf16_kernel<NIT, A, true>), grid, block, 0,    \
+                     stream, (const __hip_bfloat16*)x, nullptr,                \
+                     (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,       \
+                     (const __hip_bfloat16*)in2, (__hip_bfloat16*)sum_out,     \
+                     n_rows, dim, eps, q_scale, q_amax, (unsigned char*)y8)
+#define AM_LN8_DISPATCH(NIT)                                                   \
+  do {                                                                         \
+    if (add) AM_LN8_CASE(NIT, true);                                           \
+    else AM_LN8_CASE(NIT, false);                                              \
+  } while (0)
+  const int nit = (dim + 255) / 256;
+  if (nit <= 1) AM_LN8_DISPATCH(1);
+  else if (nit <= 2) AM_LN8_DISPATCH(2);
+  else if (nit <= 4) AM_LN8_DISPATCH(4);
+  else if (nit <= 8) AM_LN8_DISPATCH(8);
+  else AM_LN8_DISPATCH(16);
+#undef AM_LN8_DISPATCH
+#undef AM_LN8_CASE
 }
 
 void launch_layernorm_bf16(const void* x, void* y, const void* w,

@@ -77,6 +77,59 @@ def quantize_weight(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     return wq, s
 
 
+# -- producer-fused path (the profitable one) -------------------------------
+# LayerNorm emits e4m3 directly with a DELAYED per-tensor scale: the
+# kernel quantizes with the previous step's amax while atomically
+# recording the current one (transformer-engine-style). The standalone
+# quantize pass — which made unfused fp8 a net loss — disappears.
+
+_E4M3_MAX = 448.0
+_ln_state: Dict[int, Tuple[torch.Tensor, torch.Tensor, list]] = {}
+
+
+def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
+    key = id(module)
+    st = _ln_state.get(key)
+    if st is None or st[0].device != device:
+        # init scale for |LN out| up to ~8 (gamma ~ 1); self-corrects
+        # from the recorded amax after the first step
+        scale = torch.full((), 8.0 / _E4M3_MAX, device=device)
+        amax = torch.zeros((), device=device)
+        st = (scale, amax, [False])
+        _ln_state[key] = st
+    scale, amax, warm = st
+    if warm[0]:
+        scale.copy_((amax.clamp(min=1e-6) * 1.05) / _E4M3_MAX)
+        amax.zero_()
+    warm[0] = True
+    return scale, amax
+
+
+def ln_fp8(norm, x: torch.Tensor):
+    """FusedLayerNorm -> (y_fp8, scale) via the fused-quantize kernel."""
+    from audiomuse_amd.ops import _ext
+
+    ext = _ext.require()
+    scale, amax = _state_for(norm, x.device)
+    y8 = ext.layernorm_bf16_fp8(
+        x.contiguous(), norm.weight.to(torch.bfloat16).contiguous(),
+        norm.bias.to(torch.bfloat16).contiguous(), norm.eps, scale, amax)
+    return y8, scale
+
+
+def add_ln_fp8(norm, x: torch.Tensor, other: torch.Tensor):
+    """Residual add + LN -> (sum_bf16, y_fp8, scale)."""
+    from audiomuse_amd.ops import _ext
+
+    ext = _ext.require()
+    scale, amax = _state_for(norm, x.device)
+    s, y8 = ext.add_layernorm_bf16_fp8(
+        x.contiguous(), other.contiguous(),
+        norm.weight.to(torch.bfloat16).contiguous(),
+        norm.bias.to(torch.bfloat16).contiguous(), norm.eps, scale, amax)
+    return s, y8, scale
+
+
 def scaled_linear(x: torch.Tensor, weight: torch.Tensor,
                   bias: Optional[torch.Tensor]) -> torch.Tensor:
     """Linear in fp8: x (..., K) bf16 @ weight (N, K) -> (..., N) bf16.
