@@ -109,7 +109,9 @@ __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
       *reinterpret_cast<uchar4*>(y8 + row * dim + i) = o8;
       for (int off = 16; off > 0; off >>= 1)
         amax = fmaxf(amax, __shfl_xor(amax, off, 32));
-      if (sl == 0) am_atomic_fmax(q_amax, amax);
+      // 256-slot amax vector: same-address atomics would serialize
+      // (~0.5M waves -> one cacheline); slots spread the contention
+      if (sl == 0) am_atomic_fmax(q_amax + (blockIdx.x & 255), amax);
     } else {
       short4 out;
       __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
@@ -219,7 +221,7 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
   if (F8) {
     for (int off = 32; off > 0; off >>= 1)
       amax = fmaxf(amax, __shfl_xor(amax, off, 64));
-    if (lane == 0) am_atomic_fmax(q_amax, amax);
+    if (lane == 0) am_atomic_fmax(q_amax + (blockIdx.x & 255), amax);
   }
 }
 

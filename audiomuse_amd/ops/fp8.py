@@ -94,12 +94,12 @@ def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
         # init scale for |LN out| up to ~8 (gamma ~ 1); self-corrects
         # from the recorded amax after the first step
         scale = torch.full((), 8.0 / _E4M3_MAX, device=device)
-        amax = torch.zeros((), device=device)
+        amax = torch.zeros(256, device=device)   # slot-spread (see kernel)
         st = (scale, amax, [False])
         _ln_state[key] = st
     scale, amax, warm = st
     if warm[0]:
-        scale.copy_((amax.clamp(min=1e-6) * 1.05) / _E4M3_MAX)
+        scale.copy_((amax.max().clamp(min=1e-6) * 1.05) / _E4M3_MAX)
         amax.zero_()
     warm[0] = True
     return scale, amax
