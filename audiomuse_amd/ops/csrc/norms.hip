@@ -26,6 +26,16 @@ __device__ __forceinline__ void am_atomic_fmax(float* addr, float v) {
   atomicMax(reinterpret_cast<int*>(addr), __float_as_int(v));
 }
 
+// 4 floats -> 4 OCP e4m3 bytes via the gfx950 packed-convert
+// instruction (the hip_fp8.h helper is software emulation)
+__device__ __forceinline__ unsigned int am_pack_fp8x4(float f0, float f1,
+                                                      float f2, float f3) {
+  int packed = 0;
+  packed = __builtin_amdgcn_cvt_pk_fp8_f32(f0, f1, packed, false);
+  packed = __builtin_amdgcn_cvt_pk_fp8_f32(f2, f3, packed, true);
+  return (unsigned int)packed;
+}
+
 __device__ __forceinline__ float wrsum(float v) {
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
   return __shfl(v, 0, 64);
@@ -197,16 +207,16 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
     const __hip_bfloat16* wb = reinterpret_cast<const __hip_bfloat16*>(&pw);
     const __hip_bfloat16* bb = reinterpret_cast<const __hip_bfloat16*>(&pbv);
     if (F8) {
-      uchar4 o8;
-      unsigned char* o8b = reinterpret_cast<unsigned char*>(&o8);
+      float fq[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const float f = (vals[t * 4 + j] - mean) * rstd *
                         __bfloat162float(wb[j]) + __bfloat162float(bb[j]);
         amax = fmaxf(amax, fabsf(f));
-        o8b[j] = __hip_cvt_float_to_fp8(f * inv, __HIP_SATFINITE, __HIP_E4M3);
+        fq[j] = __builtin_amdgcn_fmed3f(f * inv, 448.0f, -448.0f);
       }
-      *reinterpret_cast<uchar4*>(y8 + row * dim + i) = o8;
+      *reinterpret_cast<unsigned int*>(y8 + row * dim + i) =
+          am_pack_fp8x4(fq[0], fq[1], fq[2], fq[3]);
     } else {
       short4 out;
       __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
@@ -218,7 +228,7 @@ __global__ __launch_bounds__(256) void layernorm_bf16_kernel(
       *reinterpret_cast<short4*>(yr + i) = out;
     }
   }
-  if (F8) {
+  if (F8 && (blockIdx.x & 63) == 0) {
     for (int off = 32; off > 0; off >>= 1)
       amax = fmaxf(amax, __shfl_xor(amax, off, 64));
     if (lane == 0) am_atomic_fmax(q_amax + (blockIdx.x & 255), amax);

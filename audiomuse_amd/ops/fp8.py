@@ -64,7 +64,10 @@ def quantize_weight(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     Cached by (data_ptr, _version): serving weights are frozen, so the
     quantization runs once per weight tensor.
     """
-    key = (w.data_ptr(), w._version)
+    try:
+        key = (w.data_ptr(), w._version)
+    except RuntimeError:          # inference tensors track no version
+        key = (w.data_ptr(), -1)
     hit = _wcache.get(key)
     if hit is not None:
         return hit
