@@ -107,21 +107,24 @@ __global__ __launch_bounds__(256) void layernorm_bf16_half_kernel(
     if (F8) {
       const float inv = 1.0f / q_scale[0];
       float amax = 0.0f;
-      uchar4 o8;
-      unsigned char* o8b = reinterpret_cast<unsigned char*>(&o8);
+      float fq[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const float f = ((vals[j] - mean) * rstd) * __bfloat162float(wb[j]) +
                         __bfloat162float(bb[j]);
         amax = fmaxf(amax, fabsf(f));
-        o8b[j] = __hip_cvt_float_to_fp8(f * inv, __HIP_SATFINITE, __HIP_E4M3);
+        fq[j] = __builtin_amdgcn_fmed3f(f * inv, 448.0f, -448.0f);
       }
-      *reinterpret_cast<uchar4*>(y8 + row * dim + i) = o8;
-      for (int off = 16; off > 0; off >>= 1)
-        amax = fmaxf(amax, __shfl_xor(amax, off, 32));
-      // 256-slot amax vector: same-address atomics would serialize
-      // (~0.5M waves -> one cacheline); slots spread the contention
-      if (sl == 0) am_atomic_fmax(q_amax + (blockIdx.x & 255), amax);
+      *reinterpret_cast<unsigned int*>(y8 + row * dim + i) =
+          am_pack_fp8x4(fq[0], fq[1], fq[2], fq[3]);
+      // amax is SAMPLED (1/64 of blocks) into a 256-slot vector: a full
+      // per-wave atomic on one address serializes; the delayed scale
+      // only needs a statistical amax and the e4m3 cast saturates
+      if ((blockIdx.x & 63) == 0) {
+        for (int off = 16; off > 0; off >>= 1)
+          amax = fmaxf(amax, __shfl_xor(amax, off, 32));
+        if (sl == 0) am_atomic_fmax(q_amax + (blockIdx.x & 255), amax);
+      }
     } else {
       short4 out;
       __hip_bfloat16* ob = reinterpret_cast<__hip_bfloat16*>(&out);
