@@ -18,19 +18,25 @@ The reference has no quantized serving path (its ONNX students run
 fp32 CUDA/DML — /root/reference/tasks/clap_analyzer.py); this is an
 MI355X-native extra, not a parity item.
 
-MEASURED STATUS (scripts/fp8_shapes.py on MI355X, 2026-09): the fp8
-GEMM alone beats bf16 at most encoder shapes (1.4-1.9x) but the
-unfused dynamic quantization costs ~2/3 of a full bf16 GEMM per input
-(it re-reads the very activation bytes fp8 saves), and hipBLASLt's fp8
-algo at the stage-1 qkv shape (M=2.1M, K=128, N=384) is 2.1x SLOWER
-than bf16. Net: this mode currently trades ~2.7x throughput for the
-accuracy headroom experiment. Making fp8 pay requires (a) quantization
-fused into the producing kernels (LayerNorm/attention emit e4m3 +
-running amax, delayed scaling) and (b) an algo-searched fp8 GEMM for
-the K=128 shapes — both round-2 items. gfx950 note: non-scaled fp8
-MFMA runs at the bf16 rate; fp8's entire win at these memory-bound
-shapes is the halved operand traffic, which is exactly what an
-unfused quantize pass spends back.
+MEASURED STATUS (MI355X, end of round 1): the fused design WINS —
+10 724 clips/s vs 9 621 bf16 same box (+11.5%), embedding cosine >
+0.98 vs bf16. What made it pay (each step A/B-measured, see
+profiles/r01_final_profile.md):
+- quantization fused into the producing LayerNorm kernels
+  (norms.hip F8 variants): e4m3 emitted via the gfx950 packed-convert
+  instruction with a DELAYED per-tensor scale; amax is sampled (1/64
+  of blocks) into a 256-slot vector — a naive per-wave atomic on one
+  address serialized the whole kernel;
+- fp8 GEMMs through the extension's timed algo search
+  (gemm_gelu.cpp::linear_fp8): torch._scaled_mm's heuristic pick was
+  2.1x SLOWER than bf16 at the stage-1 qkv shape, the searched algo is
+  1.4x FASTER;
+- only the profitable GEMMs convert (qkv + mlp0 with fused GELU_BIAS
+  epilogue); proj and mlp2 stay bf16 (their gain is smaller than any
+  quantize cost).
+gfx950 note: non-scaled fp8 MFMA runs at the bf16 rate; the entire fp8
+win at these memory-bound shapes is halved operand traffic, which is
+why the quantize pass must ride a producer kernel, not its own.
 """
 
 from __future__ import annotations
