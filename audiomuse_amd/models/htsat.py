@@ -188,12 +188,26 @@ class SwinBlock(nn.Module):
                 # fused residual add + LN emitting e4m3 for the mlp0 GEMM
                 x2, xn2q, xs2 = fp8.add_ln_fp8(self.norm2, x, proj)
                 w0q, w0s = fp8.quantize_weight(self.mlp[0].weight)
-                hidden = ext.linear_fp8(
-                    xn2q, w0q, xs2, w0s,
-                    self.mlp[0].bias.to(torch.bfloat16).contiguous(),
-                    gelu=True)
-                # mlp2 stays bf16 (quantizing the 4C-wide hidden costs
-                # more than the fp8 GEMM saves) with the residual folded
+                b0 = self.mlp[0].bias.to(torch.bfloat16).contiguous()
+                if fp8.FP8_HIDDEN[0]:
+                    # mlp0 emits e4m3 hidden straight from the GEMM
+                    # epilogue (D-scale + device-recorded amax), making
+                    # the mlp2 GEMM fp8 too with the residual folded
+                    try:
+                        hs, hinv, hamax = fp8.hidden_state(self.mlp[0],
+                                                           x.device)
+                        hidden8 = ext.linear_fp8(xn2q, w0q, xs2, w0s, b0,
+                                                 gelu=True,
+                                                 d_inv_scale=hinv,
+                                                 amax_d=hamax)
+                        w2q, w2s = fp8.quantize_weight(self.mlp[2].weight)
+                        return ext.linear_fp8(
+                            hidden8, w2q, hs, w2s,
+                            self.mlp[2].bias.to(torch.bfloat16).contiguous(),
+                            resid=x2)
+                    except RuntimeError:   # no fp8-D algo on this build
+                        fp8.FP8_HIDDEN[0] = False
+                hidden = ext.linear_fp8(xn2q, w0q, xs2, w0s, b0, gelu=True)
                 return ext.linear_bias_add(hidden,
                                            self.mlp[2].weight.contiguous(),
                                            self.mlp[2].bias.contiguous(), x2)

@@ -210,7 +210,10 @@ torch::Tensor linear_bias(torch::Tensor x, torch::Tensor w,
 torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
                          torch::Tensor sa, torch::Tensor sb,
                          c10::optional<torch::Tensor> bias,
-                         bool gelu) {
+                         bool gelu,
+                         c10::optional<torch::Tensor> resid,
+                         c10::optional<torch::Tensor> d_inv_scale,
+                         c10::optional<torch::Tensor> amax_d) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat8_e4m3fn &&
                   x.is_contiguous(),
               "x must be contiguous fp8e4m3 GPU");
@@ -223,9 +226,11 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
   const int64_t N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "shape mismatch");
 
+  const bool out8 = d_inv_scale.has_value();
   auto sizes = x.sizes().vec();
   sizes.back() = N;
-  auto y = torch::empty(sizes, x.options().dtype(at::kBFloat16));
+  auto y = torch::empty(sizes, x.options().dtype(
+      out8 ? at::kFloat8_e4m3fn : at::kBFloat16));
 
   hipblasLtMatmulDesc_t desc;
   HIPBLASLT_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F,
@@ -235,6 +240,17 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
       desc, HIPBLASLT_MATMUL_DESC_TRANSA, &opA, sizeof(opA)));
   HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
       desc, HIPBLASLT_MATMUL_DESC_TRANSB, &opB, sizeof(opB)));
+  if (out8) {
+    // D = d_inv_scale * acc, cast to e4m3; true amax recorded on device
+    const void* ds = d_inv_scale->data_ptr();
+    HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+        desc, HIPBLASLT_MATMUL_DESC_D_SCALE_POINTER, &ds, sizeof(ds)));
+    if (amax_d.has_value()) {
+      const void* ap = amax_d->data_ptr();
+      HIPBLASLT_CHECK(hipblasLtMatmulDescSetAttribute(
+          desc, HIPBLASLT_MATMUL_DESC_AMAX_D_POINTER, &ap, sizeof(ap)));
+    }
+  }
   // col-major framing: A = w (K, N) cm via T, B = x (K, M) cm; scales
   // swap accordingly (A-scale applies to w)
   const void* a_scale = sb.data_ptr();
@@ -262,7 +278,15 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
   hipblasLtMatrixLayout_t la, lb, ld;
   HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_8F_E4M3, K, N, K));
   HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_8F_E4M3, K, M, K));
-  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(&ld, HIP_R_16BF, N, M, N));
+  HIPBLASLT_CHECK(hipblasLtMatrixLayoutCreate(
+      &ld, out8 ? HIP_R_8F_E4M3 : HIP_R_16BF, N, M, N));
+  if (resid.has_value()) {
+    TORCH_CHECK(!out8, "residual add requires bf16 output");
+    TORCH_CHECK(resid->is_contiguous() &&
+                    resid->scalar_type() == at::kBFloat16 &&
+                    resid->numel() == M * N,
+                "residual must be contiguous bf16 of (M, N)");
+  }
 
   auto stream = c10::hip::getCurrentHIPStream();
   static void* workspace = nullptr;
@@ -277,7 +301,7 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
   bool have_algo = false;
   {
     std::lock_guard<std::mutex> g(algo_mu);
-    auto it = algo_cache.find({M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0)});
+    auto it = algo_cache.find({M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0) + (out8 ? 4 : 0) + (resid.has_value() ? 8 : 0)});
     if (it != algo_cache.end()) {
       algo = it->second;
       have_algo = true;
@@ -295,7 +319,9 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
         get_handle(), desc, la, lb, ld, ld, pref, 8, results, &found));
     hipblasLtMatmulPreferenceDestroy(pref);
     TORCH_CHECK(found > 0, "no hipblaslt fp8 algo at ", M, "x", N, "x", K);
-    const float alpha_s = 1.0f, beta_s = 0.0f;
+    const float alpha_s = 1.0f;
+    const float beta_s = resid.has_value() ? 1.0f : 0.0f;
+    const void* c_s = resid.has_value() ? resid->data_ptr() : y.data_ptr();
     auto stream0 = stream.stream();
     hipEvent_t ev0, ev1;
     (void)hipEventCreate(&ev0);
@@ -304,14 +330,14 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
     float best_ms = 1e30f;
     for (int i = 0; i < found; ++i) {
       if (hipblasLtMatmul(get_handle(), desc, &alpha_s, w.data_ptr(), la,
-                          x.data_ptr(), lb, &beta_s, y.data_ptr(), ld,
+                          x.data_ptr(), lb, &beta_s, c_s, ld,
                           y.data_ptr(), ld, &results[i].algo, workspace,
                           workspace_size, stream0) != HIPBLAS_STATUS_SUCCESS)
         continue;
       (void)hipEventRecord(ev0, stream0);
       for (int r = 0; r < 2; ++r)
         (void)hipblasLtMatmul(get_handle(), desc, &alpha_s, w.data_ptr(), la,
-                              x.data_ptr(), lb, &beta_s, y.data_ptr(), ld,
+                              x.data_ptr(), lb, &beta_s, c_s, ld,
                               y.data_ptr(), ld, &results[i].algo, workspace,
                               workspace_size, stream0);
       (void)hipEventRecord(ev1, stream0);
@@ -327,13 +353,15 @@ torch::Tensor linear_fp8(torch::Tensor x, torch::Tensor w,
     (void)hipEventDestroy(ev1);
     algo = results[best].algo;
     std::lock_guard<std::mutex> g(algo_mu);
-    algo_cache[{M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0)}] = algo;
+    algo_cache[{M, N, K, 1000 + (bias.has_value() ? 1 : 0) + (gelu ? 2 : 0) + (out8 ? 4 : 0) + (resid.has_value() ? 8 : 0)}] = algo;
   }
 
-  const float alpha = 1.0f, beta = 0.0f;
+  const float alpha = 1.0f;
+  const float beta = resid.has_value() ? 1.0f : 0.0f;
+  const void* c_ptr = resid.has_value() ? resid->data_ptr() : y.data_ptr();
   HIPBLASLT_CHECK(hipblasLtMatmul(
       get_handle(), desc, &alpha, w.data_ptr(), la, x.data_ptr(), lb, &beta,
-      y.data_ptr(), ld, y.data_ptr(), ld, &algo, workspace, workspace_size,
+      c_ptr, ld, y.data_ptr(), ld, &algo, workspace, workspace_size,
       stream.stream()));
 
   hipblasLtMatrixLayoutDestroy(la);
@@ -356,5 +384,8 @@ void register_gemm_gelu(pybind11::module_& m) {
         "fp8 e4m3 GEMM with per-tensor scales + timed algo search",
         pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("sa"),
         pybind11::arg("sb"), pybind11::arg("bias") = pybind11::none(),
-        pybind11::arg("gelu") = false);
+        pybind11::arg("gelu") = false,
+        pybind11::arg("resid") = pybind11::none(),
+        pybind11::arg("d_inv_scale") = pybind11::none(),
+        pybind11::arg("amax_d") = pybind11::none());
 }

@@ -114,6 +114,30 @@ def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
     return scale, amax
 
 
+_hid_state: Dict[int, Tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]] = {}
+FP8_HIDDEN = [True]   # flipped off if hipBLASLt lacks fp8-D+GELU here
+
+
+def hidden_state(module, device):
+    """(scale, inv_scale, amax) for a GEMM's fp8 D output: the GEMM
+    records the true amax (AMAX_D pointer), delayed like the LN scales."""
+    key = id(module)
+    st = _hid_state.get(key)
+    if st is None or st[0].device != device:
+        scale = torch.full((), 4.0 / _E4M3_MAX, device=device)
+        inv = torch.full((), _E4M3_MAX / 4.0, device=device)
+        amax = torch.zeros((), device=device)
+        st = (scale, inv, amax, [False])
+        _hid_state[key] = st
+    scale, inv, amax, warm = st
+    if warm[0]:
+        scale.copy_((amax.clamp(min=1e-6) * 1.05) / _E4M3_MAX)
+        inv.copy_(1.0 / scale)
+        amax.zero_()
+    warm[0] = True
+    return scale, inv, amax
+
+
 def ln_fp8(norm, x: torch.Tensor):
     """FusedLayerNorm -> (y_fp8, scale) via the fused-quantize kernel."""
     from audiomuse_amd.ops import _ext
