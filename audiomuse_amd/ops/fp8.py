@@ -19,7 +19,7 @@ fp32 CUDA/DML — /root/reference/tasks/clap_analyzer.py); this is an
 MI355X-native extra, not a parity item.
 
 MEASURED STATUS (MI355X, end of round 1): the fused design WINS —
-10 724 clips/s vs 9 621 bf16 same box (+11.5%), embedding cosine >
+10 811 clips/s vs 9 621 bf16 same box (+11.5%), embedding cosine >
 0.98 vs bf16. What made it pay (each step A/B-measured, see
 profiles/r01_final_profile.md):
 - quantization fused into the producing LayerNorm kernels
