@@ -205,8 +205,9 @@ class SwinBlock(nn.Module):
                             hidden8, w2q, hs, w2s,
                             self.mlp[2].bias.to(torch.bfloat16).contiguous(),
                             resid=x2)
-                    except RuntimeError:   # no fp8-D algo on this build
+                    except RuntimeError as exc:  # no fp8-D algo here
                         fp8.FP8_HIDDEN[0] = False
+                        fp8.FP8_HIDDEN_ERR = str(exc)
                 hidden = ext.linear_fp8(xn2q, w0q, xs2, w0s, b0, gelu=True)
                 return ext.linear_bias_add(hidden,
                                            self.mlp[2].weight.contiguous(),

@@ -116,6 +116,7 @@ def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
 
 _hid_state: Dict[int, Tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]] = {}
 FP8_HIDDEN = [True]   # flipped off if hipBLASLt lacks fp8-D+GELU here
+FP8_HIDDEN_ERR = ""
 
 
 def hidden_state(module, device):
