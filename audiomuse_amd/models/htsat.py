@@ -196,10 +196,13 @@ class SwinBlock(nn.Module):
                     try:
                         hs, hinv, hamax = fp8.hidden_state(self.mlp[0],
                                                            x.device)
+                        # amax_d + GELU has no algo at large M on this
+                        # hipBLASLt: estimate amax from a sampled slice
                         hidden8 = ext.linear_fp8(xn2q, w0q, xs2, w0s, b0,
                                                  gelu=True,
-                                                 d_inv_scale=hinv,
-                                                 amax_d=hamax)
+                                                 d_inv_scale=hinv)
+                        samp = hidden8.reshape(-1)[: 1 << 22]
+                        hamax.copy_(samp.float().abs().amax() * hs)
                         w2q, w2s = fp8.quantize_weight(self.mlp[2].weight)
                         return ext.linear_fp8(
                             hidden8, w2q, hs, w2s,
