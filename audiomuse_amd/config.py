@@ -200,6 +200,7 @@ CLAP_GPU_BATCH = _env_int("CLAP_GPU_BATCH", 256)
 # opt-in fp8 e4m3 serving path for encoder GEMMs (ops/fp8.py); the
 # headline bench stays bf16 regardless of this flag unless --fp8 is passed
 CLAP_FP8_SERVING = _env_bool("AUDIOMUSE_FP8_SERVING", False)
+FP8_HIDDEN_ENABLE = _env_bool("AUDIOMUSE_FP8_HIDDEN", False)
 HIP_REQUIRE_NATIVE = _env_bool("HIP_REQUIRE_NATIVE", True)  # fail loudly on GPU without .so
 RCCL_BUCKET_CAP_MB = _env_int("RCCL_BUCKET_CAP_MB", 64)
 
