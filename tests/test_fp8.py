@@ -52,3 +52,17 @@ def test_fp8_encoder_embeddings_match_bf16(monkeypatch):
         out = model(mel).float()
     cos = torch.nn.functional.cosine_similarity(ref, out, dim=1)
     assert float(cos.min()) > 0.98
+
+
+def test_quantize_weight_under_inference_mode():
+    # inference tensors track no version counter (the soak hit this)
+    with torch.inference_mode():
+        w = torch.randn(16, 8) * 0.2
+        wq, s = fp8.quantize_weight(w)
+        assert wq.dtype == torch.float8_e4m3fn
+        wq2, _ = fp8.quantize_weight(w)
+        assert wq2 is wq  # cached under the fallback key
+
+
+def test_hidden_chain_off_by_default():
+    assert fp8.FP8_HIDDEN[0] is False
