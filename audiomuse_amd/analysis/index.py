@@ -166,7 +166,8 @@ def build_artist_index(conn: sqlite3.Connection, seed: int = 0) -> int:
         return 0
     sim = ArtistSimilarity()
     sim.fit_catalogue(per_artist, seed=seed)
-    payload = {name: {"means": m.means, "weights": m.weights,
+    payload = {name: {"means": torch.from_numpy(np.ascontiguousarray(m.means)),
+                      "weights": torch.from_numpy(np.ascontiguousarray(m.weights)),
                       "n_tracks": m.n_tracks}
                for name, m in sim.models.items()}
     buf = io.BytesIO()
@@ -181,13 +182,14 @@ def load_artist_similarity(conn: sqlite3.Connection) -> Optional[ArtistSimilarit
     if got is None:
         return None
     payload = torch.load(io.BytesIO(got[0]), map_location="cpu",
-                         weights_only=False)
+                         weights_only=True)
     from audiomuse_amd.engines.artist_gmm import ArtistModel
 
     sim = ArtistSimilarity()
     for name, d in payload.items():
-        sim.models[name] = ArtistModel(name=name, means=d["means"],
-                                       weights=d["weights"],
+        sim.models[name] = ArtistModel(name=name,
+                                       means=d["means"].numpy(),
+                                       weights=d["weights"].numpy(),
                                        n_tracks=d["n_tracks"])
     sim._names = list(sim.models)
     cents = [m.means.mean(axis=0) for m in sim.models.values()]

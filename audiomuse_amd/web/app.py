@@ -82,7 +82,7 @@ class AppState:
             import io
             import torch
             eng = torch.load(io.BytesIO(got[0]), map_location="cpu",
-                             weights_only=False)  # our own trusted blob
+                             weights_only=True)
         else:
             eng = idx.load_ivf_engine(self.conn(), name, device=self.device,
                                       meta_fn=self.meta_fn)
