@@ -33,6 +33,8 @@ class SubsonicProvider(Provider):
         self.http = session or requests.Session()
 
     def _params(self) -> Dict[str, str]:
+        # md5(password + salt) is the Subsonic API's own token scheme
+        # (protocol-mandated; not a choice of hash for security here)
         salt = secrets.token_hex(8)
         token = hashlib.md5((self.credential + salt).encode()).hexdigest()
         return {"u": self.username, "t": token, "s": salt, "v": "1.16.1",
