@@ -197,6 +197,19 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             pass
         return resp
 
+    # background engine warm (reference: startup index load + map cache
+    # build, app.py:1244): the first query should not pay the blob
+    # deserialize + device upload (~8 s at 1M tracks on GPU)
+    def _warm_engines():
+        for name in (idx.AUDIO_INDEX, idx.CLAP_INDEX, idx.SONG_MAP):
+            try:
+                state.engine(name)
+            except Exception:  # noqa: BLE001 — warm-up must never crash boot
+                pass
+
+    threading.Thread(target=_warm_engines, daemon=True,
+                     name="audiomuse-engine-warm").start()
+
     # cron scheduler thread (reference: app.py cron loop)
     import threading as _threading
 
