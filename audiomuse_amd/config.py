@@ -214,6 +214,9 @@ AI_MAX_TOOL_CALLS = _env_int("AI_MAX_TOOL_CALLS", 4)
 # Web
 JWT_SECRET = _env("AUDIOMUSE_JWT_SECRET", "")
 API_TOKEN = _env("AUDIOMUSE_API_TOKEN", "")
+# Opt-in trust of X-Forwarded-Prefix/Proto (reference: proxy_prefix.py
+# is mounted only under a known proxy deployment)
+BEHIND_PROXY = _env_bool("AUDIOMUSE_BEHIND_PROXY", False)
 
 
 def set_db_override_provider(provider: Optional[Callable[[], Dict[str, str]]]) -> None:

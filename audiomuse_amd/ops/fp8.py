@@ -41,6 +41,7 @@ why the quantize pass must ride a producer kernel, not its own.
 
 from __future__ import annotations
 
+import weakref
 from typing import Dict, Optional, Tuple
 
 import torch
@@ -48,7 +49,24 @@ import torch
 from audiomuse_amd import config as C
 
 _E4M3 = torch.float8_e4m3fn
-_wcache: Dict[Tuple[int, int], Tuple[torch.Tensor, torch.Tensor]] = {}
+# Keyed by id(owner) with a validating weakref: allocator/GC id reuse
+# after a model swap must never return another tensor's quantized data
+# (ADVICE r1). The weakref callback evicts entries when the owner dies.
+_wcache: Dict[int, tuple] = {}
+
+
+def _weak_entry(cache: dict, owner, *payload):
+    key = id(owner)
+    cache[key] = (weakref.ref(owner, lambda _r, k=key, c=cache: c.pop(k, None)),
+                  *payload)
+    return cache[key]
+
+
+def _weak_get(cache: dict, owner):
+    ent = cache.get(id(owner))
+    if ent is None or ent[0]() is not owner:
+        return None
+    return ent[1:]
 
 
 def serving_enabled() -> bool:
@@ -67,22 +85,21 @@ def available(device: Optional[torch.device] = None) -> bool:
 def quantize_weight(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """(N, K) bf16/f32 -> (wq (N, K) e4m3 contiguous, scale 0-d f32).
 
-    Cached by (data_ptr, _version): serving weights are frozen, so the
-    quantization runs once per weight tensor.
+    Cached per weight tensor (weakref-validated): serving weights are
+    frozen, so the quantization runs once per weight tensor and the
+    entry dies with it.
     """
     try:
-        key = (w.data_ptr(), w._version)
+        ver = w._version
     except RuntimeError:          # inference tensors track no version
-        key = (w.data_ptr(), -1)
-    hit = _wcache.get(key)
-    if hit is not None:
-        return hit
+        ver = -1
+    hit = _weak_get(_wcache, w)
+    if hit is not None and hit[0] == ver:
+        return hit[1], hit[2]
     fmax = torch.finfo(_E4M3).max
     s = (w.detach().abs().amax().float() / fmax).clamp(min=1e-12)
     wq = (w.detach().float() / s).clamp(-fmax, fmax).to(_E4M3).contiguous()
-    if len(_wcache) > 256:   # bounded: model swaps must not leak memory
-        _wcache.clear()
-    _wcache[key] = (wq, s)
+    _weak_entry(_wcache, w, ver, wq, s)
     return wq, s
 
 
@@ -93,19 +110,19 @@ def quantize_weight(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
 # quantize pass — which made unfused fp8 a net loss — disappears.
 
 _E4M3_MAX = 448.0
-_ln_state: Dict[int, Tuple[torch.Tensor, torch.Tensor, list]] = {}
+_ln_state: Dict[int, tuple] = {}
 
 
 def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
-    key = id(module)
-    st = _ln_state.get(key)
+    hit = _weak_get(_ln_state, module)
+    st = hit[0] if hit is not None else None
     if st is None or st[0].device != device:
         # init scale for |LN out| up to ~8 (gamma ~ 1); self-corrects
         # from the recorded amax after the first step
         scale = torch.full((), 8.0 / _E4M3_MAX, device=device)
         amax = torch.zeros(256, device=device)   # slot-spread (see kernel)
         st = (scale, amax, [False])
-        _ln_state[key] = st
+        _weak_entry(_ln_state, module, st)
     scale, amax, warm = st
     if warm[0]:
         scale.copy_((amax.max().clamp(min=1e-6) * 1.05) / _E4M3_MAX)
@@ -114,7 +131,7 @@ def _state_for(module, device) -> Tuple[torch.Tensor, torch.Tensor]:
     return scale, amax
 
 
-_hid_state: Dict[int, Tuple[torch.Tensor, torch.Tensor, torch.Tensor, list]] = {}
+_hid_state: Dict[int, tuple] = {}
 # fp8 hidden chain (mlp0 emits e4m3 for an fp8 mlp2): measured NEUTRAL
 # (10 739 vs 10 811 clips/s) while costing embedding accuracy (the
 # e4m3-quantized 4C hidden fails the cosine>0.98 gate), so it is OFF by
@@ -126,14 +143,14 @@ FP8_HIDDEN_ERR = ""
 def hidden_state(module, device):
     """(scale, inv_scale, amax) for a GEMM's fp8 D output: the GEMM
     records the true amax (AMAX_D pointer), delayed like the LN scales."""
-    key = id(module)
-    st = _hid_state.get(key)
+    hit = _weak_get(_hid_state, module)
+    st = hit[0] if hit is not None else None
     if st is None or st[0].device != device:
         scale = torch.full((), 4.0 / _E4M3_MAX, device=device)
         inv = torch.full((), _E4M3_MAX / 4.0, device=device)
         amax = torch.zeros((), device=device)
         st = (scale, inv, amax, [False])
-        _hid_state[key] = st
+        _weak_entry(_hid_state, module, st)
     scale, inv, amax, warm = st
     if warm[0]:
         scale.copy_((amax.clamp(min=1e-6) * 1.05) / _E4M3_MAX)

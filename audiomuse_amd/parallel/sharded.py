@@ -75,8 +75,13 @@ def allgather_embeddings(local: torch.Tensor, group=None) -> torch.Tensor:
     if not dist.is_initialized():
         return local
     world = dist.get_world_size(group)
-    n_local = torch.tensor([local.shape[0]], dtype=torch.long)
-    counts = [torch.zeros(1, dtype=torch.long) for _ in range(world)]
+    # Count exchange must ride the same device as the payload: with the
+    # nccl/RCCL backend (the GPU deployment shape) collectives on CPU
+    # tensors fail outright.
+    n_local = torch.tensor([local.shape[0]], dtype=torch.long,
+                           device=local.device)
+    counts = [torch.zeros(1, dtype=torch.long, device=local.device)
+              for _ in range(world)]
     dist.all_gather(counts, n_local, group=group)
     maxn = int(max(c.item() for c in counts))
     padded = torch.zeros(maxn, local.shape[1], dtype=local.dtype,

@@ -25,14 +25,16 @@ def setup_status():
 def setup_admin():
     """First-boot admin creation (reference: setup wizard step)."""
     conn = _state().conn()
-    if not check_setup_needed(conn):
-        return jsonify({"error": "already configured"}), 409
     body = request.get_json(force=True, silent=True) or {}
     username = (body.get("username") or "").strip()
     password = body.get("password") or ""
     if not username or len(password) < 8:
         return jsonify({"error": "username and password (>=8 chars) required"}), 400
+    # COUNT check and INSERT share one write transaction so two racing
+    # first-boot requests cannot both create an admin (ADVICE r1 TOCTOU).
     with write_txn(conn):
+        if not check_setup_needed(conn):
+            return jsonify({"error": "already configured"}), 409
         conn.execute(
             "INSERT INTO audiomuse_users (username, password_hash, role) "
             "VALUES (?,?, 'admin')", (username, hash_password(password)))
@@ -85,7 +87,8 @@ def get_config():
     from audiomuse_amd import config as C
     from audiomuse_amd.db.store import get_app_config
 
-    overrides = get_app_config(_state().conn())
+    overrides = {k: v for k, v in get_app_config(_state().conn()).items()
+                 if not k.startswith("_")}  # underscore keys are internal
     safe = {k: getattr(C, k) for k in dir(C)
             if k.isupper() and isinstance(getattr(C, k), (int, float, str, bool))
             and "SECRET" not in k and "TOKEN" not in k and "PASSWORD" not in k}

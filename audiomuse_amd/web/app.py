@@ -99,7 +99,11 @@ class AppState:
 
 class ProxyPrefixMiddleware:
     """Honor X-Forwarded-Prefix from a reverse proxy so url_for and
-    redirects carry the mount prefix (reference: proxy_prefix.py:31)."""
+    redirects carry the mount prefix (reference: proxy_prefix.py:31).
+
+    Opt-in via AUDIOMUSE_BEHIND_PROXY: without a proxy in front, these
+    headers arrive attacker-controlled and could rewrite SCRIPT_NAME /
+    url scheme in generated URLs."""
 
     def __init__(self, wsgi_app):
         self.wsgi_app = wsgi_app
@@ -125,7 +129,8 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
                 static_folder=_os.path.join(_os.path.dirname(
                     _os.path.abspath(__file__)), "static"),
                 static_url_path="/static")
-    app.wsgi_app = ProxyPrefixMiddleware(app.wsgi_app)
+    if getattr(C, "BEHIND_PROXY", False):
+        app.wsgi_app = ProxyPrefixMiddleware(app.wsgi_app)
     state = AppState(db_url or C.DATABASE_URL, device=device)
     app.extensions["audiomuse"] = state
     app.config["AUTH_DISABLED"] = auth_disabled

@@ -50,9 +50,31 @@ def verify_password(password: str, stored: str) -> bool:
 
 
 def _secret() -> bytes:
+    """Session-signing secret. Priority: env/config > cached > persisted.
+
+    A generated secret is persisted to app_config on first boot (key
+    ``_jwt_secret``, underscore = internal, never surfaced via
+    /api/config) so sessions survive restarts and every WSGI worker
+    signs/verifies with the same key (a per-process random
+    secret broke multi-process deployments).
+    """
     s = C.JWT_SECRET or current_app.config.get("JWT_SECRET", "")
     if not s:
-        s = current_app.config["JWT_SECRET"] = secrets.token_hex(32)
+        from audiomuse_amd.db.store import get_app_config
+        conn = current_app.extensions["audiomuse"].conn()
+        s = get_app_config(conn).get("_jwt_secret", "")
+        if not s:
+            with write_txn(conn):
+                conn.execute(
+                    "INSERT INTO app_config (key, value) VALUES (?,?) "
+                    "ON CONFLICT(key) DO NOTHING",
+                    ("_jwt_secret", secrets.token_hex(32)))
+            s = get_app_config(conn)["_jwt_secret"]  # racing boots converge
+            import logging
+            logging.getLogger(__name__).warning(
+                "AUDIOMUSE_JWT_SECRET unset; generated one and persisted "
+                "it to app_config (set the env var to rotate)")
+        current_app.config["JWT_SECRET"] = s
     return s.encode()
 
 

@@ -337,12 +337,28 @@ def test_lyrics_axes_similar(client_ids):
         f"/api/lyrics_axes_similar?item_id={ids[1]}").status_code == 404
 
 
-def test_proxy_prefix_middleware(client_ids):
-    client, _ = client_ids
+def test_proxy_prefix_middleware(tmp_path, monkeypatch):
+    from audiomuse_amd import config as C
+    from audiomuse_amd.web.app import create_app
+
+    monkeypatch.setattr(C, "BEHIND_PROXY", True)
+    app = create_app(db_url=f"sqlite:///{tmp_path}/proxy.db",
+                     auth_disabled=True)
+    client = app.test_client()
     r = client.get("/am/health", headers={"X-Forwarded-Prefix": "/am"})
     assert r.status_code == 200 and r.json["status"] == "ok"
     # without the header the prefixed path does not exist
     assert client.get("/am/health").status_code == 404
+
+
+def test_proxy_prefix_headers_ignored_by_default(client_ids):
+    # Not behind a proxy (default): forwarded headers are untrusted and
+    # must not rewrite SCRIPT_NAME/scheme.
+    client, _ = client_ids
+    assert client.get("/am/health",
+                      headers={"X-Forwarded-Prefix": "/am"}).status_code == 404
+    r = client.get("/health", headers={"X-Forwarded-Prefix": "/am"})
+    assert r.status_code == 200
 
 
 def test_external_api(client_ids):
