@@ -191,13 +191,75 @@ def sonic_fingerprint_task(ctx: TaskContext, payload: Dict) -> Dict:
 
 # -- backup / restore (app_backup.py analog) --------------------------------
 
-def backup_database(conn: sqlite3.Connection, dest_path: str) -> None:
+def _schema_tables() -> list:
+    """Table names from the canonical DDL (one catalogue, both backends)."""
+    import re
+
+    from audiomuse_amd.db.schema import DDL
+    return re.findall(r"CREATE TABLE IF NOT EXISTS (\w+)", DDL)
+
+
+def backup_database(conn, dest_path: str) -> None:
+    """Portable logical backup: the artifact is always an SQLite file.
+
+    SQLite backend: the online-backup API snapshots consistently on its
+    own. PostgreSQL backend: every table is streamed into a
+    canonical-schema SQLite file inside one REPEATABLE READ snapshot
+    (the stand-in for the reference's pg_dump, app_backup.py:607 — no
+    pg_dump binary ships in this image)."""
+    from audiomuse_amd.db import backend_kind
+    if backend_kind(conn) == "sqlite":
+        dest = sqlite3.connect(dest_path)
+        try:
+            conn.backup(dest)
+        finally:
+            dest.close()
+        return
+    from audiomuse_amd.db.schema import init_db
     dest = sqlite3.connect(dest_path)
     try:
-        # sqlite's online-backup API snapshots consistently on its own
-        conn.backup(dest)
+        init_db(dest)
+        conn.execute("BEGIN ISOLATION LEVEL REPEATABLE READ")
+        try:
+            for table in _schema_tables():
+                rows = conn.execute(f"SELECT * FROM {table}").fetchall()
+                if not rows:
+                    continue
+                cols = rows[0].keys()
+                marks = ",".join("?" for _ in cols)
+                dest.executemany(
+                    f"INSERT INTO {table} ({','.join(cols)}) "
+                    f"VALUES ({marks})",
+                    [tuple(r) for r in rows])
+        finally:
+            conn.execute("COMMIT")
+        dest.commit()
     finally:
         dest.close()
+
+
+def restore_database(conn, src_sqlite_path: str) -> None:
+    """Inverse of backup_database for the PostgreSQL backend: truncate
+    and reload every table from the SQLite backup artifact in one
+    transaction."""
+    src = sqlite3.connect(src_sqlite_path)
+    src.row_factory = sqlite3.Row
+    try:
+        with write_txn(conn):
+            for table in reversed(_schema_tables()):  # FK-safe delete order
+                conn.execute(f"DELETE FROM {table}")
+            for table in _schema_tables():
+                rows = src.execute(f"SELECT * FROM {table}").fetchall()
+                if not rows:
+                    continue
+                cols = rows[0].keys()
+                marks = ",".join("?" for _ in cols)
+                for r in rows:
+                    conn.execute(
+                        f"INSERT INTO {table} ({','.join(cols)}) "
+                        f"VALUES ({marks})", tuple(r))
+    finally:
+        src.close()
 
 
 def refresh_dashboard_stats(conn: sqlite3.Connection) -> Dict[str, int]:

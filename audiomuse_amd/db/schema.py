@@ -181,5 +181,27 @@ CREATE TABLE IF NOT EXISTS control_ack (
 """
 
 
-def init_db(conn: sqlite3.Connection) -> None:
-    conn.executescript(DDL)
+def to_postgres(ddl: str) -> str:
+    """Derive the PostgreSQL DDL from the canonical (SQLite-typed) DDL.
+
+    The reference's schema is native PG (database.py:1286-1651); here the
+    translation is mechanical so both backends share one table catalogue
+    and the conventions test can diff them column by column.
+    """
+    out = ddl.replace("INTEGER PRIMARY KEY AUTOINCREMENT",
+                      "BIGSERIAL PRIMARY KEY")
+    out = out.replace("BLOB", "BYTEA")
+    out = out.replace("REAL", "DOUBLE PRECISION")
+    out = out.replace("(julianday('now') - 2440587.5) * 86400.0",
+                      "EXTRACT(EPOCH FROM now())")
+    return out
+
+
+DDL_PG = to_postgres(DDL)
+
+
+def init_db(conn) -> None:
+    if getattr(conn, "kind", "sqlite") == "postgres":
+        conn.executescript(DDL_PG)
+    else:
+        conn.executescript(DDL)

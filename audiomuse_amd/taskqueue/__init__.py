@@ -26,7 +26,7 @@ import uuid
 from typing import Any, Dict, Optional
 
 from audiomuse_amd import config as C
-from audiomuse_amd.db import write_txn
+from audiomuse_amd.db import CHAN_CANCEL, CHAN_JOB, notify, write_txn
 
 QUEUE_HIGH = "high"
 QUEUE_DEFAULT = "default"
@@ -68,6 +68,9 @@ def enqueue(conn: sqlite3.Connection, task_type: str,
             (task_id, task_type, parent_task_id, queue, PENDING, priority,
              json.dumps(payload or {}), token,
              max_attempts if max_attempts is not None else C.QUEUE_MAX_ATTEMPTS))
+    # wake idle workers (PG NOTIFY, reference sql.py:399; no-op on SQLite
+    # where workers poll)
+    notify(conn, CHAN_JOB, task_id)
     return task_id
 
 
@@ -113,4 +116,5 @@ def cancel_task_recursive(conn: sqlite3.Connection, task_id: str) -> int:
                 "SELECT task_id FROM task_status WHERE parent_task_id=?",
                 (tid,)).fetchall()
             frontier.extend(k["task_id"] for k in kids)
+    notify(conn, CHAN_CANCEL, task_id)
     return n

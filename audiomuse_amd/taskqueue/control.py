@@ -26,12 +26,14 @@ def publish_control_request(conn: sqlite3.Connection, action: str,
                             window_seconds: float = DEFAULT_WINDOW_SECONDS
                             ) -> int:
     """reference: control.publish_control_request :115"""
+    from audiomuse_amd.db import insert_returning_id
     with write_txn(conn):
-        cur = conn.execute(
+        rid = insert_returning_id(
+            conn,
             "INSERT INTO control_request (action, payload, expires_at) "
             "VALUES (?,?,?)",
             (action, json.dumps(payload or {}), time.time() + window_seconds))
-    return int(cur.lastrowid)
+    return rid
 
 
 def pending_requests(conn: sqlite3.Connection, listener: str,

@@ -1,4 +1,15 @@
-"""Queue SQL operations (reference: taskqueue/sql.py)."""
+"""Queue SQL operations (reference: taskqueue/sql.py).
+
+Liveness model by backend:
+
+- PostgreSQL: the claim is ``FOR UPDATE SKIP LOCKED`` (reference
+  sql.py:415-430) and the worker connection takes a per-task advisory
+  lock (sql.py:452-462) that dies with the connection — the primary
+  orphan signal. Lease expiry remains as a belt-and-braces fallback for
+  a wedged-but-connected worker.
+- SQLite: the claim is one exclusive write transaction; lease +
+  heartbeat is the only liveness signal.
+"""
 
 from __future__ import annotations
 
@@ -8,32 +19,92 @@ import time
 from typing import Optional, Sequence
 
 from audiomuse_amd import config as C
-from audiomuse_amd.db import write_txn
+from audiomuse_amd.db import backend_kind, write_txn
 from audiomuse_amd.taskqueue import (FAILURE, PENDING, REVOKED, RUNNING,
                                      SUCCESS)
 
+# advisory-lock key namespace: (class, hashtext(task_id)); class 1=task,
+# 2=maintenance election, 3=admission gate, 4=cron minute
+LOCK_CLASS_TASK = 1
+LOCK_CLASS_MAINT = 2
+LOCK_CLASS_ADMIT = 3
+LOCK_CLASS_CRON = 4
 
-def claim(conn: sqlite3.Connection, worker_id: str,
+
+def try_advisory_lock(conn, lock_class: int, key: str) -> bool:
+    """PG: session advisory lock (dies with the connection). SQLite:
+    vacuously True — liveness is lease-based there. Held keys are
+    tracked on the connection because PG advisory locks are
+    session-reentrant: a liveness probe must know which locks ITS OWN
+    session already owns (those tasks are trivially alive)."""
+    if backend_kind(conn) != "postgres":
+        return True
+    row = conn.execute(
+        "SELECT pg_try_advisory_lock(?, hashtext(?)) AS got",
+        (lock_class, key)).fetchone()
+    if bool(row["got"]):
+        held = getattr(conn, "_held_advisory", None)
+        if held is None:
+            held = conn._held_advisory = set()
+        held.add((lock_class, key))
+        return True
+    return False
+
+
+def holds_advisory_lock(conn, lock_class: int, key: str) -> bool:
+    return (lock_class, key) in getattr(conn, "_held_advisory", ())
+
+
+def advisory_unlock(conn, lock_class: int, key: str) -> None:
+    if backend_kind(conn) != "postgres":
+        return
+    conn.execute("SELECT pg_advisory_unlock(?, hashtext(?))",
+                 (lock_class, key))
+    getattr(conn, "_held_advisory", set()).discard((lock_class, key))
+
+
+def claim(conn, worker_id: str,
           queues: Sequence[str] = ("high", "default"),
           lease_seconds: Optional[float] = None) -> Optional[sqlite3.Row]:
-    """Atomically claim the next PENDING job (reference: sql.py:430,
-    FOR UPDATE SKIP LOCKED; here one exclusive write txn)."""
+    """Atomically claim the next PENDING job.
+
+    PG: FOR UPDATE SKIP LOCKED subselect + UPDATE ... RETURNING, then the
+    advisory liveness lock (reference: sql.py:430,452). SQLite: one
+    exclusive write txn.
+    """
     lease = lease_seconds if lease_seconds is not None else C.QUEUE_LEASE_SECONDS
     now = time.time()
     qmarks = ",".join("?" for _ in queues)
-    with write_txn(conn):
-        row = conn.execute(
-            f"""SELECT task_id FROM task_status
-                WHERE status = ? AND queue IN ({qmarks})
-                ORDER BY priority DESC, created_at LIMIT 1""",
-            (PENDING, *queues)).fetchone()
+    if backend_kind(conn) == "postgres":
+        with write_txn(conn):
+            row = conn.execute(
+                f"""UPDATE task_status SET status=?, worker_id=?,
+                        lease_expires=?, started_at=?, attempts=attempts+1
+                    WHERE task_id = (
+                        SELECT task_id FROM task_status
+                        WHERE status = ? AND queue IN ({qmarks})
+                        ORDER BY priority DESC, created_at
+                        FOR UPDATE SKIP LOCKED LIMIT 1)
+                    RETURNING task_id""",
+                (RUNNING, worker_id, now + lease, now, PENDING,
+                 *queues)).fetchone()
         if row is None:
             return None
-        conn.execute(
-            """UPDATE task_status SET status=?, worker_id=?,
-                   lease_expires=?, started_at=?, attempts=attempts+1
-               WHERE task_id=?""",
-            (RUNNING, worker_id, now + lease, now, row["task_id"]))
+        try_advisory_lock(conn, LOCK_CLASS_TASK, row["task_id"])
+    else:
+        with write_txn(conn):
+            row = conn.execute(
+                f"""SELECT task_id FROM task_status
+                    WHERE status = ? AND queue IN ({qmarks})
+                    ORDER BY priority DESC, created_at LIMIT 1""",
+                (PENDING, *queues)).fetchone()
+            if row is None:
+                return None
+            conn.execute(
+                """UPDATE task_status SET status=?, worker_id=?,
+                       lease_expires=?, started_at=?, attempts=attempts+1
+                   WHERE task_id=?""",
+                (RUNNING, worker_id, now + lease, now, row["task_id"]))
     return conn.execute("SELECT * FROM task_status WHERE task_id=?",
                         (row["task_id"],)).fetchone()
 
@@ -73,6 +144,8 @@ def finish(conn: sqlite3.Connection, task_id: str, worker_id: str,
                 "VALUES (?,?,?,?)",
                 (task_id, row["task_type"] if row else "", status,
                  (json.dumps(result)[:500] if result else None)))
+    if cur.rowcount == 1:
+        advisory_unlock(conn, LOCK_CLASS_TASK, task_id)
     return cur.rowcount == 1
 
 
@@ -104,18 +177,49 @@ def is_cancelled(conn: sqlite3.Connection, task_id: str) -> bool:
     return False
 
 
-def reclaim_orphans(conn: sqlite3.Connection) -> int:
+def _orphaned_running_rows(conn, now: float) -> list:
+    """RUNNING rows whose worker is provably gone.
+
+    SQLite: lease expired. PG: lease expired OR the per-task advisory
+    lock is free (the owner's connection died — reference
+    maintenance.py:177: orphan = RUNNING and lock acquirable). A lock
+    probed free is released immediately; the row decides the outcome.
+    """
+    rows = conn.execute(
+        """SELECT task_id, attempts, max_attempts, lease_expires
+           FROM task_status WHERE status=?""", (RUNNING,)).fetchall()
+    out = []
+    is_pg = backend_kind(conn) == "postgres"
+    for row in rows:
+        expired = (row["lease_expires"] is not None
+                   and row["lease_expires"] < now)
+        if expired:
+            out.append(row)
+        elif (is_pg
+              and not holds_advisory_lock(conn, LOCK_CLASS_TASK,
+                                          row["task_id"])  # ours = alive
+              and try_advisory_lock(conn, LOCK_CLASS_TASK,
+                                    row["task_id"])):
+            advisory_unlock(conn, LOCK_CLASS_TASK, row["task_id"])
+            out.append(row)
+    return out
+
+
+def reclaim_orphans(conn) -> int:
     """Maintenance pass (reference: maintenance.py:177 + sql.py:579-641):
-    RUNNING + expired lease => re-PENDING while attempts remain, else
-    FAILURE for good."""
+    orphaned RUNNING rows => re-PENDING while attempts remain, else
+    FAILURE for good. Only one node runs this at a time on PG (the
+    maintenance-election advisory lock, reference sql.py:472)."""
     now = time.time()
     n = 0
+    if not try_advisory_lock(conn, LOCK_CLASS_MAINT, "maintenance"):
+        return 0
+    try:
+        orphans = _orphaned_running_rows(conn, now)
+    finally:
+        advisory_unlock(conn, LOCK_CLASS_MAINT, "maintenance")
     with write_txn(conn):
-        rows = conn.execute(
-            """SELECT task_id, attempts, max_attempts FROM task_status
-               WHERE status=? AND lease_expires IS NOT NULL AND lease_expires < ?""",
-            (RUNNING, now)).fetchall()
-        for row in rows:
+        for row in orphans:
             if row["attempts"] >= row["max_attempts"]:
                 conn.execute(
                     """UPDATE task_status SET status=?, worker_id=NULL,

@@ -141,6 +141,13 @@ class Worker:
         conn = connect(self.db_url)
         from audiomuse_amd.db.schema import init_db
         init_db(conn)
+        # PG: park on LISTEN channels instead of polling (reference:
+        # worker.py:197 + listen.py); SQLite: bounded polling.
+        from audiomuse_amd import db as dbmod
+
+        dbmod.listen(conn, dbmod.CHAN_JOB)
+        dbmod.listen(conn, dbmod.CHAN_CANCEL)
+        dbmod.listen(conn, dbmod.CHAN_CONTROL)
         try:
             import_builtin_handlers()
         except Exception:
@@ -188,7 +195,10 @@ class Worker:
                     continue
                 if idle_timeout is not None and time.time() - idle_since > idle_timeout:
                     return
-                self._stop.wait(self.poll_seconds)
+                if getattr(conn, "kind", "sqlite") == "postgres":
+                    conn.wait_notify(self.poll_seconds)  # NOTIFY wakes us early
+                else:
+                    self._stop.wait(self.poll_seconds)
         finally:
             conn.close()
 

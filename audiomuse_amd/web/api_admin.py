@@ -56,12 +56,31 @@ def restore():
     blob = request.get_data()
     if not blob.startswith(b"SQLite format 3"):
         return jsonify({"error": "not an SQLite backup"}), 400
+    import sqlite3 as s3
+
+    from audiomuse_amd.db import backend_kind
+    if backend_kind(state.db_url) == "postgres":
+        # logical reload through the live connection (pg_dump analog)
+        fd, tmp = tempfile.mkstemp(suffix=".restore.db")
+        os.close(fd)
+        try:
+            with open(tmp, "wb") as fh:
+                fh.write(blob)
+            check = s3.connect(tmp)
+            check.execute("SELECT COUNT(*) FROM score")
+            check.close()
+            from audiomuse_amd.analysis.maintenance import restore_database
+            restore_database(conn, tmp)
+        except Exception as exc:  # noqa: BLE001
+            return jsonify({"error": f"backup failed validation: {exc}"}), 400
+        finally:
+            os.unlink(tmp)
+        state.invalidate()
+        return jsonify({"restored": True})
     db_path = state.db_url[len("sqlite:///"):]
     tmp = db_path + ".restore"
     with open(tmp, "wb") as fh:
         fh.write(blob)
-    import sqlite3 as s3
-
     try:
         check = s3.connect(tmp)
         check.execute("SELECT COUNT(*) FROM score")

@@ -260,14 +260,16 @@ def cron_add():
     if not validate_cron(schedule):
         return jsonify({"error": f"invalid cron expression {schedule!r}"}), 400
     conn = _state().conn()
+    from audiomuse_amd.db import insert_returning_id
     with write_txn(conn):
-        cur = conn.execute(
+        rid = insert_returning_id(
+            conn,
             "INSERT INTO cron (name, schedule, task_type, payload, enabled) "
             "VALUES (?,?,?,?,1)",
             (body.get("name", ""), schedule,
              body.get("task_type", "rebuild_indexes"),
              json.dumps(body.get("payload", {}))))
-    return jsonify({"id": cur.lastrowid})
+    return jsonify({"id": rid})
 
 
 @bp.delete("/api/cron/<int:cron_id>")

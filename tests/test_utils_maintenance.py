@@ -137,7 +137,7 @@ def test_dashboard_stats(db):
     assert row["value"] == "5"
 
 
-def test_catalogue_never_deleted_by_cleaning(db):
+def test_catalogue_never_deleted_by_cleaning(db, tmp_db_url):
     """The reference's guarded invariant (test_catalogue_is_never_deleted):
     cleaning removes mappings only."""
     import json
@@ -152,7 +152,7 @@ def test_catalogue_never_deleted_by_cleaning(db):
         db.execute(
             """INSERT INTO track_server_map (provider_id, server_id, item_id)
                VALUES ('orphan', 'srv1', 'fp_4dead')""")
-    url = "sqlite:///" + db.execute("PRAGMA database_list").fetchone()[2]
+    url = tmp_db_url
     tid = enqueue(db, "clean_orphans", {"delete": True})
     Worker(db_url=url, max_jobs=1).run_forever(idle_timeout=3.0)
     row = task_row(db, tid)
