@@ -47,10 +47,11 @@ def _env_bool(name: str, default: bool) -> bool:
 # Storage / control plane
 # --------------------------------------------------------------------------
 # The reference uses PostgreSQL as both store and queue (database.py,
-# taskqueue/sql.py). This build ships an SQLite backend with the same
-# schema and queue semantics (claim, lease-liveness, cooperative cancel);
-# DATABASE_URL switches backends ("sqlite:///path" in-tree; "postgresql://"
-# reserved for a psycopg backend where available).
+# taskqueue/sql.py). DATABASE_URL switches backends: "postgresql://..."
+# runs the deployment contract (first-party wire driver, SKIP LOCKED
+# claims, advisory-lock liveness, LISTEN/NOTIFY — docs/POSTGRES.md);
+# "sqlite:///path" is the zero-dependency single-box mode with the same
+# schema and observable queue semantics.
 DATABASE_URL = _env("DATABASE_URL", "sqlite:///" + os.path.join(
     os.environ.get("AUDIOMUSE_DATA_DIR", os.path.expanduser("~/.audiomuse-amd")),
     "audiomuse.db"))
@@ -64,6 +65,17 @@ QUEUE_POLL_SECONDS = _env_float("QUEUE_POLL_SECONDS", 0.25)
 MAX_QUEUED_ANALYSIS_JOBS = _env_int("MAX_QUEUED_ANALYSIS_JOBS", 30)
 REBUILD_INDEX_BATCH_SIZE = _env_int("REBUILD_INDEX_BATCH_SIZE", 500)
 WORKER_MAX_JOBS = _env_int("WORKER_MAX_JOBS", 50)
+
+# Media-server HTTP behavior (reference: per-provider request helpers,
+# navidrome.py:161-220 / jellyfin.py:310-359; centralised here in
+# mediaserver/http.py)
+MEDIASERVER_TIMEOUT_SECONDS = _env_float("MEDIASERVER_TIMEOUT_SECONDS", 30.0)
+MEDIASERVER_DOWNLOAD_TIMEOUT_SECONDS = _env_float(
+    "MEDIASERVER_DOWNLOAD_TIMEOUT_SECONDS", 300.0)
+MEDIASERVER_RETRIES = _env_int("MEDIASERVER_RETRIES", 3)
+MEDIASERVER_RETRY_BACKOFF_SECONDS = _env_float(
+    "MEDIASERVER_RETRY_BACKOFF_SECONDS", 1.0)
+MEDIASERVER_PAGE_SIZE = _env_int("MEDIASERVER_PAGE_SIZE", 500)
 
 # --------------------------------------------------------------------------
 # Audio front-end

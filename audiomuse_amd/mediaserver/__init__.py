@@ -75,5 +75,6 @@ def active_server() -> Optional[BoundServer]:
 from audiomuse_amd.mediaserver import synthetic  # noqa: E402,F401
 from audiomuse_amd.mediaserver import subsonic  # noqa: E402,F401
 from audiomuse_amd.mediaserver import jellyfin  # noqa: E402,F401
+from audiomuse_amd.mediaserver import emby  # noqa: E402,F401
 from audiomuse_amd.mediaserver import plex  # noqa: E402,F401
 from audiomuse_amd.mediaserver import lyrion  # noqa: E402,F401

