@@ -40,10 +40,35 @@ _DEFAULT_MODEL = {
 
 
 def _base_and_key(provider: str) -> tuple:
-    base = (C.AI_BASE_URL or _DEFAULT_BASE[provider]).rstrip("/")
-    key = C.AI_API_KEY
+    """Per-vendor URL/key config (reference providers/* +
+    PARAMETERS.md: OPENAI_SERVER_URL/OLLAMA_SERVER_URL/*_API_KEY);
+    AI_BASE_URL / AI_API_KEY are the generic overrides."""
+    vendor_base = {
+        "openai": C.OPENAI_SERVER_URL,
+        "ollama": C.OLLAMA_SERVER_URL,
+        "mistral": _DEFAULT_BASE["mistral"],
+        "gemini": _DEFAULT_BASE["gemini"],
+    }.get(provider) or _DEFAULT_BASE.get(provider, "")
+    vendor_key = {
+        "openai": C.OPENAI_API_KEY,
+        "gemini": C.GEMINI_API_KEY,
+        "mistral": C.MISTRAL_API_KEY,
+    }.get(provider, "")
+    base = (C.AI_BASE_URL or vendor_base).rstrip("/")
+    key = C.AI_API_KEY or vendor_key
     validate_outbound_url(base)
     return base, key
+
+
+def _model_for(provider: str) -> str:
+    vendor_model = {
+        "openai": C.OPENAI_MODEL_NAME,
+        "ollama": C.OLLAMA_MODEL_NAME,
+        "gemini": C.GEMINI_MODEL_NAME,
+        "mistral": C.MISTRAL_MODEL_NAME,
+    }.get(provider, "")
+    return C.AI_MODEL_NAME or vendor_model or _DEFAULT_MODEL.get(
+        provider, "")
 
 
 def build_openai_request(provider: str, prompt: str,
@@ -51,9 +76,12 @@ def build_openai_request(provider: str, prompt: str,
     """OpenAI-compatible chat/completions body (openai.py + mistral.py:
     Mistral's chat API is OpenAI-wire-compatible)."""
     return {
-        "model": C.AI_MODEL_NAME or _DEFAULT_MODEL[provider],
+        "model": _model_for(provider),
         "messages": [{"role": "system", "content": _SYSTEM},
                      {"role": "user", "content": prompt}],
+        "temperature": C.AI_TOOLCALL_TEMPERATURE,
+        "top_p": C.AI_TOOLCALL_TOP_P,
+        "max_tokens": C.AI_TOOLCALL_NUM_PREDICT,
         "tools": [{"type": "function",
                    "function": {"name": name,
                                 "parameters": schema or {"type": "object"}}}
@@ -108,7 +136,7 @@ def plan_with_llm(prompt: str, tools: Dict[str, Dict],
     provider = (provider or C.AI_PROVIDER or "none").lower()
     if provider in ("", "none"):
         return None
-    if provider not in _DEFAULT_BASE:
+    if provider not in _DEFAULT_BASE and provider != "ollama":
         logger.warning("unknown AI provider %r", provider)
         return None
     try:
@@ -118,16 +146,17 @@ def plan_with_llm(prompt: str, tools: Dict[str, Dict],
             post = requests.post
         base, key = _base_and_key(provider)
         if provider == "gemini":
-            model = C.AI_MODEL_NAME or _DEFAULT_MODEL[provider]
+            model = _model_for(provider)
             r = post(f"{base}/models/{model}:generateContent",
                      headers={"x-goog-api-key": key},
-                     json=build_gemini_request(prompt, tools), timeout=30)
+                     json=build_gemini_request(prompt, tools),
+                     timeout=C.AI_REQUEST_TIMEOUT_SECONDS)
             r.raise_for_status()
             return parse_gemini_response(r.json())
         r = post(f"{base}/chat/completions",
                  headers={"Authorization": f"Bearer {key}"},
                  json=build_openai_request(provider, prompt, tools),
-                 timeout=30)
+                 timeout=C.AI_REQUEST_TIMEOUT_SECONDS)
         r.raise_for_status()
         return parse_openai_response(r.json())
     except Exception:

@@ -22,11 +22,10 @@ import sqlite3
 import time
 from typing import Dict, List
 
+from audiomuse_amd import config as C
 from audiomuse_amd.db import connect, write_txn
 from audiomuse_amd.mediaserver import make_provider
 from audiomuse_amd.taskqueue.worker import TaskContext, task_handler
-
-SWEEP_PRUNE_MIN_FETCH_RATIO = 0.5
 
 _NOISE = re.compile(r"\s*[\(\[].*?[\)\]]\s*|\s*(feat\.|ft\.)\s.*$|[^\w\s]",
                     re.IGNORECASE)
@@ -98,7 +97,7 @@ def multiserver_sync_task(ctx: TaskContext, payload: Dict) -> Dict:
         "SELECT COUNT(*) AS n FROM track_server_map WHERE server_id=?",
         (server_id,)).fetchone()["n"]
     pruned = 0
-    if mapped and len(tracks) / mapped >= SWEEP_PRUNE_MIN_FETCH_RATIO:
+    if mapped and len(tracks) / mapped >= C.SWEEP_PRUNE_MIN_FETCH_RATIO:
         live = {t.provider_id for t in tracks}
         rows = conn.execute(
             "SELECT provider_id FROM track_server_map WHERE server_id=?",
@@ -129,7 +128,10 @@ def clean_orphans_task(ctx: TaskContext, payload: Dict) -> Dict:
         # triple guard: explicit flag + bounded fraction + re-check
         total = conn.execute(
             "SELECT COUNT(*) AS n FROM track_server_map").fetchone()["n"]
-        if total and len(orphans) / total <= payload.get("max_fraction", 0.2):
+        # triple guard part 2: bounded fraction AND absolute safety cap
+        # (reference CLEANING_SAFETY_LIMIT)
+        if (total and len(orphans) / total <= payload.get("max_fraction", 0.2)
+                and len(orphans) <= C.CLEANING_SAFETY_LIMIT):
             with write_txn(conn):
                 for r in orphans:
                     conn.execute(
@@ -160,7 +162,8 @@ def sonic_fingerprint_task(ctx: TaskContext, payload: Dict) -> Dict:
     eng = load_ivf_engine(conn, AUDIO_INDEX)
     if eng is None:
         return {"error": "audio index not built"}
-    top = provider.get_top_played_songs(payload.get("top_n", 100))
+    top = provider.get_top_played_songs(
+        payload.get("top_n", C.SONIC_FINGERPRINT_TOP_PLAYED))
     mapped = {r["provider_id"]: r["item_id"] for r in conn.execute(
         "SELECT provider_id, item_id FROM track_server_map WHERE server_id=?",
         (server_id,))}

@@ -50,25 +50,38 @@ class IterationResult:
 
 
 def _param_space(algorithm: str, n: int, rng: random.Random) -> Dict:
+    """Explore ranges come from config (reference PARAMETERS.md:
+    NUM_CLUSTERS_MIN/MAX, DBSCAN_*, GMM_*, SPECTRAL_*, PCA_*)."""
     if algorithm == "kmeans":
-        return {"n_clusters": rng.randint(2, max(3, min(50, n // 10)))}
+        hi = max(C.NUM_CLUSTERS_MIN + 1, min(C.NUM_CLUSTERS_MAX, n // 10))
+        return {"n_clusters": rng.randint(C.NUM_CLUSTERS_MIN, hi)}
     if algorithm == "dbscan":
-        return {"eps": rng.uniform(0.2, 2.5), "min_samples": rng.randint(3, 15)}
+        return {"eps": rng.uniform(C.DBSCAN_EPS_MIN, C.DBSCAN_EPS_MAX),
+                "min_samples": rng.randint(C.DBSCAN_MIN_SAMPLES_MIN,
+                                           C.DBSCAN_MIN_SAMPLES_MAX)}
     if algorithm == "gmm":
-        return {"n_components": rng.randint(2, max(3, min(40, n // 10)))}
+        hi = max(C.GMM_N_COMPONENTS_MIN + 1,
+                 min(C.GMM_N_COMPONENTS_MAX, n // 10))
+        return {"n_components": rng.randint(C.GMM_N_COMPONENTS_MIN, hi)}
     if algorithm == "spectral":
-        return {"n_clusters": rng.randint(2, max(3, min(30, n // 20)))}
+        hi = max(C.SPECTRAL_N_CLUSTERS_MIN + 1,
+                 min(C.SPECTRAL_N_CLUSTERS_MAX, n // 20))
+        return {"n_clusters": rng.randint(C.SPECTRAL_N_CLUSTERS_MIN, hi)}
     raise ValueError(algorithm)
 
 
 def _mutate(params: Dict, algorithm: str, n: int, rng: random.Random) -> Dict:
+    """Elite mutation, deltas from config (reference PARAMETERS.md:
+    MUTATION_INT_ABS_DELTA / MUTATION_FLOAT_ABS_DELTA)."""
     p = dict(params)
+    di = C.MUTATION_INT_ABS_DELTA
+    df = C.MUTATION_FLOAT_ABS_DELTA
     if algorithm in ("kmeans", "spectral"):
-        p["n_clusters"] = max(2, p["n_clusters"] + rng.randint(-3, 3))
+        p["n_clusters"] = max(2, p["n_clusters"] + rng.randint(-di, di))
     elif algorithm == "gmm":
-        p["n_components"] = max(2, p["n_components"] + rng.randint(-3, 3))
+        p["n_components"] = max(2, p["n_components"] + rng.randint(-di, di))
     else:
-        p["eps"] = max(0.05, p["eps"] * rng.uniform(0.8, 1.25))
+        p["eps"] = max(0.05, p["eps"] * rng.uniform(1.0 - df, 1.0 + df))
         p["min_samples"] = max(2, p["min_samples"] + rng.randint(-2, 2))
     return p
 
@@ -154,15 +167,22 @@ def run_iteration(x: torch.Tensor, rows: Sequence[TrackRow], algorithm: str,
 
 def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
                         algorithm: Optional[str] = None, *,
-                        runs: Optional[int] = None, elite_size: int = 5,
-                        exploit_prob: float = 0.6, stall_limit: int = 30,
+                        runs: Optional[int] = None,
+                        elite_size: Optional[int] = None,
+                        exploit_prob: Optional[float] = None,
+                        stall_limit: int = 30,
                         seed: int = 0, subset: Optional[int] = None,
-                        max_songs_per_cluster: int = 0,
+                        max_songs_per_cluster: Optional[int] = None,
                         progress_cb=None) -> List[IterationResult]:
     """Explore/exploit search with an elite pool and a stall valve
     (clustering.py:383-1449). Returns elites sorted best-first."""
     algorithm = algorithm or C.CLUSTER_ALGORITHM
     runs = runs or C.CLUSTERING_RUNS
+    elite_size = elite_size if elite_size is not None else C.TOP_N_ELITES
+    exploit_prob = (exploit_prob if exploit_prob is not None
+                    else C.EXPLOITATION_PROBABILITY_CONFIG)
+    if max_songs_per_cluster is None:
+        max_songs_per_cluster = C.MAX_SONGS_PER_CLUSTER
     rng = random.Random(seed)
     n = x.shape[0]
     sub = min(subset or C.CLUSTERING_SUBSET_SONGS, n)
@@ -175,7 +195,8 @@ def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
             rs = [rows[i] for i in pick.tolist()]
         else:
             xs, rs = x, rows
-        if elites and rng.random() < exploit_prob:
+        exploit_allowed = it >= runs * C.EXPLOITATION_START_FRACTION
+        if elites and exploit_allowed and rng.random() < exploit_prob:
             params = _mutate(rng.choice(elites).params, algorithm, sub, rng)
         else:
             params = _param_space(algorithm, sub, rng)
@@ -198,11 +219,13 @@ def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
 
 
 def diverse_top_n(elite: IterationResult, n: Optional[int] = None,
-                  min_size: int = 3) -> Dict[str, List[str]]:
+                  min_size: Optional[int] = None) -> Dict[str, List[str]]:
     """Winner post-processing (clustering_postprocessing.py:66): drop tiny
     playlists, then pick Top-N diverse by centroid max-min distance
     (the reference's "6+4": half largest, half most diverse)."""
     n = n or C.TOP_N_PLAYLISTS
+    min_size = (min_size if min_size is not None
+                else C.MIN_PLAYLIST_SIZE_FOR_TOP_N)
     items = [(name, ids) for name, ids in elite.playlists.items()
              if len(ids) >= min_size]
     if len(items) <= n:

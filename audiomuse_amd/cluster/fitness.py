@@ -115,11 +115,20 @@ def purity_diversity(score_vectors: torch.Tensor, labels: torch.Tensor
     }
 
 
-_DEFAULT_WEIGHTS = {
-    "silhouette": 1.0, "davies_bouldin": 1.0, "calinski_harabasz": 1.0,
-    "mood_purity": 1.0, "mood_diversity": 1.0,
-    "other_purity": 1.0, "other_diversity": 1.0,
-}
+def _default_weights() -> Dict[str, float]:
+    """Config-driven fitness weights (reference PARAMETERS.md
+    SCORE_WEIGHT_*; clustering_helper.py:689). Read at call time so DB
+    overrides apply without a restart."""
+    from audiomuse_amd import config as C
+    return {
+        "silhouette": C.SCORE_WEIGHT_SILHOUETTE,
+        "davies_bouldin": C.SCORE_WEIGHT_DAVIES_BOULDIN,
+        "calinski_harabasz": C.SCORE_WEIGHT_CALINSKI_HARABASZ,
+        "mood_purity": C.SCORE_WEIGHT_PURITY,
+        "mood_diversity": C.SCORE_WEIGHT_DIVERSITY,
+        "other_purity": C.SCORE_WEIGHT_OTHER_FEATURE_PURITY,
+        "other_diversity": C.SCORE_WEIGHT_OTHER_FEATURE_DIVERSITY,
+    }
 
 
 def fitness(x_metrics: torch.Tensor, labels: torch.Tensor,
@@ -128,9 +137,20 @@ def fitness(x_metrics: torch.Tensor, labels: torch.Tensor,
             weights: Optional[Dict[str, float]] = None) -> Dict[str, float]:
     """Composite fitness (reference combines with log1p + z-ish scaling:
     raw metrics are squashed to comparable ranges before weighting)."""
-    w = dict(_DEFAULT_WEIGHTS)
+    w = _default_weights()
     if weights:
         w.update(weights)
+    # the configured weights may zero every term that is computable for
+    # this call (e.g. geometry-only scoring with the reference defaults,
+    # which put all weight on mood purity/diversity) — fall back to
+    # equal weights over the available terms so fitness stays ordered
+    relevant = ["silhouette", "davies_bouldin", "calinski_harabasz"]
+    if mood_scores is not None:
+        relevant += ["mood_purity", "mood_diversity"]
+    if other_scores is not None:
+        relevant += ["other_purity", "other_diversity"]
+    if all(w.get(k, 0.0) == 0.0 for k in relevant):
+        w = {k: 1.0 for k in w}
     uniq = labels[labels >= 0].unique()
     out: Dict[str, float] = {}
     if uniq.numel() < 2:

@@ -78,12 +78,21 @@ def temperature_sample(results: List[Dict], n: int, temperature: float,
 
 
 def alchemy_query(engine: SimilarityEngine, add: Sequence[np.ndarray],
-                  subtract: Sequence[np.ndarray] = (), *, n: int = 25,
-                  subtract_radius: float = 0.0, temperature: float = 0.0,
+                  subtract: Sequence[np.ndarray] = (), *,
+                  n: Optional[int] = None,
+                  subtract_radius: Optional[float] = None,
+                  temperature: Optional[float] = None,
                   exclude: Sequence[str] = (), seed: Optional[int] = None,
                   **filters) -> List[Dict]:
     """Full alchemy pipeline: combine -> multi-query -> subtract-radius
-    -> temperature sample."""
+    -> temperature sample. Defaults from config (reference ALCHEMY_*)."""
+    from audiomuse_amd import config as C
+    n = min(n if n is not None else C.ALCHEMY_DEFAULT_N_RESULTS,
+            C.ALCHEMY_MAX_N_RESULTS)
+    if subtract_radius is None:
+        subtract_radius = C.ALCHEMY_SUBTRACT_RADIUS if subtract else 0.0
+    if temperature is None:
+        temperature = C.ALCHEMY_TEMPERATURE
     combined = combine_vectors(add, subtract)
     if combined is None:
         return []

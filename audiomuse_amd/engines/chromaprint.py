@@ -15,6 +15,8 @@ import zlib
 from typing import Optional
 
 import numpy as np
+
+from audiomuse_amd import config as C
 import torch
 
 from audiomuse_amd.ops.features import chroma_from_stft
@@ -53,9 +55,14 @@ def _unpack(fp: bytes) -> Optional[np.ndarray]:
     return np.frombuffer(raw, dtype=np.uint32)
 
 
-def bit_match_ratio(fp_a: bytes, fp_b: bytes, max_offset: int = 40) -> float:
+def bit_match_ratio(fp_a: bytes, fp_b: bytes,
+                    max_offset: Optional[int] = None) -> float:
     """Best aligned per-bit agreement over the overlap (the reference's
-    align/overlap/bit-match comparison)."""
+    align/overlap/bit-match comparison). Alignment range and minimum
+    overlap come from config (CHROMAPRINT_ALIGN_RANGE /
+    CHROMAPRINT_MIN_OVERLAP)."""
+    if max_offset is None:
+        max_offset = C.CHROMAPRINT_ALIGN_RANGE // 2
     a = _unpack(fp_a)
     b = _unpack(fp_b)
     if a is None or b is None or a.size == 0 or b.size == 0:
@@ -67,7 +74,7 @@ def bit_match_ratio(fp_a: bytes, fp_b: bytes, max_offset: int = 40) -> float:
         else:
             aa, bb = a, b[-off:]
         n = min(aa.size, bb.size)
-        if n < 8:
+        if n < max(8, C.CHROMAPRINT_MIN_OVERLAP // 8):
             continue
         diff = np.bitwise_xor(aa[:n], bb[:n])
         bits = np.unpackbits(diff.view(np.uint8)).sum()

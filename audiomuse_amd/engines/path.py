@@ -43,9 +43,13 @@ def interpolate(a: np.ndarray, b: np.ndarray, k: int,
 
 
 def find_path(engine: SimilarityEngine, start_id: str, end_id: str,
-              length: int = 12, mode: str = "slerp",
+              length: Optional[int] = None, mode: str = "slerp",
               max_per_artist: Optional[int] = None) -> List[Dict]:
-    """Path of ~`length` tracks from start to end (path_manager entry)."""
+    """Path of ~`length` tracks from start to end (path_manager entry).
+    Default length from config (reference PATH_DEFAULT_LENGTH)."""
+    from audiomuse_amd import config as C
+    if length is None:
+        length = C.PATH_DEFAULT_LENGTH
     va = engine.vector_for_id(start_id)
     vb = engine.vector_for_id(end_id)
     if va is None or vb is None:

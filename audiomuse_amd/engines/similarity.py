@@ -26,7 +26,12 @@ from audiomuse_amd.index.ivf import IVFIndex
 class ResultCache:
     """TTL LRU cache (reference: ivf_manager._ResultCache :73)."""
 
-    def __init__(self, max_items: int = 256, ttl: float = 300.0):
+    def __init__(self, max_items: Optional[int] = None,
+                 ttl: Optional[float] = None):
+        if max_items is None:
+            max_items = C.IVF_RESULT_CACHE_MAX
+        if ttl is None:
+            ttl = C.IVF_RESULT_CACHE_SECONDS
         self.max_items = max_items
         self.ttl = ttl
         self._data: OrderedDict = OrderedDict()
@@ -93,12 +98,14 @@ class SimilarityEngine:
 
     def _apply_filters(self, cands: List[Tuple[str, float]], n: int, *,
                        exclude: Sequence[str] = (),
-                       eliminate_duplicates: bool = False,
+                       eliminate_duplicates: Optional[bool] = None,
                        max_per_artist: Optional[int] = None,
                        mood_filter: Optional[str] = None
                        ) -> List[Tuple[str, float]]:
         """Near-dup lookback filter + artist cap + mood filter
         (ivf_manager.py:419-502, 652, 935)."""
+        if eliminate_duplicates is None:
+            eliminate_duplicates = C.SIMILARITY_ELIMINATE_DUPLICATES_DEFAULT
         lookback = C.DUPLICATE_DISTANCE_CHECK_LOOKBACK
         thresh = C.DUPLICATE_DISTANCE_THRESHOLD_COSINE
         exclude_set = set(exclude)
@@ -140,7 +147,7 @@ class SimilarityEngine:
 
     def find_similar_by_vector(self, vec: torch.Tensor, n: int, *,
                                exclude: Sequence[str] = (),
-                               eliminate_duplicates: bool = False,
+                               eliminate_duplicates: Optional[bool] = None,
                                max_per_artist: Optional[int] = None,
                                mood_filter: Optional[str] = None,
                                radius: bool = False,

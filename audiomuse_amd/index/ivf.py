@@ -62,9 +62,10 @@ def default_nlist(n: int) -> int:
 class IVFIndex:
     """Packed-cell IVF index. All tensors live on `self.device`."""
 
-    def __init__(self, dim: int, metric: str = "angular",
+    def __init__(self, dim: int, metric: Optional[str] = None,
                  storage: Optional[str] = None,
                  device: str | torch.device = "cpu"):
+        metric = metric or C.IVF_METRIC
         if metric not in _METRIC_CODE:
             raise ValueError(f"unknown metric {metric}")
         self.dim = dim
@@ -85,7 +86,7 @@ class IVFIndex:
 
     @classmethod
     def build(cls, vectors: torch.Tensor, ids: Optional[torch.Tensor] = None,
-              metric: str = "angular", storage: Optional[str] = None,
+              metric: Optional[str] = None, storage: Optional[str] = None,
               nlist: Optional[int] = None, device: str | torch.device = "cpu",
               seed: int = 0, keep_f32: bool = True,
               group: Optional[object] = None) -> "IVFIndex":
@@ -96,6 +97,7 @@ class IVFIndex:
         vectors = torch.as_tensor(vectors, dtype=torch.float32).to(device)
         n, dim = vectors.shape
         idx = cls(dim, metric=metric, storage=storage, device=device)
+        metric = idx.metric  # None resolved to C.IVF_METRIC
 
         if metric == "angular":
             norms = vectors.norm(dim=1, keepdim=True).clamp(min=1e-12)
@@ -108,6 +110,7 @@ class IVFIndex:
         g = torch.Generator().manual_seed(seed)
         sample = unit[torch.randperm(n, generator=g)[:n_train].to(device)]
         centroids = minibatch_kmeans(sample, nlist, iters=C.IVF_KMEANS_ITERS,
+                                     batch=C.IVF_KMEANS_BATCH,
                                      seed=seed, group=group)
         assign = assign_to_centroids(unit, centroids)
 

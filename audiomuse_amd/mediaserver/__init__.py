@@ -38,12 +38,41 @@ def provider_types() -> List[str]:
     return sorted(_PROVIDERS)
 
 
+def config_defaults(server_type: str) -> Dict:
+    """Connection kwargs for `server_type` from the legacy single-server
+    env/config surface (reference PARAMETERS.md "Mediaserver General":
+    NAVIDROME_URL/JELLYFIN_TOKEN/... + MUSIC_LIBRARIES). Registry rows
+    in music_servers override these per server."""
+    from audiomuse_amd import config as C
+
+    libs = [x.strip() for x in (C.MUSIC_LIBRARIES or "").split(",")
+            if x.strip()]
+    if server_type in ("navidrome", "subsonic"):
+        return {"base_url": C.NAVIDROME_URL, "username": C.NAVIDROME_USER,
+                "credential": C.NAVIDROME_API_KEY or C.NAVIDROME_PASSWORD,
+                "music_folder_ids": libs}
+    if server_type == "jellyfin":
+        return {"base_url": C.JELLYFIN_URL, "credential": C.JELLYFIN_TOKEN,
+                "user_id": C.JELLYFIN_USER_ID, "library_ids": libs}
+    if server_type == "emby":
+        return {"base_url": C.EMBY_URL, "credential": C.EMBY_TOKEN,
+                "user_id": C.EMBY_USER_ID, "library_ids": libs}
+    if server_type == "lyrion":
+        return {"base_url": C.LYRION_URL, "target_paths": libs}
+    if server_type == "plex":
+        return {"base_url": C.PLEX_URL, "credential": C.PLEX_TOKEN,
+                "section_ids": libs}
+    return {}
+
+
 def make_provider(server_type: str, **kwargs) -> Provider:
     cls = _PROVIDERS.get(server_type)
     if cls is None:
         raise ValueError(f"unsupported media server type {server_type!r} "
                          f"(supported: {', '.join(provider_types())})")
-    return cls(**kwargs)
+    merged = config_defaults(server_type)
+    merged.update(kwargs)
+    return cls(**merged)
 
 
 class BoundServer:

@@ -23,6 +23,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from audiomuse_amd import config as C
 from audiomuse_amd.ops.norms import FusedLayerNorm
 
 # special tokens (our own vocabulary layout)
@@ -335,15 +336,21 @@ def detect_language(model: WhisperModel, enc: torch.Tensor) -> int:
 @torch.inference_mode()
 def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
                   max_tokens: Optional[int] = None,
-                  repetition_penalty: float = 1.2,
-                  no_repeat_ngram: int = 3,
+                  repetition_penalty: Optional[float] = None,
+                  no_repeat_ngram: Optional[int] = None,
                   use_graph: Optional[bool] = None) -> List[int]:
     """Greedy KV-cache decode of one chunk. mel (n_mels, T).
 
     On GPU the per-token step runs as one hipGraph replay
     (GraphedDecoder) unless use_graph=False; CPU always runs eager."""
     cfg = model.cfg
-    max_tokens = min(max_tokens or cfg.max_tokens - 4, cfg.max_tokens - 4)
+    if repetition_penalty is None:
+        repetition_penalty = C.WHISPER_REPETITION_PENALTY
+    if no_repeat_ngram is None:
+        no_repeat_ngram = C.WHISPER_NO_REPEAT_NGRAM
+    max_tokens = min(max_tokens or min(C.WHISPER_MAX_NEW_TOKENS,
+                                       cfg.max_tokens - 4),
+                     cfg.max_tokens - 4)
     enc = model.encode(mel.unsqueeze(0))
     if use_graph is None:
         use_graph = enc.is_cuda
@@ -391,7 +398,13 @@ def beam_decode(model: WhisperModel, mel: torch.Tensor, beam: int = 2, *,
                              repetition_penalty=repetition_penalty,
                              no_repeat_ngram=no_repeat_ngram)
     cfg = model.cfg
-    max_tokens = min(max_tokens or cfg.max_tokens - 4, cfg.max_tokens - 4)
+    if repetition_penalty is None:
+        repetition_penalty = C.WHISPER_REPETITION_PENALTY
+    if no_repeat_ngram is None:
+        no_repeat_ngram = C.WHISPER_NO_REPEAT_NGRAM
+    max_tokens = min(max_tokens or min(C.WHISPER_MAX_NEW_TOKENS,
+                                       cfg.max_tokens - 4),
+                     cfg.max_tokens - 4)
     enc = model.encode(mel.unsqueeze(0)).expand(beam, -1, -1).contiguous()
     caches = model.make_caches(beam, enc.device, enc.dtype)
     ckv = model.cross_kvs(enc)

@@ -116,7 +116,11 @@ class PluginManager:
         """Load every enabled plugin stored in the `plugin` table (the
         reference keeps plugin zips in the DB and loads them at web and
         worker boot), then sync their cron tasks. Individual plugin
-        failures are logged and skipped. Returns plugins loaded."""
+        failures are logged and skipped. Returns plugins loaded.
+        PLUGINS_ENABLED=0 disables the whole subsystem."""
+        from audiomuse_amd import config as C
+        if not C.PLUGINS_ENABLED:
+            return 0
         rows = conn.execute(
             "SELECT name, blob FROM plugin WHERE enabled = 1").fetchall()
         n = 0

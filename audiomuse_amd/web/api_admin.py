@@ -221,6 +221,12 @@ def upload_plugin():
         return jsonify({"error": "plugin name must be alphanumeric"}), 400
     if not blob:
         return jsonify({"error": "empty plugin zip"}), 400
+    from audiomuse_amd import config as C
+    if not C.PLUGINS_ENABLED:
+        return jsonify({"error": "plugins disabled (PLUGINS_ENABLED=0)"}), 403
+    if len(blob) > C.PLUGIN_MAX_DOWNLOAD_MB * 1024 * 1024:
+        return jsonify({"error": f"plugin exceeds PLUGIN_MAX_DOWNLOAD_MB "
+                                 f"({C.PLUGIN_MAX_DOWNLOAD_MB} MB)"}), 413
     try:
         plugin_manager.load_zip(name, blob)   # validate before persisting
     except Exception as exc:  # noqa: BLE001 — surface the load error

@@ -100,9 +100,11 @@ def verify_session_token(token: str) -> Optional[dict]:
 
 
 def seed_admin_from_env(conn: sqlite3.Connection) -> None:
-    """reference: app_auth.seed_admin_from_env :417"""
-    user = os.environ.get("AUDIOMUSE_ADMIN_USER")
-    pw = os.environ.get("AUDIOMUSE_ADMIN_PASSWORD")
+    """reference: app_auth.seed_admin_from_env :417 — reference env
+    names AUDIOMUSE_USER/AUDIOMUSE_PASSWORD (PARAMETERS.md), with the
+    ADMIN_-prefixed variants kept as aliases."""
+    user = os.environ.get("AUDIOMUSE_ADMIN_USER") or C.AUDIOMUSE_USER
+    pw = os.environ.get("AUDIOMUSE_ADMIN_PASSWORD") or C.AUDIOMUSE_PASSWORD
     if not user or not pw:
         return
     with write_txn(conn):
