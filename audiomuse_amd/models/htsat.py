@@ -134,7 +134,10 @@ class SwinBlock(nn.Module):
         return ext is not None and hasattr(ext, "window_attn_fwd")
 
     def _fused_attn_available(self, x: torch.Tensor) -> bool:
-        return (self._fused_available(x) and self.window == 8
+        # window 8 -> window_attn_fwd (one 64-token window per wave);
+        # window 4 -> window_attn4_fwd (stage 4's 16-token windows,
+        # VERDICT r1 item 5 — was the eager SDPA chain)
+        return (self._fused_available(x) and self.window in (4, 8)
                 and self.attn.heads % 2 == 0
                 and self.attn.dim // self.attn.heads == 32)
 
@@ -175,7 +178,9 @@ class SwinBlock(nn.Module):
                 qkv = ext.linear_bias(xn.contiguous(),
                                       self.attn.qkv.weight.contiguous(),
                                       self.attn.qkv.bias.contiguous())
-            out = ext.window_attn_fwd(
+            attn_fwd = (ext.window_attn_fwd if self.window == 8
+                        else ext.window_attn4_fwd)
+            out = attn_fwd(
                 qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
                 self.attn.heads, self.shift, self.attn.scale)
             out = out.view(B, L, C)

@@ -295,6 +295,154 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused 4x4-window attention (stage 4: grid 32x4, heads 32, window 4).
+//
+// 16 tokens x head_dim 32 per (window, head) wave: S = QK^T is ONE
+// 16x16 MFMA (K=32); softmax in registers; O = PV is two 16x16 tiles
+// whose K (=16 tokens) rides a zero-padded K=32 MFMA. Replaces the
+// eager roll/partition/SDPA/reverse chain that r1 left on stage 4
+// (models/htsat.py:231-246).
+// ---------------------------------------------------------------------------
+
+constexpr int A4_LDS = 32 * 32 + 16 * 32;   // VT[32][32] + P[16][32] bf16
+
+__global__ __launch_bounds__(64, 6) void window_attn4_kernel(
+    const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
+    __bf16* __restrict__ out,        // (B, H, W, C)
+    const float* __restrict__ bias,  // (heads, 16, 16)
+    int Bn, int H, int W, int C, int heads, int shift, float scale) {
+  extern __shared__ __bf16 lds[];
+  const int lane = threadIdx.x & 63;
+
+  const int nWw = W >> 2;
+  const int nWh = H >> 2;
+  const int win = blockIdx.x;
+  const int b = win / (nWh * nWw);
+  const int wrem = win - b * (nWh * nWw);
+  const int wh = wrem / nWw;
+  const int ww = wrem - wh * nWw;
+
+  __bf16* VT = lds;            // [dim 32][token 32 padded]
+  __bf16* P = VT + 32 * 32;    // [row 16][k 32 padded]
+
+  auto src_of = [&](int t, int& si, int& sj, int& wrap) {
+    const int ri = t >> 2, ci = t & 3;
+    int gi = wh * 4 + ri + shift;
+    int gj = ww * 4 + ci + shift;
+    const int wr = gi >= H;
+    const int wc = gj >= W;
+    si = wr ? gi - H : gi;
+    sj = wc ? gj - W : gj;
+    wrap = (wr << 1) | wc;
+  };
+
+  // wrap bits of the 16 tokens (lanes 0-15 vote; mask replicated)
+  int si0, sj0, wrap0;
+  src_of(lane & 15, si0, sj0, wrap0);
+  const unsigned long long wrap_r_mask = __ballot(wrap0 & 2) & 0xffffull;
+  const unsigned long long wrap_c_mask = __ballot(wrap0 & 1) & 0xffffull;
+
+  const int h = blockIdx.y;
+
+  // ---- zero the K-padding once (tokens 16..31) ----
+  // VT cols 16..31 and P cols 16..31 must be zero for the padded-K MFMA
+  {
+    const int r = lane >> 1, c0 = 16 + 8 * (lane & 1);
+    if (r < 32) *(bf16x8*)(VT + r * 32 + c0) = bf16x8{};
+    if (r < 16) *(bf16x8*)(P + r * 32 + c0) = bf16x8{};
+  }
+
+  // ---- Q/K fragments from global ----
+  // A/B-frag: row|col = lane&15, k = 8*(lane>>4)+j
+  const int kk = 8 * (lane >> 4);
+  const int t16 = lane & 15;
+  int si, sj, w_;
+  src_of(t16, si, sj, w_);
+  const long long base = (((long long)b * H + si) * W + sj) * 3 * C + h * 32;
+  const bf16x8 qf = *(const bf16x8*)(qkv + base + kk);
+  const bf16x8 kf = *(const bf16x8*)(qkv + base + C + kk);
+
+  // ---- stage V transposed: VT[dim][token] (16 valid tokens) ----
+  if (lane < 16) {
+    const __bf16* vptr = qkv + base + 2 * C;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const bf16x8 vv = *(const bf16x8*)(vptr + g * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) VT[(g * 8 + j) * 32 + t16] = vv[j];
+    }
+  }
+
+  // ---- S = QK^T (one MFMA), bias prefetched alongside ----
+  const float* bias_base = bias + (h * 16 + (lane >> 4) * 4) * 16 + t16;
+  float bv[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) bv[reg] = bias_base[reg * 16];
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  f32x4 s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc, 0, 0, 0);
+
+  // ---- scale + bias + shift mask + row softmax ----
+  // C-frag: col = lane&15, row = (lane>>4)*4 + reg; the 16 lanes of a
+  // row live in the SAME lane group? No: a row's 16 cols spread over
+  // lanes with identical (lane>>4, reg) — reduce across lane&15.
+  float ex[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = (lane >> 4) * 4 + reg;
+    const int col = t16;
+    const int rwrap = (((wrap_r_mask >> row) & 1ull) << 1) |
+                      ((wrap_c_mask >> row) & 1ull);
+    const int cwrap = (((wrap_r_mask >> col) & 1ull) << 1) |
+                      ((wrap_c_mask >> col) & 1ull);
+    float v = s[reg] * scale + bv[reg];
+    if (shift && rwrap != cwrap) v = -1e30f;
+    float m = v;
+#pragma unroll
+    for (int d = 1; d < 16; d <<= 1) m = fmaxf(m, __shfl_xor(m, d, 64));
+    float e = __expf(v - m);
+    float sum = e;
+#pragma unroll
+    for (int d = 1; d < 16; d <<= 1) sum += __shfl_xor(sum, d, 64);
+    ex[reg] = e * __frcp_rn(sum + 1e-20f);
+  }
+
+  __syncthreads();   // V staged before PV reads; P pad settled
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = (lane >> 4) * 4 + reg;
+    P[row * 32 + t16] = (__bf16)ex[reg];
+  }
+  __syncthreads();
+
+  // ---- O = P @ V: 2 tiles (dims 0-15, 16-31), zero-padded K ----
+  const bf16x8 pa = *(const bf16x8*)(P + t16 * 32 + kk);
+#pragma unroll
+  for (int tc = 0; tc < 2; ++tc) {
+    const bf16x8 vb = *(const bf16x8*)(VT + (tc * 16 + t16) * 32 + kk);
+    f32x4 z = {0.f, 0.f, 0.f, 0.f};
+    const f32x4 o = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, z,
+                                                            0, 0, 0);
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int t = (lane >> 4) * 4 + reg;
+      int osi, osj, ow_;
+      src_of(t, osi, osj, ow_);
+      out[(((long long)b * H + osi) * W + osj) * C + h * 32 + tc * 16 +
+          t16] = (__bf16)o[reg];
+    }
+  }
+}
+
+void launch_window_attn4(const void* qkv, void* out, const float* bias,
+                         int Bn, int H, int W, int C, int heads, int shift,
+                         float scale, hipStream_t stream) {
+  const int n_windows = Bn * (H >> 2) * (W >> 2);
+  hipLaunchKernelGGL(window_attn4_kernel, dim3(n_windows, heads), dim3(64),
+                     A4_LDS * sizeof(__bf16), stream, (const __bf16*)qkv,
+                     (__bf16*)out, bias, Bn, H, W, C, heads, shift, scale);
+}
+
 void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream) {

@@ -32,6 +32,9 @@ void launch_layernorm_bf16_fp8_impl(const void* x, void* y8, const void* w,
                                     float eps, hipStream_t stream);
 void launch_mfma_probe(const void* A, const void* B, float* D,
                        hipStream_t stream);
+void launch_window_attn4(const void* qkv, void* out, const float* bias,
+                         int Bn, int H, int W, int C, int heads, int shift,
+                         float scale, hipStream_t stream);
 void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream);
@@ -171,6 +174,30 @@ static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
   return out;
 }
 
+static torch::Tensor window_attn4_fwd(torch::Tensor qkv, torch::Tensor bias,
+                                      int64_t heads, int64_t shift,
+                                      double scale) {
+  AM_CHECK(qkv.is_cuda() && qkv.scalar_type() == at::kBFloat16 &&
+               qkv.is_contiguous() && qkv.dim() == 4,
+           "qkv must be (B, H, W, 3C) bf16 contiguous GPU");
+  const int64_t Bn = qkv.size(0), H = qkv.size(1), W = qkv.size(2);
+  const int64_t C = qkv.size(3) / 3;
+  AM_CHECK(qkv.size(3) == 3 * C && C == heads * 32,
+           "C must be heads*32 and last dim 3C");
+  AM_CHECK(H % 4 == 0 && W % 4 == 0, "H, W must be multiples of 4");
+  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kFloat &&
+               bias.is_contiguous() && bias.numel() == heads * 16 * 16,
+           "bias must be (heads, 16, 16) f32 contiguous");
+  auto out = torch::empty({Bn, H, W, C}, qkv.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_window_attn4(qkv.data_ptr(), out.data_ptr(),
+                                 bias.data_ptr<float>(), (int)Bn, (int)H,
+                                 (int)W, (int)C, (int)heads, (int)shift,
+                                 (float)scale, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 static std::vector<torch::Tensor> add_layernorm_bf16(torch::Tensor x,
                                                      torch::Tensor other,
                                                      torch::Tensor w,
@@ -253,6 +280,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("window_attn_fwd", &window_attn_fwd,
         "Fused shifted-window attention (qkv BHW3C bf16, bias, heads, "
         "shift, scale) -> (B,H,W,C)");
+  m.def("window_attn4_fwd", &window_attn4_fwd,
+        "Fused 4x4-window attention for stage 4 (qkv BHW3C bf16, "
+        "bias (heads,16,16), heads, shift, scale) -> (B,H,W,C)");
   m.def("layernorm_bf16", &layernorm_bf16,
         "Fused LayerNorm forward, bf16 in/out, fp32 stats (x, w, b, eps)");
   m.def("mel_fwd", &mel_fwd,
