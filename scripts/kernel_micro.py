@@ -58,6 +58,12 @@ def main():
     bias3 = torch.randn(16, 64, 64, device=dev)
     t = timeit(lambda: ext.window_attn_fwd(qkv3, bias3, 16, 4, 0.176), iters)
     print(f"window_attn s3 : {t*1000:8.2f} ms")
+    # stage-4 shape: 4x32 grid, window 4, C=1024, h=32 (window_attn4)
+    qkv4 = torch.randn(B, 4, 32, 3 * 1024, device=dev, dtype=torch.bfloat16)
+    bias4 = torch.randn(32, 16, 16, device=dev)
+    t = timeit(lambda: ext.window_attn4_fwd(qkv4, bias4, 32, 2, 0.176),
+               iters)
+    print(f"window_attn4 s4: {t*1000:8.2f} ms")
 
     # layernorm: stage-1 tokens
     x = torch.randn(B * 8192, 128, device=dev, dtype=torch.bfloat16)
