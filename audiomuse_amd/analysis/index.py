@@ -337,7 +337,8 @@ def run_all_index_builds(conn: sqlite3.Connection, device: str = "cpu",
                      ("semgrove", build_semgrove_index),
                      ("artist", lambda c, **k: build_artist_index(c)),
                      ("song_map", build_song_map),
-                     ("artist_map", build_artist_map)]:
+                     ("artist_map", build_artist_map),
+                     ("hyperbolic_tree", build_hyperbolic_tree_cache)]:
         try:
             results[name] = fn(conn, device=device) if name != "artist" \
                 else build_artist_index(conn)

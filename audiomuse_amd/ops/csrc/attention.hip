@@ -74,17 +74,20 @@ void launch_mfma_probe(const void* A, const void* B, float* D,
 // Fused window attention
 // ---------------------------------------------------------------------------
 
-// per-wave LDS: VT[32][64] + P[64][64] bf16, XOR-swizzled instead of
-// padded (swizzle: element col ^ ((row&7)<<3) keeps 16 B reads aligned
-// and spreads the 16-lane column reads across all banks). 12 KiB/wave;
-// 1-wave blocks -> ~13 blocks/CU, registers fit 4 waves/SIMD with the
-// launch-bounds hint (was: 181 regs -> 2 waves/SIMD, 11.9% occupancy
-// measured — this kernel is gather-latency bound, occupancy is the lever).
+// per-wave LDS: VT[32][64] + a HALF P tile [64][32] bf16, XOR-swizzled
+// instead of padded (swizzle: element col ^ (f(row)<<3) keeps 16 B
+// reads aligned and spreads the 16-lane column reads across banks).
+// 8 KiB/wave: r2 PMC showed the old 12 KiB (full 64x64 P) capped the
+// CU at 13 waves -> 30.7% occupancy on a latency-bound kernel. The PV
+// K loop now runs in two 32-column halves, the second half's softmax
+// rows parked in registers while the first half's MFMAs drain.
 constexpr int ATTN_WAVES = 1;
-constexpr int WAVE_LDS_HALF = 32 * 64 + 64 * 64;  // bf16 elems (VT + P)
+constexpr int WAVE_LDS_HALF = 32 * 64 + 64 * 32;  // bf16 elems (VT + P half)
 #define AM_SWZ(row, col) (((row) << 6) + ((col) ^ (((row) & 7) << 3)))
+// 32-wide swizzle for the P half-tile (8-element vectors stay aligned)
+#define AM_SWZ32(row, col) (((row) << 5) + ((col) ^ (((row) & 3) << 3)))
 
-__global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
+__global__ __launch_bounds__(64 * ATTN_WAVES, 4) void window_attn_kernel(
     const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
     __bf16* __restrict__ out,        // (B, H, W, C)
     const float* __restrict__ bias,          // (heads, 64, 64)
@@ -241,7 +244,11 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
         rsum[tr][reg] = sum;
       }
 
-    // ---- write P = softmax(S) to LDS (bf16) ----
+    // ---- write P = softmax(S) to LDS in TWO 32-column halves ----
+    // Half 0 (key tokens 0..31 = tc 0,1) goes to the 4 KiB P buffer;
+    // half 1 (tc 2,3) stays normalized in registers (s[][2..3]) and is
+    // written after the first PV K-step drains — halving LDS per wave
+    // was worth more than avoiding one extra single-wave barrier.
     __syncthreads();  // VT writes (and previous round's P reads) settled
 #pragma unroll
     for (int tr = 0; tr < 4; ++tr)
@@ -251,30 +258,54 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 3) void window_attn_kernel(
         const float inv = __frcp_rn(rsum[tr][reg] + 1e-20f);
 #pragma unroll
         for (int tc = 0; tc < 4; ++tc) {
-          const int col = tc * 16 + col_in_tile;
-          P[AM_SWZ(row, col)] = (__bf16)(s[tr][tc][reg] * inv);
+          const float pv = s[tr][tc][reg] * inv;
+          if (tc < 2)
+            P[AM_SWZ32(row, tc * 16 + col_in_tile)] = (__bf16)pv;
+          else
+            s[tr][tc][reg] = pv;  // parked for half 1
         }
       }
     __syncthreads();
 
-    // ---- O = P @ V: 4x2 tiles of 16x16, K = 64 (2 MFMA each) ----
+    // ---- O = P @ V, K-step 0 (tokens 0..31) ----
     f32x4 o[4][2];
 #pragma unroll
     for (int tr = 0; tr < 4; ++tr)
 #pragma unroll
       for (int tc = 0; tc < 2; ++tc) {
+        bf16x8 pa = *(const bf16x8*)(
+            P + AM_SWZ32(tr * 16 + (lane & 15), kk));
+        bf16x8 vb = *(const bf16x8*)(
+            VT + AM_SWZ(tc * 16 + (lane & 15), kk));
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        o[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc,
+                                                            0, 0, 0);
+      }
+
+    // ---- park half 1 into the P buffer, then K-step 1 (tokens 32..63) --
+    __syncthreads();
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          // A = P tile: row = tr*16 + (lane&15), k = ks*32 + kk + j
-          bf16x8 pa = *(const bf16x8*)(
-              P + AM_SWZ(tr * 16 + (lane & 15), ks * 32 + kk));
-          // B = V tile: col(dim) = tc*16 + (lane&15), k(token) = ks*32+kk+j
-          bf16x8 vb = *(const bf16x8*)(
-              VT + AM_SWZ(tc * 16 + (lane & 15), ks * 32 + kk));
-          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc, 0, 0, 0);
-        }
-        o[tr][tc] = acc;
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+#pragma unroll
+        for (int tc = 2; tc < 4; ++tc)
+          P[AM_SWZ32(row, (tc - 2) * 16 + col_in_tile)] =
+              (__bf16)s[tr][tc][reg];
+      }
+    __syncthreads();
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 2; ++tc) {
+        bf16x8 pa = *(const bf16x8*)(
+            P + AM_SWZ32(tr * 16 + (lane & 15), kk));
+        bf16x8 vb = *(const bf16x8*)(
+            VT + AM_SWZ(tc * 16 + (lane & 15), 32 + kk));
+        o[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb,
+                                                            o[tr][tc],
+                                                            0, 0, 0);
       }
 
     // ---- scatter O back to (B, H, W, C) with the inverse roll ----
