@@ -197,6 +197,34 @@ def load_artist_similarity(conn: sqlite3.Connection) -> Optional[ArtistSimilarit
     return sim
 
 
+def build_hyperbolic_tree_cache(conn: sqlite3.Connection,
+                                device: str = "cpu") -> int:
+    """Precompute + persist the explorer tree (full + skeleton blobs;
+    reference: hyperbolic_manager.build_hyperbolic_tree_cache :613)."""
+    import json as _json
+
+    from audiomuse_amd.engines.hyperbolic_tree import build_tree, persist_tree
+
+    ids, mat = load_all_embeddings(conn, "embedding")
+    if not ids:
+        return 0
+    meta_rows = {r["item_id"]: r for r in conn.execute(
+        "SELECT item_id, title, author, mood_vector FROM score")}
+
+    def meta_fn(item_id):
+        row = meta_rows.get(item_id)
+        if row is None:
+            return None
+        moods = row["mood_vector"]
+        return {"title": row["title"], "author": row["author"],
+                "mood_vector": _json.loads(moods) if isinstance(moods, str)
+                and moods else (moods or {})}
+
+    tree = build_tree(torch.from_numpy(mat).to(device), ids, meta_fn)
+    persist_tree(conn, tree)
+    return tree["track_count"]
+
+
 def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
     """2-D map projection (reference: app_helper.build_and_store_map_
     projection :340 — UMAP with PCA fallback; engines/projection.py

@@ -95,6 +95,19 @@ class WindowAttention(nn.Module):
         bias = self.rel_bias[self.rel_index.view(-1)].view(T, T, self.heads)
         return bias.permute(2, 0, 1).float().contiguous()
 
+    def full_bias_bf16(self) -> torch.Tensor:
+        """Serving-path bias: gathered ONCE per frozen weight version
+        and kept bf16 (the fused kernels read bf16 — half the per-wave
+        L2 bias stream; the gather used to run on every block call)."""
+        ver = self.rel_bias._version
+        cached = getattr(self, "_bias_cache", None)
+        if cached is not None and cached[0] == ver \
+                and cached[1].device == self.rel_bias.device:
+            return cached[1]
+        table = self.full_bias().to(torch.bfloat16).contiguous()
+        self._bias_cache = (ver, table)
+        return table
+
     def forward(self, x: torch.Tensor, mask: torch.Tensor | None) -> torch.Tensor:
         """x: (nW, T, C) with T = window*window; mask: (groups, T, T) or None."""
         nW, T, C = x.shape
@@ -181,7 +194,7 @@ class SwinBlock(nn.Module):
             attn_fwd = (ext.window_attn_fwd if self.window == 8
                         else ext.window_attn4_fwd)
             out = attn_fwd(
-                qkv.view(B, H, W, 3 * C), self.attn.full_bias(),
+                qkv.view(B, H, W, 3 * C), self.attn.full_bias_bf16(),
                 self.attn.heads, self.shift, self.attn.scale)
             out = out.view(B, L, C)
             # proj stays bf16 in fp8 mode too: its GEMM gain is smaller

@@ -90,7 +90,8 @@ constexpr int WAVE_LDS_HALF = 32 * 64 + 64 * 32;  // bf16 elems (VT + P half)
 __global__ __launch_bounds__(64 * ATTN_WAVES, 4) void window_attn_kernel(
     const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
     __bf16* __restrict__ out,        // (B, H, W, C)
-    const float* __restrict__ bias,          // (heads, 64, 64)
+    const __bf16* __restrict__ bias,         // (heads, 64, 64) bf16:
+    // halves the per-wave 16 KiB L2 bias stream (r2: latency-bound)
     int Bn, int H, int W, int C, int heads, int shift, float scale) {
   extern __shared__ __bf16 lds[];
   const int lane = threadIdx.x & 63;
@@ -166,13 +167,13 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 4) void window_attn_kernel(
     // (double-buffered software pipeline).
     const int col_in_tile = lane & 15;
     const int row_grp = (lane >> 4) * 4;
-    const float* bias_base = bias + (h * 64 + row_grp) * 64 + col_in_tile;
+    const __bf16* bias_base = bias + (h * 64 + row_grp) * 64 + col_in_tile;
     float bv[2][4][4];  // [buf][reg][tc]
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg)
 #pragma unroll
       for (int tc = 0; tc < 4; ++tc)
-        bv[0][reg][tc] = bias_base[reg * 64 + tc * 16];
+        bv[0][reg][tc] = (float)bias_base[reg * 64 + tc * 16];
 
     f32x4 s[4][4];
 #pragma unroll
@@ -195,7 +196,7 @@ __global__ __launch_bounds__(64 * ATTN_WAVES, 4) void window_attn_kernel(
 #pragma unroll
           for (int tc = 0; tc < 4; ++tc)
             bv[(tr + 1) & 1][reg][tc] =
-                bias_base[((tr + 1) * 16 + reg) * 64 + tc * 16];
+                (float)bias_base[((tr + 1) * 16 + reg) * 64 + tc * 16];
       }
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
@@ -341,7 +342,7 @@ constexpr int A4_LDS = 32 * 32 + 16 * 32;   // VT[32][32] + P[16][32] bf16
 __global__ __launch_bounds__(64, 6) void window_attn4_kernel(
     const __bf16* __restrict__ qkv,  // (B, H, W, 3C)
     __bf16* __restrict__ out,        // (B, H, W, C)
-    const float* __restrict__ bias,  // (heads, 16, 16)
+    const __bf16* __restrict__ bias,  // (heads, 16, 16) bf16
     int Bn, int H, int W, int C, int heads, int shift, float scale) {
   extern __shared__ __bf16 lds[];
   const int lane = threadIdx.x & 63;
@@ -406,10 +407,10 @@ __global__ __launch_bounds__(64, 6) void window_attn4_kernel(
   }
 
   // ---- S = QK^T (one MFMA), bias prefetched alongside ----
-  const float* bias_base = bias + (h * 16 + (lane >> 4) * 4) * 16 + t16;
+  const __bf16* bias_base = bias + (h * 16 + (lane >> 4) * 4) * 16 + t16;
   float bv[4];
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) bv[reg] = bias_base[reg * 16];
+  for (int reg = 0; reg < 4; ++reg) bv[reg] = (float)bias_base[reg * 16];
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   f32x4 s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc, 0, 0, 0);
 
@@ -465,16 +466,17 @@ __global__ __launch_bounds__(64, 6) void window_attn4_kernel(
   }
 }
 
-void launch_window_attn4(const void* qkv, void* out, const float* bias,
+void launch_window_attn4(const void* qkv, void* out, const void* bias,
                          int Bn, int H, int W, int C, int heads, int shift,
                          float scale, hipStream_t stream) {
   const int n_windows = Bn * (H >> 2) * (W >> 2);
   hipLaunchKernelGGL(window_attn4_kernel, dim3(n_windows, heads), dim3(64),
                      A4_LDS * sizeof(__bf16), stream, (const __bf16*)qkv,
-                     (__bf16*)out, bias, Bn, H, W, C, heads, shift, scale);
+                     (__bf16*)out, (const __bf16*)bias, Bn, H, W, C, heads,
+                     shift, scale);
 }
 
-void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
+void launch_window_attn(const void* qkv, void* out, const void* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream) {
   const int n_windows = Bn * (H >> 3) * (W >> 3);
@@ -483,7 +485,7 @@ void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
   hipLaunchKernelGGL(window_attn_kernel, dim3(n_windows, head_groups),
                      dim3(64 * ATTN_WAVES), lds_bytes,
                      stream, (const __bf16*)qkv, (__bf16*)out,
-                     bias, Bn, H, W, C, heads, shift, scale);
+                     (const __bf16*)bias, Bn, H, W, C, heads, shift, scale);
 }
 
 }  // namespace audiomuse

@@ -32,10 +32,10 @@ void launch_layernorm_bf16_fp8_impl(const void* x, void* y8, const void* w,
                                     float eps, hipStream_t stream);
 void launch_mfma_probe(const void* A, const void* B, float* D,
                        hipStream_t stream);
-void launch_window_attn4(const void* qkv, void* out, const float* bias,
+void launch_window_attn4(const void* qkv, void* out, const void* bias,
                          int Bn, int H, int W, int C, int heads, int shift,
                          float scale, hipStream_t stream);
-void launch_window_attn(const void* qkv, void* out, const float* bias, int Bn,
+void launch_window_attn(const void* qkv, void* out, const void* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream);
 }
@@ -161,13 +161,13 @@ static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
            "C must be heads*32 and last dim 3C");
   AM_CHECK(H % 8 == 0 && W % 8 == 0, "H, W must be multiples of 8");
   AM_CHECK(heads % 2 == 0, "heads must be a multiple of 2");
-  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kFloat &&
+  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kBFloat16 &&
                bias.is_contiguous() && bias.numel() == heads * 64 * 64,
-           "bias must be (heads, 64, 64) f32 contiguous");
+           "bias must be (heads, 64, 64) bf16 contiguous");
   auto out = torch::empty({Bn, H, W, C}, qkv.options());
   auto stream = c10::hip::getCurrentHIPStream();
   audiomuse::launch_window_attn(qkv.data_ptr(), out.data_ptr(),
-                                bias.data_ptr<float>(), (int)Bn, (int)H,
+                                bias.data_ptr(), (int)Bn, (int)H,
                                 (int)W, (int)C, (int)heads, (int)shift,
                                 (float)scale, stream.stream());
   C10_HIP_CHECK(hipGetLastError());
@@ -185,13 +185,13 @@ static torch::Tensor window_attn4_fwd(torch::Tensor qkv, torch::Tensor bias,
   AM_CHECK(qkv.size(3) == 3 * C && C == heads * 32,
            "C must be heads*32 and last dim 3C");
   AM_CHECK(H % 4 == 0 && W % 4 == 0, "H, W must be multiples of 4");
-  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kFloat &&
+  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kBFloat16 &&
                bias.is_contiguous() && bias.numel() == heads * 16 * 16,
-           "bias must be (heads, 16, 16) f32 contiguous");
+           "bias must be (heads, 16, 16) bf16 contiguous");
   auto out = torch::empty({Bn, H, W, C}, qkv.options());
   auto stream = c10::hip::getCurrentHIPStream();
   audiomuse::launch_window_attn4(qkv.data_ptr(), out.data_ptr(),
-                                 bias.data_ptr<float>(), (int)Bn, (int)H,
+                                 bias.data_ptr(), (int)Bn, (int)H,
                                  (int)W, (int)C, (int)heads, (int)shift,
                                  (float)scale, stream.stream());
   C10_HIP_CHECK(hipGetLastError());

@@ -354,40 +354,64 @@ def lyrics_axes_similar():
     return jsonify(_with_meta(res))
 
 
+def _tree_cache():
+    """Process-wide TreeCache (skeleton resident, full tree lazy-warm;
+    reference hyperbolic_manager.py:832-897)."""
+    from audiomuse_amd.engines.hyperbolic_tree import TreeCache
+
+    cache = current_app.extensions.get("hyperbolic_tree")
+    if cache is None:
+        cache = current_app.extensions["hyperbolic_tree"] = TreeCache()
+        cache.load_skeleton(_state().conn())
+    return cache
+
+
 @bp.get("/api/hyperbolic_tree")
 @require_auth
 def hyperbolic_tree():
-    """Mood-rooted explorer tree (reference: hyperbolic_manager genre/
-    mood tree cache :613): root -> predominant moods -> nearest tracks
-    in hyperbolic space."""
-    from audiomuse_amd.engines.hyperbolic import HyperbolicSpace
+    """Explorer tree root (reference: hyperbolic_manager tree cache
+    :613): served from the persisted SKELETON — opening the explorer
+    never forces the full tree into memory."""
+    cache = _tree_cache()
+    root = cache.node(_state().conn(), "root")
+    if root is None:
+        return jsonify({"error": "hyperbolic tree not built; run "
+                                 "analysis/index rebuild"}), 503
+    return jsonify(root)
 
-    eng = _state().engine(idx.AUDIO_INDEX)
-    if eng is None:
-        return jsonify({"error": "audio index not built"}), 503
-    per_node = int(request.args.get("per_node", 8))
+
+@bp.get("/api/hyperbolic_tree/node/<path:node_id>")
+@require_auth
+def hyperbolic_tree_node(node_id):
+    """One tree node. Folder nodes come from the skeleton; leaf nodes
+    lazily warm the full tree (warm timer unloads it again after
+    HYPERBOLIC_TREE_WARMUP_DURATION)."""
+    cache = _tree_cache()
+    node = cache.node(_state().conn(), node_id)
+    if node is None:
+        return jsonify({"error": f"unknown node {node_id!r}"}), 404
+    return jsonify(node)
+
+
+@bp.get("/api/hyperbolic_tree/status")
+@require_auth
+def hyperbolic_tree_status():
+    return jsonify(_tree_cache().status())
+
+
+@bp.post("/api/hyperbolic_tree/build")
+@require_auth
+def hyperbolic_tree_build():
+    """Inline (re)build — normally the analysis run's index-build phase
+    does this (analysis/index.py run_all_index_builds)."""
+    from audiomuse_amd.analysis.index import build_hyperbolic_tree_cache
+
     state = _state()
-    space = current_app.extensions.get("hyperbolic")
-    if space is None or current_app.extensions.get("hyperbolic_n") != eng.index.n:
-        space = HyperbolicSpace(eng.index.vectors_f32)
-        current_app.extensions["hyperbolic"] = space
-        current_app.extensions["hyperbolic_n"] = eng.index.n
-    by_mood = {}
-    for pos, item_id in enumerate(eng.item_ids):
-        meta = state.meta_fn(item_id) or {}
-        moods = meta.get("mood_vector") or {}
-        if not moods:
-            continue
-        top = max(moods, key=moods.get)
-        by_mood.setdefault(top, []).append((pos, item_id, moods[top]))
-    tree = []
-    for mood, members in sorted(by_mood.items(), key=lambda kv: -len(kv[1])):
-        members.sort(key=lambda t: -t[2])
-        kids = [{"item_id": iid,
-                 "radius": float(space.points[pos].norm())}
-                for pos, iid, _s in members[:per_node]]
-        tree.append({"mood": mood, "count": len(members), "children": kids})
-    return jsonify(tree)
+    n = build_hyperbolic_tree_cache(state.conn(), device=state.device)
+    cache = _tree_cache()
+    cache.load_skeleton(state.conn())
+    cache._unload_full()
+    return jsonify({"tracks": n})
 
 
 @bp.get("/api/sonic_fingerprint")

@@ -49,18 +49,18 @@ def main():
     # window attention: stage-1 shape (B=128, 32x256 grid, C=128, h=4)
     B, H, W, C, heads = 128, 32, 256, 128, 4
     qkv = torch.randn(B, H, W, 3 * C, device=dev, dtype=torch.bfloat16)
-    bias = torch.randn(heads, 64, 64, device=dev)
+    bias = torch.randn(heads, 64, 64, device=dev).to(torch.bfloat16)
     t = timeit(lambda: ext.window_attn_fwd(qkv, bias, heads, 0, 0.176), iters)
     toks = B * H * W
     print(f"window_attn s1 : {t*1000:8.2f} ms ({toks/t/1e6:6.1f} Mtok/s)")
     # stage-3 shape (C=512, h=16, 8x64 grid)
     qkv3 = torch.randn(B, 8, 64, 3 * 512, device=dev, dtype=torch.bfloat16)
-    bias3 = torch.randn(16, 64, 64, device=dev)
+    bias3 = torch.randn(16, 64, 64, device=dev).to(torch.bfloat16)
     t = timeit(lambda: ext.window_attn_fwd(qkv3, bias3, 16, 4, 0.176), iters)
     print(f"window_attn s3 : {t*1000:8.2f} ms")
     # stage-4 shape: 4x32 grid, window 4, C=1024, h=32 (window_attn4)
     qkv4 = torch.randn(B, 4, 32, 3 * 1024, device=dev, dtype=torch.bfloat16)
-    bias4 = torch.randn(32, 16, 16, device=dev)
+    bias4 = torch.randn(32, 16, 16, device=dev).to(torch.bfloat16)
     t = timeit(lambda: ext.window_attn4_fwd(qkv4, bias4, 32, 2, 0.176),
                iters)
     print(f"window_attn4 s4: {t*1000:8.2f} ms")

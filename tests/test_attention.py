@@ -144,10 +144,12 @@ def test_window8_kernel_vs_fp32_reference(shift):
     dim = heads * 32
     qkv = (torch.randn(B, H, W, 3 * dim, device="cuda") * 0.5).to(
         torch.bfloat16).contiguous()
-    bias = torch.randn(heads, 64, 64, device="cuda") * 0.1
+    bias = (torch.randn(heads, 64, 64, device="cuda") * 0.1).to(
+        torch.bfloat16)
     scale = 32 ** -0.5
     out = C.window_attn_fwd(qkv, bias.contiguous(), heads, shift, scale)
-    ref = _fp32_window_attention_reference(qkv, bias, heads, shift, scale, 8)
+    ref = _fp32_window_attention_reference(qkv, bias.float(), heads, shift,
+                                           scale, 8)
     diff = (out.float() - ref).abs()
     # bf16 inputs + fp32 accum kernel vs fp32 reference: per-element
     # bound a few bf16 ulps of the output scale
@@ -168,10 +170,12 @@ def test_window4_kernel_vs_fp32_reference(shift):
     dim = heads * 32
     qkv = (torch.randn(B, H, W, 3 * dim, device="cuda") * 0.5).to(
         torch.bfloat16).contiguous()
-    bias = torch.randn(heads, 16, 16, device="cuda") * 0.1
+    bias = (torch.randn(heads, 16, 16, device="cuda") * 0.1).to(
+        torch.bfloat16)
     scale = 32 ** -0.5
     out = C.window_attn4_fwd(qkv, bias.contiguous(), heads, shift, scale)
-    ref = _fp32_window_attention_reference(qkv, bias, heads, shift, scale, 4)
+    ref = _fp32_window_attention_reference(qkv, bias.float(), heads, shift,
+                                           scale, 4)
     diff = (out.float() - ref).abs()
     assert float(diff.max()) < 0.06, f"max abs err {float(diff.max()):.4f}"
     rel = diff.mean() / ref.abs().mean().clamp(min=1e-6)

@@ -212,13 +212,30 @@ def test_lyrics_axes_endpoint(client_ids):
     assert client.get("/api/lyrics_axes?axis=bogus").status_code == 400
 
 
-def test_hyperbolic_tree_endpoint(client_ids):
+def test_hyperbolic_tree_endpoints(client_ids):
     client, _ = client_ids
-    r = client.get("/api/hyperbolic_tree?per_node=3")
-    assert r.status_code == 200
-    tree = r.json
-    assert tree and all("mood" in n and "children" in n for n in tree)
-    assert all(len(n["children"]) <= 3 for n in tree)
+    # the seeding index rebuild persisted the tree already (the
+    # run_all_index_builds hook); the inline endpoint rebuilds it
+    r = client.post("/api/hyperbolic_tree/build")
+    assert r.status_code == 200 and r.json["tracks"] > 0
+    # root serves from the SKELETON: no full-tree warm happened
+    status = client.get("/api/hyperbolic_tree/status").json
+    assert status["skeleton_loaded"] and not status["full_loaded"]
+    root = client.get("/api/hyperbolic_tree").json
+    assert root["id"] == "root" and root["children_count"] >= 1
+    mood_id = root["items"][0]["id"]
+    folder = client.get(f"/api/hyperbolic_tree/node/{mood_id}").json
+    assert folder["type"] == "folder" and not folder["leaf"]
+    assert client.get("/api/hyperbolic_tree/status"
+                      ).json["full_loaded"] is False  # still skeleton-only
+    # a LEAF node lazily warms the full tree
+    leaf_id = folder["items"][0]["id"]
+    leaf = client.get(f"/api/hyperbolic_tree/node/{leaf_id}").json
+    assert leaf["leaf"] and leaf["items"]
+    assert all("item_id" in t and "radius" in t for t in leaf["items"])
+    status = client.get("/api/hyperbolic_tree/status").json
+    assert status["full_loaded"] and status["warm_seconds_left"] > 0
+    assert client.get("/api/hyperbolic_tree/node/bogus").status_code == 404
 
 
 def test_users_endpoint(client_ids):
