@@ -187,11 +187,11 @@ LYRICS_ASR_MIN_AVG_LOGPROB = _env_float("LYRICS_ASR_MIN_AVG_LOGPROB", -1.0)
 LYRICS_ASR_NON_ENGLISH_MIN_LOGPROB = _env_float(
     "LYRICS_ASR_NON_ENGLISH_MIN_LOGPROB", -0.7)
 LYRICS_TEXT_MAX_COMPRESSION_RATIO = _env_float(
-    "LYRICS_TEXT_MAX_COMPRESSION_RATIO", 2.4)
+    "LYRICS_TEXT_MAX_COMPRESSION_RATIO", 15.0)
 LYRICS_MIN_CHARS_FOR_EMBEDDING = _env_int("LYRICS_MIN_CHARS_FOR_EMBEDDING",
-                                          40)
-LYRICS_LANG_CONFIDENCE_MIN = _env_float("LYRICS_LANG_CONFIDENCE_MIN", 0.5)
-LYRICS_CJK_SCRIPT_MIN_RATIO = _env_float("LYRICS_CJK_SCRIPT_MIN_RATIO", 0.3)
+                                          250)
+LYRICS_LANG_CONFIDENCE_MIN = _env_float("LYRICS_LANG_CONFIDENCE_MIN", 0.7)
+LYRICS_CJK_SCRIPT_MIN_RATIO = _env_float("LYRICS_CJK_SCRIPT_MIN_RATIO", 0.1)
 LYRICS_GTE_WARMUP_DURATION = _env_float("LYRICS_GTE_WARMUP_DURATION", 300.0)
 LYRICS_MAX_WORDS = _env_int("LYRICS_MAX_WORDS", 300)
 

@@ -37,25 +37,119 @@ _LANG_STOPWORDS = {
 }
 
 
-def detect_language(text: str) -> str:
-    words = set(w.lower() for w in _WORD_RE.findall(text or ""))
-    best, best_n = "unknown", 0
-    for lang, sw in _LANG_STOPWORDS.items():
-        n = len(words & sw)
-        if n > best_n:
-            best, best_n = lang, n
-    return best if best_n >= 2 else "unknown"
+_CJK_RANGES = ((0x3040, 0x30FF),    # hiragana/katakana
+               (0x4E00, 0x9FFF),    # CJK unified
+               (0xAC00, 0xD7AF),    # hangul
+               (0x3400, 0x4DBF))
+
+
+def _cjk_ratio(text: str) -> float:
+    chars = [c for c in text if not c.isspace()]
+    if not chars:
+        return 0.0
+    n = sum(1 for c in chars
+            if any(lo <= ord(c) <= hi for lo, hi in _CJK_RANGES))
+    return n / len(chars)
+
+
+def detect_language(text: str, with_confidence: bool = False):
+    """Script-aware language gate (reference: langdetect +
+    LYRICS_CJK_SCRIPT_MIN_RATIO / LYRICS_LANG_CONFIDENCE_MIN gates,
+    lyrics_transcriber.py language stage). CJK scripts identify by
+    codepoint ratio; Latin languages by stopword evidence with a
+    confidence = hit fraction among known stopwords."""
+    ratio = _cjk_ratio(text or "")
+    if ratio >= C.LYRICS_CJK_SCRIPT_MIN_RATIO:
+        han = sum(1 for c in text if 0x4E00 <= ord(c) <= 0x9FFF)
+        kana = sum(1 for c in text if 0x3040 <= ord(c) <= 0x30FF)
+        hangul = sum(1 for c in text if 0xAC00 <= ord(c) <= 0xD7AF)
+        lang = ("ja" if kana > 0 else "ko" if hangul > han else "zh")
+        return (lang, 1.0) if with_confidence else lang
+    words = [w.lower() for w in _WORD_RE.findall(text or "")]
+    uniq = set(words)
+    scores = {lang: len(uniq & sw) for lang, sw in _LANG_STOPWORDS.items()}
+    ranked = sorted(scores.items(), key=lambda kv: -kv[1])
+    best, hits = ranked[0] if ranked else ("unknown", 0)
+    runner_up = ranked[1][1] if len(ranked) > 1 else 0
+    # confidence = margin over the runner-up language (shared stopwords
+    # like es/pt "que" depress it exactly when the call is ambiguous)
+    conf = hits / max(hits + runner_up, 1) if hits >= 2 else 0.0
+    lang = best if conf >= C.LYRICS_LANG_CONFIDENCE_MIN and hits >= 2 \
+        else "unknown"
+    return (lang, conf) if with_confidence else lang
+
+
+def compression_ratio(text: str) -> float:
+    """zlib length ratio — ASR loops compress extremely well (reference
+    LYRICS_TEXT_MAX_COMPRESSION_RATIO gate)."""
+    import zlib
+    data = (text or "").encode("utf-8")
+    if not data:
+        return 0.0
+    return len(data) / max(len(zlib.compress(data, 6)), 1)
 
 
 def quality_gate(text: str, min_words: int = 8,
                  max_repeat_ratio: float = 0.6) -> bool:
-    """Reject junk transcripts: too short, or dominated by one token
-    (reference quality gates)."""
+    """Reject junk transcripts: too short, dominated by one token, or
+    degenerate-repetitive (compression gate; reference quality gates +
+    LYRICS_TEXT_MAX_COMPRESSION_RATIO)."""
     words = [w.lower() for w in _WORD_RE.findall(text or "")]
     if len(words) < min_words:
         return False
     top = max(words.count(w) for w in set(words))
-    return top / len(words) <= max_repeat_ratio
+    if top / len(words) > max_repeat_ratio:
+        return False
+    return compression_ratio(text) <= C.LYRICS_TEXT_MAX_COMPRESSION_RATIO
+
+
+def fetch_external_lyrics(title: str, artist: str,
+                          http_get=None) -> Optional[str]:
+    """External lyrics APIs, tried in order before ASR (reference:
+    lyrics_transcriber stages 1-2 + LYRICS_API_{1,2}_* config).
+    URL templates may carry {artist}/{title} placeholders; otherwise the
+    configured param names ride the query string. SSRF-guarded."""
+    if not C.LYRICS_API_ENABLE:
+        return None
+    from audiomuse_amd.utils.logging_utils import validate_outbound_url
+    if http_get is None:
+        import requests
+        http_get = requests.get
+    for n in (1, 2):
+        tmpl = getattr(C, f"LYRICS_API_{n}_URL_TEMPLATE", "")
+        if not tmpl:
+            continue
+        field_path = getattr(C, f"LYRICS_API_{n}_LYRICS_FIELD")
+        params = {}
+        url = tmpl
+        if "{artist}" in tmpl or "{title}" in tmpl:
+            from urllib.parse import quote
+            url = tmpl.replace("{artist}", quote(artist)).replace(
+                "{title}", quote(title))
+        else:
+            params[getattr(C, f"LYRICS_API_{n}_ARTIST_PARAM")] = artist
+            params[getattr(C, f"LYRICS_API_{n}_TITLE_PARAM")] = title
+        key_param = getattr(C, f"LYRICS_API_{n}_APIKEY_PARAM")
+        if key_param:
+            params[key_param] = getattr(C, f"LYRICS_API_{n}_APIKEY_VALUE")
+        try:
+            validate_outbound_url(url)
+            r = http_get(url, params=params,
+                         timeout=getattr(C, f"LYRICS_API_{n}_TIMEOUT"))
+            if r.status_code != 200:
+                continue
+            body = r.json()
+            value = body
+            for part in field_path.split("."):
+                if not isinstance(value, dict):
+                    value = None
+                    break
+                value = value.get(part)
+            if isinstance(value, str) and value.strip():
+                return value.strip()
+        except Exception:
+            continue
+    return None
 
 
 def score_axes(embedding: np.ndarray, axis_label_embeddings: Dict[str, np.ndarray],
@@ -121,14 +215,22 @@ class LyricsPipeline:
             axis_scores={a: INSTRUMENTAL_AXIS_FILL for a in self.axis_labels})
 
     def analyze(self, audio: Optional[torch.Tensor] = None,
-                provided_lyrics: Optional[str] = None) -> LyricsResult:
-        """analyze_lyrics (:1137): provided text wins; else VAD -> ASR ->
-        gates -> embed + axes."""
+                provided_lyrics: Optional[str] = None,
+                title: str = "", artist: str = "",
+                http_get=None) -> LyricsResult:
+        """analyze_lyrics (:1137): provided/server text wins; else the
+        external lyrics APIs (stages 1-2); else VAD -> ASR -> gates ->
+        embed + axes."""
         text: Optional[str] = None
         source = "none"
         if provided_lyrics and quality_gate(provided_lyrics, min_words=4):
             text, source = provided_lyrics, "provided"
-        elif audio is not None and self.asr_fn is not None:
+        if text is None and (title or artist):
+            api_text = fetch_external_lyrics(title, artist,
+                                             http_get=http_get)
+            if api_text and quality_gate(api_text, min_words=4):
+                text, source = api_text, "api"
+        if text is None and audio is not None and self.asr_fn is not None:
             if audio.shape[-1] > C.LYRICS_MAX_AUDIO_SECONDS * 16000:
                 audio = audio[..., : C.LYRICS_MAX_AUDIO_SECONDS * 16000]
             if self.vad is not None:
@@ -139,8 +241,12 @@ class LyricsPipeline:
                     return self._instrumental()
             transcript = self.asr_fn(audio) or ""
             words = _WORD_RE.findall(transcript)
-            transcript = " ".join(words[:300])   # reference 300-word cap
-            if quality_gate(transcript):
+            transcript = " ".join(words[:C.LYRICS_MAX_WORDS])
+            # ASR junk guard: embed only transcripts of real length
+            # (reference LYRICS_MIN_CHARS_FOR_EMBEDDING; provided/API
+            # lyrics are trusted sources and skip this gate)
+            if quality_gate(transcript) and \
+                    len(transcript) >= C.LYRICS_MIN_CHARS_FOR_EMBEDDING:
                 text, source = transcript, "asr"
         if not text:
             return self._instrumental()

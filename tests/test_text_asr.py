@@ -138,7 +138,14 @@ def test_lyrics_pipeline_vad_gates_asr():
 
     def asr(audio):
         calls.append(1)
-        return "we sing the words of a long and meaningful chorus tonight"
+        # long enough to clear LYRICS_MIN_CHARS_FOR_EMBEDDING (the
+        # reference's 250-char ASR junk gate)
+        return ("we sing the words of a long and meaningful chorus "
+                "tonight under silver city lights while the band keeps "
+                "playing that familiar melody and you have this feeling "
+                "that the night could last forever with every voice "
+                "rising higher and higher over the rooftops of the town "
+                "until morning finds us still singing")
 
     pipe = LyricsPipeline(emb, vad=vad, asr_fn=asr, vad_speech_threshold=1.1,
                           axis_labels=["love"])
@@ -150,6 +157,12 @@ def test_lyrics_pipeline_vad_gates_asr():
                            axis_labels=["love"])
     res2 = pipe2.analyze(audio=torch.randn(16000) * 0.01)
     assert calls and res2.source == "asr"
+
+    # a too-short transcript is ASR junk -> instrumental (reference
+    # LYRICS_MIN_CHARS_FOR_EMBEDDING gate applies to ASR only)
+    pipe3 = LyricsPipeline(emb, vad=vad, asr_fn=lambda a: "short words only",
+                           vad_speech_threshold=0.0, axis_labels=["love"])
+    assert pipe3.analyze(audio=torch.randn(16000) * 0.01).instrumental
     assert len(res2.text.split()) <= 300
 
 
@@ -199,3 +212,106 @@ def test_whisper_graphed_decoder_reset_across_chunks():
     assert getattr(m, "_graphed_decoder") is first          # reused, no recapture
     e2 = greedy_decode(m, mel2, max_tokens=12, use_graph=False)
     assert g2 == e2
+
+
+def test_external_lyrics_api_stage(monkeypatch):
+    """Stages 1-2 of the reference pipeline: external lyrics APIs are
+    tried before ASR (lyrics_transcriber.py:1137; LYRICS_API_* config)."""
+    from audiomuse_amd import config as C
+    from audiomuse_amd.engines.lyrics import fetch_external_lyrics
+
+    calls = []
+
+    class R:
+        def __init__(self, status, body):
+            self.status_code = status
+            self._body = body
+
+        def json(self):
+            return self._body
+
+    def fake_get(url, params=None, timeout=None):
+        calls.append((url, dict(params or {}), timeout))
+        if "lrclib" in url:
+            return R(200, {"plainLyrics": "  real lyrics text here  "})
+        return R(404, {})
+
+    # disabled -> no network at all
+    monkeypatch.setattr(C, "LYRICS_API_ENABLE", False)
+    assert fetch_external_lyrics("T", "A", http_get=fake_get) is None
+    assert not calls
+
+    monkeypatch.setattr(C, "LYRICS_API_ENABLE", True)
+    out = fetch_external_lyrics("Song", "Artist", http_get=fake_get)
+    assert out == "real lyrics text here"
+    url, params, timeout = calls[0]
+    assert params[C.LYRICS_API_1_ARTIST_PARAM] == "Artist"
+    assert params[C.LYRICS_API_1_TITLE_PARAM] == "Song"
+    assert timeout == C.LYRICS_API_1_TIMEOUT
+
+
+def test_external_lyrics_api_fallback_to_second(monkeypatch):
+    from audiomuse_amd import config as C
+    from audiomuse_amd.engines.lyrics import fetch_external_lyrics
+
+    class R:
+        def __init__(self, status, body):
+            self.status_code = status
+            self._body = body
+
+        def json(self):
+            return self._body
+
+    def fake_get(url, params=None, timeout=None):
+        if "first" in url:
+            return R(500, {})
+        return R(200, {"data": {"lyrics": "from api two"}})
+
+    monkeypatch.setattr(C, "LYRICS_API_ENABLE", True)
+    monkeypatch.setattr(C, "LYRICS_API_1_URL_TEMPLATE",
+                        "https://first.example/get")
+    monkeypatch.setattr(C, "LYRICS_API_2_URL_TEMPLATE",
+                        "https://second.example/{artist}/{title}")
+    monkeypatch.setattr(C, "LYRICS_API_2_LYRICS_FIELD", "data.lyrics")
+    out = fetch_external_lyrics("My Song", "Some Artist", http_get=fake_get)
+    assert out == "from api two"
+
+
+def test_pipeline_uses_api_before_asr(monkeypatch):
+    from audiomuse_amd import config as C
+    from audiomuse_amd.models.text import TextEmbedder
+    from tests.test_text_asr import tiny_text_cfg  # self-import safe
+
+    monkeypatch.setattr(C, "LYRICS_API_ENABLE", True)
+    asr_calls = []
+
+    class R:
+        status_code = 200
+
+        def json(self):
+            return {"plainLyrics": "you and me dancing all night long in "
+                                   "the summer rain with every star above"}
+
+    emb = TextEmbedder(tiny_text_cfg(out_dim=0, pool="cls"), seed=1)
+    pipe = LyricsPipeline(emb, vad=None,
+                          asr_fn=lambda a: asr_calls.append(1) or "x",
+                          axis_labels=["love"])
+    res = pipe.analyze(audio=torch.randn(16000), title="T", artist="A",
+                       http_get=lambda *a, **k: R())
+    assert res.source == "api" and not asr_calls
+
+
+def test_language_gates_cjk_and_confidence():
+    from audiomuse_amd.engines.lyrics import compression_ratio
+
+    assert detect_language("桜の花が風に舞い散る春の夜に君を想う") == "ja"
+    assert detect_language("我们一起走过春天的街道看花开花落") == "zh"
+    assert detect_language("사랑해요 그대와 함께 걷던 그 길을 기억해요") == "ko"
+    # ambiguous es/pt text with shared stopwords -> confident winner
+    lang, conf = detect_language(
+        "los ninos que cantan por las calles con una cancion",
+        with_confidence=True)
+    assert lang == "es" and conf >= 0.7
+    # degenerate repetition compresses extremely well -> gate fires
+    assert compression_ratio("la " * 400) > 15.0
+    assert not quality_gate("la la " * 200)
