@@ -30,14 +30,23 @@ _GENRE_WORDS = ["Dawn", "Static", "Harbor", "Neon", "Pines", "Orbit",
 class SyntheticProvider(Provider):
     def __init__(self, n_albums: int = 4, tracks_per_album: int = 5,
                  seconds: float = 12.0, sr: int = 44100, seed: int = 0,
+                 path_prefix: str = "/music", id_prefix: str = "",
                  **_ignored):
+        """path_prefix/id_prefix simulate a second server exposing the
+        SAME library under different ids and mount points — the shape
+        the provider-migration wizard has to bridge."""
         self.n_albums = n_albums
         self.tracks_per_album = tracks_per_album
         self.seconds = seconds
         self.sr = sr
         self.seed = seed
+        self.path_prefix = path_prefix.rstrip("/")
+        self.id_prefix = id_prefix
         self._playlists: Dict[str, Dict] = {}
         self._next_pl = 1
+
+    def list_libraries(self) -> List[Dict]:
+        return [{"id": "lib1", "name": "Music", "path": self.path_prefix}]
 
     # -- catalogue -----------------------------------------------------
 
@@ -47,22 +56,24 @@ class SyntheticProvider(Provider):
     def _album(self, ai: int) -> Album:
         artist = _ARTISTS[(self.seed + ai) % len(_ARTISTS)]
         name = f"{_GENRE_WORDS[ai % len(_GENRE_WORDS)]} {ai + 1}"
-        tracks = [f"a{ai}t{ti}" for ti in range(self.tracks_per_album)]
-        return Album(provider_id=f"a{ai}", name=name, author=artist,
-                     track_ids=tracks)
+        tracks = [f"{self.id_prefix}a{ai}t{ti}"
+                  for ti in range(self.tracks_per_album)]
+        return Album(provider_id=f"{self.id_prefix}a{ai}", name=name,
+                     author=artist, track_ids=tracks)
 
     def get_recent_albums(self, limit: int = 0) -> List[Album]:
         albums = [self._album(i) for i in range(self.n_albums)]
         return albums[:limit] if limit else albums
 
     def get_tracks_from_album(self, album_id: str) -> List[Track]:
-        ai = int(album_id[1:])
+        ai = int(album_id[len(self.id_prefix) + 1:])
         album = self._album(ai)
         return [
             Track(provider_id=tid, title=f"Track {ti + 1} of {album.name}",
                   author=album.author, album=album.name,
                   duration=self.seconds,
-                  file_path=f"/music/{album.author}/{album.name}/{ti + 1}.wav")
+                  file_path=f"{self.path_prefix}/{album.author}/"
+                            f"{album.name}/{ti + 1}.wav")
             for ti, tid in enumerate(album.track_ids)
         ]
 
@@ -73,7 +84,7 @@ class SyntheticProvider(Provider):
         return out
 
     def download_track(self, track_id: str) -> Optional[bytes]:
-        ai, ti = track_id[1:].split("t")
+        ai, ti = track_id[len(self.id_prefix) + 1:].split("t")
         seed = self.seed * 100003 + int(ai) * 101 + int(ti)
         audio = synthetic_track(seed, seconds=self.seconds, sr=self.sr)
         pcm = (torch.clamp(audio, -1, 1) * 32767.0).to(torch.int16).numpy()

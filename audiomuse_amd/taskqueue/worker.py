@@ -46,6 +46,7 @@ def get_handler(task_type: str) -> Optional[Callable]:
 def import_builtin_handlers() -> None:
     """Register the built-in task modules (idempotent)."""
     import audiomuse_amd.analysis.maintenance  # noqa: F401
+    import audiomuse_amd.analysis.migration  # noqa: F401
     import audiomuse_amd.analysis.tasks  # noqa: F401
     import audiomuse_amd.cluster.tasks  # noqa: F401
 

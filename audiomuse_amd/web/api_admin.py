@@ -171,21 +171,85 @@ def save_radio():
     return jsonify({"saved": name})
 
 
-# -- provider migration ------------------------------------------------------
+# -- provider migration wizard -----------------------------------------------
+# (reference: app_provider_migration.py:2828 — probe, library select,
+# path-format detection, match preview, transactional rewrite, restart
+# handshake; condensed into 4 endpoints + one queued task)
+
+@bp.post("/api/migration/probe")
+@require_auth
+def migration_probe():
+    """Wizard step 1: can we reach the target; which libraries; what do
+    its paths look like."""
+    from audiomuse_amd.analysis.migration import probe_server
+
+    body = request.get_json(force=True, silent=True) or {}
+    try:
+        out = probe_server(body.get("server_type", "synthetic"),
+                           body.get("server_config", {}))
+    except Exception as exc:  # noqa: BLE001 — wizard shows the reason
+        return jsonify({"reachable": False, "error": str(exc)}), 502
+    return jsonify(out)
+
+
+@bp.post("/api/migration/preview")
+@require_auth
+def migration_preview():
+    """Wizard steps 2-3: path rule + tiered match preview. Read-only —
+    nothing is written until /api/migration/start with apply=true."""
+    from audiomuse_amd.analysis.migration import (build_match_preview,
+                                                  propose_path_rule)
+    from audiomuse_amd.mediaserver import make_provider
+
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    provider = make_provider(body.get("server_type", "synthetic"),
+                             **body.get("server_config", {}))
+    tracks = provider.get_all_songs()
+    source = body.get("source_server_id", "default")
+    src_paths = [r["file_path"] for r in conn.execute(
+        "SELECT file_path FROM track_server_map WHERE server_id=? "
+        "AND file_path != ''", (source,)).fetchall()]
+    rule = propose_path_rule([t.file_path for t in tracks], src_paths)
+    preview = build_match_preview(conn, tracks, source, path_rule=rule)
+    return jsonify({"path_rule": rule, "tiers": preview["tiers"],
+                    "total": preview["total"],
+                    "matched": preview["matched"],
+                    "match_ratio": round(preview["match_ratio"], 4),
+                    "unmatched": preview["unmatched"][:50]})
+
 
 @bp.post("/api/migration/start")
 @require_auth
 def migration_start():
-    """Move a library's mappings to another provider (reference:
-    app_provider_migration wizard -> planner task). Runs the metadata
-    alignment sweep against the target server as a queued task."""
+    """Wizard step 4: queue the full wizard task. apply=false stops at
+    the preview; apply=true performs the transactional rewrite and the
+    restart handshake."""
     body = request.get_json(force=True, silent=True) or {}
-    tid = enqueue(_state().conn(), "multiserver_sync", {
+    tid = enqueue(_state().conn(), "provider_migration", {
         "server_type": body.get("server_type", "synthetic"),
         "server_config": body.get("server_config", {}),
-        "server_id": body.get("server_id", "migrated"),
+        "source_server_id": body.get("source_server_id", "default"),
+        "target_server_id": body.get("target_server_id", "migrated"),
+        "apply": bool(body.get("apply", False)),
+        "remove_source": bool(body.get("remove_source", False)),
+        "min_match_ratio": float(body.get("min_match_ratio", 0.5)),
     }, queue="high")
     return jsonify({"task_id": tid}), 202
+
+
+@bp.get("/api/migration/status/<task_id>")
+@require_auth
+def migration_status(task_id):
+    from audiomuse_amd.taskqueue import task_row
+
+    row = task_row(_state().conn(), task_id)
+    if row is None:
+        return jsonify({"error": "unknown task"}), 404
+    result = row["result"]
+    return jsonify({"status": row["status"], "progress": row["progress"],
+                    "details": row["details"],
+                    "result": json.loads(result) if result else None})
 
 
 @bp.get("/api/plugins")
