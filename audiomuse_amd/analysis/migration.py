@@ -38,7 +38,9 @@ def probe_server(server_type: str, server_config: Optional[Dict] = None,
                  sample: int = 50) -> Dict:
     """Reachability + libraries + a small track sample (reference:
     wizard probe step)."""
-    provider = make_provider(server_type, **(server_config or {}))
+    cfg = {k: v for k, v in (server_config or {}).items()
+           if k != "server_type"}  # callers may echo the type back
+    provider = make_provider(server_type, **cfg)
     if not provider.test_connection():
         return {"reachable": False}
     libraries = provider.list_libraries()

@@ -248,7 +248,7 @@ def test_ui_served(client_ids):
     client, _ = client_ids
     r = client.get("/")
     assert r.status_code == 200
-    assert b"AudioMuse-AMD" in r.data and b"/api/map" in r.data
+    assert b"AudioMuse-AMD" in r.data and b"app.js" in r.data
 
 
 def test_config_override_round_trip(client_ids):
