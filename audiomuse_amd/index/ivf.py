@@ -54,6 +54,14 @@ def encode_vectors(vecs: torch.Tensor, storage: str) -> torch.Tensor:
     return vecs
 
 
+def decode_vectors(enc: torch.Tensor, storage: str) -> torch.Tensor:
+    """Inverse of encode_vectors (the codec contract the scan kernels
+    assume; i8 rows are x*127 rounded, reference ivf_quant.py:42)."""
+    if storage == "i8":
+        return enc.float() / 127.0
+    return enc.float()
+
+
 def default_nlist(n: int) -> int:
     """min(8*sqrt(N), IVF_NLIST_MAX), >= 1 (paged_ivf.py:1412-1413)."""
     return max(1, min(int(8 * math.sqrt(max(n, 1))), C.IVF_NLIST_MAX))
