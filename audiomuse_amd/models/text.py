@@ -97,6 +97,8 @@ class TextEncoderConfig:
     max_len: int = 128
     out_dim: int = 512          # projection dim; 0 = raw CLS
     pool: str = "cls"           # cls | mean
+    tokenizer_json: str = ""    # path to a published tokenizer.json
+                                # (RoBERTa/GTE interop; "" = hashed)
 
 
 class TextEncoder(nn.Module):
@@ -139,15 +141,57 @@ def gte_config() -> TextEncoderConfig:
                              pool="cls", max_len=C.LYRICS_GTE_MAX_TOKENS)
 
 
+class TrainedTokenizer:
+    """Adapter for a published BPE/WordPiece vocabulary via the
+    `tokenizers` wheel (the reference loads RoBERTa/GTE vocab files,
+    clap_analyzer.py:534 / gte_onnx.py:55). Drop a HuggingFace
+    tokenizer.json next to the checkpoint and pass its path through
+    TextEncoderConfig.tokenizer_json (or AUDIOMUSE_TOKENIZER_JSON) —
+    the hashed stand-in is only the no-vocab-files fallback."""
+
+    def __init__(self, path: str, vocab_size: int, max_len: int = 128):
+        from tokenizers import Tokenizer
+        self.tk = Tokenizer.from_file(path)
+        if self.tk.get_vocab_size() > vocab_size:
+            raise ValueError(
+                f"tokenizer vocab {self.tk.get_vocab_size()} exceeds the "
+                f"model's embedding table ({vocab_size}); re-init the "
+                "model with vocab_size >= the tokenizer's")
+        self.max_len = max_len
+
+    def encode(self, text: str, max_len: Optional[int] = None) -> List[int]:
+        max_len = max_len or self.max_len
+        ids = self.tk.encode(text or "").ids[: max_len - 2]
+        return [CLS] + ids + [SEP]
+
+    def batch(self, texts: Sequence[str], max_len: Optional[int] = None
+              ) -> Tuple[torch.Tensor, torch.Tensor]:
+        rows = [self.encode(t, max_len) for t in texts]
+        L = max(len(r) for r in rows)
+        ids = torch.full((len(rows), L), PAD, dtype=torch.long)
+        mask = torch.zeros(len(rows), L, dtype=torch.bool)
+        for i, r in enumerate(rows):
+            ids[i, : len(r)] = torch.tensor(r)
+            mask[i, : len(r)] = True
+        return ids, mask
+
+
 class TextEmbedder:
     """Model + tokenizer wrapper: texts -> L2-normed embeddings
     (reference: get_text_embeddings_batch, clap_analyzer.py:534)."""
 
     def __init__(self, cfg: TextEncoderConfig, device: str = "cpu",
                  dtype: torch.dtype = torch.float32, seed: int = 0):
+        import os
         torch.manual_seed(seed)
         self.model = TextEncoder(cfg).to(device=device, dtype=dtype).eval()
-        self.tokenizer = HashTokenizer(cfg.vocab_size, cfg.max_len)
+        tok_path = (getattr(cfg, "tokenizer_json", "")
+                    or os.environ.get("AUDIOMUSE_TOKENIZER_JSON", ""))
+        if tok_path:
+            self.tokenizer = TrainedTokenizer(tok_path, cfg.vocab_size,
+                                              cfg.max_len)
+        else:
+            self.tokenizer = HashTokenizer(cfg.vocab_size, cfg.max_len)
         self.device = device
 
     @torch.inference_mode()

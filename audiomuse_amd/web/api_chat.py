@@ -83,9 +83,14 @@ def tool_text_match(args: Dict) -> List[Dict]:
 
 
 def tool_search_database(args: Dict) -> List[Dict]:
-    """Read-only metadata filters (tool_impl search_database via the
-    low-privilege role, mcp_helper)."""
-    conn = _state().conn()
+    """Read-only metadata filters under the low-privilege connection
+    (reference: tool_impl search_database via mcp_helper._ensure_ai_
+    chat_db_user :63 — PG role with SELECT-only grants; SQLite mode=ro
+    connection). AI-shaped queries can never write."""
+    from audiomuse_amd.ai.dbrole import readonly_connection
+
+    state = _state()
+    conn = readonly_connection(state.db_url, admin_conn=state.conn())
     clauses, params = [], []
     if args.get("mood"):
         clauses.append("mood_vector LIKE ?")

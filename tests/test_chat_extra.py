@@ -59,3 +59,44 @@ def test_ordered_playlist_option(chat_client):
                             json={"prompt": "12 rock songs",
                                   "order": True}).json
     assert len(body["tracks"]) >= 10
+
+
+def test_ai_readonly_connection_cannot_write(tmp_sqlite_url):
+    """The AI tool connection is a real privilege boundary (reference
+    mcp_helper low-privilege role): SELECT works, writes raise."""
+    import sqlite3
+
+    import pytest as _pytest
+
+    from audiomuse_amd.ai.dbrole import readonly_connection, reset_cache
+    from audiomuse_amd.db import connect
+    from audiomuse_amd.db.schema import init_db
+
+    admin = connect(tmp_sqlite_url)
+    init_db(admin)
+    from audiomuse_amd.db import write_txn
+    with write_txn(admin):
+        admin.execute("INSERT INTO score (item_id, title) VALUES "
+                      "('fp_1', 'Song')")
+    reset_cache()
+    ro = readonly_connection(tmp_sqlite_url)
+    assert ro.execute("SELECT title FROM score").fetchone()["title"] == "Song"
+    with _pytest.raises(sqlite3.OperationalError):
+        ro.execute("INSERT INTO score (item_id) VALUES ('evil')")
+    with _pytest.raises(sqlite3.OperationalError):
+        ro.execute("DELETE FROM score")
+    reset_cache()
+    admin.close()
+
+
+def test_search_database_tool_runs_readonly(chat_client):
+    """The chat search_database tool path goes through the read-only
+    connection and still answers filters."""
+    r = chat_client.post("/chat/api/chatPlaylist",
+                         json={"prompt": "10 rock songs faster than 90 bpm"})
+    assert r.status_code == 200
+    body = r.json
+    # the plan includes a database filter call and the read-only path
+    # produced candidates
+    assert any(c["tool"] == "search_database" for c in body["plan"])
+    assert body["tracks"]
