@@ -34,6 +34,14 @@ def main() -> None:
     an.add_argument("--server-type", default="synthetic")
     an.add_argument("--server-id", default="default")
 
+    sa = sub.add_parser("standalone")
+    sa.add_argument("--host", default="0.0.0.0")
+    sa.add_argument("--port", type=int, default=8000)
+    sa.add_argument("--db", default=None)
+    sa.add_argument("--workers", type=int, default=None,
+                    help="worker processes (default: one per GPU)")
+    sa.add_argument("--no-auth", action="store_true")
+
     sub.add_parser("bench")
 
     args, rest = ap.parse_known_args()
@@ -57,6 +65,11 @@ def main() -> None:
                       {"server_type": args.server_type,
                        "server_id": args.server_id}, queue="high")
         print(f"enqueued analysis task {tid}")
+    elif args.cmd == "standalone":
+        from audiomuse_amd.standalone import run_standalone
+
+        run_standalone(db_url=args.db, host=args.host, port=args.port,
+                       workers=args.workers, no_auth=args.no_auth)
     elif args.cmd == "bench":
         sys.argv = [sys.argv[0]] + rest
         import runpy
