@@ -215,6 +215,10 @@ LYRICS_API_2_APIKEY_PARAM = _env("LYRICS_API_2_APIKEY_PARAM", "")
 LYRICS_API_2_APIKEY_VALUE = _env("LYRICS_API_2_APIKEY_VALUE", "")
 LYRICS_API_2_TIMEOUT = _env_float("LYRICS_API_2_TIMEOUT", 5.0)
 
+# Published-vocabulary interop (models/text.py TrainedTokenizer):
+# path to a HuggingFace tokenizer.json; empty = hashed stand-in
+TOKENIZER_JSON = _env("AUDIOMUSE_TOKENIZER_JSON", "")
+
 # Whisper decode behavior (models/whisper.py decoder loop)
 WHISPER_MAX_NEW_TOKENS = _env_int("WHISPER_MAX_NEW_TOKENS", 224)
 WHISPER_NO_REPEAT_NGRAM = _env_int("WHISPER_NO_REPEAT_NGRAM", 3)

@@ -182,11 +182,11 @@ class TextEmbedder:
 
     def __init__(self, cfg: TextEncoderConfig, device: str = "cpu",
                  dtype: torch.dtype = torch.float32, seed: int = 0):
-        import os
+        from audiomuse_amd import config as C
         torch.manual_seed(seed)
         self.model = TextEncoder(cfg).to(device=device, dtype=dtype).eval()
         tok_path = (getattr(cfg, "tokenizer_json", "")
-                    or os.environ.get("AUDIOMUSE_TOKENIZER_JSON", ""))
+                    or getattr(C, "TOKENIZER_JSON", ""))
         if tok_path:
             self.tokenizer = TrainedTokenizer(tok_path, cfg.vocab_size,
                                               cfg.max_len)

@@ -21,6 +21,9 @@ _ENV_ALLOWED = {
     os.path.join("audiomuse_amd", "web", "api_chat.py"),  # AI provider creds
     os.path.join("audiomuse_amd", "parallel", "dist.py"), # torchrun env
     os.path.join("audiomuse_amd", "__main__.py"),
+    # passes the process env through to spawned worker subprocesses
+    # (HIP_VISIBLE_DEVICES pinning) — not a config read
+    os.path.join("audiomuse_amd", "standalone.py"),
 }
 
 
