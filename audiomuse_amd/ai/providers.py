@@ -82,6 +82,10 @@ def build_openai_request(provider: str, prompt: str,
         "temperature": C.AI_TOOLCALL_TEMPERATURE,
         "top_p": C.AI_TOOLCALL_TOP_P,
         "max_tokens": C.AI_TOOLCALL_NUM_PREDICT,
+        # ollama's OpenAI-compatible endpoint honors these; cloud
+        # vendors ignore unknown sampler fields
+        **({"top_k": C.AI_TOOLCALL_TOP_K, "min_p": C.AI_TOOLCALL_MIN_P}
+           if provider == "ollama" else {}),
         "tools": [{"type": "function",
                    "function": {"name": name,
                                 "parameters": schema or {"type": "object"}}}

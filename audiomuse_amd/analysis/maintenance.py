@@ -139,7 +139,25 @@ def clean_orphans_task(ctx: TaskContext, payload: Dict) -> Dict:
                         "AND server_id=?",
                         (r["provider_id"], r["server_id"]))
                     deleted += 1
-    return {"orphans": len(orphans), "deleted": deleted}
+    # CLEANING_CATALOGUE (opt-in, default OFF — the catalogue is
+    # append-only otherwise; reference keeps the same explicit switch):
+    # catalogue rows with NO mapping on ANY server may be purged.
+    purged = 0
+    if do_delete and C.CLEANING_CATALOGUE:
+        rows = conn.execute(
+            """SELECT s.item_id FROM score s
+               LEFT JOIN track_server_map m ON m.item_id = s.item_id
+               WHERE m.item_id IS NULL LIMIT ?""",
+            (C.CLEANING_SAFETY_LIMIT,)).fetchall()
+        with write_txn(conn):
+            for r in rows:
+                for table in ("embedding", "clap_embedding",
+                              "lyrics_embedding", "chromaprint", "score"):
+                    conn.execute(f"DELETE FROM {table} WHERE item_id=?",
+                                 (r["item_id"],))
+                purged += 1
+    return {"orphans": len(orphans), "deleted": deleted,
+            "catalogue_purged": purged}
 
 
 @task_handler("sonic_fingerprint")
@@ -168,7 +186,17 @@ def sonic_fingerprint_task(ctx: TaskContext, payload: Dict) -> Dict:
         "SELECT provider_id, item_id FROM track_server_map WHERE server_id=?",
         (server_id,))}
     vecs, times = [], []
+    per_album: Dict[str, int] = {}
+    cap_album = int(payload.get("max_per_album",
+                                C.SONIC_FINGERPRINT_MAX_SONGS_PER_ALBUM))
     for t in top:
+        # per-album cap on the SEEDS (reference
+        # SONIC_FINGERPRINT_MAX_SONGS_PER_ALBUM: one heavy-rotation
+        # album must not dominate the taste vector)
+        if cap_album and t.album:
+            if per_album.get(t.album, 0) >= cap_album:
+                continue
+            per_album[t.album] = per_album.get(t.album, 0) + 1
         iid = mapped.get(t.provider_id)
         v = eng.vector_for_id(iid) if iid else None
         if v is None:
@@ -178,9 +206,22 @@ def sonic_fingerprint_task(ctx: TaskContext, payload: Dict) -> Dict:
     if not vecs:
         return {"tracks": 0}
     fp = sonic_fingerprint(np.stack(vecs), times)
-    res = eng.find_similar_by_vector(torch.from_numpy(fp),
-                                     payload.get("n", 30))
-    name = payload.get("name", "Sonic Fingerprint_automatic")
+    # result size + per-seed neighborhood expansion (reference
+    # SONIC_FINGERPRINT_TOP_N_SONGS / SONIC_FINGERPRINT_NEIGHBORS: the
+    # taste-vector hits are widened with each seed's own neighbors)
+    n_out = int(payload.get("n", C.SONIC_FINGERPRINT_TOP_N_SONGS))
+    res = eng.find_similar_by_vector(torch.from_numpy(fp), n_out)
+    if C.SONIC_FINGERPRINT_NEIGHBORS > 0:
+        seen = {r["item_id"] for r in res}
+        extras = []
+        for v in vecs[: max(n_out // 5, 3)]:
+            for r in eng.find_similar_by_vector(
+                    torch.from_numpy(v), C.SONIC_FINGERPRINT_NEIGHBORS):
+                if r["item_id"] not in seen:
+                    seen.add(r["item_id"])
+                    extras.append(r)
+        res = (res + sorted(extras, key=lambda r: r["distance"]))[:n_out]
+    name = payload.get("name", C.SONIC_FINGERPRINT_CRON_PLAYLIST_NAME)
     with write_txn(conn):
         conn.execute("DELETE FROM playlist WHERE name=?", (name,))
         conn.execute(

@@ -38,6 +38,7 @@ class TrackAnalysis:
     moods: Dict[str, float] = field(default_factory=dict)
     clap_embedding: Optional[np.ndarray] = None     # 512-d
     other_features: Dict[str, float] = field(default_factory=dict)
+    chromaprint: Optional[bytes] = None             # compressed fingerprint
 
 
 class AnalysisRuntime:
@@ -46,18 +47,43 @@ class AnalysisRuntime:
     def __init__(self, device: str = "cpu", seed: int = 0,
                  enable_clap: Optional[bool] = None):
         self.device = torch.device(device)
-        self.dtype = (torch.bfloat16 if self.device.type == "cuda"
-                      else torch.float32)
+        # GPU_DTYPE (PARAMETERS.md): bf16 default; f16 for experiments
+        self.dtype = ((torch.float16 if C.GPU_DTYPE == "f16"
+                       else torch.bfloat16)
+                      if self.device.type == "cuda" else torch.float32)
         self.enable_clap = C.CLAP_ENABLED if enable_clap is None else enable_clap
         torch.manual_seed(seed)
         self.musicnn_emb = MusiCNNEmbedding().to(self.device).eval()
         self.musicnn_pred = MusiCNNPrediction().to(self.device).eval()
-        self.htsat = (HTSATEncoder(HTSATConfig())
+        self.htsat = (HTSATEncoder(
+                          HTSATConfig(out_dim=C.CLAP_EMBEDDING_DIMENSION))
                       .to(self.device, self.dtype).eval()
                       if self.enable_clap else None)
         self._clap_text: Optional[TextEmbedder] = None
         self._other_label_embs: Optional[np.ndarray] = None
         self._lyrics = None
+        self._seed = seed
+
+    def recycle_models(self) -> None:
+        """PER_Song_MODEL_RELOAD compatibility switch (reference
+        PER_SONG_MODEL_RELOAD: small-VRAM deployments drop and reload
+        models between songs; with 288 GB HBM the default keeps them
+        resident). Called by the analysis task after each album batch
+        when the flag is on."""
+        if not C.PER_SONG_MODEL_RELOAD:
+            return
+        import gc
+        torch.manual_seed(self._seed)
+        self.musicnn_emb = MusiCNNEmbedding().to(self.device).eval()
+        self.musicnn_pred = MusiCNNPrediction().to(self.device).eval()
+        if self.enable_clap:
+            self.htsat = HTSATEncoder(
+                HTSATConfig(out_dim=C.CLAP_EMBEDDING_DIMENSION)
+            ).to(self.device, self.dtype).eval()
+        self._lyrics = None
+        gc.collect()
+        if self.device.type == "cuda":
+            torch.cuda.empty_cache()
 
     def lyrics_pipeline(self):
         """Lazy lyrics pipeline (GTE embedder + VAD; ASR off by default —
@@ -86,9 +112,10 @@ class AnalysisRuntime:
         model = WhisperModel().to(self.device, self.dtype).eval()
         mel_cfg = whisper_mel_config()
 
-        def asr(audio16k: torch.Tensor) -> str:
+        def asr(audio16k: torch.Tensor):
             chunk = C.WHISPER_CHUNK_SECONDS * C.WHISPER_SAMPLE_RATE
             tokens = []
+            logprobs = []
             for s0 in range(0, audio16k.shape[-1], chunk):
                 seg = audio16k[s0 : s0 + chunk].to(self.device)
                 mel = hip_ops.mel_spectrogram(seg, mel_cfg,
@@ -98,10 +125,17 @@ class AnalysisRuntime:
                     toks = beam_decode(model, mel,
                                        beam=C.LYRICS_ASR_BEAM_SIZE)
                 else:
-                    toks = greedy_decode(model, mel)
+                    toks = greedy_decode(model, mel, return_logprob=True)
+                if isinstance(toks, tuple):
+                    toks, lp = toks
+                    logprobs.append(lp)
                 tokens.extend(toks)
             # token ids -> placeholder wordpieces (no trained vocab in-image)
-            return " ".join(f"tok{t}" for t in tokens)
+            text = " ".join(f"tok{t}" for t in tokens)
+            avg_lp = (sum(logprobs) / len(logprobs)) if logprobs else None
+            # confidence rides along for the reference's
+            # LYRICS_ASR_MIN_AVG_LOGPROB gate (engines/lyrics.py)
+            return (text, avg_lp) if avg_lp is not None else text
 
         return asr
 
@@ -137,6 +171,19 @@ class AnalysisRuntime:
         a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE)
         out.tempo, out.energy, out.key, out.scale = \
             features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+        if C.CHROMAPRINT_COLLECTION_ENABLED:
+            # acoustic fingerprint for the identity confirm gate
+            # (reference: _stage_collect_chromaprint, album.py:120)
+            from audiomuse_amd.engines import chromaprint as cp
+            try:
+                out.chromaprint = cp.compute(a16.cpu(),
+                                             C.MUSICNN_SAMPLE_RATE)
+            except Exception:  # noqa: BLE001 — fingerprint is best-effort
+                out.chromaprint = None
+        if C.LYRICS_MUSICNN_SKIP and C.LYRICS_ENABLED:
+            # lyrics-only pass (reference LYRICS_MUSICNN_SKIP): keep the
+            # cheap scalar features, skip the MusiCNN/CLAP embeddings
+            return out
 
         # MusiCNN patches (song.py:240-256): 187-frame log-mel windows
         mel = hip_ops.mel_spectrogram(a16, dsp.musicnn_mel_config())
@@ -146,7 +193,13 @@ class AnalysisRuntime:
             patches = torch.stack([mel[:, i : i + P]
                                    for i in range(0, frames - P + 1, P)])
             patches = patches.transpose(1, 2)          # (N, 187, 96)
-            emb = self.musicnn_emb(patches.float())
+            # chunked inference (reference MUSICNN_BATCH_SIZE,
+            # song.py:374-390 — bounded activations on small GPUs; the
+            # 288 GB default makes the chunks large no-ops)
+            bs = max(int(C.MUSICNN_BATCH_SIZE), 1)
+            embs = [self.musicnn_emb(patches[i : i + bs].float())
+                    for i in range(0, patches.shape[0], bs)]
+            emb = torch.cat(embs, dim=0)
             logits = self.musicnn_pred(emb)
             track_emb, moods = aggregate_track(emb, logits)
             out.embedding = track_emb.cpu().numpy().astype(np.float32)
@@ -161,8 +214,8 @@ class AnalysisRuntime:
     def _clap_embed(self, a48: torch.Tensor) -> np.ndarray:
         """Segments -> fused mel -> HTSAT -> mean + L2
         (clap_analyzer.py:432-511; int16 round-trip fused in kernel)."""
-        segs = dsp.segment_audio(a48, C.CLAP_SEGMENT_SAMPLES,
-                                 C.CLAP_SEGMENT_HOP_SAMPLES)
+        seg_len = int(C.CLAP_SEGMENT_SECONDS * C.CLAP_SAMPLE_RATE)
+        segs = dsp.segment_audio(a48, seg_len, C.CLAP_SEGMENT_HOP_SAMPLES)
         mel = hip_ops.mel_spectrogram(segs, dsp.clap_mel_config(),
                                       quantize_int16=True)
         emb = self.htsat(mel.to(self.dtype)).float()
@@ -211,7 +264,8 @@ class AnalysisRuntime:
             results.append(res)
             if self.htsat is not None:
                 a48 = resample(audio, in_sr, C.CLAP_SAMPLE_RATE)
-                segs = dsp.segment_audio(a48, C.CLAP_SEGMENT_SAMPLES,
+                segs = dsp.segment_audio(
+                    a48, int(C.CLAP_SEGMENT_SECONDS * C.CLAP_SAMPLE_RATE),
                                          C.CLAP_SEGMENT_HOP_SAMPLES)
                 seg_batches.append(segs)
                 seg_owner.extend([i] * segs.shape[0])

@@ -113,6 +113,7 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
         return {"analyzed": 0, "skipped": len(tracks) - len(todo)}
 
     results = runtime.analyze_album_batch([b for _, b in valid])
+    runtime.recycle_models()        # PER_SONG_MODEL_RELOAD compat switch
     n = 0
     # one write transaction per album (write_txn is reentrant: the
     # nested save_* helpers join it) — one fsync instead of ~3/track
@@ -129,8 +130,34 @@ def _persist_album(ctx, conn, valid, results, resolver, server_id,
         ctx.check_cancelled()
         if res is None:
             continue
+
+        def _fp_confirm(cand_id: str) -> bool:
+            """Chromaprint confirm gate (reference
+            CHROMAPRINT_GATE_ENABLED + MATCH_THRESHOLD): merging two
+            provider tracks into one recording requires the acoustic
+            fingerprints to agree when both exist."""
+            if not C.CHROMAPRINT_GATE_ENABLED or res.chromaprint is None:
+                return True
+            row = conn.execute(
+                "SELECT fingerprint FROM chromaprint WHERE item_id=?",
+                (cand_id,)).fetchone()
+            if row is None or row["fingerprint"] is None:
+                return True
+            from audiomuse_amd.engines.chromaprint import bit_match_ratio
+            return bit_match_ratio(res.chromaprint, row["fingerprint"]) \
+                >= C.CHROMAPRINT_MATCH_THRESHOLD
+
         item_id, _matched = resolver.resolve(
-            res.embedding, res.duration, server_id, track.provider_id)
+            res.embedding, res.duration, server_id, track.provider_id,
+            confirm_fn=_fp_confirm)
+        if res.chromaprint is not None:
+            conn.execute(
+                """INSERT INTO chromaprint (item_id, fingerprint, duration)
+                   VALUES (?,?,?)
+                   ON CONFLICT(item_id) DO UPDATE SET
+                       fingerprint=excluded.fingerprint,
+                       duration=excluded.duration""",
+                (item_id, res.chromaprint, res.duration))
         global _RESOLVER_EXPECT_N
         if not _matched:
             _RESOLVER_EXPECT_N += 1     # we will insert one embedding row
@@ -144,7 +171,8 @@ def _persist_album(ctx, conn, valid, results, resolver, server_id,
         if C.LYRICS_ENABLED:
             # stage 8 (album.py:276): provided lyrics win; ASR optional
             lyr = runtime.lyrics_pipeline().analyze(
-                provided_lyrics=provider.get_lyrics(track.provider_id))
+                provided_lyrics=provider.get_lyrics(track.provider_id),
+                title=track.title, artist=track.author)
             save_lyrics_embedding(conn, item_id, lyr.embedding,
                                   axis_scores=lyr.axis_scores,
                                   lyrics_text=lyr.text, language=lyr.language,
@@ -180,11 +208,17 @@ def run_analysis_task(ctx: TaskContext, payload: Dict) -> Dict:
     if not provider.test_connection():        # main.py:294 preflight
         raise RuntimeError(f"media server {server_id!r} unreachable")
 
-    albums = provider.get_recent_albums(limit=payload.get("album_limit", 0))
+    # NUM_RECENT_ALBUMS (reference): 0 = whole library, else only the
+    # newest N albums are scanned (the incremental-analysis mode).
+    # Standalone/loose tracks ride provider.get_recent_music_items
+    # (Emby exposes them as single-track pseudo-albums).
+    limit = int(payload.get("album_limit", C.NUM_RECENT_ALBUMS))
+    albums = provider.get_recent_music_items(limit=limit)
     ctx.report(5.0, f"{len(albums)} albums to scan")
 
     child_ids: List[str] = []
     dispatched = 0
+    rebuild_every = max(int(C.REBUILD_INDEX_BATCH_SIZE), 1)
     for album in albums:
         ctx.check_cancelled()
         # back-pressure (main.py: <= MAX_QUEUED_ANALYSIS_JOBS live children)
@@ -198,10 +232,15 @@ def run_analysis_task(ctx: TaskContext, payload: Dict) -> Dict:
              "server_id": server_id, "album_id": album.provider_id},
             parent_task_id=ctx.task_id))
         dispatched += 1
+        # mid-run index rebuild every REBUILD_INDEX_BATCH_SIZE albums
+        # (main.py: features go live before the full scan finishes)
+        if dispatched % rebuild_every == 0:
+            enqueue(conn, "rebuild_indexes", {},
+                    parent_task_id=ctx.task_id)
         ctx.report(5.0 + 60.0 * dispatched / max(len(albums), 1),
                    f"dispatched {dispatched}/{len(albums)}")
 
-    # drain loop (main.py stage 4)
+    # drain loop (main.py stage 4; cadence ANALYSIS_MONITOR_DB_INTERVAL)
     deadline = time.time() + payload.get("drain_timeout", 3600.0)
     while time.time() < deadline:
         ctx.check_cancelled()
@@ -210,13 +249,56 @@ def run_analysis_task(ctx: TaskContext, payload: Dict) -> Dict:
             break
         ctx.report(65.0 + 25.0 * (1 - left / max(dispatched, 1)),
                    f"{left} album jobs outstanding")
-        time.sleep(C.QUEUE_POLL_SECONDS)
+        time.sleep(min(C.ANALYSIS_MONITOR_DB_INTERVAL,
+                       C.QUEUE_POLL_SECONDS * 8))
 
     ctx.report(92.0, "rebuilding indexes")
     device = "cuda" if torch.cuda.is_available() else "cpu"
     built = run_all_index_builds(conn, device=device)
     ctx.report(100.0, "done")
     return {"albums": dispatched, "indexes": built}
+
+
+@task_handler("chromaprint_backfill")
+def chromaprint_backfill_task(ctx: TaskContext, payload: Dict) -> Dict:
+    """Backfill fingerprints for tracks analyzed before collection was
+    enabled (reference CHROMAPRINT_BACKFILL_ALBUMS_PER_RUN: bounded
+    batches per run so the cron job never monopolizes a worker)."""
+    from audiomuse_amd.engines import chromaprint as cp
+    from audiomuse_amd.ops.audio_io import load_audio
+
+    conn = ctx.conn
+    server_id = payload.get("server_id", "default")
+    provider = make_provider(payload.get("server_type", C.MEDIASERVER_TYPE),
+                             **payload.get("server_config", {}))
+    limit = int(payload.get("albums_per_run",
+                            C.CHROMAPRINT_BACKFILL_ALBUMS_PER_RUN))
+    rows = conn.execute(
+        """SELECT m.provider_id, m.item_id FROM track_server_map m
+           LEFT JOIN chromaprint c ON c.item_id = m.item_id
+           WHERE m.server_id = ? AND c.item_id IS NULL
+           LIMIT ?""", (server_id, limit * 10)).fetchall()
+    done = 0
+    for r in rows:
+        ctx.check_cancelled()
+        blob = provider.download_track(r["provider_id"])
+        if blob is None:
+            continue
+        audio, sr = load_audio(blob)
+        if audio is None:
+            continue
+        try:
+            fp = cp.compute(audio, sr)
+        except Exception:  # noqa: BLE001
+            continue
+        with write_txn(conn):
+            conn.execute(
+                """INSERT INTO chromaprint (item_id, fingerprint, duration)
+                   VALUES (?,?,?) ON CONFLICT(item_id) DO NOTHING""",
+                (r["item_id"], fp, audio.shape[-1] / sr))
+        done += 1
+        ctx.report(100.0 * done / max(len(rows), 1), f"{done} fingerprints")
+    return {"backfilled": done}
 
 
 @task_handler("rebuild_indexes")

@@ -183,10 +183,16 @@ class GMMResult:
 
 
 def gmm_fit(x: torch.Tensor, k: int, iters: int = 60, seed: int = 0,
-            reg: float = 1e-6, tol: float = 1e-4) -> GMMResult:
+            reg: float = 1e-6, tol: float = 1e-4,
+            covariance_type: str = "diag") -> GMMResult:
     """Diagonal-covariance EM (reference: artist_gmm_manager.fit_best_gmm
     uses sklearn diag GMM; clustering uses GMM option). Batched over all
-    components: responsibilities are one GEMM-shaped log-prob pass."""
+    components: responsibilities are one GEMM-shaped log-prob pass.
+    covariance_type: only the reference default "diag" is implemented
+    (GMM_COVARIANCE_TYPE); anything else fails loudly."""
+    if covariance_type != "diag":
+        raise ValueError(f"covariance_type {covariance_type!r} not "
+                         "implemented (diag only, the reference default)")
     x = x.float()
     n, d = x.shape
     k = max(1, min(k, n))
@@ -264,11 +270,17 @@ def fit_predict(algorithm: str, x: torch.Tensor, params: Dict) -> Tuple[torch.Te
                             int(params.get("min_samples", 5)))
         return labels, _centers_from_labels(x, labels)
     if algorithm == "gmm":
+        from audiomuse_amd import config as C
         r = gmm_fit(x, int(params.get("n_components", 8)),
-                    seed=int(params.get("seed", 0)))
+                    seed=int(params.get("seed", 0)),
+                    covariance_type=params.get("covariance_type",
+                                               C.GMM_COVARIANCE_TYPE))
         return r.labels, r.means
     if algorithm == "spectral":
+        from audiomuse_amd import config as C
         labels = spectral_fit(x, int(params.get("n_clusters", 8)),
+                              n_neighbors=int(params.get(
+                                  "n_neighbors", C.SPECTRAL_N_NEIGHBORS)),
                               seed=int(params.get("seed", 0)))
         return labels, _centers_from_labels(x, labels)
     raise ValueError(f"unknown algorithm {algorithm!r}")

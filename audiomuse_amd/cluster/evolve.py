@@ -62,12 +62,23 @@ def _param_space(algorithm: str, n: int, rng: random.Random) -> Dict:
     if algorithm == "gmm":
         hi = max(C.GMM_N_COMPONENTS_MIN + 1,
                  min(C.GMM_N_COMPONENTS_MAX, n // 10))
-        return {"n_components": rng.randint(C.GMM_N_COMPONENTS_MIN, hi)}
+        return {"n_components": rng.randint(C.GMM_N_COMPONENTS_MIN, hi),
+                "covariance_type": C.GMM_COVARIANCE_TYPE}
     if algorithm == "spectral":
         hi = max(C.SPECTRAL_N_CLUSTERS_MIN + 1,
                  min(C.SPECTRAL_N_CLUSTERS_MAX, n // 20))
-        return {"n_clusters": rng.randint(C.SPECTRAL_N_CLUSTERS_MIN, hi)}
+        return {"n_clusters": rng.randint(C.SPECTRAL_N_CLUSTERS_MIN, hi),
+                "n_neighbors": C.SPECTRAL_N_NEIGHBORS}
     raise ValueError(algorithm)
+
+
+def _sample_pca(rng: random.Random, dim: int) -> int:
+    """Optional PCA stage in the explored space (reference
+    PCA_COMPONENTS_MIN/MAX; 0 = off)."""
+    lo, hi = C.PCA_COMPONENTS_MIN, min(C.PCA_COMPONENTS_MAX, dim - 1)
+    if hi <= 0 or hi < lo:
+        return 0
+    return rng.randint(lo, hi)
 
 
 def _mutate(params: Dict, algorithm: str, n: int, rng: random.Random) -> Dict:
@@ -77,7 +88,13 @@ def _mutate(params: Dict, algorithm: str, n: int, rng: random.Random) -> Dict:
     di = C.MUTATION_INT_ABS_DELTA
     df = C.MUTATION_FLOAT_ABS_DELTA
     if algorithm in ("kmeans", "spectral"):
-        p["n_clusters"] = max(2, p["n_clusters"] + rng.randint(-di, di))
+        # kmeans additionally jitters proportionally to the current k
+        # (reference MUTATION_KMEANS_COORD_FRACTION's role: small
+        # relative perturbations of the solution between generations)
+        rel = (max(1, int(p["n_clusters"] * C.MUTATION_KMEANS_COORD_FRACTION))
+               if algorithm == "kmeans" else di)
+        p["n_clusters"] = max(2, p["n_clusters"]
+                              + rng.randint(-max(di, rel), max(di, rel)))
     elif algorithm == "gmm":
         p["n_components"] = max(2, p["n_components"] + rng.randint(-di, di))
     else:
@@ -165,6 +182,40 @@ def run_iteration(x: torch.Tensor, rows: Sequence[TrackRow], algorithm: str,
                            centroids=centroids)
 
 
+def _stratified_subset(rows: Sequence[TrackRow], sub: int,
+                       rng: random.Random) -> List[int]:
+    """Stratified sample by predominant mood (reference
+    _get_stratified_song_subset :1211): genres with at least
+    MIN_SONGS_PER_GENRE_FOR_STRATIFICATION members are each sampled
+    toward the STRATIFIED_SAMPLING_TARGET_PERCENTILE of genre sizes so
+    giant genres cannot crowd out the rest of the subset."""
+    by_mood: Dict[str, List[int]] = {}
+    for i, r in enumerate(rows):
+        top = max(r.mood_vector, key=r.mood_vector.get) \
+            if r.mood_vector else "unknown"
+        by_mood.setdefault(top, []).append(i)
+    big = {m: idxs for m, idxs in by_mood.items()
+           if len(idxs) >= C.MIN_SONGS_PER_GENRE_FOR_STRATIFICATION}
+    if len(big) < 2:
+        return rng.sample(range(len(rows)), sub)
+    sizes = sorted(len(v) for v in big.values())
+    pct = min(max(C.STRATIFIED_SAMPLING_TARGET_PERCENTILE, 0.0), 100.0)
+    target = sizes[min(len(sizes) - 1, int(len(sizes) * pct / 100.0))]
+    picked: List[int] = []
+    for idxs in big.values():
+        take = min(len(idxs), target)
+        picked.extend(rng.sample(idxs, take))
+    small = [i for m, idxs in by_mood.items() if m not in big
+             for i in idxs]
+    picked.extend(small)
+    if len(picked) > sub:
+        picked = rng.sample(picked, sub)
+    elif len(picked) < sub:
+        rest = list(set(range(len(rows))) - set(picked))
+        picked.extend(rng.sample(rest, min(sub - len(picked), len(rest))))
+    return picked
+
+
 def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
                         algorithm: Optional[str] = None, *,
                         runs: Optional[int] = None,
@@ -172,6 +223,7 @@ def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
                         exploit_prob: Optional[float] = None,
                         stall_limit: int = 30,
                         seed: int = 0, subset: Optional[int] = None,
+                        seed_params: Optional[Dict] = None,
                         max_songs_per_cluster: Optional[int] = None,
                         progress_cb=None) -> List[IterationResult]:
     """Explore/exploit search with an elite pool and a stall valve
@@ -189,8 +241,15 @@ def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
     elites: List[IterationResult] = []
     stall = 0
     for it in range(runs):
-        if sub < n:
-            pick = torch.tensor(rng.sample(range(n), sub), device=x.device)
+        # subset size wobbles per run (reference
+        # SAMPLING_PERCENTAGE_CHANGE_PER_RUN) so elites do not overfit
+        # one sample
+        wobble = 1.0 + rng.uniform(-C.SAMPLING_PERCENTAGE_CHANGE_PER_RUN,
+                                   C.SAMPLING_PERCENTAGE_CHANGE_PER_RUN)
+        sub_it = max(16, min(n, int(sub * wobble)))
+        if sub_it < n:
+            pick = torch.tensor(_stratified_subset(rows, sub_it, rng),
+                                device=x.device)
             xs = x[pick]
             rs = [rows[i] for i in pick.tolist()]
         else:
@@ -198,11 +257,15 @@ def evolutionary_search(x: torch.Tensor, rows: Sequence[TrackRow],
         exploit_allowed = it >= runs * C.EXPLOITATION_START_FRACTION
         if elites and exploit_allowed and rng.random() < exploit_prob:
             params = _mutate(rng.choice(elites).params, algorithm, sub, rng)
+        elif seed_params and it == 0:
+            params = dict(seed_params)   # calibration probe's winner
         else:
             params = _param_space(algorithm, sub, rng)
         params["seed"] = rng.randint(0, 2**31 - 1)
+        pca_k = _sample_pca(rng, int(x.shape[1]))
         try:
             result = run_iteration(xs, rs, algorithm, params,
+                                   pca_components=pca_k,
                                    max_songs_per_cluster=max_songs_per_cluster)
         except Exception:
             continue

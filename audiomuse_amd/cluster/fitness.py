@@ -92,10 +92,14 @@ def calinski_harabasz(x: torch.Tensor, labels: torch.Tensor) -> float:
     return (bss / (k - 1)) / (wss / (n - k))
 
 
-def purity_diversity(score_vectors: torch.Tensor, labels: torch.Tensor
-                     ) -> Dict[str, float]:
-    """Mood/other-feature purity (mean max class share per cluster) and
-    diversity (distinct predominant classes across clusters / k)."""
+def purity_diversity(score_vectors: torch.Tensor, labels: torch.Tensor,
+                     top_k: Optional[int] = None) -> Dict[str, float]:
+    """Mood/other-feature purity (mean top-k class share per cluster —
+    reference TOP_K_MOODS_FOR_PURITY_CALCULATION) and diversity
+    (distinct predominant classes across clusters / k)."""
+    if top_k is None:
+        from audiomuse_amd import config as C
+        top_k = C.TOP_K_MOODS_FOR_PURITY_CALCULATION
     m = _valid(labels)
     sv, labels = score_vectors[m].float(), labels[m]
     uniq = labels.unique()
@@ -103,12 +107,13 @@ def purity_diversity(score_vectors: torch.Tensor, labels: torch.Tensor
         return {"purity": 0.0, "diversity": 0.0}
     predominant = []
     purities = []
+    k_eff = max(1, min(int(top_k), sv.shape[1]))
     for c in uniq.tolist():
         mean_scores = sv[labels == c].mean(dim=0)
         total = float(mean_scores.sum())
-        top = int(mean_scores.argmax())
-        predominant.append(top)
-        purities.append(float(mean_scores[top]) / max(total, 1e-12))
+        topv = mean_scores.topk(k_eff).values
+        predominant.append(int(mean_scores.argmax()))
+        purities.append(float(topv.sum()) / max(total, 1e-12))
     return {
         "purity": float(torch.tensor(purities).mean()),
         "diversity": len(set(predominant)) / max(len(predominant), 1),

@@ -29,22 +29,60 @@ from audiomuse_amd.taskqueue.worker import TaskContext, task_handler
 
 
 def _load_catalogue(conn):
+    """Clustering inputs: the 200-d embeddings when
+    ENABLE_CLUSTERING_EMBEDDINGS (reference default), else the score
+    feature vector (moods + other features + tempo/energy — the
+    reference's pre-embedding clustering mode)."""
     rows = conn.execute(
         """SELECT s.item_id, s.title, s.author, s.mood_vector,
-               s.other_features, e.embedding
+               s.other_features, s.tempo, s.energy, e.embedding
            FROM score s JOIN embedding e ON e.item_id = s.item_id""").fetchall()
     tracks: List[TrackRow] = []
     vecs: List[np.ndarray] = []
+    use_emb = C.ENABLE_CLUSTERING_EMBEDDINGS
     for r in rows:
+        moods = json.loads(r["mood_vector"] or "{}")
+        others = json.loads(r["other_features"] or "{}")
         tracks.append(TrackRow(
             item_id=r["item_id"], title=r["title"] or "",
             author=r["author"] or "",
-            mood_vector=json.loads(r["mood_vector"] or "{}"),
-            other_features=json.loads(r["other_features"] or "{}")))
-        vecs.append(np.frombuffer(r["embedding"], dtype=np.float32))
+            mood_vector=moods, other_features=others))
+        if use_emb:
+            vecs.append(np.frombuffer(r["embedding"], dtype=np.float32))
+        else:
+            feat = ([moods.get(m, 0.0) for m in C.MOOD_LABELS]
+                    + [others.get(m, 0.0) for m in C.OTHER_FEATURE_LABELS]
+                    + [(r["tempo"] or 0.0) / 200.0, r["energy"] or 0.0])
+            vecs.append(np.asarray(feat, dtype=np.float32))
     if not vecs:
         return [], torch.zeros(0, 0)
     return tracks, torch.from_numpy(np.stack(vecs))
+
+
+def _calibrate(x, tracks, algorithm: str, device: str) -> Dict:
+    """Calibration probes (reference _calibrate_cluster_params :775):
+    up to CLUSTERING_CALIBRATION_MAX_TRIES quick single iterations; the
+    first parameter set that yields >= 2 clusters seeds the batches."""
+    import random as _random
+
+    if not C.CLUSTERING_AUTO_CALIBRATION or x.numel() == 0:
+        return {}
+    from audiomuse_amd.cluster.evolve import _param_space, run_iteration
+    rng = _random.Random(7)
+    sub = min(512, x.shape[0])
+    for _ in range(max(1, C.CLUSTERING_CALIBRATION_MAX_TRIES)):
+        params = _param_space(algorithm, sub, rng)
+        params["seed"] = rng.randint(0, 2**31 - 1)
+        try:
+            pick = rng.sample(range(x.shape[0]), sub)
+            res = run_iteration(x[pick].to(device),
+                                [tracks[i] for i in pick],
+                                algorithm, dict(params))
+            if len(res.playlists) >= 2:
+                return params
+        except Exception:
+            continue
+    return {}
 
 
 @task_handler("run_clustering_batch")
@@ -55,12 +93,14 @@ def run_clustering_batch_task(ctx: TaskContext, payload: Dict) -> Dict:
     tracks, x = _load_catalogue(conn)
     if x.numel() == 0:
         return {"best": None}
-    device = "cuda" if torch.cuda.is_available() else "cpu"
+    device = ("cuda" if torch.cuda.is_available()
+              and C.USE_GPU_CLUSTERING else "cpu")
     elites = evolutionary_search(
         x.to(device), tracks, payload.get("algorithm", C.CLUSTER_ALGORITHM),
         runs=payload.get("iterations", C.ITERATIONS_PER_BATCH_JOB),
         seed=payload.get("seed", 0), stall_limit=10**9,
         subset=payload.get("subset"),
+        seed_params=payload.get("seed_params") or None,
         max_songs_per_cluster=payload.get("max_songs_per_cluster", 0),
         progress_cb=lambda i, n, s: ctx.report(100.0 * i / n, f"best {s:.4f}"))
     if not elites:
@@ -81,43 +121,86 @@ def run_clustering_task(ctx: TaskContext, payload: Dict) -> Dict:
                                 C.ITERATIONS_PER_BATCH_JOB))
     n_batches = max(1, (runs + per_batch - 1) // per_batch)
     algorithm = payload.get("algorithm", C.CLUSTER_ALGORITHM)
+    max_songs = int(payload.get("max_songs_per_cluster",
+                                C.CLUSTERING_MAX_PLAYLIST_SONGS))
 
+    # calibration probes seed the search (reference :775)
+    tracks, x = _load_catalogue(conn)
+    device = ("cuda" if torch.cuda.is_available()
+              and C.USE_GPU_CLUSTERING else "cpu")
+    seed_params = _calibrate(x, tracks, algorithm, device)
+    if seed_params:
+        ctx.report(5.0, f"calibrated {seed_params}")
+
+    # dispatch with back-pressure, absorbing incrementally; no
+    # improvement across CLUSTERING_EARLY_STOP_BATCHES absorbed batches
+    # stops dispatching (reference early-stop), and the stall valve
+    # bounds total wall time (CLUSTERING_STALL_TIMEOUT_MINUTES,
+    # reference :1426-1449)
     child_ids: List[str] = []
+    absorbed: set = set()
+    best = None
+    failed = 0
+    no_improve = 0
+    stall_deadline = time.time() + payload.get(
+        "drain_timeout", C.CLUSTERING_STALL_TIMEOUT_MINUTES * 60.0)
+
+    def _absorb() -> None:
+        nonlocal best, failed, no_improve
+        for tid in child_ids:
+            if tid in absorbed:
+                continue
+            row = task_row(conn, tid)
+            if row is None:
+                absorbed.add(tid)
+                failed += 1
+                continue
+            if row["status"] not in ("SUCCESS", "FAILURE", "REVOKED"):
+                continue
+            absorbed.add(tid)
+            if row["status"] != SUCCESS:
+                failed += 1
+                continue
+            result = json.loads(row["result"] or "{}").get("best")
+            if not result:
+                no_improve += 1
+                continue
+            score = result["fitness"].get("fitness_score", -1.0)
+            if best is None or score > best["fitness"].get(
+                    "fitness_score", -1.0):
+                best = result
+                no_improve = 0
+            else:
+                no_improve += 1
+
     for b in range(n_batches):
         while qsql.pending_children(conn, ctx.task_id) >= C.MAX_CONCURRENT_BATCH_JOBS:
             ctx.check_cancelled()
+            _absorb()
             time.sleep(C.QUEUE_POLL_SECONDS)
+        _absorb()
+        if best is not None and no_improve >= C.CLUSTERING_EARLY_STOP_BATCHES:
+            ctx.report(50.0, f"early stop after {len(absorbed)} batches")
+            break
+        if time.time() > stall_deadline:
+            ctx.report(50.0, "stall valve: forcing completion")
+            break
         child_ids.append(enqueue(
             conn, "run_clustering_batch",
             {"algorithm": algorithm, "iterations": per_batch,
-             "seed": 1000 + b,
-             "max_songs_per_cluster": payload.get("max_songs_per_cluster", 0)},
+             "seed": 1000 + b, "seed_params": seed_params,
+             "max_songs_per_cluster": max_songs},
             parent_task_id=ctx.task_id))
         ctx.report(10.0 + 40.0 * (b + 1) / n_batches,
                    f"batch {b + 1}/{n_batches} dispatched")
 
-    deadline = time.time() + payload.get("drain_timeout", 3600.0)
-    while time.time() < deadline:
+    while time.time() < stall_deadline:
         ctx.check_cancelled()
         if qsql.pending_children(conn, ctx.task_id) == 0:
             break
         time.sleep(C.QUEUE_POLL_SECONDS)
+    _absorb()
 
-    # absorb (clustering.py:1581): fold batch-best results; tolerate
-    # CLUSTERING_MAX_FAILED_BATCHES dead batches
-    best = None
-    failed = 0
-    for tid in child_ids:
-        row = task_row(conn, tid)
-        if row is None or row["status"] != SUCCESS:
-            failed += 1
-            continue
-        result = json.loads(row["result"] or "{}").get("best")
-        if not result:
-            continue
-        score = result["fitness"].get("fitness_score", -1.0)
-        if best is None or score > best["fitness"].get("fitness_score", -1.0):
-            best = result
     if failed > C.CLUSTERING_MAX_FAILED_BATCHES:
         raise RuntimeError(f"{failed} clustering batches failed")
     if best is None:
@@ -126,7 +209,29 @@ def run_clustering_task(ctx: TaskContext, payload: Dict) -> Dict:
     winner = IterationResult(params=best["params"], fitness=best["fitness"],
                              playlists=best["playlists"],
                              centroids=best["centroids"])
-    top = diverse_top_n(winner, n=payload.get("top_n", C.TOP_N_PLAYLISTS))
+    if C.CLUSTERING_CLEANING:
+        top = diverse_top_n(winner, n=payload.get("top_n", C.TOP_N_PLAYLISTS))
+    else:
+        top = dict(winner.playlists)   # cleaning off: keep the raw winner
+
+    # playlist-name history: avoid reusing names from the last
+    # PLAYLIST_NAME_HISTORY_ROUNDS finalizations (reference
+    # CLUSTER_NAMING_AI_HISTORY + playlist_name_history table)
+    if C.CLUSTER_NAMING_AI_HISTORY:
+        hist_n = C.PLAYLIST_NAME_HISTORY_ROUNDS * max(len(top), 1)
+        recent = {r["name"] for r in conn.execute(
+            "SELECT name FROM playlist_name_history "
+            "ORDER BY id DESC LIMIT ?", (hist_n,)).fetchall()}
+        renamed = {}
+        for name, ids in top.items():
+            new, i = name, 2
+            while new in recent or new in renamed:
+                base = name[:-len("_automatic")] \
+                    if name.endswith("_automatic") else name
+                new = f"{base} ({i})_automatic"
+                i += 1
+            renamed[new] = ids
+        top = renamed
 
     # persist playlists (reference also pushes to the media server and
     # deletes old _automatic ones, mediaserver/__init__.py:321)

@@ -70,6 +70,14 @@ if POSTGRES_HOST:
 DATABASE_URL = _env("DATABASE_URL", _default_db)
 DATA_DIR = _env("AUDIOMUSE_DATA_DIR", os.path.expanduser("~/.audiomuse-amd"))
 TZ = _env("TZ", "UTC")
+# apply the zone process-wide (reference: TZ is env-only and applied at
+# boot, docs/ALGORITHM.md:129-135)
+os.environ.setdefault("TZ", TZ)
+try:
+    import time as _time
+    _time.tzset()
+except Exception:   # pragma: no cover — tzset is POSIX-only
+    pass
 
 # Queue semantics (reference: taskqueue/sql.py, maintenance.py)
 QUEUE_MAX_ATTEMPTS = _env_int("QUEUE_MAX_ATTEMPTS", 3)

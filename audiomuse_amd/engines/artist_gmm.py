@@ -95,10 +95,28 @@ class ArtistSimilarity:
 
     def fit_catalogue(self, per_artist: Dict[str, np.ndarray],
                       seed: int = 0) -> None:
-        for name, embs in per_artist.items():
-            if len(embs) == 0:
-                continue
-            self.models[name] = fit_artist(name, embs, seed=seed)
+        """Per-artist BIC-selected GMMs. On CPU the independent fits run
+        on an INDEX_BUILD_WORKERS thread pool (the reference's process
+        pool, artist_gmm_manager.py:219-336); on GPU the batched EM per
+        fit already saturates the device, so fits stay sequential."""
+        import torch as _torch
+
+        from audiomuse_amd import config as C
+
+        items = [(n, e) for n, e in per_artist.items() if len(e) > 0]
+        if items and not _torch.cuda.is_available() \
+                and C.INDEX_BUILD_WORKERS > 1 and len(items) > 2:
+            from concurrent.futures import ThreadPoolExecutor
+            with ThreadPoolExecutor(
+                    max_workers=C.INDEX_BUILD_WORKERS) as pool:
+                fitted = list(pool.map(
+                    lambda ne: (ne[0], fit_artist(ne[0], ne[1], seed=seed)),
+                    items))
+            for name, model in fitted:
+                self.models[name] = model
+        else:
+            for name, embs in items:
+                self.models[name] = fit_artist(name, embs, seed=seed)
         self._names = list(self.models)
         cents = [m.means.mean(axis=0) for m in self.models.values()]
         self._centroids = (np.stack(cents).astype(np.float32)

@@ -22,10 +22,16 @@ _MAX_NORM = 1.0 - 1e-5
 
 def calibrate_scale(norms: torch.Tensor,
                     percentile: Optional[float] = None,
-                    target_radius: float = 0.85) -> float:
+                    target_radius: Optional[float] = None) -> float:
     """Scale s so the percentile-P norm projects to target_radius
-    (hyperbolic_geometry.py:101)."""
-    percentile = percentile if percentile is not None else C.HYPERBOLIC_SCALE_PERCENTILE
+    (hyperbolic_geometry.py:101). Defaults: HYPERBOLIC_RADIUS_PERCENTILE
+    picks the calibrated percentile and HYPERBOLIC_RADIUS_SCALE scales
+    the 0.85 target radius (both reference knobs)."""
+    if percentile is None:
+        percentile = getattr(C, "HYPERBOLIC_RADIUS_PERCENTILE",
+                             C.HYPERBOLIC_SCALE_PERCENTILE)
+    if target_radius is None:
+        target_radius = min(0.85 * C.HYPERBOLIC_RADIUS_SCALE, 0.999)
     if norms.numel() == 0:
         return 1.0
     p = float(torch.quantile(norms.float(), percentile / 100.0))

@@ -66,12 +66,16 @@ def build_tree(embeddings: torch.Tensor, item_ids: List[str],
         top = max(moods, key=moods.get) if moods else "unknown"
         by_mood.setdefault(top, []).append(pos)
 
+    # HYPERBOLIC_RADIAL_SPREAD: display-space exponent that spreads the
+    # dense outer shell of the ball (< 1 pushes mid-radii outward)
+    spread = max(C.HYPERBOLIC_RADIAL_SPREAD, 1e-3)
+
     def track_item(pos: int) -> Dict:
         meta = metas[pos]
         return {"item_id": item_ids[pos], "type": "track",
                 "title": meta.get("title", ""),
                 "author": meta.get("author", ""),
-                "radius": round(float(radii[pos]), 4)}
+                "radius": round(float(radii[pos]) ** spread, 4)}
 
     nodes: Dict[str, Dict] = {}
     flat_ids: Dict[str, List[str]] = {}

@@ -233,13 +233,27 @@ class LyricsPipeline:
         if text is None and audio is not None and self.asr_fn is not None:
             if audio.shape[-1] > C.LYRICS_MAX_AUDIO_SECONDS * 16000:
                 audio = audio[..., : C.LYRICS_MAX_AUDIO_SECONDS * 16000]
-            if self.vad is not None:
+            if self.vad is not None and C.VAD_VOICE_RECOGNITION:
                 vdev = next(self.vad.parameters()).device
                 probs = speech_probabilities(self.vad,
                                              audio.to(vdev, torch.float32))
                 if speech_ratio(probs) < self.vad_speech_threshold:
                     return self._instrumental()
-            transcript = self.asr_fn(audio) or ""
+            asr_out = self.asr_fn(audio) or ""
+            # ASR confidence gate (reference LYRICS_ASR_MIN_AVG_LOGPROB /
+            # NON_ENGLISH_MIN_LOGPROB): asr_fn may return
+            # (text, avg_logprob); low-confidence decodes are junk
+            avg_logprob = None
+            if isinstance(asr_out, tuple):
+                transcript, avg_logprob = asr_out[0] or "", asr_out[1]
+            else:
+                transcript = asr_out
+            if avg_logprob is not None:
+                lang_guess = detect_language(transcript)
+                floor = (C.LYRICS_ASR_MIN_AVG_LOGPROB if lang_guess == "en"
+                         else C.LYRICS_ASR_NON_ENGLISH_MIN_LOGPROB)
+                if avg_logprob < floor:
+                    return self._instrumental()
             words = _WORD_RE.findall(transcript)
             transcript = " ".join(words[:C.LYRICS_MAX_WORDS])
             # ASR junk guard: embed only transcripts of real length

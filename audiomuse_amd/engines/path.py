@@ -43,13 +43,19 @@ def interpolate(a: np.ndarray, b: np.ndarray, k: int,
 
 
 def find_path(engine: SimilarityEngine, start_id: str, end_id: str,
-              length: Optional[int] = None, mode: str = "slerp",
+              length: Optional[int] = None, mode: Optional[str] = None,
               max_per_artist: Optional[int] = None) -> List[Dict]:
     """Path of ~`length` tracks from start to end (path_manager entry).
-    Default length from config (reference PATH_DEFAULT_LENGTH)."""
+    Defaults from config: PATH_DEFAULT_LENGTH; interpolation follows
+    PATH_DISTANCE_METRIC (angular -> slerp on the unit sphere,
+    euclidean/dot -> linear); PATH_FIX_SIZE backfills starved waypoints
+    so the playlist comes out at exactly `length` (the reference's
+    merge-on-starve behavior)."""
     from audiomuse_amd import config as C
     if length is None:
         length = C.PATH_DEFAULT_LENGTH
+    if mode is None:
+        mode = "slerp" if C.PATH_DISTANCE_METRIC == "angular" else "linear"
     va = engine.vector_for_id(start_id)
     vb = engine.vector_for_id(end_id)
     if va is None or vb is None:
@@ -74,8 +80,33 @@ def find_path(engine: SimilarityEngine, start_id: str, end_id: str,
                 artist_counts[author] = artist_counts.get(author, 0) + 1
             break
         if picked is None:
-            continue  # waypoint merges into its neighbor (path_fix_size)
+            continue  # waypoint merges into its neighbor
         used.add(picked["item_id"])
         path.append(picked)
     path.append({"item_id": end_id, "distance": 0.0})
+    # PATH_FIX_SIZE: starved waypoints shortened the path — backfill by
+    # re-querying the midpoints of the largest gaps until `length` is
+    # met (reference: merge-on-starve keeps the playlist size fixed)
+    if C.PATH_FIX_SIZE:
+        guard = 0
+        while len(path) < length and guard < length * 2:
+            guard += 1
+            grew = False
+            for i in range(len(path) - 1):
+                a = engine.vector_for_id(path[i]["item_id"])
+                b = engine.vector_for_id(path[i + 1]["item_id"])
+                if a is None or b is None:
+                    continue
+                mid = interpolate(a.cpu().numpy(), b.cpu().numpy(), 1,
+                                  mode=mode)[0]
+                cands = engine.find_similar_by_vector(
+                    torch.from_numpy(mid), 5, exclude=tuple(used))
+                if cands:
+                    used.add(cands[0]["item_id"])
+                    path.insert(i + 1, cands[0])
+                    grew = True
+                    if len(path) >= length:
+                        break
+            if not grew:
+                break
     return path

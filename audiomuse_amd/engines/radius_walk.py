@@ -16,7 +16,16 @@ from typing import Callable, Dict, List, Optional
 
 import numpy as np
 
-BUCKET_SIZE = 50
+BUCKET_SIZE = 50  # legacy constant; RADIUS_WALK_BUCKETS overrides
+
+
+def _bucket_size(n_cands: int) -> int:
+    """Candidates split into ~RADIUS_WALK_BUCKETS buckets (reference
+    RADIUS_WALK_BUCKETS; the 50-item bucket is its shape at the default
+    candidate count)."""
+    from audiomuse_amd import config as C
+    k = max(int(C.RADIUS_WALK_BUCKETS), 1)
+    return max(1, (n_cands + k - 1) // k)
 
 
 def _cosine_distance(a: np.ndarray, b: np.ndarray) -> float:
@@ -53,8 +62,9 @@ def execute_radius_walk(candidates: List[Dict], n: int,
         return []
     dist_fn = get_distance_fn or _cosine_distance
     cands = sorted(candidates, key=lambda c: c["dist_anchor"])
-    buckets = [cands[i : i + BUCKET_SIZE]
-               for i in range(0, len(cands), BUCKET_SIZE)]
+    bs = _bucket_size(len(cands))
+    buckets = [cands[i : i + bs]
+               for i in range(0, len(cands), bs)]
 
     cap_active = bool(eliminate_duplicates and max_songs_per_artist
                       and max_songs_per_artist > 0)
@@ -66,8 +76,13 @@ def execute_radius_walk(candidates: List[Dict], n: int,
     if first_author:
         artist_counts[first_author] = 1
 
-    window = max(3, math.ceil(n / BUCKET_SIZE))
+    window = max(3, math.ceil(n / max(bs, 1)))
     processed = 0
+    # RADIUS_INSTRUMENTATION (reference radius_walk_helper.py:18-32):
+    # per-bucket pick counts logged for walk tuning
+    from audiomuse_amd import config as C
+    instrument = C.RADIUS_INSTRUMENTATION
+    bucket_picks: Dict[int, int] = {}
     while len(picked) < n and processed < len(buckets):
         target = min(len(buckets), window)
         for bi in range(processed, target):
@@ -87,6 +102,8 @@ def execute_radius_walk(candidates: List[Dict], n: int,
                 if best is None:
                     break
                 picked.append(best["item_id"])
+                if instrument:
+                    bucket_picks[bi] = bucket_picks.get(bi, 0) + 1
                 used.add(best["item_id"])
                 prev_vec = np.asarray(best["vector"], dtype=np.float32)
                 author = best.get("author")
@@ -100,6 +117,12 @@ def execute_radius_walk(candidates: List[Dict], n: int,
         if len(picked) < n and target < len(buckets):
             window = min(len(buckets), max(window + 1, window * 2))
 
+    if instrument:
+        import logging
+        logging.getLogger(__name__).info(
+            "radius walk: %d picked over %d/%d buckets (size %d): %s",
+            len(picked), len(bucket_picks), len(buckets), bs,
+            dict(sorted(bucket_picks.items())))
     id_to_author = {c["item_id"]: c.get("author") for c in cands}
     picked = avoid_triple_adjacent(picked, id_to_author)[:n]
     dist_map = {c["item_id"]: float(c["dist_anchor"]) for c in cands}

@@ -27,12 +27,16 @@ import numpy as np
 
 from audiomuse_amd import config as C
 
-SIGNATURE_BITS = 200
-SIGNATURE_BYTES = SIGNATURE_BITS // 8          # 25
-SIGNATURE_MATCH_MAX_HAMMING = 10
+# SIMHASH_BITS / SIMHASH_BANDS / CATALOGUE_ID_SCHEME_VERSION come from
+# config so the id scheme is a deployment choice, as in the reference
+# (config.py CATALOGUE_ID_SCHEME_VERSION; 200-bit scheme 4 default —
+# changing them is an id-scheme migration, not a tuning knob)
+SIGNATURE_BITS = int(C.SIMHASH_BITS)
+SIGNATURE_BYTES = SIGNATURE_BITS // 8          # 25 at the default 200
+SIGNATURE_MATCH_MAX_HAMMING = max(int(C.SIMHASH_BANDS) - 1, 1)
 _BAND_COUNT = SIGNATURE_MATCH_MAX_HAMMING + 1
 _ID_PREFIX = "fp_"
-_ID_SCHEME = "4"
+_ID_SCHEME = str(C.CATALOGUE_ID_SCHEME_VERSION)
 _HEX_LEN = SIGNATURE_BYTES * 2                 # 50
 CANONICAL_ID_LEN = len(_ID_PREFIX) + 1 + _HEX_LEN
 _SIGNATURE_MASK = (1 << SIGNATURE_BITS) - 1
@@ -211,8 +215,15 @@ class CatalogResolver:
             self.vectors[item_id] = np.asarray(embedding, dtype=np.float32)
 
     def resolve(self, embedding: Optional[np.ndarray], duration: float,
-                server_id: str, provider_track_id: str) -> Tuple[str, bool]:
-        """Returns (canonical_id, matched_existing)."""
+                server_id: str, provider_track_id: str,
+                confirm_fn=None) -> Tuple[str, bool]:
+        """Returns (canonical_id, matched_existing).
+
+        confirm_fn(candidate_id) -> bool is the optional LAST confirm
+        gate (reference: the chromaprint bit-match gate,
+        CHROMAPRINT_GATE_ENABLED) — a candidate that passes simhash +
+        cosine + duration but fails it is treated as a different
+        recording."""
         sig = embedding_signature(embedding)
         if sig is None:
             return unsignable_id(server_id, provider_track_id), False
@@ -221,6 +232,8 @@ class CatalogResolver:
             if vec is None:
                 continue
             if cosine_distance(embedding, vec) < C.SIMHASH_CONFIRM_COSINE:
+                if confirm_fn is not None and not confirm_fn(cand_id):
+                    continue       # acoustic fingerprint disagrees
                 return cand_id, True
         item_id = mint_canonical_id(sig, self.taken)
         self.register_existing(item_id, embedding, duration)

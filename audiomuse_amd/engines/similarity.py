@@ -107,11 +107,23 @@ class SimilarityEngine:
         if eliminate_duplicates is None:
             eliminate_duplicates = C.SIMILARITY_ELIMINATE_DUPLICATES_DEFAULT
         lookback = C.DUPLICATE_DISTANCE_CHECK_LOOKBACK
-        thresh = C.DUPLICATE_DISTANCE_THRESHOLD_COSINE
+        # threshold follows the index metric (reference keeps separate
+        # cosine/euclidean knobs, ivf_manager.py:419)
+        thresh = (C.DUPLICATE_DISTANCE_THRESHOLD_EUCLIDEAN
+                  if self.index.metric == "euclidean"
+                  else C.DUPLICATE_DISTANCE_THRESHOLD_COSINE)
         exclude_set = set(exclude)
         accepted: List[Tuple[str, float]] = []
         accepted_vecs: List[torch.Tensor] = []
         artist_counts: Dict[str, int] = {}
+        # seed-mood similarity gate (reference MOOD_SIMILARITY_ENABLE:
+        # neighbors must share the seed's mood profile, not just its
+        # embedding neighborhood)
+        seed_moods = None
+        if C.MOOD_SIMILARITY_ENABLE and getattr(self, "_seed_moods", None):
+            seed_moods = self._seed_moods
+            seed_top = max(seed_moods, key=seed_moods.get)
+            top_keys = sorted(seed_moods, key=seed_moods.get)[-5:]
         for item_id, dist in cands:
             if item_id in exclude_set:
                 continue
@@ -120,6 +132,15 @@ class SimilarityEngine:
                 moods = meta.get("mood_vector") or {}
                 if moods and moods.get(mood_filter, 0.0) <= 0.0:
                     continue
+            if seed_moods:
+                moods = meta.get("mood_vector") or {}
+                if moods:
+                    if moods.get(seed_top, 0.0) < C.MOOD_SCORE_MATCH_THRESHOLD:
+                        continue
+                    drift = sum(abs(moods.get(m, 0.0) - seed_moods[m])
+                                for m in top_keys) / max(len(top_keys), 1)
+                    if drift > C.MOOD_SIMILARITY_THRESHOLD:
+                        continue
             author = (meta.get("author") or "").strip().lower()
             cap = max_per_artist if max_per_artist is not None else C.MAX_SONGS_PER_ARTIST
             if cap and author and artist_counts.get(author, 0) >= cap:
@@ -156,6 +177,8 @@ class SimilarityEngine:
         fetch = max(n * 4 + len(exclude), 32)
         cands = self._query_candidates(vec, fetch, nprobe=nprobe)
         if radius:
+            nprobe = nprobe or C.IVF_MAX_DISTANCE_NPROBE  # wider probe
+            cands = self._query_candidates(vec, fetch, nprobe=nprobe)
             cdata = []
             for item_id, dist in cands:
                 if item_id in set(exclude):
@@ -188,7 +211,14 @@ class SimilarityEngine:
         if vec is None:
             return []
         kw.setdefault("exclude", (item_id,))
-        out = self.find_similar_by_vector(vec, n, **kw)
+        # stash the seed's mood profile for the MOOD_SIMILARITY_ENABLE
+        # gate (id-anchored queries only — raw vectors carry no moods)
+        self._seed_moods = (self.meta_fn(item_id) or {}).get("mood_vector") \
+            if C.MOOD_SIMILARITY_ENABLE else None
+        try:
+            out = self.find_similar_by_vector(vec, n, **kw)
+        finally:
+            self._seed_moods = None
         self.cache.put(key, out)
         return out
 

@@ -122,6 +122,41 @@ class IVFIndex:
                                      seed=seed, group=group)
         assign = assign_to_centroids(unit, centroids)
 
+        # oversized-cell split (reference paged_ivf.py:1337-1343): any
+        # cell beyond IVF_MAX_CELL_ROWS is re-clustered with a
+        # sub-k-means; its members spread over the new sub-centroids so
+        # a scan's per-cell work stays bounded
+        max_rows = int(C.IVF_MAX_CELL_ROWS)
+        if max_rows > 0 and group is None:
+            counts0 = torch.bincount(assign, minlength=nlist)
+            over = (counts0 > max_rows).nonzero(as_tuple=True)[0]
+            if over.numel():
+                import math as _math
+                new_cents = [centroids]
+                next_id = nlist
+                for c in over.tolist():
+                    idxs = (assign == c).nonzero(as_tuple=True)[0]
+                    k_sub = min(int(_math.ceil(idxs.numel() / max_rows)),
+                                idxs.numel())
+                    if k_sub < 2:
+                        continue
+                    sub_c = minibatch_kmeans(unit[idxs], k_sub, iters=15,
+                                             batch=C.IVF_KMEANS_BATCH,
+                                             seed=seed + c + 1)
+                    sub_assign = assign_to_centroids(unit[idxs], sub_c)
+                    # sub-cluster 0 keeps cell id c; the rest append
+                    centroids[c] = sub_c[0]
+                    remap = torch.full((k_sub,), c, dtype=assign.dtype,
+                                       device=device)
+                    remap[1:] = torch.arange(
+                        next_id, next_id + k_sub - 1, device=device)
+                    assign[idxs] = remap[sub_assign]
+                    new_cents.append(sub_c[1:])
+                    next_id += k_sub - 1
+                if next_id > nlist:
+                    centroids = torch.cat(new_cents, dim=0).contiguous()
+                    nlist = next_id
+
         order = torch.argsort(assign, stable=True)
         counts = torch.bincount(assign, minlength=nlist)
         cell_off = torch.zeros(nlist + 1, dtype=torch.int32, device=device)

@@ -20,6 +20,7 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional
 
+from audiomuse_amd import config as C
 from audiomuse_amd.mediaserver import register_provider
 from audiomuse_amd.mediaserver.base import Album, Provider, Track
 from audiomuse_amd.mediaserver.http import paged
@@ -66,7 +67,10 @@ class EmbyProvider(JellyfinProvider):
 
     def get_lyrics(self, track_id: str) -> Optional[str]:
         try:
-            body = self._get(f"/Items/{track_id}/Lyrics")
+            r = self.http.get(f"{self.base_url}/Items/{track_id}/Lyrics",
+                              headers=self._headers(),
+                              timeout=C.MUSICSERVER_LYRICS_TIMEOUT)
+            body = r.json()
             lines = [l.get("Text", "") for l in (body.get("Lyrics") or [])]
             text = "\n".join(x for x in lines if x)
             return text or None

@@ -193,7 +193,10 @@ class JellyfinProvider(Provider):
 
     def get_lyrics(self, track_id: str) -> Optional[str]:
         try:
-            body = self._get(f"/Audio/{track_id}/Lyrics")
+            r = self.http.get(f"{self.base_url}/Audio/{track_id}/Lyrics",
+                              headers=self._headers(),
+                              timeout=C.MUSICSERVER_LYRICS_TIMEOUT)
+            body = r.json()
             lines = [l.get("Text", "") for l in (body.get("Lyrics") or [])]
             text = "\n".join(x for x in lines if x)
             return text or None

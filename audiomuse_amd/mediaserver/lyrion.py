@@ -163,8 +163,14 @@ class LyrionProvider(Provider):
     def get_lyrics(self, track_id: str) -> Optional[str]:
         """songinfo carries lyrics when tagged (lyrion.py:1103)."""
         try:
-            body = self._rpc("songinfo", 0, 100,
-                             f"track_id:{track_id}", "tags:L")
+            from audiomuse_amd import config as C
+            old = self.http.timeout
+            self.http.timeout = C.MUSICSERVER_LYRICS_TIMEOUT
+            try:
+                body = self._rpc("songinfo", 0, 100,
+                                 f"track_id:{track_id}", "tags:L")
+            finally:
+                self.http.timeout = old
             for entry in body.get("songinfo_loop", []) or []:
                 if "lyrics" in entry:
                     return entry["lyrics"] or None

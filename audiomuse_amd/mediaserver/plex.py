@@ -165,8 +165,10 @@ class PlexProvider(Provider):
             part = ((meta.get("Media") or [{}])[0].get("Part") or [{}])[0]
             for stream in part.get("Stream", []) or []:
                 if stream.get("streamType") == 4 and stream.get("key"):
+                    from audiomuse_amd import config as C
                     r = self.http.get(f"{self.base_url}{stream['key']}",
-                                      params={"X-Plex-Token": self.token})
+                                      params={"X-Plex-Token": self.token},
+                                      timeout=C.MUSICSERVER_LYRICS_TIMEOUT)
                     return r.text or None
             return None
         except Exception:

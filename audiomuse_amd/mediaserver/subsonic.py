@@ -19,6 +19,7 @@ import hashlib
 import secrets
 from typing import Dict, List, Optional
 
+from audiomuse_amd import config as C
 from audiomuse_amd.mediaserver import register_provider
 from audiomuse_amd.mediaserver.base import Album, Provider, Track
 from audiomuse_amd.mediaserver.http import MediaHttp
@@ -167,7 +168,13 @@ class SubsonicProvider(Provider):
 
     def get_lyrics(self, track_id: str) -> Optional[str]:
         try:
-            body = self._get("getLyrics", id=track_id)
+            p = self._params()
+            p["id"] = track_id
+            r = self.http.get(f"{self.base_url}/rest/getLyrics", params=p,
+                              timeout=C.MUSICSERVER_LYRICS_TIMEOUT)
+            body = r.json().get("subsonic-response", {})
+            if body.get("status") != "ok":
+                return None
             lyr = body.get("lyrics") or {}
             return lyr.get("value")
         except Exception:

@@ -105,4 +105,9 @@ def aggregate_track(per_patch_emb: torch.Tensor, mood_logits: torch.Tensor
     emb = per_patch_emb.mean(dim=0)
     scores = torch.sigmoid(torch.sigmoid(mood_logits).mean(dim=0))
     moods = {label: float(s) for label, s in zip(C.MOOD_LABELS, scores)}
+    # persist only the strongest TOP_N_MOODS (reference: mood_vector is
+    # a sparse top-N map, song.py mood aggregation + TOP_N_MOODS)
+    if C.TOP_N_MOODS and len(moods) > C.TOP_N_MOODS:
+        keep = sorted(moods, key=moods.get, reverse=True)[: C.TOP_N_MOODS]
+        moods = {k: moods[k] for k in keep}
     return emb, moods

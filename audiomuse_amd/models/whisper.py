@@ -338,7 +338,8 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
                   max_tokens: Optional[int] = None,
                   repetition_penalty: Optional[float] = None,
                   no_repeat_ngram: Optional[int] = None,
-                  use_graph: Optional[bool] = None) -> List[int]:
+                  use_graph: Optional[bool] = None,
+                  return_logprob: bool = False):
     """Greedy KV-cache decode of one chunk. mel (n_mels, T).
 
     On GPU the per-token step runs as one hipGraph replay
@@ -367,6 +368,7 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
     logits = model.decode_step(
         torch.tensor([prompt], device=enc.device), 0, caches, ckv)
     seq: List[int] = []
+    logprob_sum = 0.0
     cache_len = len(prompt)
     step_logits = logits[0, -1].float()
     for _ in range(max_tokens):
@@ -374,6 +376,10 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
         nxt = int(step_logits.argmax())
         if nxt == TOK_EOT:
             break
+        if return_logprob:
+            # avg token logprob feeds the reference's ASR confidence
+            # gates (LYRICS_ASR_MIN_AVG_LOGPROB)
+            logprob_sum += float(torch.log_softmax(step_logits, dim=-1)[nxt])
         seq.append(nxt)
         if gdec is not None:
             out = gdec.step(nxt, cache_len)
@@ -383,6 +389,8 @@ def greedy_decode(model: WhisperModel, mel: torch.Tensor, *,
                 ckv)
         cache_len += 1
         step_logits = out[0, -1].float()
+    if return_logprob:
+        return seq, (logprob_sum / max(len(seq), 1))
     return seq
 
 

@@ -22,6 +22,8 @@ import zipfile
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional
 
+from audiomuse_amd import config as C
+
 logger = logging.getLogger(__name__)
 
 
@@ -92,12 +94,26 @@ class PluginManager:
     def load_zip(self, name: str, blob: bytes) -> PluginAPI:
         """Extract + import plugin.py + call register(api)
         (reference: PluginManager.load :537)."""
-        dest = tempfile.mkdtemp(prefix=f"audiomuse-plugin-{name}-")
+        base = os.path.join(C.DATA_DIR, "plugins")
+        os.makedirs(base, exist_ok=True)
+        dest = tempfile.mkdtemp(prefix=f"{name}-", dir=base)
         with zipfile.ZipFile(io.BytesIO(blob)) as zf:
             _safe_extract(zf, dest)
         mod_path = os.path.join(dest, "plugin.py")
         if not os.path.exists(mod_path):
             raise FileNotFoundError("plugin zip must contain plugin.py")
+        # pip-install of plugin requirements is intentionally
+        # unsupported in this offline build; PLUGIN_ALLOW_PIP=1 fails
+        # loudly instead of silently skipping the dependency step
+        if os.path.exists(os.path.join(dest, "requirements.txt")):
+            if C.PLUGIN_ALLOW_PIP:
+                raise RuntimeError(
+                    "PLUGIN_ALLOW_PIP is set but pip installs are "
+                    "unsupported in this offline build; vendor the "
+                    "dependencies inside the plugin zip instead")
+            logger.warning("plugin %s ships requirements.txt; pip install "
+                           "is disabled (vendor dependencies in the zip)",
+                           name)
         spec = importlib.util.spec_from_file_location(
             f"audiomuse_plugin_{name}", mod_path)
         module = importlib.util.module_from_spec(spec)

@@ -18,14 +18,21 @@ from audiomuse_amd.db import write_txn
 
 ACTION_STOP_WORKERS = "stop_workers"
 ACTION_RESTART = "restart"
-DEFAULT_WINDOW_SECONDS = 60.0
+def _default_window() -> float:
+    from audiomuse_amd import config as C
+    return C.CONTROL_WINDOW_SECONDS
+
+
+DEFAULT_WINDOW_SECONDS = 60.0  # fallback when config is unavailable
 
 
 def publish_control_request(conn: sqlite3.Connection, action: str,
                             payload: Optional[dict] = None,
-                            window_seconds: float = DEFAULT_WINDOW_SECONDS
+                            window_seconds: Optional[float] = None
                             ) -> int:
     """reference: control.publish_control_request :115"""
+    if window_seconds is None:
+        window_seconds = _default_window()
     from audiomuse_amd.db import insert_returning_id
     with write_txn(conn):
         rid = insert_returning_id(

@@ -180,6 +180,8 @@ def cron_loop(conn: sqlite3.Connection, stop_event,
               poll_seconds: float = 20.0) -> None:
     """Background scheduler thread (reference: app.py cron loop)."""
     while not stop_event.wait(poll_seconds):
+        if not C.CRON_ENABLED:
+            continue  # scheduler parked (reference CRON kill switch)
         try:
             run_due_cron_jobs(conn)
         except Exception:  # noqa: BLE001

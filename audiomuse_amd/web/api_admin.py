@@ -100,6 +100,36 @@ def dashboard():
     return jsonify(refresh_dashboard_stats(_state().conn()))
 
 
+@bp.get("/api/dashboard/browse")
+@require_auth
+def dashboard_browse():
+    """Paged catalogue browser (reference: app_dashboard.py browse with
+    DASHBOARD_BROWSE_PAGE_SIZE / DASHBOARD_BROWSE_MAX_OFFSET caps)."""
+    from audiomuse_amd import config as C
+
+    offset = min(max(int(request.args.get("offset", 0)), 0),
+                 C.DASHBOARD_BROWSE_MAX_OFFSET)
+    limit = min(int(request.args.get("limit",
+                                     C.DASHBOARD_BROWSE_PAGE_SIZE)),
+                C.DASHBOARD_BROWSE_PAGE_SIZE)
+    q = (request.args.get("q") or "").strip().lower()
+    conn = _state().conn()
+    if q:
+        rows = conn.execute(
+            """SELECT item_id, title, author, album, tempo, energy
+               FROM score WHERE LOWER(title) LIKE ? OR LOWER(author) LIKE ?
+               ORDER BY item_id LIMIT ? OFFSET ?""",
+            (f"%{q}%", f"%{q}%", limit, offset)).fetchall()
+    else:
+        rows = conn.execute(
+            """SELECT item_id, title, author, album, tempo, energy
+               FROM score ORDER BY item_id LIMIT ? OFFSET ?""",
+            (limit, offset)).fetchall()
+    total = conn.execute("SELECT COUNT(*) AS n FROM score").fetchone()["n"]
+    return jsonify({"rows": [dict(r) for r in rows], "offset": offset,
+                    "limit": limit, "total": total})
+
+
 # -- alchemy anchors / radios ----------------------------------------------
 
 @bp.get("/api/alchemy/anchors")

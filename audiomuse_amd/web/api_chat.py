@@ -198,12 +198,16 @@ def run_plan(prompt: str, hints: Dict):
     planner 'one replan' rule) -> (plan, per-tool results)."""
     plan = llm_plan(prompt, hints) or heuristic_plan(hints)
     plan = validate_and_normalize_plan(plan)
-    results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
+    # per-tool candidate pools are capped at MAX_SONGS_IN_AI_PROMPT
+    # (reference: the planner context/candidate budget)
+    cap = max(int(C.MAX_SONGS_IN_AI_PROMPT), 1)
+    results = [TOOLS[c["tool"]](c.get("args", {}))[:cap] for c in plan]
     if not any(results):
         replan = validate_and_normalize_plan(heuristic_plan(hints))
         if replan != plan:
             plan = replan
-            results = [TOOLS[c["tool"]](c.get("args", {})) for c in plan]
+            results = [TOOLS[c["tool"]](c.get("args", {}))[:cap]
+                       for c in plan]
     return plan, results
 
 

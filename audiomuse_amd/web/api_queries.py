@@ -70,7 +70,10 @@ def similar_tracks():
         return jsonify({"error": "audio index not built"}), 503
     item_id = request.args.get("item_id", "")
     n = int(request.args.get("n", 10))
-    radius = request.args.get("radius_similarity", "0") in ("1", "true")
+    radius_default = "1" if C.SIMILARITY_RADIUS_DEFAULT else "0"
+    radius = request.args.get("radius_similarity",
+                              request.args.get("radius", radius_default)
+                              ) in ("1", "true")
     mood = request.args.get("mood_filter") or None
     cap = request.args.get("max_per_artist")
     # over-fetch when a server scope will drop unmapped tracks
@@ -146,9 +149,11 @@ def alchemy():
 
     res = alchemy_query(
         eng, _vecs(add_ids), _vecs(sub_ids),
-        n=int(body.get("n", 25)),
-        subtract_radius=float(body.get("subtract_radius", 0.0)),
-        temperature=float(body.get("temperature", 0.0)),
+        n=int(body["n"]) if "n" in body else None,      # ALCHEMY_DEFAULT/MAX
+        subtract_radius=(float(body["subtract_radius"])
+                         if "subtract_radius" in body else None),
+        temperature=(float(body["temperature"])
+                     if "temperature" in body else None),
         exclude=tuple(add_ids), seed=body.get("seed"))
     return jsonify(_with_meta(res))
 
@@ -245,15 +250,29 @@ def lyrics_search():
     q = request.args.get("q", "")
     if not q:
         return jsonify([])
-    emb = current_app.extensions.setdefault("gte_text", _make_gte_embedder())
+    emb = _gte_lifecycle().get()    # warm/unload countdown
     vec = emb.embed([q])[0]
     res = eng.find_similar_by_vector(vec, int(request.args.get("n", 20)))
     return jsonify(_with_meta(res))
 
 
-def _make_gte_embedder():
-    from audiomuse_amd.models.text import TextEmbedder, gte_config
-    return TextEmbedder(gte_config(), device=_state().device)
+def _gte_lifecycle():
+    """GTE embedder warm/unload cycle (reference: tasks.gte_warm_cache,
+    same shape as the CLAP text lifecycle; countdown
+    LYRICS_GTE_WARMUP_DURATION)."""
+    from audiomuse_amd.utils.resources import ModelLifecycle
+
+    lc = current_app.extensions.get("gte_text_lc")
+    if lc is None:
+        def factory():
+            from audiomuse_amd.models.text import TextEmbedder, gte_config
+            return TextEmbedder(gte_config(), device=_state().device)
+
+        lc = ModelLifecycle(factory,
+                            idle_seconds=C.LYRICS_GTE_WARMUP_DURATION)
+        current_app.extensions["gte_text_lc"] = lc
+    lc.maybe_unload()
+    return lc
 
 
 # category-weighted suggested queries (reference: tasks/query.json used by
@@ -459,9 +478,22 @@ def hyperbolic_similar():
         space = HyperbolicSpace(eng.index.vectors_f32)
         current_app.extensions["hyperbolic"] = space
         current_app.extensions["hyperbolic_n"] = eng.index.n
-    d, indices = space.similar(pos, int(request.args.get("n", 10)))
-    res = [{"item_id": eng.item_ids[int(i)], "distance": float(dd)}
-           for dd, i in zip(d, indices)]
+    # defaults/caps + candidate overfetch before meta filtering
+    # (reference HYPERBOLIC_DEFAULT_LIMIT / MAX_LIMIT /
+    # CANDIDATE_OVERFETCH)
+    n = min(int(request.args.get("n", C.HYPERBOLIC_DEFAULT_LIMIT)),
+            C.HYPERBOLIC_MAX_LIMIT)
+    fetch = min(n * max(C.HYPERBOLIC_CANDIDATE_OVERFETCH, 1),
+                eng.index.n - 1)
+    d, indices = space.similar(pos, fetch)
+    res = []
+    for dd, i in zip(d, indices):
+        iid = eng.item_ids[int(i)]
+        if _state().meta_fn(iid) is None:
+            continue   # dropped catalogue rows never surface
+        res.append({"item_id": iid, "distance": float(dd)})
+        if len(res) >= n:
+            break
     return jsonify(_with_meta(res))
 
 
@@ -558,5 +590,7 @@ def order():
         meta = state.meta_fn(i) or {}
         tracks.append({"item_id": i, **{k: meta.get(k) for k in
                                         ("tempo", "energy", "key", "scale")}})
-    ordered = order_playlist(tracks, energy_arc=bool(body.get("energy_arc")))
+    arc = (bool(body["energy_arc"]) if "energy_arc" in body
+           else C.PLAYLIST_ENERGY_ARC)
+    ordered = order_playlist(tracks, energy_arc=arc)
     return jsonify([t["item_id"] for t in ordered])

@@ -43,7 +43,7 @@ def analysis_start():
     body = request.get_json(force=True, silent=True) or {}
     conn = _state().conn()
     tid, existing = _admit_and_enqueue(conn, "run_analysis", {
-        "server_type": body.get("server_type", "synthetic"),
+        "server_type": body.get("server_type", C.MEDIASERVER_TYPE),
         "server_config": body.get("server_config", {}),
         "server_id": body.get("server_id", "default"),
         "album_limit": int(body.get("album_limit", 0)),
@@ -152,8 +152,19 @@ def create_playlist():
         (server_id,) if server_id else ()).fetchone()
     if row is None:
         return jsonify({"error": "no configured media server"}), 404
+    # per-artist cap for created playlists (reference
+    # MAX_SONGS_PER_ARTIST_PLAYLIST; 0 = uncapped)
+    cap = int(body.get("max_per_artist", C.MAX_SONGS_PER_ARTIST_PLAYLIST))
+    per_artist = {}
     provider_ids, missing = [], 0
     for iid in item_ids:
+        if cap:
+            meta = _state().meta_fn(iid) or {}
+            a = (meta.get("author") or "").strip().lower()
+            if a and per_artist.get(a, 0) >= cap:
+                continue
+            if a:
+                per_artist[a] = per_artist.get(a, 0) + 1
         m = conn.execute(
             "SELECT provider_id FROM track_server_map WHERE item_id = ? "
             "AND server_id = ?", (iid, row["server_id"])).fetchone()

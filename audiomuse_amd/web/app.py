@@ -122,7 +122,7 @@ class ProxyPrefixMiddleware:
 
 
 def create_app(db_url: Optional[str] = None, device: str = "cpu",
-               auth_disabled: bool = False) -> Flask:
+               auth_disabled: Optional[bool] = None) -> Flask:
     import os as _os
 
     app = Flask("audiomuse_amd",
@@ -133,6 +133,10 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
         app.wsgi_app = ProxyPrefixMiddleware(app.wsgi_app)
     state = AppState(db_url or C.DATABASE_URL, device=device)
     app.extensions["audiomuse"] = state
+    # AUTH_ENABLED (reference PARAMETERS.md): config-level kill switch
+    # for the auth layer; the explicit argument wins when given
+    if auth_disabled is None:
+        auth_disabled = not C.AUTH_ENABLED
     app.config["AUTH_DISABLED"] = auth_disabled
 
     from audiomuse_amd.web.api_auth import bp as auth_bp
@@ -230,6 +234,22 @@ def create_app(db_url: Optional[str] = None, device: str = "cpu",
             cron_loop(conn, _stop)
 
         _threading.Thread(target=_cron_thread, daemon=True).start()
+
+        # dashboard snapshot refresher (reference: app.py:1363-1394;
+        # cadence DASHBOARD_REFRESH_SECONDS)
+        def _dash_thread():
+            from audiomuse_amd.analysis.maintenance import \
+                refresh_dashboard_stats
+            from audiomuse_amd.db import connect as _connect
+
+            conn = _connect(state.db_url)
+            while not _stop.wait(C.DASHBOARD_REFRESH_SECONDS):
+                try:
+                    refresh_dashboard_stats(conn)
+                except Exception:  # noqa: BLE001 — stats must never kill it
+                    pass
+
+        _threading.Thread(target=_dash_thread, daemon=True).start()
 
     @app.get("/")
     def ui_index():  # L7: minimal first-party UI over the API
