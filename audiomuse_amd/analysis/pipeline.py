@@ -252,6 +252,13 @@ class AnalysisRuntime:
             a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE)
             res.tempo, res.energy, res.key, res.scale = \
                 features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+            if C.CHROMAPRINT_COLLECTION_ENABLED:
+                from audiomuse_amd.engines import chromaprint as cp
+                try:
+                    res.chromaprint = cp.compute(a16.cpu(),
+                                                 C.MUSICNN_SAMPLE_RATE)
+                except Exception:  # noqa: BLE001 — best-effort
+                    res.chromaprint = None
             mel = hip_ops.mel_spectrogram(a16, dsp.musicnn_mel_config())
             P = C.MUSICNN_PATCH_FRAMES
             if mel.shape[-1] >= P:
