@@ -69,5 +69,10 @@ def whisper_rate():
 
 
 if __name__ == "__main__":
-    soak()
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--tracks", type=int, default=100,
+                    help="total synthetic tracks (albums of 4)")
+    args = ap.parse_args()
+    soak(n_albums=max(1, args.tracks // 4))
     whisper_rate()
