@@ -176,8 +176,9 @@ class AnalysisRuntime:
             # (reference: _stage_collect_chromaprint, album.py:120)
             from audiomuse_amd.engines import chromaprint as cp
             try:
-                out.chromaprint = cp.compute(a16.cpu(),
-                                             C.MUSICNN_SAMPLE_RATE)
+                # stays on-device: only the ~(frames,12) chroma matrix
+                # crosses back to the host
+                out.chromaprint = cp.compute(a16, C.MUSICNN_SAMPLE_RATE)
             except Exception:  # noqa: BLE001 — fingerprint is best-effort
                 out.chromaprint = None
         if C.LYRICS_MUSICNN_SKIP and C.LYRICS_ENABLED:
@@ -255,7 +256,7 @@ class AnalysisRuntime:
             if C.CHROMAPRINT_COLLECTION_ENABLED:
                 from audiomuse_amd.engines import chromaprint as cp
                 try:
-                    res.chromaprint = cp.compute(a16.cpu(),
+                    res.chromaprint = cp.compute(a16,
                                                  C.MUSICNN_SAMPLE_RATE)
                 except Exception:  # noqa: BLE001 — best-effort
                     res.chromaprint = None

@@ -14,6 +14,7 @@ import sqlite3
 import time
 from typing import List, Optional
 
+from audiomuse_amd import config as C
 from audiomuse_amd.db import write_txn
 from audiomuse_amd.taskqueue import enqueue
 

@@ -20,7 +20,9 @@ def test_analysis_runtime_on_gpu():
     assert res.embedding is not None and res.embedding.shape == (200,)
     assert res.clap_embedding is not None and res.clap_embedding.shape == (512,)
     assert abs(float(np.linalg.norm(res.clap_embedding)) - 1.0) < 1e-3
-    assert len(res.moods) == 50 and len(res.other_features) == 6
+    from audiomuse_amd import config as C
+    assert len(res.moods) == C.TOP_N_MOODS    # sparse top-N mood_vector
+    assert len(res.other_features) == 6
     assert res.tempo == 0.0 or 40.0 <= res.tempo <= 200.0
 
 
