@@ -32,7 +32,14 @@ const Admin = {
           <p class="muted">Backups are portable logical snapshots (SQLite
             artifact) on both storage backends; restore refuses while
             tasks run.</p>
-          <h3>Users</h3><ul class="list" id="ad-users"></ul></section>
+          <h3>Users</h3><ul class="list" id="ad-users"></ul>
+          <h3>Catalogue browser</h3>
+          <div class="row"><input id="br-q" size="14" placeholder="filter">
+            <button onclick="Admin.browse(0)">Browse</button>
+            <button onclick="Admin.browse(Admin.brOffset +
+              Admin.brLimit)">Next page</button>
+            <span id="br-info" class="muted"></span></div>
+          <table id="ad-browse"></table></section>
 
         <section class="wide"><h2>Provider migration wizard</h2>
           <div class="steps"><span class="on">Target</span><span>Probe
@@ -189,6 +196,22 @@ const Admin = {
     await AM.api(`/api/plugins/${encodeURIComponent(name)}`,
       { method: "DELETE" });
     this.refresh();
+  },
+
+  brOffset: 0, brLimit: 0,
+
+  async browse(offset) {
+    const q = document.getElementById("br-q").value;
+    const out = await AM.api(`/api/dashboard/browse?offset=${offset}` +
+      (q ? `&q=${encodeURIComponent(q)}` : ""));
+    this.brOffset = out.offset; this.brLimit = out.limit;
+    document.getElementById("br-info").textContent =
+      `${out.offset}-${out.offset + out.rows.length} of ${out.total}`;
+    document.getElementById("ad-browse").innerHTML =
+      `<tr><th>title</th><th>artist</th><th>album</th><th>bpm</th></tr>` +
+      out.rows.map(r => `<tr><td>${AM.esc(r.title)}</td>
+        <td>${AM.esc(r.author)}</td><td>${AM.esc(r.album)}</td>
+        <td>${Math.round(r.tempo || 0)}</td></tr>`).join("");
   },
 
   async loadConfig() {

@@ -32,7 +32,8 @@ const Tasks = {
             <select id="cron-type"><option>rebuild_indexes</option>
               <option>analysis</option><option>clustering</option>
               <option>sonic_fingerprint</option><option>clean_orphans</option>
-              <option>multiserver_sync</option></select>
+              <option>multiserver_sync</option>
+              <option>chromaprint_backfill</option></select>
             <button onclick="Tasks.addCron()">Add</button></div>
           <table id="cron-table"></table></section>
       </div>`;

@@ -184,6 +184,11 @@ async function main() {
   if (!E("mg-out").textContent.includes("reachable"))
     throw new Error("migration probe failed");
 
+  // catalogue browser paging
+  await Admin.browse(0);
+  if (!E("ad-browse").innerHTML.includes("Song"))
+    throw new Error("catalogue browser empty");
+
   console.log("UI_SMOKE_OK views=" + Object.keys(AM.views).length);
 }
 
