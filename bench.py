@@ -126,12 +126,11 @@ def main() -> None:
         launch_mel(0)
         out = pipelined_step(0)        # pipeline warm (untimed)
         torch.cuda.synchronize()
-        # hipGraph the encoder inside the pipeline (launch-bound Python
-        # dispatch leaves ~0.5 ms/step on the table; the graph replays
-        # the whole encoder as one submission while the next step's mel
-        # still overlaps on its own stream). Static input buffer; mel
-        # copies in device-to-device (65 MB ~ 8 us at HBM rate).
-        if os.environ.get("AUDIOMUSE_BENCH_GRAPHS", "1") == "1":
+        # hipGraph-captured encoder inside the pipeline: measured
+        # NEUTRAL-to-NEGATIVE on hardware (same box: 9495 graphed vs
+        # 9573 eager-pipelined, gpurun_out/r2_bench6*) — the pipelined
+        # eager path is not launch-bound. Kept opt-in for experiments.
+        if os.environ.get("AUDIOMUSE_BENCH_ENCGRAPH", "0") == "1":
             try:
                 g = torch.cuda.CUDAGraph()
                 static_mel = mels[1 % 2].clone()
