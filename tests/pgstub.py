@@ -114,9 +114,11 @@ def _encode_cell(v) -> Optional[bytes]:
 class StubServer:
     """Threaded PG-wire server over one shared SQLite file."""
 
-    def __init__(self, db_path: str, require_auth: bool = True):
+    def __init__(self, db_path: str, require_auth: bool = True,
+                 auth_mode: str = "scram"):
         self.db_path = str(db_path)
         self.require_auth = require_auth
+        self.auth_mode = auth_mode  # scram | md5 | cleartext
         self._srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         self._srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
         self._srv.bind(("127.0.0.1", 0))
@@ -215,7 +217,9 @@ class StubServer:
             break
         out = b""
         if self.require_auth:
-            if not self._scram(sock):
+            handler = {"scram": self._scram, "md5": self._md5,
+                       "cleartext": self._cleartext}[self.auth_mode]
+            if not handler(sock):
                 return False
         out += _msg(b"R", struct.pack("!I", 0))
         out += _msg(b"S", b"server_version\x00pgstub 15.0\x00")
@@ -223,6 +227,37 @@ class StubServer:
         out += _msg(b"Z", b"I")
         with wlock:
             sock.sendall(out)
+        return True
+
+    def _auth_failed(self, sock: socket.socket) -> bool:
+        sock.sendall(_msg(b"E", b"SFATAL\x00C28P01\x00"
+                          b"Mpassword authentication failed\x00\x00"))
+        return False
+
+    def _read_password_msg(self, sock: socket.socket) -> bytes:
+        tag = self._read_exact(sock, 1)
+        (length,) = struct.unpack("!I", self._read_exact(sock, 4))
+        body = self._read_exact(sock, length - 4)
+        if tag != b"p":
+            raise _ClientGone()
+        return body.rstrip(b"\x00")
+
+    def _md5(self, sock: socket.socket) -> bool:
+        salt = secrets.token_bytes(4)
+        sock.sendall(_msg(b"R", struct.pack("!I", 5) + salt))
+        body = self._read_password_msg(sock)
+        inner = hashlib.md5(
+            STUB_PASSWORD.encode() + STUB_USER.encode()).hexdigest()
+        expect = b"md5" + hashlib.md5(
+            inner.encode() + salt).hexdigest().encode()
+        if not hmac.compare_digest(body, expect):
+            return self._auth_failed(sock)
+        return True
+
+    def _cleartext(self, sock: socket.socket) -> bool:
+        sock.sendall(_msg(b"R", struct.pack("!I", 3)))
+        if self._read_password_msg(sock) != STUB_PASSWORD.encode():
+            return self._auth_failed(sock)
         return True
 
     def _scram(self, sock: socket.socket) -> bool:

@@ -68,6 +68,30 @@ def test_scram_auth_and_bad_password(stub):
         connect_url(bad)
 
 
+@pytest.mark.parametrize("mode", ["md5", "cleartext"])
+def test_legacy_auth_modes(tmp_path, mode):
+    """Servers configured for md5 or password auth (pgwire.py:343-352)
+    still connect; a wrong password is refused the same way."""
+    srv = StubServer(tmp_path / f"{mode}.db", auth_mode=mode).start()
+    try:
+        conn = connect_url(srv.url)
+        assert conn.execute("SELECT 7 AS x").fetchone()["x"] == 7
+        conn.close()
+        with pytest.raises((PGError, ProtocolError, OSError)):
+            connect_url(srv.url.replace("audiomuse-test", "wrong"))
+    finally:
+        srv.stop()
+
+
+def test_executemany_accumulates_rowcount(pg):
+    pg.execute("CREATE TABLE em (v INTEGER)")
+    cur = pg.executemany("INSERT INTO em VALUES (?)",
+                         [(i,) for i in range(5)])
+    assert cur.rowcount == 5
+    rows = pg.execute("SELECT v FROM em ORDER BY v").fetchall()
+    assert [r["v"] for r in rows] == list(range(5))
+
+
 def test_type_round_trip(pg):
     pg.execute("CREATE TABLE t (i INTEGER, f DOUBLE PRECISION, s TEXT, "
                "b BYTEA, n TEXT)")
