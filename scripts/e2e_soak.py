@@ -73,6 +73,20 @@ if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--tracks", type=int, default=100,
                     help="total synthetic tracks (albums of 4)")
+    ap.add_argument("--profile", action="store_true",
+                    help="cProfile the soak and print top cumulative")
     args = ap.parse_args()
-    soak(n_albums=max(1, args.tracks // 4))
-    whisper_rate()
+    if args.profile:
+        import cProfile
+        import pstats
+
+        pr = cProfile.Profile()
+        pr.enable()
+        soak(n_albums=max(1, args.tracks // 4))
+        pr.disable()
+        stats = pstats.Stats(pr)
+        stats.sort_stats("cumulative")
+        stats.print_stats(35)
+    else:
+        soak(n_albums=max(1, args.tracks // 4))
+        whisper_rate()
