@@ -104,7 +104,9 @@ def aggregate_track(per_patch_emb: torch.Tensor, mood_logits: torch.Tensor
     MOOD_LABELS."""
     emb = per_patch_emb.mean(dim=0)
     scores = torch.sigmoid(torch.sigmoid(mood_logits).mean(dim=0))
-    moods = {label: float(s) for label, s in zip(C.MOOD_LABELS, scores)}
+    # one D2H copy for the whole vector (a per-label float() was one
+    # device sync per mood — 50 syncs/track in the album loop)
+    moods = dict(zip(C.MOOD_LABELS, map(float, scores.cpu().tolist())))
     # persist only the strongest TOP_N_MOODS (reference: mood_vector is
     # a sparse top-N map, song.py mood aggregation + TOP_N_MOODS)
     if C.TOP_N_MOODS and len(moods) > C.TOP_N_MOODS:

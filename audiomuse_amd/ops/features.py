@@ -37,6 +37,42 @@ _KS_MINOR = torch.tensor([6.33, 2.68, 3.52, 5.38, 2.60, 3.53,
 TEMPO_MIN_BPM = 40.0
 TEMPO_MAX_BPM = 200.0
 
+# constant per-(sr, fft, device) tensors, built once: rebuilding the
+# mel filterbank / chroma projection per call was a host-side matrix
+# build + H2D upload for EVERY track of the album loop
+_FB_CACHE: dict = {}
+_CHROMA_W_CACHE: dict = {}
+
+
+def _onset_fb(sr: int, n_fft: int, n_mels: int,
+              device: torch.device) -> torch.Tensor:
+    key = (sr, n_fft, n_mels, str(device))
+    fb = _FB_CACHE.get(key)
+    if fb is None:
+        fb = torch.from_numpy(
+            mel_filterbank(sr, n_fft, n_mels, 0.0, sr / 2.0)).to(device)
+        _FB_CACHE[key] = fb
+    return fb
+
+
+def _chroma_weights(sr: int, n_fft: int,
+                    device: torch.device) -> torch.Tensor:
+    key = (sr, n_fft, str(device))
+    wt = _CHROMA_W_CACHE.get(key)
+    if wt is None:
+        freqs = np.linspace(0, sr / 2.0, 1 + n_fft // 2)
+        with np.errstate(divide="ignore"):
+            midi = 69.0 + 12.0 * np.log2(np.maximum(freqs, 1e-9) / 440.0)
+        pc = np.mod(midi, 12.0)
+        w = np.zeros((12, len(freqs)), dtype=np.float32)
+        usable = (freqs >= 30.0) & (freqs <= 5000.0)
+        for c in range(12):
+            d = np.minimum(np.abs(pc - c), 12.0 - np.abs(pc - c))
+            w[c] = np.exp(-0.5 * (d / 1.0) ** 2) * usable
+        wt = torch.from_numpy(w).to(device)
+        _CHROMA_W_CACHE[key] = wt
+    return wt
+
 
 def estimate_energy(audio: torch.Tensor, frame: int = 2048,
                     hop: int = 512) -> float:
@@ -60,8 +96,7 @@ def onset_envelope(audio: torch.Tensor, sr: int, n_fft: int = 2048,
                    hop: int = 512, n_mels: int = 128) -> torch.Tensor:
     """Spectral-flux onset strength (librosa.onset.onset_strength analog)."""
     power = power_spectrogram(audio.float().flatten(), n_fft, hop, center=True)
-    fb = torch.from_numpy(mel_filterbank(sr, n_fft, n_mels, 0.0, sr / 2.0)
-                          ).to(audio.device)
+    fb = _onset_fb(sr, n_fft, n_mels, audio.device)
     mel = fb @ power
     log_mel = torch.log10(torch.clamp(mel, min=1e-10))
     flux = torch.clamp(log_mel[:, 1:] - log_mel[:, :-1], min=0.0)
@@ -89,9 +124,12 @@ def estimate_tempo(audio: torch.Tensor, sr: int, hop: int = 512) -> float:
     prior = torch.exp(-0.5 * ((torch.log2(bpm) - math.log2(120.0)) / 1.0) ** 2)
     valid = (bpm >= 20.0) & (bpm <= 400.0)
     score = ac[1:] * prior * valid
-    if float(score.max()) <= 0:
+    # one D2H transfer for (best score, best bpm) instead of three
+    best = score.argmax()
+    pair = torch.stack([score[best], bpm[best]]).cpu()
+    if float(pair[0]) <= 0:
         return 0.0
-    tempo = float(bpm[int(score.argmax())])
+    tempo = float(pair[1])
     if tempo <= 0:
         return 0.0
     while tempo < TEMPO_MIN_BPM:
@@ -106,17 +144,9 @@ def chroma_from_stft(audio: torch.Tensor, sr: int, n_fft: int = 4096,
     """(12, frames) chroma via pitch-class projection of |STFT|."""
     power = power_spectrogram(audio.float().flatten(), n_fft, hop, center=True)
     mag = power.sqrt()
-    freqs = np.linspace(0, sr / 2.0, 1 + n_fft // 2)
-    with np.errstate(divide="ignore"):
-        midi = 69.0 + 12.0 * np.log2(np.maximum(freqs, 1e-9) / 440.0)
-    pc = np.mod(midi, 12.0)
     # gaussian weighting of each bin onto its nearest pitch classes
-    w = np.zeros((12, len(freqs)), dtype=np.float32)
-    usable = (freqs >= 30.0) & (freqs <= 5000.0)
-    for c in range(12):
-        d = np.minimum(np.abs(pc - c), 12.0 - np.abs(pc - c))
-        w[c] = np.exp(-0.5 * (d / 1.0) ** 2) * usable
-    wt = torch.from_numpy(w).to(audio.device)
+    # (constant per (sr, n_fft, device) — cached)
+    wt = _chroma_weights(sr, n_fft, audio.device)
     chroma = wt @ mag
     return chroma
 
