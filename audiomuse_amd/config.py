@@ -438,6 +438,9 @@ CLAP_GPU_BATCH = _env_int("CLAP_GPU_BATCH", 256)
 # headline bench stays bf16 regardless of this flag unless --fp8 is passed
 CLAP_FP8_SERVING = _env_bool("AUDIOMUSE_FP8_SERVING", False)
 FP8_HIDDEN_ENABLE = _env_bool("AUDIOMUSE_FP8_HIDDEN", False)
+# fp8-ingest attention inside the fp8 serving mode (QKV GEMM emits e4m3,
+# the window kernel reads 8-byte fragments); self-disables without an algo
+FP8_ATTN_ENABLE = _env_bool("AUDIOMUSE_FP8_ATTN", True)
 HIP_REQUIRE_NATIVE = _env_bool("HIP_REQUIRE_NATIVE", True)  # fail loudly on GPU without .so
 RCCL_BUCKET_CAP_MB = _env_int("RCCL_BUCKET_CAP_MB", 64)
 

@@ -177,25 +177,46 @@ class SwinBlock(nn.Module):
             from audiomuse_amd.ops import _ext, fp8
             ext = _ext.require()
             use_fp8 = fp8.serving_enabled() and fp8.available(x.device)
+            out = None
             if use_fp8:
                 # LN emits e4m3 directly (delayed scale); fp8 GEMM with
                 # searched algo — no standalone quantize pass
                 xq, xs = fp8.ln_fp8(self.norm1, x)
                 wq, ws = fp8.quantize_weight(self.attn.qkv.weight)
-                qkv = ext.linear_fp8(
-                    xq, wq, xs, ws,
-                    self.attn.qkv.bias.to(torch.bfloat16).contiguous())
+                qkv_bias = self.attn.qkv.bias.to(torch.bfloat16).contiguous()
+                if self.window == 8 and fp8.FP8_ATTN[0]:
+                    # fp8-ingest attention: the QKV GEMM emits e4m3
+                    # (D-scale epilogue), halving the gather bytes that
+                    # bound the latency-limited window kernel; MFMAs and
+                    # P stay bf16 inside the kernel.
+                    try:
+                        qsc, qinv, qamax = fp8.hidden_state(self.attn.qkv,
+                                                            x.device)
+                        qkv8 = ext.linear_fp8(xq, wq, xs, ws, qkv_bias,
+                                              d_inv_scale=qinv)
+                        samp = qkv8.reshape(-1)[: 1 << 22]
+                        qamax.copy_(samp.float().abs().amax() * qsc)
+                        out = ext.window_attn_fp8_fwd(
+                            qkv8.view(B, H, W, 3 * C),
+                            self.attn.full_bias_bf16(), qsc,
+                            self.attn.heads, self.shift, self.attn.scale)
+                    except RuntimeError as exc:  # no fp8-D algo here
+                        fp8.FP8_ATTN[0] = False
+                        fp8.FP8_ATTN_ERR = str(exc)
+                if out is None:
+                    qkv = ext.linear_fp8(xq, wq, xs, ws, qkv_bias)
             else:
                 xn = self.norm1(x)
                 # routed through the extension for the timed algo search
                 qkv = ext.linear_bias(xn.contiguous(),
                                       self.attn.qkv.weight.contiguous(),
                                       self.attn.qkv.bias.contiguous())
-            attn_fwd = (ext.window_attn_fwd if self.window == 8
-                        else ext.window_attn4_fwd)
-            out = attn_fwd(
-                qkv.view(B, H, W, 3 * C), self.attn.full_bias_bf16(),
-                self.attn.heads, self.shift, self.attn.scale)
+            if out is None:
+                attn_fwd = (ext.window_attn_fwd if self.window == 8
+                            else ext.window_attn4_fwd)
+                out = attn_fwd(
+                    qkv.view(B, H, W, 3 * C), self.attn.full_bias_bf16(),
+                    self.attn.heads, self.shift, self.attn.scale)
             out = out.view(B, L, C)
             # proj stays bf16 in fp8 mode too: its GEMM gain is smaller
             # than any quantize cost at C x C shapes (fp8_shapes.py)

@@ -488,4 +488,273 @@ void launch_window_attn(const void* qkv, void* out, const void* bias, int Bn,
                      (const __bf16*)bias, Bn, H, W, C, heads, shift, scale);
 }
 
+// ---------------------------------------------------------------------------
+// fp8-ingest variant (opt-in serving mode): QKV arrives as OCP e4m3
+// straight from the projection GEMM's fp8-D epilogue, HALVING the
+// global gather bytes that bound this kernel (r2 PMC: latency-bound at
+// MfmaUtil 3.3% — compute is free, bytes are not). The MFMAs stay bf16:
+// e4m3 -> bf16 is exact, P keeps full bf16 precision (no fp8 P
+// quantization), and only the Q/K/V loads shrink. The per-tensor
+// dequant scale is read from device memory (delayed scaling — no host
+// sync); S folds qs^2 into the softmax scale, O folds qs once.
+// ---------------------------------------------------------------------------
+
+typedef float f32x2 __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ bf16x8 am_fp8x8_to_bf16(unsigned long long v) {
+  const unsigned int lo = (unsigned int)v, hi = (unsigned int)(v >> 32);
+  const f32x2 a = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+  const f32x2 b = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+  const f32x2 c = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+  const f32x2 d = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+  bf16x8 r;
+  r[0] = (__bf16)a[0]; r[1] = (__bf16)a[1];
+  r[2] = (__bf16)b[0]; r[3] = (__bf16)b[1];
+  r[4] = (__bf16)c[0]; r[5] = (__bf16)c[1];
+  r[6] = (__bf16)d[0]; r[7] = (__bf16)d[1];
+  return r;
+}
+
+// LDS per wave: VT bytes [d 32][t 64] (2 KiB) + P half-tile bf16 (4 KiB)
+constexpr int WAVE_LDS_FP8 = 32 * 64 + (64 * 32) * 2;  // bytes
+
+__global__ __launch_bounds__(64, 4) void window_attn_fp8_kernel(
+    const unsigned char* __restrict__ qkv,  // (B, H, W, 3C) e4m3
+    __bf16* __restrict__ out,               // (B, H, W, C) bf16
+    const __bf16* __restrict__ bias,        // (heads, 64, 64) bf16
+    const float* __restrict__ qs_ptr,       // per-tensor dequant scale
+    int Bn, int H, int W, int C, int heads, int shift, float sm_scale) {
+  extern __shared__ unsigned char lds8[];
+  const int lane = threadIdx.x & 63;
+
+  const float qs = *qs_ptr;
+  const float eff_scale = sm_scale * qs * qs;
+
+  const int nWw = W >> 3;
+  const int nWh = H >> 3;
+  const int win = blockIdx.x;
+  const int b = win / (nWh * nWw);
+  const int wrem = win - b * (nWh * nWw);
+  const int wh = wrem / nWw;
+  const int ww = wrem - wh * nWw;
+
+  unsigned char* VTb = lds8;                       // [32][64] bytes
+  __bf16* P = (__bf16*)(lds8 + 32 * 64);           // [64][32] bf16 half
+
+  auto src_of = [&](int t, int& si, int& sj, int& wrap) {
+    const int ri = t >> 3, ci = t & 7;
+    int gi = wh * 8 + ri + shift;
+    int gj = ww * 8 + ci + shift;
+    const int wr = gi >= H;
+    const int wc = gj >= W;
+    si = wr ? gi - H : gi;
+    sj = wc ? gj - W : gj;
+    wrap = (wr << 1) | wc;
+  };
+
+  int my_si, my_sj, my_wrap;
+  src_of(lane, my_si, my_sj, my_wrap);
+  const long long my_base = (((long long)b * H + my_si) * W + my_sj) * 3 * C;
+  const unsigned long long wrap_r_mask = __ballot(my_wrap & 2);
+  const unsigned long long wrap_c_mask = __ballot(my_wrap & 1);
+
+  const int h = blockIdx.y;
+  if (h < heads) {
+    // ---- Q/K fragments: 8 BYTES per frag (was 16), converted to bf16
+    // in registers (exact) ----
+    bf16x8 qf[4], kf[4];
+    const int kk = 8 * (lane >> 4);
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr) {
+      const int t = tr * 16 + (lane & 15);
+      int si, sj, wr_;
+      src_of(t, si, sj, wr_);
+      const long long base =
+          (((long long)b * H + si) * W + sj) * 3 * C + h * 32;
+      qf[tr] = am_fp8x8_to_bf16(*(const unsigned long long*)(qkv + base + kk));
+      kf[tr] = am_fp8x8_to_bf16(
+          *(const unsigned long long*)(qkv + base + C + kk));
+    }
+
+    // ---- stage V transposed as raw e4m3 bytes: VT[d][t] ----
+    {
+      const unsigned char* vptr = qkv + my_base + 2 * C + h * 32;
+      unsigned long long vv[4];
+#pragma unroll
+      for (int g = 0; g < 4; ++g)
+        vv[g] = *(const unsigned long long*)(vptr + g * 8);
+#pragma unroll
+      for (int g = 0; g < 4; ++g)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          VTb[AM_SWZ(g * 8 + j, lane)] = (unsigned char)(vv[g] >> (8 * j));
+    }
+
+    // ---- S = QK^T, bias double-buffered exactly as the bf16 kernel ----
+    const int col_in_tile = lane & 15;
+    const int row_grp = (lane >> 4) * 4;
+    const __bf16* bias_base = bias + (h * 64 + row_grp) * 64 + col_in_tile;
+    float bv[2][4][4];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc)
+        bv[0][reg][tc] = (float)bias_base[reg * 64 + tc * 16];
+
+    f32x4 s[4][4];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 4; ++tc) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        s[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[tr], kf[tc],
+                                                            acc, 0, 0, 0);
+      }
+
+    float rmax[4][4];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr) {
+      if (tr < 3) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+#pragma unroll
+          for (int tc = 0; tc < 4; ++tc)
+            bv[(tr + 1) & 1][reg][tc] =
+                (float)bias_base[((tr + 1) * 16 + reg) * 64 + tc * 16];
+      }
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+        const int rwrap = (((wrap_r_mask >> row) & 1ull) << 1) |
+                          ((wrap_c_mask >> row) & 1ull);
+        float m = -1e30f;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const int col = tc * 16 + col_in_tile;
+          const int cwrap = (((wrap_r_mask >> col) & 1ull) << 1) |
+                            ((wrap_c_mask >> col) & 1ull);
+          float v = s[tr][tc][reg] * eff_scale + bv[tr & 1][reg][tc];
+          if (shift && rwrap != cwrap) v = -1e30f;
+          s[tr][tc][reg] = v;
+          m = fmaxf(m, v);
+        }
+        rmax[tr][reg] = m;
+      }
+    }
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float m = rmax[tr][reg];
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1)
+          m = fmaxf(m, __shfl_xor(m, d, 64));
+        rmax[tr][reg] = m;
+      }
+    float rsum[4][4];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float sum = 0.f;
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const float e = __expf(s[tr][tc][reg] - rmax[tr][reg]);
+          s[tr][tc][reg] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int d = 1; d < 16; d <<= 1) sum += __shfl_xor(sum, d, 64);
+        rsum[tr][reg] = sum;
+      }
+
+    // ---- P half 0 to LDS (bf16), half 1 parked in registers ----
+    __syncthreads();
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+        const float inv = __frcp_rn(rsum[tr][reg] + 1e-20f);
+#pragma unroll
+        for (int tc = 0; tc < 4; ++tc) {
+          const float pv = s[tr][tc][reg] * inv;
+          if (tc < 2)
+            P[AM_SWZ32(row, tc * 16 + col_in_tile)] = (__bf16)pv;
+          else
+            s[tr][tc][reg] = pv;
+        }
+      }
+    __syncthreads();
+
+    // ---- O = P @ V, K-step 0 (V fragments convert from LDS bytes) ----
+    f32x4 o[4][2];
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 2; ++tc) {
+        bf16x8 pa = *(const bf16x8*)(
+            P + AM_SWZ32(tr * 16 + (lane & 15), kk));
+        bf16x8 vb = am_fp8x8_to_bf16(*(const unsigned long long*)(
+            VTb + AM_SWZ(tc * 16 + (lane & 15), kk)));
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        o[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc,
+                                                            0, 0, 0);
+      }
+
+    __syncthreads();
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = tr * 16 + row_grp + reg;
+#pragma unroll
+        for (int tc = 2; tc < 4; ++tc)
+          P[AM_SWZ32(row, (tc - 2) * 16 + col_in_tile)] =
+              (__bf16)s[tr][tc][reg];
+      }
+    __syncthreads();
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int tc = 0; tc < 2; ++tc) {
+        bf16x8 pa = *(const bf16x8*)(
+            P + AM_SWZ32(tr * 16 + (lane & 15), kk));
+        bf16x8 vb = am_fp8x8_to_bf16(*(const unsigned long long*)(
+            VTb + AM_SWZ(tc * 16 + (lane & 15), 32 + kk)));
+        o[tr][tc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb,
+                                                            o[tr][tc],
+                                                            0, 0, 0);
+      }
+
+    // ---- scatter O (dequantized by qs) ----
+#pragma unroll
+    for (int tr = 0; tr < 4; ++tr)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int t = tr * 16 + row_grp + reg;
+        int si, sj, w_;
+        src_of(t, si, sj, w_);
+        __bf16* op =
+            out + (((long long)b * H + si) * W + sj) * C + h * 32;
+#pragma unroll
+        for (int tc = 0; tc < 2; ++tc)
+          op[tc * 16 + col_in_tile] = (__bf16)(o[tr][tc][reg] * qs);
+      }
+    __syncthreads();
+  }
+}
+
+void launch_window_attn_fp8(const void* qkv, void* out, const void* bias,
+                            const void* qs_ptr, int Bn, int H, int W, int C,
+                            int heads, int shift, float sm_scale,
+                            hipStream_t stream) {
+  const int n_windows = Bn * (H >> 3) * (W >> 3);
+  hipLaunchKernelGGL(window_attn_fp8_kernel, dim3(n_windows, heads),
+                     dim3(64), WAVE_LDS_FP8, stream,
+                     (const unsigned char*)qkv, (__bf16*)out,
+                     (const __bf16*)bias, (const float*)qs_ptr, Bn, H, W, C,
+                     heads, shift, sm_scale);
+}
+
 }  // namespace audiomuse

@@ -38,6 +38,10 @@ void launch_window_attn4(const void* qkv, void* out, const void* bias,
 void launch_window_attn(const void* qkv, void* out, const void* bias, int Bn,
                         int H, int W, int C, int heads, int shift, float scale,
                         hipStream_t stream);
+void launch_window_attn_fp8(const void* qkv, void* out, const void* bias,
+                            const void* qs_ptr, int Bn, int H, int W, int C,
+                            int heads, int shift, float sm_scale,
+                            hipStream_t stream);
 }
 
 #define AM_CHECK(x, msg) TORCH_CHECK(x, msg)
@@ -174,6 +178,36 @@ static torch::Tensor window_attn_fwd(torch::Tensor qkv, torch::Tensor bias,
   return out;
 }
 
+static torch::Tensor window_attn_fp8_fwd(torch::Tensor qkv,
+                                         torch::Tensor bias,
+                                         torch::Tensor q_scale,
+                                         int64_t heads, int64_t shift,
+                                         double sm_scale) {
+  AM_CHECK(qkv.is_cuda() && qkv.scalar_type() == at::kFloat8_e4m3fn &&
+               qkv.is_contiguous() && qkv.dim() == 4,
+           "qkv must be (B, H, W, 3C) fp8e4m3 contiguous GPU");
+  const int64_t Bn = qkv.size(0), H = qkv.size(1), W = qkv.size(2);
+  const int64_t C = qkv.size(3) / 3;
+  AM_CHECK(qkv.size(3) == 3 * C && C == heads * 32,
+           "C must be heads*32 and last dim 3C");
+  AM_CHECK(H % 8 == 0 && W % 8 == 0, "H, W must be multiples of 8");
+  AM_CHECK(bias.is_cuda() && bias.scalar_type() == at::kBFloat16 &&
+               bias.is_contiguous() && bias.numel() == heads * 64 * 64,
+           "bias must be (heads, 64, 64) bf16 contiguous");
+  AM_CHECK(q_scale.is_cuda() && q_scale.scalar_type() == at::kFloat &&
+               q_scale.numel() == 1,
+           "q_scale must be a f32 device scalar");
+  auto out = torch::empty({Bn, H, W, C},
+                          qkv.options().dtype(at::kBFloat16));
+  auto stream = c10::hip::getCurrentHIPStream();
+  audiomuse::launch_window_attn_fp8(
+      qkv.data_ptr(), out.data_ptr(), bias.data_ptr(), q_scale.data_ptr(),
+      (int)Bn, (int)H, (int)W, (int)C, (int)heads, (int)shift,
+      (float)sm_scale, stream.stream());
+  C10_HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 static torch::Tensor window_attn4_fwd(torch::Tensor qkv, torch::Tensor bias,
                                       int64_t heads, int64_t shift,
                                       double scale) {
@@ -277,6 +311,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_layernorm_bf16_fp8", &add_layernorm_bf16_fp8,
         "Residual add + LN with fused e4m3 quantize: (sum bf16, y fp8)");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("window_attn_fp8_fwd", &window_attn_fp8_fwd,
+        "fused shifted-window attention, fp8-ingest QKV (e4m3 + device "
+        "dequant scale), bf16 MFMAs/out");
   m.def("window_attn_fwd", &window_attn_fwd,
         "Fused shifted-window attention (qkv BHW3C bf16, bias, heads, "
         "shift, scale) -> (B,H,W,C)");

@@ -139,6 +139,12 @@ _hid_state: Dict[int, tuple] = {}
 FP8_HIDDEN = [bool(getattr(C, "FP8_HIDDEN_ENABLE", False))]
 FP8_HIDDEN_ERR = ""
 
+# fp8-ingest attention (QKV GEMM emits e4m3, window kernel reads 8-byte
+# fragments): ON by default in fp8 serving mode; self-disables if
+# hipBLASLt has no fp8-D algo at the QKV shape.
+FP8_ATTN = [bool(getattr(C, "FP8_ATTN_ENABLE", True))]
+FP8_ATTN_ERR = ""
+
 
 def hidden_state(module, device):
     """(scale, inv_scale, amax) for a GEMM's fp8 D output: the GEMM

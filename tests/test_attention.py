@@ -159,6 +159,35 @@ def test_window8_kernel_vs_fp32_reference(shift):
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("shift", [0, 4])
+def test_window8_fp8_kernel_vs_fp32_reference(shift):
+    """fp8-ingest variant: QKV quantized to e4m3 with a known dequant
+    scale; the reference runs on the DEQUANTIZED values, so this bounds
+    the kernel's own error (conversion is exact, P/accum as bf16 kernel)
+    rather than the quantization error the fp8 mode already accepts."""
+    torch.manual_seed(31 + shift)
+    import audiomuse_amd._C as C
+
+    B, H, W, heads = 2, 16, 32, 8
+    dim = heads * 32
+    raw = torch.randn(B, H, W, 3 * dim, device="cuda") * 0.5
+    qs = torch.full((), 0.03, device="cuda", dtype=torch.float32)
+    qkv8 = (raw / qs).clamp(-448, 448).to(torch.float8_e4m3fn).contiguous()
+    deq = qkv8.float() * qs
+    bias = (torch.randn(heads, 64, 64, device="cuda") * 0.1).to(
+        torch.bfloat16)
+    scale = 32 ** -0.5
+    out = C.window_attn_fp8_fwd(qkv8, bias.contiguous(), qs, heads, shift,
+                                scale)
+    ref = _fp32_window_attention_reference(deq, bias.float(), heads, shift,
+                                           scale, 8)
+    diff = (out.float() - ref).abs()
+    assert float(diff.max()) < 0.06, f"max abs err {float(diff.max()):.4f}"
+    rel = diff.mean() / ref.abs().mean().clamp(min=1e-6)
+    assert float(rel) < 1.5e-2, f"mean rel err {float(rel):.4f}"
+
+
+@pytest.mark.gpu
 @pytest.mark.parametrize("shift", [0, 2])
 def test_window4_kernel_vs_fp32_reference(shift):
     """Stage-4 16-token kernel (window_attn4_fwd) against the fp32
