@@ -58,6 +58,19 @@ def main():
     bias3 = torch.randn(16, 64, 64, device=dev).to(torch.bfloat16)
     t = timeit(lambda: ext.window_attn_fwd(qkv3, bias3, 16, 4, 0.176), iters)
     print(f"window_attn s3 : {t*1000:8.2f} ms")
+    # fp8-ingest variant, same shapes (A/B isolating the kernel from the
+    # fp8-D GEMM and scale-state plumbing)
+    qs = torch.full((), 0.03, device=dev, dtype=torch.float32)
+    qkv8 = (qkv.float() / qs).clamp(-448, 448).to(
+        torch.float8_e4m3fn).contiguous()
+    t = timeit(lambda: ext.window_attn_fp8_fwd(qkv8, bias, qs, heads, 0,
+                                               0.176), iters)
+    print(f"window_attn s1 fp8: {t*1000:5.2f} ms ({toks/t/1e6:6.1f} Mtok/s)")
+    qkv38 = (qkv3.float() / qs).clamp(-448, 448).to(
+        torch.float8_e4m3fn).contiguous()
+    t = timeit(lambda: ext.window_attn_fp8_fwd(qkv38, bias3, qs, 16, 4,
+                                               0.176), iters)
+    print(f"window_attn s3 fp8: {t*1000:5.2f} ms")
     # stage-4 shape: 4x32 grid, window 4, C=1024, h=32 (window_attn4)
     qkv4 = torch.randn(B, 4, 32, 3 * 1024, device=dev, dtype=torch.bfloat16)
     bias4 = torch.randn(32, 16, 16, device=dev).to(torch.bfloat16)
