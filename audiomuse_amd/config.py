@@ -438,9 +438,12 @@ CLAP_GPU_BATCH = _env_int("CLAP_GPU_BATCH", 256)
 # headline bench stays bf16 regardless of this flag unless --fp8 is passed
 CLAP_FP8_SERVING = _env_bool("AUDIOMUSE_FP8_SERVING", False)
 FP8_HIDDEN_ENABLE = _env_bool("AUDIOMUSE_FP8_HIDDEN", False)
-# fp8-ingest attention inside the fp8 serving mode (QKV GEMM emits e4m3,
-# the window kernel reads 8-byte fragments); self-disables without an algo
-FP8_ATTN_ENABLE = _env_bool("AUDIOMUSE_FP8_ATTN", True)
+# fp8-ingest attention experiment: measured NEGATIVE on hardware (kernel
+# 0.51 vs 0.45 ms s1 / 0.11 vs 0.09 ms s3, bench 10 275 vs 10 861 —
+# profiles/r2_fp8_attn_negative.md): the window kernel is gather-LATENCY
+# bound, and the e4m3->bf16 converts sit on the load->MFMA critical path
+# while the halved bytes buy nothing. OFF by default.
+FP8_ATTN_ENABLE = _env_bool("AUDIOMUSE_FP8_ATTN", False)
 HIP_REQUIRE_NATIVE = _env_bool("HIP_REQUIRE_NATIVE", True)  # fail loudly on GPU without .so
 RCCL_BUCKET_CAP_MB = _env_int("RCCL_BUCKET_CAP_MB", 64)
 

@@ -140,9 +140,10 @@ FP8_HIDDEN = [bool(getattr(C, "FP8_HIDDEN_ENABLE", False))]
 FP8_HIDDEN_ERR = ""
 
 # fp8-ingest attention (QKV GEMM emits e4m3, window kernel reads 8-byte
-# fragments): ON by default in fp8 serving mode; self-disables if
-# hipBLASLt has no fp8-D algo at the QKV shape.
-FP8_ATTN = [bool(getattr(C, "FP8_ATTN_ENABLE", True))]
+# fragments): measured NEGATIVE — see FP8_ATTN_ENABLE in config.py for
+# the numbers. OFF by default; AUDIOMUSE_FP8_ATTN=1 re-enables for
+# experiments. Self-disables if hipBLASLt has no fp8-D algo at the shape.
+FP8_ATTN = [bool(getattr(C, "FP8_ATTN_ENABLE", False))]
 FP8_ATTN_ERR = ""
 
 
