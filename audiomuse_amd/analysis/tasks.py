@@ -13,6 +13,7 @@ analyze (MusiCNN + features + CLAP batched on GPU) -> identity
 from __future__ import annotations
 
 import logging
+import threading
 import time
 from typing import Dict, List, Optional
 
@@ -35,14 +36,19 @@ from audiomuse_amd.taskqueue.worker import TaskContext, task_handler
 logger = logging.getLogger(__name__)
 
 _RUNTIME: Optional[AnalysisRuntime] = None
+_RUNTIME_LOCK = threading.Lock()
 
 
 def get_runtime() -> AnalysisRuntime:
-    """One model runtime per worker process (one process per GPU rank)."""
+    """One model runtime per worker process (one process per GPU rank).
+    Lock guards the threaded-workers deployment (several Worker threads
+    sharing one GPU — scripts/e2e_soak.py --workers)."""
     global _RUNTIME
     if _RUNTIME is None:
-        device = "cuda" if torch.cuda.is_available() else "cpu"
-        _RUNTIME = AnalysisRuntime(device=device)
+        with _RUNTIME_LOCK:
+            if _RUNTIME is None:
+                device = "cuda" if torch.cuda.is_available() else "cpu"
+                _RUNTIME = AnalysisRuntime(device=device)
     return _RUNTIME
 
 
@@ -116,7 +122,9 @@ def analyze_album_task(ctx: TaskContext, payload: Dict) -> Dict:
     runtime.recycle_models()        # PER_SONG_MODEL_RELOAD compat switch
     n = 0
     # one write transaction per album (write_txn is reentrant: the
-    # nested save_* helpers join it) — one fsync instead of ~3/track
+    # nested save_* helpers join it) — one fsync instead of ~3/track.
+    # The txn's write lock also serializes resolver.resolve() across
+    # threaded sibling workers (resolve only runs inside this section).
     with write_txn(conn):
         n = _persist_album(ctx, conn, valid, results, resolver, server_id,
                            runtime, provider)
