@@ -46,7 +46,8 @@ def soak(n_albums=25, tracks_per_album=4, seconds=12.0, n_workers=1):
         import threading
 
         threads = [threading.Thread(
-            target=lambda: Worker(db_url=url).run_forever(idle_timeout=5.0),
+            target=lambda: Worker(db_url=url, max_jobs=n_albums + 1)
+            .run_forever(idle_timeout=5.0),
             daemon=True) for _ in range(n_workers)]
         for t in threads:
             t.start()
