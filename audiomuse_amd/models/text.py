@@ -81,8 +81,12 @@ class EncoderLayer(nn.Module):
         qkv = self.qkv(xn).view(B, L, 3, h, D // h).permute(2, 0, 3, 1, 4)
         q, k, v = qkv.unbind(0)
         attn_mask = torch.where(mask, 0.0, float("-inf"))[:, None, None, :]
-        out = F.scaled_dot_product_attention(q, k, v,
-                                             attn_mask=attn_mask.to(x.dtype))
+        # explicit math attention: torch's SDPA backend selection picked
+        # the AOTriton path (unsupported on gfx950) intermittently under
+        # threaded workers; this encoder is small and never the hot path
+        scale = (D // h) ** -0.5
+        s = (q @ k.transpose(-2, -1)) * scale + attn_mask.to(q.dtype)
+        out = torch.softmax(s, dim=-1) @ v
         out = out.transpose(1, 2).reshape(B, L, D)
         x = x + self.proj(out)
         return x + self.mlp(self.norm2(x))

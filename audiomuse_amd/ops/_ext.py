@@ -13,6 +13,7 @@ reference path is used and the extension is optional.
 from __future__ import annotations
 
 import importlib
+import threading
 from typing import Any, Optional
 
 import torch
@@ -20,17 +21,23 @@ import torch
 _EXT: Optional[Any] = None
 _TRIED = False
 _ERR: Optional[BaseException] = None
+_LOCK = threading.Lock()
 
 
 def try_load() -> Optional[Any]:
     global _EXT, _TRIED, _ERR
     if not _TRIED:
-        _TRIED = True
-        try:
-            _EXT = importlib.import_module("audiomuse_amd._C")
-        except Exception as exc:  # noqa: BLE001
-            _ERR = exc
-            _EXT = None
+        # lock + set _TRIED only after the import finishes: a sibling
+        # worker thread arriving mid-import must not observe the
+        # "tried, got None" state and raise (seen under --workers soak)
+        with _LOCK:
+            if not _TRIED:
+                try:
+                    _EXT = importlib.import_module("audiomuse_amd._C")
+                except Exception as exc:  # noqa: BLE001
+                    _ERR = exc
+                    _EXT = None
+                _TRIED = True
     return _EXT
 
 
