@@ -28,6 +28,8 @@ def main() -> None:
     wk.add_argument("--db", default=None)
     wk.add_argument("--queues", default="high,default")
     wk.add_argument("--max-jobs", type=int, default=None)
+    wk.add_argument("--idle-timeout", type=float, default=None,
+                    help="exit after this many idle seconds (default: run forever)")
 
     an = sub.add_parser("analyze")
     an.add_argument("--db", default=None)
@@ -55,7 +57,8 @@ def main() -> None:
         from audiomuse_amd.taskqueue.worker import Worker
 
         Worker(db_url=args.db, queues=tuple(args.queues.split(",")),
-               max_jobs=args.max_jobs).run_forever()
+               max_jobs=args.max_jobs).run_forever(
+            idle_timeout=args.idle_timeout)
     elif args.cmd == "analyze":
         from audiomuse_amd.db import get_db
         from audiomuse_amd.taskqueue import enqueue

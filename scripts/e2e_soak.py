@@ -16,7 +16,8 @@ import torch
 sys.path.insert(0, __file__.rsplit("/", 2)[0])
 
 
-def soak(n_albums=25, tracks_per_album=4, seconds=12.0, n_workers=1):
+def soak(n_albums=25, tracks_per_album=4, seconds=12.0, n_workers=1,
+         n_procs=0):
     import audiomuse_amd.analysis.tasks as atasks
     from audiomuse_amd.db import connect
     from audiomuse_amd.db.schema import init_db
@@ -36,7 +37,19 @@ def soak(n_albums=25, tracks_per_album=4, seconds=12.0, n_workers=1):
             for i in range(n_albums)]
     n_tracks = n_albums * tracks_per_album
     t0 = time.perf_counter()
-    if n_workers <= 1:
+    if n_procs > 1:
+        # N worker PROCESSES sharing one GPU (no GIL coupling — the
+        # reference's actual multi-worker deployment shape). Each pays
+        # its own model-load startup; WAL handles cross-process writes.
+        import subprocess
+
+        procs = [subprocess.Popen(
+            [sys.executable, "-m", "audiomuse_amd", "worker", "--db", url,
+             "--idle-timeout", "5", "--max-jobs", str(n_albums + 1)])
+            for _ in range(n_procs)]
+        for p in procs:
+            p.wait()
+    elif n_workers <= 1:
         Worker(db_url=url, max_jobs=n_albums + 1).run_forever(idle_timeout=5.0)
     else:
         # N workers sharing ONE GPU (the reference's deployment shape —
@@ -58,7 +71,8 @@ def soak(n_albums=25, tracks_per_album=4, seconds=12.0, n_workers=1):
     n_emb = conn.execute("SELECT COUNT(*) FROM track_server_map").fetchone()[0]
     print(f"soak: {n_tracks} tracks ({seconds}s each) in {wall:.1f}s wall "
           f"-> {n_tracks / wall:.2f} tracks/s ({ok}/{n_albums} albums ok, "
-          f"{n_emb} mapped, workers={n_workers})")
+          f"{n_emb} mapped, "
+          f"{f'procs={n_procs}' if n_procs > 1 else f'workers={n_workers}'})")
     print(f"  extrapolated: 100k tracks ~ {100_000 / (n_tracks / wall) / 3600:.1f} h "
           "on ONE GPU (reference FAQ: '1 week+ can be totally normal')")
 
@@ -93,6 +107,8 @@ if __name__ == "__main__":
                     help="cProfile the soak and print top cumulative")
     ap.add_argument("--workers", type=int, default=1,
                     help="worker threads sharing the GPU")
+    ap.add_argument("--procs", type=int, default=0,
+                    help="worker PROCESSES sharing the GPU (overrides --workers)")
     args = ap.parse_args()
     if args.profile:
         import cProfile
@@ -100,11 +116,13 @@ if __name__ == "__main__":
 
         pr = cProfile.Profile()
         pr.enable()
-        soak(n_albums=max(1, args.tracks // 4), n_workers=args.workers)
+        soak(n_albums=max(1, args.tracks // 4), n_workers=args.workers,
+             n_procs=args.procs)
         pr.disable()
         stats = pstats.Stats(pr)
         stats.sort_stats("cumulative")
         stats.print_stats(35)
     else:
-        soak(n_albums=max(1, args.tracks // 4), n_workers=args.workers)
+        soak(n_albums=max(1, args.tracks // 4), n_workers=args.workers,
+             n_procs=args.procs)
         whisper_rate()
