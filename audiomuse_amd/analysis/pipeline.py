@@ -241,6 +241,8 @@ class AnalysisRuntime:
         seg_owner: List[int] = []
         patch_batches: List[torch.Tensor] = []
         patch_owner: List[int] = []
+        feat_audio: List[torch.Tensor] = []
+        feat_owner: List[int] = []
         from audiomuse_amd.ops.audio_io import resample
 
         for i, blob in enumerate(wav_blobs):
@@ -251,8 +253,10 @@ class AnalysisRuntime:
             audio = audio.to(self.device)      # resample + DSP on-device
             res = TrackAnalysis(duration=audio.shape[-1] / in_sr)
             a16 = resample(audio, in_sr, C.MUSICNN_SAMPLE_RATE)
-            res.tempo, res.energy, res.key, res.scale = \
-                features.extract_basic_features(a16, C.MUSICNN_SAMPLE_RATE)
+            # tempo/energy/key batch across the album after this loop
+            # (per-track DSP was kernel-launch bound)
+            feat_audio.append(a16)
+            feat_owner.append(i)
             if C.CHROMAPRINT_COLLECTION_ENABLED:
                 from audiomuse_amd.engines import chromaprint as cp
                 try:
@@ -277,6 +281,15 @@ class AnalysisRuntime:
                                          C.CLAP_SEGMENT_HOP_SAMPLES)
                 seg_batches.append(segs)
                 seg_owner.extend([i] * segs.shape[0])
+
+        # one batched DSP pass for tempo/energy/key (exact per-track math)
+        if feat_audio:
+            feats = features.extract_basic_features_batch(
+                feat_audio, C.MUSICNN_SAMPLE_RATE)
+            for owner, (tempo, energy, key, scale) in zip(feat_owner, feats):
+                res = results[owner]
+                res.tempo, res.energy, res.key, res.scale = (
+                    tempo, energy, key, scale)
 
         # one MusiCNN pass for the whole album
         if patch_batches:

@@ -107,7 +107,7 @@ def onset_envelope(audio: torch.Tensor, sr: int, n_fft: int = 2048,
 def estimate_tempo(audio: torch.Tensor, sr: int, hop: int = 512) -> float:
     """Autocorrelation tempogram + log-normal 120 BPM prior, folded into
     [40, 200] BPM (song.py:186-199)."""
-    if audio.numel() == 0:
+    if audio.numel() <= 1024:   # shorter than the reflect pad: no tempo
         return 0.0
     env = onset_envelope(audio, sr, hop=hop)
     n = env.shape[0]
@@ -151,12 +151,9 @@ def chroma_from_stft(audio: torch.Tensor, sr: int, n_fft: int = 4096,
     return chroma
 
 
-def estimate_key_scale(audio: torch.Tensor, sr: int) -> Tuple[str, str]:
-    """Krumhansl-Schmuckler correlation (song.py:213-231, exact math)."""
-    if audio.numel() == 0:
-        return "C", "major"
-    chroma = chroma_from_stft(audio, sr)
-    cm = chroma.mean(dim=1)
+def _key_from_chroma_mean(cm: torch.Tensor) -> Tuple[str, str]:
+    """Krumhansl-Schmuckler correlation from a mean chroma vector
+    (song.py:213-231, exact math). cm: (12,) on any device."""
     if float(cm.sum()) <= 0:
         return "C", "major"
     c = (cm / (cm.norm() + 1e-9)).cpu()
@@ -170,9 +167,123 @@ def estimate_key_scale(audio: torch.Tensor, sr: int) -> Tuple[str, str]:
     return _KEYS[ni], "minor"
 
 
+def estimate_key_scale(audio: torch.Tensor, sr: int) -> Tuple[str, str]:
+    """Krumhansl-Schmuckler correlation (song.py:213-231, exact math)."""
+    if audio.numel() <= 2048:   # shorter than the reflect pad: no key
+        return "C", "major"
+    chroma = chroma_from_stft(audio, sr)
+    return _key_from_chroma_mean(chroma.mean(dim=1))
+
+
 def extract_basic_features(audio: torch.Tensor, sr: int):
     """song.py:233-238: (tempo, energy, key, scale)."""
     tempo = estimate_tempo(audio, sr)
     energy = estimate_energy(audio)
     key, scale = estimate_key_scale(audio, sr)
     return tempo, energy, key, scale
+
+
+def _reflect_rows(audios, lens, pad: int, device) -> torch.Tensor:
+    """Stack variable-length tracks into rows pre-padded with each
+    track's OWN reflection (so batched center=False framing reproduces
+    the per-track center=True frames exactly); zeros beyond."""
+    L = max(lens)
+    rows = torch.zeros(len(audios), L + 2 * pad, device=device)
+    for i, a in enumerate(audios):
+        rows[i, : lens[i] + 2 * pad] = torch.nn.functional.pad(
+            a[None], (pad, pad), mode="reflect")[0]
+    return rows
+
+
+def extract_basic_features_batch(audios, sr: int):
+    """Batched (tempo, energy, key, scale) for an album's tracks.
+
+    Same math as the per-track functions — framing is reproduced
+    exactly (reflect pre-pad per track; masked means; zero-padded
+    linear autocorrelation is lag-wise identical) — but one
+    STFT/FFT/matmul launch set per album instead of per track
+    (the per-track loop was kernel-launch bound on GPU).
+    """
+    audios = [a.float().flatten() for a in audios]
+    if not audios:
+        return []
+    lens = [a.numel() for a in audios]
+    # tiny or single tracks: the per-track path handles edge cases
+    if len(audios) == 1 or min(lens) <= 4096:
+        return [extract_basic_features(a, sr) for a in audios]
+    device = audios[0].device
+    B = len(audios)
+    L = max(lens)
+    nT = torch.tensor(lens, device=device)[:, None]
+
+    # -- energy (frame 2048, hop 512, zero pad — zeros beyond each
+    # track ARE the per-track end padding) --
+    frame, hop = 2048, 512
+    p = frame // 2
+    xb = torch.zeros(B, L, device=device)
+    for i, a in enumerate(audios):
+        xb[i, : lens[i]] = a
+    frames = torch.nn.functional.pad(xb, (p, p)).unfold(1, frame, hop)
+    rms = frames.square().mean(dim=2).sqrt()
+    rms_db = 20.0 * torch.log10(torch.clamp(rms, min=1e-9))
+    en = torch.clamp((rms_db + 60.0) / 60.0, 0.0, 1.0)
+    n_en = 1 + (nT + 2 * p - frame) // hop
+    eidx = torch.arange(en.shape[1], device=device)[None, :]
+    emask = (eidx < n_en).float()
+    energy = ((en * emask).sum(1) / emask.sum(1).clamp(min=1.0))
+
+    # -- tempo (onset flux 2048/512/128-mel -> autocorr + prior) --
+    pf = 2048 // 2
+    rows = _reflect_rows(audios, lens, pf, device)
+    pw = power_spectrogram(rows, 2048, 512, center=False)
+    fb = _onset_fb(sr, 2048, 128, device)
+    log_mel = torch.log10(torch.clamp(fb @ pw, min=1e-10))
+    flux = torch.clamp(log_mel[:, :, 1:] - log_mel[:, :, :-1], min=0.0)
+    env = flux.mean(dim=1)                       # (B, Tmax-1)
+    n_env = (1 + torch.div(nT, 512, rounding_mode="floor")) - 1
+    vmask = (torch.arange(env.shape[1], device=device)[None, :] < n_env)
+    env = env * vmask
+    env = (env - (env.sum(1, keepdim=True)
+                  / n_env.clamp(min=1).float())) * vmask
+    n2 = 2 * env.shape[1]
+    f = torch.fft.rfft(env, n=n2)
+    ac = torch.fft.irfft(f * f.conj(), n=n2)[:, : env.shape[1]]
+    ac = ac / (ac[:, :1] + 1e-12)
+    fps = sr / 512
+    lags = torch.arange(1, env.shape[1], device=device, dtype=torch.float32)
+    bpm = 60.0 * fps / lags
+    prior = torch.exp(-0.5 * ((torch.log2(bpm) - math.log2(120.0)) / 1.0) ** 2)
+    score = ac[:, 1:] * prior * ((bpm >= 20.0) & (bpm <= 400.0))
+    best = score.argmax(dim=1)
+    smax = score.gather(1, best[:, None]).squeeze(1)
+
+    # -- key (chroma 4096/1024, masked frame mean) --
+    pc = 4096 // 2
+    rows2 = _reflect_rows(audios, lens, pc, device)
+    mag = power_spectrogram(rows2, 4096, 1024, center=False).sqrt()
+    wt = _chroma_weights(sr, 4096, device)
+    ch = wt @ mag                                # (B, 12, T2)
+    n_ch = 1 + torch.div(nT, 1024, rounding_mode="floor")
+    cmask = (torch.arange(ch.shape[2], device=device)[None, :]
+             < n_ch).float()[:, None, :]
+    cm = (ch * cmask).sum(2) / cmask.sum(2).clamp(min=1.0)
+
+    # one D2H for everything, then cheap per-track CPU tail
+    packed = torch.cat([energy[:, None], smax[:, None], bpm[best][:, None],
+                        n_env.float(), cm], dim=1).cpu()
+    out = []
+    for i in range(B):
+        row = packed[i]
+        e = float(row[0])
+        s, t, nv = float(row[1]), float(row[2]), int(row[3])
+        if nv < 4 or s <= 0 or t <= 0:
+            tempo = 0.0
+        else:
+            tempo = t
+            while tempo < TEMPO_MIN_BPM:
+                tempo *= 2.0
+            while tempo > TEMPO_MAX_BPM:
+                tempo /= 2.0
+        key, scale = _key_from_chroma_mean(row[4:16])
+        out.append((tempo, e, key, scale))
+    return out

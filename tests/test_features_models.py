@@ -55,6 +55,41 @@ def test_key_detection_major_triad():
     assert features.estimate_key_scale(torch.zeros(0), sr) == ("C", "major")
 
 
+def test_batched_features_match_per_track():
+    """extract_basic_features_batch reproduces the per-track math
+    exactly on mixed-length tracks (reflect pre-pad framing, masked
+    means, zero-padded autocorrelation)."""
+    torch.manual_seed(7)
+    sr = 16000
+    tracks = [
+        _tone(440, sr, seconds=4.0) + torch.randn(4 * sr) * 0.02,
+        torch.randn(int(5.5 * sr)) * 0.1,
+        _tone(261.63, sr, seconds=3.2, amp=0.3),
+    ]
+    # a click track (strong tempo) with a different length
+    bpm_audio = torch.zeros(int(4.7 * sr))
+    step = int(sr * 60 / 120.0)
+    for s in range(0, bpm_audio.numel(), step):
+        e = min(s + 200, bpm_audio.numel())
+        bpm_audio[s:e] = torch.randn(e - s) * 0.8
+    tracks.append(bpm_audio)
+
+    batched = features.extract_basic_features_batch(tracks, sr)
+    singles = [features.extract_basic_features(a, sr) for a in tracks]
+    for (bt, be, bk, bs), (st, se, sk, ss) in zip(batched, singles):
+        assert bk == sk and bs == ss
+        assert abs(be - se) < 1e-5
+        # identical argmax lag -> identical folded tempo
+        assert abs(bt - st) < 1e-3, (bt, st)
+
+
+def test_batched_features_tiny_track_fallback():
+    sr = 16000
+    out = features.extract_basic_features_batch(
+        [torch.randn(1024), torch.randn(3 * sr) * 0.1], sr)
+    assert len(out) == 2 and out[0][0] == 0.0   # tiny track: tempo 0
+
+
 def test_musicnn_shapes_and_aggregation():
     torch.manual_seed(0)
     emb_model = MusiCNNEmbedding()
