@@ -83,6 +83,21 @@ class SimilarityEngine:
             return None
         return self.index.vector_for_id(p)
 
+    def max_distance_for_id(self, item_id: str) -> Optional[Dict]:
+        """Distance to the farthest catalogue track (reference:
+        get_max_distance_for_id ivf_manager.py:1177 — UI slider
+        normalization). Cached per item."""
+        hit = self.cache.get(("maxd", item_id))
+        if hit is not None:
+            return dict(hit)
+        vec = self.vector_for_id(item_id)
+        if vec is None:
+            return None
+        d, fid = self.index.max_distance(vec)
+        out = {"max_distance": d, "farthest_item_id": self.item_ids[fid]}
+        self.cache.put(("maxd", item_id), dict(out))
+        return out
+
     # -- core query ------------------------------------------------------
 
     def _query_candidates(self, vec: torch.Tensor, fetch: int,

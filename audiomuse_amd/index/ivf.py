@@ -278,6 +278,23 @@ class IVFIndex:
         v = self.data[:, : self.dim].float()
         return v / 127.0 if self.storage == "i8" else v
 
+    def max_distance(self, q: torch.Tensor) -> Tuple[float, int]:
+        """Distance to the FARTHEST row and its id (reference:
+        PagedIvfIndex.get_max_distance via ivf_manager.py:1177 — the UI
+        normalizes its similarity sliders with this). One full pass over
+        the decoded rows; callers cache per item."""
+        x = self._decode_unit()
+        qv = q.flatten().float().to(x.device)
+        if self.metric == "angular":
+            qv = qv / qv.norm().clamp(min=1e-12)
+            d = 1.0 - x @ qv
+        elif self.metric == "dot":
+            d = -(x @ qv)
+        else:
+            d = (x - qv[None, :]).square().sum(dim=1)
+        row = int(d.argmax())
+        return float(d[row]), int(self.ids[row])
+
     @property
     def n(self) -> int:
         return 0 if self.ids is None else int(self.ids.shape[0])

@@ -104,6 +104,100 @@ def search_tracks():
     return jsonify([dict(r) for r in rows])
 
 
+@bp.get("/api/search_artists")
+@require_auth
+def search_artists():
+    """Artist name search (reference: app_music_search /api/search_artists)."""
+    q = (request.args.get("q") or "").strip().lower()
+    if not q:
+        return jsonify([])
+    conn = _state().conn()
+    rows = conn.execute(
+        """SELECT author, COUNT(*) AS n_tracks FROM score
+           WHERE LOWER(author) LIKE ? GROUP BY author
+           ORDER BY n_tracks DESC LIMIT ?""",
+        (f"%{q}%", int(request.args.get("n", 25)))).fetchall()
+    return jsonify([{"artist": r["author"], "n_tracks": r["n_tracks"]}
+                    for r in rows])
+
+
+@bp.get("/api/artist_tracks")
+@require_auth
+def artist_tracks():
+    """All catalogue tracks of one artist (reference: /api/artist_tracks)."""
+    artist = (request.args.get("artist") or "").strip()
+    if not artist:
+        return jsonify({"error": "artist required"}), 400
+    conn = _state().conn()
+    rows = conn.execute(
+        """SELECT item_id, title, album, tempo, energy FROM score
+           WHERE author = ? ORDER BY album, title LIMIT ?""",
+        (artist, int(request.args.get("n", 200)))).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
+@bp.get("/api/track")
+@require_auth
+def track_detail():
+    """Full catalogue row for one track (reference: /api/track,
+    /get_score + /get_embedding companions exist separately)."""
+    item_id = request.args.get("item_id", "")
+    conn = _state().conn()
+    row = conn.execute("SELECT * FROM score WHERE item_id = ?",
+                       (item_id,)).fetchone()
+    if row is None:
+        return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
+    d = dict(row)
+    for k in ("mood_vector", "other_features"):
+        if d.get(k):
+            try:
+                d[k] = json.loads(d[k])
+            except (TypeError, ValueError):
+                pass
+    servers = conn.execute(
+        "SELECT server_id, provider_id FROM track_server_map "
+        "WHERE item_id = ?", (item_id,)).fetchall()
+    d["servers"] = [dict(s) for s in servers]
+    return jsonify(d)
+
+
+@bp.get("/api/max_distance")
+@require_auth
+def max_distance():
+    """Per-item distance ceiling for UI sliders (reference:
+    app_ivf.py:515 -> get_max_distance_for_id)."""
+    eng = _state().engine(idx.AUDIO_INDEX)
+    if eng is None:
+        return jsonify({"error": "audio index not built"}), 503
+    item_id = request.args.get("item_id", "")
+    out = eng.max_distance_for_id(item_id)
+    if out is None:
+        return jsonify({"error": f"unknown item_id {item_id!r}"}), 404
+    return jsonify({**out, "metric": C.IVF_METRIC})
+
+
+@bp.get("/api/mood_centroids")
+@require_auth
+def mood_centroids():
+    """Per-mood centroid positions on the 2-D song map (reference:
+    app_map.py /api/mood_centroids — map overlay labels)."""
+    entry = _map_bucket("song", 100, "")
+    if entry is None:
+        return jsonify({"error": "map not built"}), 503
+    pts = json.loads(entry[0])
+    acc: dict = {}
+    for p in pts:
+        mood = p.get("mood") or ""
+        if not mood:
+            continue
+        a = acc.setdefault(mood, [0.0, 0.0, 0])
+        a[0] += p["x"]
+        a[1] += p["y"]
+        a[2] += 1
+    return jsonify([{"mood": m, "x": a[0] / a[2], "y": a[1] / a[2],
+                     "count": a[2]} for m, a in sorted(acc.items())])
+
+
 @bp.get("/api/path")
 @require_auth
 def song_path():

@@ -235,6 +235,175 @@ def delete_server(server_id: str):
     return jsonify({"deleted": cur.rowcount})
 
 
+def _server_row(conn, server_id: str):
+    return conn.execute("SELECT * FROM music_servers WHERE server_id = ?",
+                        (server_id,)).fetchone()
+
+
+def _provider_from_row(row):
+    from audiomuse_amd.mediaserver import make_provider
+
+    cfg = json.loads(row["config"] or "{}")
+    return make_provider(row["server_type"], base_url=row["base_url"],
+                         username=row["username"],
+                         credential=row["credential"], **cfg)
+
+
+@bp.post("/api/servers/test")
+@require_auth
+def server_test():
+    """Connectivity probe for a server config BEFORE saving it
+    (reference: app_music_servers.py /api/servers/test)."""
+    from audiomuse_amd.analysis.migration import probe_server
+
+    body = request.get_json(force=True, silent=True) or {}
+    try:
+        out = probe_server(body.get("server_type", ""),
+                           body.get("server_config", body))
+    except Exception as exc:  # unsupported type / connection refused
+        out = {"reachable": False, "error": str(exc)}
+    out["ok"] = bool(out.get("reachable"))
+    return jsonify(out), 200 if out["ok"] else 502
+
+
+@bp.get("/api/servers/<server_id>/libraries")
+@require_auth
+def server_libraries(server_id: str):
+    """Music libraries/folders of a configured server (reference:
+    /api/servers/libraries — scoping for MUSIC_LIBRARIES)."""
+    conn = _state().conn()
+    row = _server_row(conn, server_id)
+    if row is None:
+        return jsonify({"error": f"unknown server {server_id!r}"}), 404
+    provider = _provider_from_row(row)
+    libs = getattr(provider, "list_libraries", lambda: [])()
+    return jsonify(libs)
+
+
+@bp.post("/api/servers/<server_id>/sweep")
+@require_auth
+def server_sweep(server_id: str):
+    """Metadata-only alignment sweep of ONE server (reference:
+    /api/servers/<id>/sweep -> multiserver_sync for that server)."""
+    conn = _state().conn()
+    row = _server_row(conn, server_id)
+    if row is None:
+        return jsonify({"error": f"unknown server {server_id!r}"}), 404
+    cfg = json.loads(row["config"] or "{}")
+    tid, existing = _admit_and_enqueue(conn, "multiserver_sync", {
+        "server_type": row["server_type"], "server_id": server_id,
+        "server_config": {"base_url": row["base_url"],
+                          "username": row["username"],
+                          "credential": row["credential"], **cfg}})
+    if tid is None:
+        return jsonify({"error": "sweep already running",
+                        "task_id": existing}), 409
+    return jsonify({"task_id": tid}), 202
+
+
+@bp.post("/api/sync")
+@require_auth
+def sync_all():
+    """Sweep every enabled server (reference: app_sync.py /api/sync)."""
+    conn = _state().conn()
+    rows = conn.execute(
+        "SELECT server_id FROM music_servers WHERE enabled = 1").fetchall()
+    if not rows:
+        return jsonify({"error": "no configured media server"}), 404
+    tids = []
+    for r in rows:
+        row = _server_row(conn, r["server_id"])
+        cfg = json.loads(row["config"] or "{}")
+        tid = enqueue(conn, "multiserver_sync", {
+            "server_type": row["server_type"],
+            "server_id": row["server_id"],
+            "server_config": {"base_url": row["base_url"],
+                              "username": row["username"],
+                              "credential": row["credential"], **cfg}},
+            queue="high")
+        tids.append(tid)
+    return jsonify({"task_ids": tids}), 202
+
+
+@bp.post("/api/cleaning/start")
+@require_auth
+def cleaning_start():
+    """Orphan report/delete task (reference: app.py /api/cleaning/start
+    -> cleaning.py:48). delete=false previews only."""
+    body = request.get_json(force=True, silent=True) or {}
+    conn = _state().conn()
+    tid, existing = _admit_and_enqueue(conn, "clean_orphans", {
+        "delete": bool(body.get("delete", False))})
+    if tid is None:
+        return jsonify({"error": "cleaning already running",
+                        "task_id": existing}), 409
+    return jsonify({"task_id": tid}), 202
+
+
+@bp.post("/api/cancel_all/<prefix>")
+@require_auth
+def cancel_all(prefix: str):
+    """Cancel every live task whose type starts with prefix (reference:
+    app.py /api/cancel_all/<task_type_prefix>; recursive per task)."""
+    conn = _state().conn()
+    rows = conn.execute(
+        "SELECT task_id FROM task_status WHERE task_type LIKE ? "
+        "AND status IN (?,?)", (prefix + "%", PENDING, RUNNING)).fetchall()
+    for r in rows:
+        cancel_task_recursive(conn, r["task_id"])
+    return jsonify({"cancelled": len(rows)})
+
+
+@bp.get("/api/last_task")
+@require_auth
+def last_task():
+    """Most recent task of a type (reference: app.py /api/last_task)."""
+    ttype = request.args.get("task_type")
+    conn = _state().conn()
+    row = conn.execute(
+        "SELECT task_id, task_type, status, progress, created_at "
+        "FROM task_status"
+        + (" WHERE task_type = ?" if ttype else "")
+        + " ORDER BY created_at DESC LIMIT 1",
+        (ttype,) if ttype else ()).fetchone()
+    if row is None:
+        return jsonify({}), 404
+    return jsonify(dict(row))
+
+
+@bp.get("/api/playlists")
+@require_auth
+def playlists():
+    """Stored playlists (reference: /api/playlists)."""
+    conn = _state().conn()
+    rows = conn.execute(
+        "SELECT id, name, server_id, kind, item_ids, created_at "
+        "FROM playlist ORDER BY created_at DESC LIMIT ?",
+        (int(request.args.get("n", 100)),)).fetchall()
+    out = []
+    for r in rows:
+        d = dict(r)
+        ids = json.loads(d.pop("item_ids") or "[]")
+        d["n_tracks"] = len(ids)
+        if request.args.get("include_tracks") in ("1", "true"):
+            d["item_ids"] = ids
+        out.append(d)
+    return jsonify(out)
+
+
+@bp.get("/api/search_playlists")
+@require_auth
+def search_playlists():
+    """Name search over stored playlists (reference: /api/search_playlists)."""
+    q = (request.args.get("q") or "").strip().lower()
+    conn = _state().conn()
+    rows = conn.execute(
+        "SELECT id, name, kind, created_at FROM playlist "
+        "WHERE LOWER(name) LIKE ? ORDER BY created_at DESC LIMIT ?",
+        (f"%{q}%", int(request.args.get("n", 50)))).fetchall()
+    return jsonify([dict(r) for r in rows])
+
+
 # -- cron (reference: app_cron.py minute-claimed scheduler) -----------------
 
 @bp.get("/api/cron")

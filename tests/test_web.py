@@ -524,3 +524,99 @@ def test_map_per_server_bucket(client_ids):
     assert all("provider_id" in p for p in pts)
     mapped = set(ids[10:16])
     assert all(p["item_id"] in mapped for p in pts)
+
+
+# -- round-2 parity endpoints (reference route sweep) -----------------------
+
+def test_search_artists_and_tracks(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/search_artists?q=artist 1")
+    assert r.status_code == 200 and len(r.json) >= 1
+    assert r.json[0]["n_tracks"] >= 1
+    r = client.get("/api/artist_tracks?artist=Artist 1")
+    assert r.status_code == 200 and len(r.json) >= 1
+    assert client.get("/api/artist_tracks").status_code == 400
+
+
+def test_track_detail(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/track?item_id={ids[3]}")
+    assert r.status_code == 200
+    assert r.json["title"] == "Song 3"
+    assert isinstance(r.json["mood_vector"], dict)
+    assert isinstance(r.json["servers"], list)
+    assert client.get("/api/track?item_id=nope").status_code == 404
+
+
+def test_max_distance(client_ids):
+    client, ids = client_ids
+    r = client.get(f"/api/max_distance?item_id={ids[0]}")
+    assert r.status_code == 200
+    assert r.json["max_distance"] > 0
+    assert r.json["farthest_item_id"] in ids
+    assert client.get("/api/max_distance?item_id=nope").status_code == 404
+
+
+def test_mood_centroids(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/mood_centroids")
+    assert r.status_code == 200
+    moods = {m["mood"] for m in r.json}
+    assert moods <= {"rock", "jazz"} and len(r.json) >= 1
+    assert all(m["count"] > 0 for m in r.json)
+
+
+def test_playlists_listing_and_search(client_ids):
+    client, ids = client_ids
+    from audiomuse_amd.db import write_txn
+    state = client.application.extensions["audiomuse"]
+    conn = state.conn()
+    with write_txn(conn):
+        conn.execute(
+            "INSERT INTO playlist (name, item_ids, kind) VALUES (?,?,?)",
+            ("Morning Mix", json.dumps(ids[:5]), "clustering"))
+    r = client.get("/api/playlists")
+    assert r.status_code == 200 and len(r.json) >= 1
+    assert r.json[0]["n_tracks"] == 5 and "item_ids" not in r.json[0]
+    r = client.get("/api/playlists?include_tracks=1")
+    assert r.json[0]["item_ids"] == ids[:5]
+    r = client.get("/api/search_playlists?q=morning")
+    assert len(r.json) == 1 and r.json[0]["name"] == "Morning Mix"
+    assert client.get("/api/search_playlists?q=zzz").json == []
+
+
+def test_cleaning_start_and_last_task(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/cleaning/start", json={"delete": False})
+    assert r.status_code == 202
+    # admission gate: a second start while pending conflicts
+    assert client.post("/api/cleaning/start", json={}).status_code == 409
+    r = client.get("/api/last_task?task_type=clean_orphans")
+    assert r.status_code == 200 and r.json["task_type"] == "clean_orphans"
+    # bulk cancel by prefix clears it
+    r = client.post("/api/cancel_all/clean")
+    assert r.status_code == 200 and r.json["cancelled"] >= 1
+
+
+def test_server_test_probe_and_sweep(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/servers/test",
+                    json={"server_type": "synthetic",
+                          "server_config": {"n_albums": 2}})
+    assert r.status_code == 200 and r.json["ok"]
+    r = client.post("/api/servers/test", json={"server_type": "bogus"})
+    assert r.status_code == 502
+    # register a synthetic server, then sweep it and sync all
+    assert client.post("/api/servers", json={
+        "server_id": "s1", "server_type": "synthetic",
+        "config": {"n_albums": 2}}).status_code == 200
+    r = client.get("/api/servers/s1/libraries")
+    assert r.status_code == 200 and isinstance(r.json, list)
+    r = client.post("/api/servers/s1/sweep")
+    assert r.status_code == 202
+    assert client.post("/api/servers/s1/sweep").status_code == 409
+    client.post("/api/cancel_all/multiserver_sync")
+    r = client.post("/api/sync")
+    assert r.status_code == 202 and len(r.json["task_ids"]) >= 1
+    client.post("/api/cancel_all/multiserver_sync")
+    assert client.get("/api/servers/nope/libraries").status_code == 404
