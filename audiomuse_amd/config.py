@@ -536,3 +536,16 @@ def refresh_config() -> None:
     (reference: config.refresh_config :1389)."""
     if _DB_OVERRIDE_PROVIDER is not None:
         apply_db_overrides(_DB_OVERRIDE_PROVIDER() or {})
+
+
+# Pristine env-resolved defaults, captured at import BEFORE any DB
+# override is layered on (reference: /api/config/defaults — the setup
+# UI shows "reset to default" values from here).
+_DEFAULTS_SNAPSHOT: Dict[str, object] = {
+    _k: _v for _k, _v in list(globals().items())
+    if _k.isupper() and isinstance(_v, (bool, int, float, str))
+}
+
+
+def defaults() -> Dict[str, object]:
+    return dict(_DEFAULTS_SNAPSHOT)

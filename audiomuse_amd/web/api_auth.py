@@ -95,6 +95,18 @@ def get_config():
     return jsonify({"config": safe, "overrides": overrides})
 
 
+@bp.get("/api/config/defaults")
+@require_auth
+def get_config_defaults():
+    """Pristine env-resolved defaults, before DB overrides (reference:
+    /api/config/defaults — the setup UI's reset-to-default values)."""
+    from audiomuse_amd import config as C
+
+    safe = {k: v for k, v in C.defaults().items()
+            if "SECRET" not in k and "TOKEN" not in k and "PASSWORD" not in k}
+    return jsonify({"defaults": safe})
+
+
 @bp.post("/api/config")
 @require_auth
 def set_config():

@@ -334,6 +334,54 @@ def clap_warmup_status():
     return jsonify({"loaded": lc.loaded, "seconds": lc.remaining()})
 
 
+@bp.post("/api/lyrics/warmup")
+@require_auth
+def lyrics_warmup():
+    """Pre-load the lyrics text embedder (reference: /api/lyrics/warmup)."""
+    lc = _gte_lifecycle()
+    lc.get()
+    return jsonify({"loaded": True, "seconds": lc.remaining()})
+
+
+@bp.get("/api/lyrics/warmup/status")
+@require_auth
+def lyrics_warmup_status():
+    lc = _gte_lifecycle()
+    return jsonify({"loaded": lc.loaded, "seconds": lc.remaining()})
+
+
+def _family_stats(index_name: str, table: str):
+    """Row count + loaded-index shape for one index family (reference:
+    /api/clap/stats, /api/lyrics/stats, /api/sem_grove/stats)."""
+    state = _state()
+    n_rows = state.conn().execute(
+        f"SELECT COUNT(*) AS n FROM {table}").fetchone()["n"]
+    eng = state.engine(index_name)
+    out = {"rows": n_rows, "index_loaded": eng is not None}
+    if eng is not None and hasattr(eng, "index"):
+        out.update({"indexed": eng.index.n, "nlist": eng.index.nlist,
+                    "dim": eng.index.dim, "storage": eng.index.storage})
+    return jsonify(out)
+
+
+@bp.get("/api/clap/stats")
+@require_auth
+def clap_stats():
+    return _family_stats(idx.CLAP_INDEX, "clap_embedding")
+
+
+@bp.get("/api/lyrics/stats")
+@require_auth
+def lyrics_stats():
+    return _family_stats(idx.LYRICS_INDEX, "lyrics_embedding")
+
+
+@bp.get("/api/semgrove/stats")
+@require_auth
+def semgrove_stats():
+    return _family_stats(idx.SEMGROVE_INDEX, "clap_embedding")
+
+
 @bp.get("/api/lyrics_search")
 @require_auth
 def lyrics_search():
@@ -671,6 +719,31 @@ def music_map():
                                  "Cache-Control": "no-store"})
     return Response(raw, mimetype="application/json",
                     headers={"Cache-Control": "no-store"})
+
+
+@bp.get("/api/map_cache_status")
+@require_auth
+def map_cache_status():
+    """Pre-gzipped map bucket cache state (reference:
+    app_map.py /api/map_cache_status)."""
+    state = _state()
+    return jsonify({
+        "buckets_cached": len(_map_cache),
+        "song_map_built": state.engine(idx.SONG_MAP) is not None,
+        "artist_map_built": state.engine(idx.ARTIST_MAP) is not None,
+        "percents": list(_MAP_PERCENTS)})
+
+
+@bp.post("/api/rebuild_map_cache")
+@require_auth
+def rebuild_map_cache():
+    """Drop the gzip bucket cache and enqueue an index refresh so the
+    projections rebuild (reference: app_map.py /api/rebuild_map_cache)."""
+    from audiomuse_amd.taskqueue import enqueue
+
+    _map_cache.clear()
+    tid = enqueue(_state().conn(), "rebuild_indexes", {}, queue="high")
+    return jsonify({"task_id": tid}), 202
 
 
 @bp.post("/api/order_playlist")

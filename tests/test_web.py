@@ -620,3 +620,30 @@ def test_server_test_probe_and_sweep(client_ids):
     assert r.status_code == 202 and len(r.json["task_ids"]) >= 1
     client.post("/api/cancel_all/multiserver_sync")
     assert client.get("/api/servers/nope/libraries").status_code == 404
+
+
+def test_config_defaults_and_family_stats(client_ids):
+    client, _ = client_ids
+    r = client.get("/api/config/defaults")
+    assert r.status_code == 200 and "CLAP_GPU_BATCH" in r.json["defaults"]
+    assert not any("PASSWORD" in k for k in r.json["defaults"])
+    for fam in ("clap", "lyrics", "semgrove"):
+        r = client.get(f"/api/{fam}/stats")
+        assert r.status_code == 200, fam
+        assert r.json["rows"] > 0 and r.json["index_loaded"], fam
+        assert r.json["indexed"] > 0
+
+
+def test_lyrics_warmup_and_map_cache(client_ids):
+    client, _ = client_ids
+    r = client.post("/api/lyrics/warmup")
+    assert r.status_code == 200 and r.json["loaded"]
+    assert client.get("/api/lyrics/warmup/status").json["loaded"]
+    client.get("/api/map")                      # populate a bucket
+    r = client.get("/api/map_cache_status")
+    assert r.status_code == 200 and r.json["song_map_built"]
+    assert r.json["buckets_cached"] >= 1
+    r = client.post("/api/rebuild_map_cache")
+    assert r.status_code == 202 and r.json["task_id"]
+    assert client.get("/api/map_cache_status").json["buckets_cached"] == 0
+    client.post("/api/cancel_all/rebuild_indexes")
