@@ -41,6 +41,54 @@ def setup_admin():
     return jsonify({"created": username})
 
 
+# -- Plex PIN link flow (reference: app_setup.py:926-1030) ------------------
+# plex.tv sends no CORS headers, so the browser cannot call it directly;
+# these two routes proxy the PIN create/poll server-side. The
+# client_id must stay constant between the two calls.
+
+PLEX_PIN_API_BASE = "https://plex.tv/api/v2/pins"
+
+
+def _plex_headers(client_id: str) -> dict:
+    return {"Accept": "application/json",
+            "X-Plex-Product": "AudioMuse-AMD",
+            "X-Plex-Client-Identifier": client_id or "audiomuse-amd"}
+
+
+@bp.post("/api/setup/plex/pin")
+def plex_pin_create():
+    """Create a plex.tv link PIN: the user enters the returned code at
+    plex.tv/link; poll the companion GET until authToken appears."""
+    from audiomuse_amd.mediaserver.http import MediaHttp
+
+    body = request.get_json(force=True, silent=True) or {}
+    client_id = body.get("client_id", "audiomuse-amd")
+    try:
+        r = MediaHttp().post(PLEX_PIN_API_BASE, params={"strong": "true"},
+                             headers=_plex_headers(client_id))
+        data = r.json()
+    except Exception as exc:  # no egress / plex.tv unreachable
+        return jsonify({"error": f"plex.tv unreachable: {exc}"}), 502
+    return jsonify({"id": data.get("id"), "code": data.get("code"),
+                    "client_id": client_id})
+
+
+@bp.get("/api/setup/plex/pin/<pin_id>")
+def plex_pin_poll(pin_id: str):
+    from audiomuse_amd.mediaserver.http import MediaHttp
+
+    client_id = request.args.get("client_id", "audiomuse-amd")
+    try:
+        r = MediaHttp().get(f"{PLEX_PIN_API_BASE}/{pin_id}",
+                            headers=_plex_headers(client_id))
+        data = r.json()
+    except Exception as exc:
+        return jsonify({"error": f"plex.tv unreachable: {exc}"}), 502
+    return jsonify({"id": data.get("id"),
+                    "auth_token": data.get("authToken") or None,
+                    "claimed": bool(data.get("authToken"))})
+
+
 @bp.post("/api/login")
 def login():
     conn = _state().conn()

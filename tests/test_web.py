@@ -647,3 +647,35 @@ def test_lyrics_warmup_and_map_cache(client_ids):
     assert r.status_code == 202 and r.json["task_id"]
     assert client.get("/api/map_cache_status").json["buckets_cached"] == 0
     client.post("/api/cancel_all/rebuild_indexes")
+
+
+def test_plex_pin_flow(client_ids, monkeypatch):
+    """Plex PIN create/poll proxy (reference app_setup.py:926-1030) —
+    plex.tv stubbed at the MediaHttp layer."""
+    client, _ = client_ids
+    from audiomuse_amd.mediaserver import http as mhttp
+
+    class FakeResp:
+        def __init__(self, body):
+            self._body = body
+
+        def json(self):
+            return self._body
+
+    calls = {}
+
+    def fake_request(self, method, url, **kw):
+        calls["last"] = (method, url,
+                         kw.get("headers", {}).get("X-Plex-Client-Identifier"))
+        if method == "POST":
+            return FakeResp({"id": 777, "code": "ABCD"})
+        return FakeResp({"id": 777, "authToken": "tok-xyz"})
+
+    monkeypatch.setattr(mhttp.MediaHttp, "request", fake_request)
+    r = client.post("/api/setup/plex/pin", json={"client_id": "cid-1"})
+    assert r.status_code == 200
+    assert r.json["code"] == "ABCD" and r.json["id"] == 777
+    assert calls["last"][2] == "cid-1"
+    r = client.get("/api/setup/plex/pin/777?client_id=cid-1")
+    assert r.status_code == 200
+    assert r.json["claimed"] and r.json["auth_token"] == "tok-xyz"
