@@ -328,7 +328,10 @@ def refresh_indexes_task(ctx: TaskContext, payload: Dict) -> Dict:
 
     device = "cuda" if torch.cuda.is_available() else "cpu"
     out: Dict[str, Dict] = {}
-    for name in idx_mod._REFRESHABLE:
+    only = payload.get("only")          # per-family refresh (cache/refresh
+    names = [n for n in idx_mod._REFRESHABLE   # routes) or all families
+             if not only or n == only]
+    for name in names:
         try:
             out[name] = idx_mod.refresh_ivf_index(ctx.conn, name,
                                                   device=device)

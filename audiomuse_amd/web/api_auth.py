@@ -89,6 +89,43 @@ def plex_pin_poll(pin_id: str):
                     "claimed": bool(data.get("authToken"))})
 
 
+@bp.post("/api/setup/providers/libraries")
+def setup_provider_libraries():
+    """Library list for an UNSAVED provider config during the setup
+    wizard (reference: /api/setup/providers/libraries)."""
+    from audiomuse_amd.mediaserver import make_provider
+
+    body = request.get_json(force=True, silent=True) or {}
+    try:
+        provider = make_provider(body.get("server_type", ""),
+                                 **(body.get("server_config") or {}))
+        libs = getattr(provider, "list_libraries", lambda: [])()
+    except Exception as exc:
+        return jsonify({"error": str(exc)}), 502
+    return jsonify(libs)
+
+
+@bp.post("/api/setup/lyrics-api/analyze")
+@require_auth
+def setup_lyrics_api_analyze():
+    """Try the configured external lyrics APIs against one artist/title
+    and return what came back (reference: /api/setup/lyrics-api/analyze
+    — lets the wizard verify LYRICS_API_* settings)."""
+    from audiomuse_amd.engines.lyrics import fetch_external_lyrics
+
+    body = request.get_json(force=True, silent=True) or {}
+    artist = body.get("artist", "")
+    title = body.get("title", "")
+    if not artist or not title:
+        return jsonify({"error": "artist and title required"}), 400
+    try:
+        text = fetch_external_lyrics(title, artist)
+    except Exception as exc:
+        return jsonify({"error": str(exc)}), 502
+    return jsonify({"found": bool(text),
+                    "preview": (text or "")[:500]})
+
+
 @bp.post("/api/login")
 def login():
     conn = _state().conn()

@@ -382,6 +382,34 @@ def semgrove_stats():
     return _family_stats(idx.SEMGROVE_INDEX, "clap_embedding")
 
 
+def _family_refresh(index_name: str):
+    """Per-family incremental refresh trigger (reference:
+    /api/clap/cache/refresh etc. — family-scoped, not a full rebuild)."""
+    from audiomuse_amd.taskqueue import enqueue
+
+    tid = enqueue(_state().conn(), "refresh_indexes", {"only": index_name},
+                  queue="high")
+    return jsonify({"task_id": tid, "index": index_name}), 202
+
+
+@bp.post("/api/clap/cache/refresh")
+@require_auth
+def clap_cache_refresh():
+    return _family_refresh(idx.CLAP_INDEX)
+
+
+@bp.post("/api/lyrics/cache/refresh")
+@require_auth
+def lyrics_cache_refresh():
+    return _family_refresh(idx.LYRICS_INDEX)
+
+
+@bp.post("/api/semgrove/cache/refresh")
+@require_auth
+def semgrove_cache_refresh():
+    return _family_refresh(idx.SEMGROVE_INDEX)
+
+
 @bp.get("/api/lyrics_search")
 @require_auth
 def lyrics_search():

@@ -679,3 +679,38 @@ def test_plex_pin_flow(client_ids, monkeypatch):
     r = client.get("/api/setup/plex/pin/777?client_id=cid-1")
     assert r.status_code == 200
     assert r.json["claimed"] and r.json["auth_token"] == "tok-xyz"
+
+
+def test_family_cache_refresh_routes(client_ids):
+    client, _ = client_ids
+    for fam in ("clap", "lyrics", "semgrove"):
+        r = client.post(f"/api/{fam}/cache/refresh")
+        assert r.status_code == 202 and r.json["task_id"], fam
+    client.post("/api/cancel_all/refresh_indexes")
+
+
+def test_setup_provider_libraries_and_lyrics_api(client_ids, monkeypatch):
+    client, _ = client_ids
+    r = client.post("/api/setup/providers/libraries",
+                    json={"server_type": "synthetic",
+                          "server_config": {"n_albums": 1}})
+    assert r.status_code == 200 and isinstance(r.json, list)
+    assert client.post("/api/setup/providers/libraries",
+                       json={"server_type": "bogus"}).status_code == 502
+    # lyrics-api analyze with a stubbed fetch
+    from audiomuse_amd import config as C2
+    monkeypatch.setattr(C2, "LYRICS_API_ENABLE", True)
+    monkeypatch.setattr(C2, "LYRICS_API_1_URL_TEMPLATE",
+                        "http://lyrics.example/{artist}/{title}")
+    import audiomuse_amd.engines.lyrics as lyr
+
+    def fake_fetch(title, artist, http_get=None):
+        return f"la la {title} by {artist}"
+
+    monkeypatch.setattr(lyr, "fetch_external_lyrics", fake_fetch)
+    r = client.post("/api/setup/lyrics-api/analyze",
+                    json={"artist": "A", "title": "T"})
+    assert r.status_code == 200 and r.json["found"]
+    assert "la la T" in r.json["preview"]
+    assert client.post("/api/setup/lyrics-api/analyze",
+                       json={}).status_code == 400
