@@ -229,6 +229,83 @@ def gmm_fit(x: torch.Tensor, k: int, iters: int = 60, seed: int = 0,
                      weights=weights, log_likelihood=ll * n)
 
 
+def gmm_fit_many(xs: "list[torch.Tensor]", k: int, iters: int = 60,
+                 seed: int = 0, reg: float = 1e-6,
+                 device: "Optional[str]" = None):
+    """Diagonal-covariance EM for MANY small datasets AT ONCE.
+
+    The artist-similarity index fits one BIC-selected GMM per artist
+    (reference: artist_gmm_manager.py:59-177 on a process pool). Per-
+    artist GPU fits are kernel-launch bound — each EM step on a
+    (50, 200) dataset is microseconds of math. This runs every
+    dataset's EM in one masked batch ("batched GPU GMM
+    EM kernel across all artists at once"): all per-iteration work is
+    three bmm's + elementwise over (A, N, k).
+
+    xs: list of (n_i, d) f32 tensors, every n_i >= k.
+    Returns (means (A,k,d), weights (A,k), bic (A,)) on `device`.
+    """
+    A = len(xs)
+    d = xs[0].shape[1]
+    dev = device or xs[0].device
+    N = max(int(x.shape[0]) for x in xs)
+    X = torch.zeros(A, N, d, device=dev)
+    M = torch.zeros(A, N, dtype=torch.bool, device=dev)
+    for i, x in enumerate(xs):
+        X[i, : x.shape[0]] = x.float().to(dev)
+        M[i, : x.shape[0]] = True
+    counts = M.sum(dim=1)                                   # (A,)
+    fM = M.float()
+
+    # init: deterministic row sample per dataset, then masked Lloyd
+    g = torch.Generator().manual_seed(seed)
+    sel = torch.stack([
+        torch.randperm(int(c), generator=g)[:k] if int(c) >= k
+        else torch.zeros(k, dtype=torch.int64)
+        for c in counts.cpu()]).to(dev)                     # (A, k)
+    means = torch.gather(X, 1, sel.unsqueeze(-1).expand(A, k, d)).clone()
+    for _ in range(10):
+        dist = (X.square().sum(-1, keepdim=True)
+                - 2 * X @ means.transpose(1, 2)
+                + means.square().sum(-1).unsqueeze(1))      # (A, N, k)
+        dist = dist.masked_fill(~M.unsqueeze(-1), float("inf"))
+        onehot = torch.nn.functional.one_hot(
+            dist.argmin(-1), k).float() * fM.unsqueeze(-1)
+        cnt = onehot.sum(dim=1)                             # (A, k)
+        new = (onehot.transpose(1, 2) @ X) / cnt.clamp(min=1).unsqueeze(-1)
+        means = torch.where((cnt > 0).unsqueeze(-1), new, means)
+
+    var = (((X - (X * fM.unsqueeze(-1)).sum(1, keepdim=True)
+             / counts.clamp(min=1).reshape(A, 1, 1)).square()
+            * fM.unsqueeze(-1)).sum(1)
+           / counts.clamp(min=1).unsqueeze(-1)).clamp(min=reg)
+    var = var.unsqueeze(1).expand(A, k, d).contiguous()     # (A, k, d)
+    weights = torch.full((A, k), 1.0 / k, device=dev)
+    log2pi = math.log(2 * math.pi)
+    ll_per = torch.zeros(A, device=dev)
+    for _ in range(iters):
+        inv = 1.0 / var
+        x_sq = X.square() @ inv.transpose(1, 2)             # (A, N, k)
+        cross = X @ (means * inv).transpose(1, 2)
+        m_sq = (means.square() * inv).sum(-1)               # (A, k)
+        quad = x_sq - 2 * cross + m_sq.unsqueeze(1)
+        logdet = torch.log(var).sum(-1)                     # (A, k)
+        logp = (-0.5 * (d * log2pi + logdet.unsqueeze(1) + quad)
+                + torch.log(weights + 1e-12).unsqueeze(1))
+        lse = torch.logsumexp(logp, dim=-1)                 # (A, N)
+        resp = torch.exp(logp - lse.unsqueeze(-1)) * fM.unsqueeze(-1)
+        nk = resp.sum(dim=1).clamp(min=1e-10)               # (A, k)
+        means = (resp.transpose(1, 2) @ X) / nk.unsqueeze(-1)
+        ex2 = (resp.transpose(1, 2) @ X.square()) / nk.unsqueeze(-1)
+        var = (ex2 - means.square()).clamp(min=reg)
+        weights = nk / counts.clamp(min=1).unsqueeze(1)
+        ll_per = (lse * fM).sum(dim=1)
+    n_params = k * d * 2 + (k - 1)
+    bic = -2.0 * ll_per + n_params * torch.log(
+        counts.clamp(min=2).float())
+    return means, weights, bic
+
+
 def _mahalanobis_diag(x, means, var):
     inv = 1.0 / var
     return (x.square() @ inv.T - 2 * (x @ (means * inv).T)

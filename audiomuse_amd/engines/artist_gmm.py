@@ -97,14 +97,17 @@ class ArtistSimilarity:
                       seed: int = 0) -> None:
         """Per-artist BIC-selected GMMs. On CPU the independent fits run
         on an INDEX_BUILD_WORKERS thread pool (the reference's process
-        pool, artist_gmm_manager.py:219-336); on GPU the batched EM per
-        fit already saturates the device, so fits stay sequential."""
+        pool, artist_gmm_manager.py:219-336); on GPU all artists batch
+        through ONE masked EM per k — per-artist GPU
+        fits are kernel-launch bound at these shapes."""
         import torch as _torch
 
         from audiomuse_amd import config as C
 
         items = [(n, e) for n, e in per_artist.items() if len(e) > 0]
-        if items and not _torch.cuda.is_available() \
+        if items and _torch.cuda.is_available() and len(items) >= 8:
+            self._fit_catalogue_batched(items, seed=seed, device="cuda")
+        elif items and not _torch.cuda.is_available() \
                 and C.INDEX_BUILD_WORKERS > 1 and len(items) > 2:
             from concurrent.futures import ThreadPoolExecutor
             with ThreadPoolExecutor(
@@ -121,6 +124,55 @@ class ArtistSimilarity:
         cents = [m.means.mean(axis=0) for m in self.models.values()]
         self._centroids = (np.stack(cents).astype(np.float32)
                            if cents else None)
+
+    def _fit_catalogue_batched(self, items, seed: int = 0,
+                               device: str = "cuda",
+                               chunk_size: int = 512) -> None:
+        """BIC selection with one gmm_fit_many call per (chunk, k):
+        artists sorted by track count and chunked so padding stays
+        bounded; every artist keeps its lowest-BIC k."""
+        import torch as _torch
+
+        from audiomuse_amd import config as C
+        from audiomuse_amd.cluster.algorithms import gmm_fit_many
+
+        singles = [(n, e) for n, e in items if len(e) < 2]
+        for name, e in singles:
+            self.models[name] = fit_artist(name, e, seed=seed)
+        big = sorted(((n, np.asarray(e, dtype=np.float32))
+                      for n, e in items if len(e) >= 2),
+                     key=lambda ne: ne[1].shape[0])
+        min_k, max_k = (C.ARTIST_GMM_MIN_COMPONENTS,
+                        C.ARTIST_GMM_MAX_COMPONENTS)
+        for c0 in range(0, len(big), chunk_size):
+            chunk = big[c0 : c0 + chunk_size]
+            names = [n for n, _ in chunk]
+            xs = [_torch.from_numpy(e) for _, e in chunk]
+            ns = [int(x.shape[0]) for x in xs]
+            best: Dict[int, Tuple[float, np.ndarray, np.ndarray]] = {}
+            for k in range(min_k, max_k + 1):
+                eligible = [i for i, n_i in enumerate(ns) if n_i >= k]
+                if not eligible:
+                    break
+                means, weights, bic = gmm_fit_many(
+                    [xs[i] for i in eligible], k, seed=seed, device=device)
+                mc, wc, bc = (means.cpu().numpy(), weights.cpu().numpy(),
+                              bic.cpu().numpy())
+                for j, i in enumerate(eligible):
+                    if i not in best or float(bc[j]) < best[i][0]:
+                        best[i] = (float(bc[j]), mc[j], wc[j])
+            # artists whose n is below min_k entirely: k=1 fit
+            leftover = [i for i in range(len(chunk)) if i not in best]
+            if leftover:
+                means, weights, bic = gmm_fit_many(
+                    [xs[i] for i in leftover], 1, seed=seed, device=device)
+                mc, wc, bc = (means.cpu().numpy(), weights.cpu().numpy(),
+                              bic.cpu().numpy())
+                for j, i in enumerate(leftover):
+                    best[i] = (float(bc[j]), mc[j], wc[j])
+            for i, (_, m, w) in best.items():
+                self.models[names[i]] = ArtistModel(
+                    name=names[i], means=m, weights=w, n_tracks=ns[i])
 
     def find_similar_artists(self, name: str, n: int = 10,
                              candidates: int = 100) -> List[Tuple[str, float]]:

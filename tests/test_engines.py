@@ -224,3 +224,32 @@ def test_order_playlist_greedy_walk():
     assert {t["item_id"] for t in out} == {"a", "b", "c", "d"}
     assert out[0]["item_id"] == "a"      # starts lowest-energy
     assert out[-1]["item_id"] == "c"     # outlier lands last
+
+
+def test_batched_catalogue_fit_matches_sequential():
+    """_fit_catalogue_batched (one masked EM per k across all artists —
+    the GPU path) produces models equivalent to the per-artist fits:
+    same similarity ordering and close chamfer distances."""
+    rng = np.random.default_rng(7)
+    base = rng.standard_normal(16).astype(np.float32) * 3
+    per_artist = {
+        "a1": base + rng.standard_normal((30, 16)).astype(np.float32) * 0.2,
+        "a2": base + rng.standard_normal((22, 16)).astype(np.float32) * 0.2,
+        "b": -base + rng.standard_normal((40, 16)).astype(np.float32) * 0.2,
+        "tiny": rng.standard_normal((1, 16)).astype(np.float32) * 3,
+        "duo": rng.standard_normal((2, 16)).astype(np.float32) * 3,
+    }
+    seq = ArtistSimilarity()
+    seq.fit_catalogue(per_artist, seed=0)
+    bat = ArtistSimilarity()
+    bat._fit_catalogue_batched(list(per_artist.items()), seed=0,
+                               device="cpu")
+    bat._names = list(bat.models)
+    cents = [m.means.mean(axis=0) for m in bat.models.values()]
+    bat._centroids = np.stack(cents).astype(np.float32)
+    assert set(bat.models) == set(per_artist)
+    for name in ("a1", "a2", "b"):
+        d = soft_chamfer_distance(seq.models[name], bat.models[name])
+        assert d < 0.05, (name, d)
+    res = bat.find_similar_artists("a1", n=2)
+    assert res[0][0] == "a2"
