@@ -253,3 +253,20 @@ def test_batched_catalogue_fit_matches_sequential():
         assert d < 0.05, (name, d)
     res = bat.find_similar_artists("a1", n=2)
     assert res[0][0] == "a2"
+
+
+@pytest.mark.gpu
+def test_batched_catalogue_fit_on_device():
+    """fit_catalogue dispatches to the batched masked-EM path on GPU
+    (>= 8 artists) and yields a queryable similarity catalogue."""
+    rng = np.random.default_rng(3)
+    base = rng.standard_normal(32).astype(np.float32) * 3
+    per_artist = {f"c{i}": base * (1 if i < 5 else -1)
+                  + rng.standard_normal((20 + i, 32)).astype(np.float32) * 0.2
+                  for i in range(10)}
+    sim = ArtistSimilarity()
+    sim.fit_catalogue(per_artist, seed=0)
+    assert len(sim.models) == 10
+    res = sim.find_similar_artists("c0", n=3)
+    assert all(name.startswith("c") for name, _ in res)
+    assert {name for name, _ in res} <= {f"c{i}" for i in range(1, 5)}
