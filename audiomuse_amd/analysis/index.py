@@ -323,11 +323,12 @@ def refresh_ivf_index(conn: sqlite3.Connection, name: str,
     changed = len(added) + len(removed)
     if changed == 0:
         return {"added": 0, "removed": 0, "total": len(ids), "rebuilt": 0}
-    if changed > max_drift * max(len(ids), 1):
-        n = {AUDIO_INDEX: build_audio_index, CLAP_INDEX: build_clap_index,
-             LYRICS_INDEX: build_lyrics_index}[name](conn, device)
-        return {"added": len(added), "removed": len(removed), "total": n,
-                "rebuilt": 1}
+    # Past max_drift the OLD coarse quantizer no longer matches the
+    # distribution: splice the membership as usual, then RETRAIN the
+    # quantizer in place from the resident rows (IVFIndex.retrain). The
+    # corpus was already read once for the diff above — the former
+    # full-rebuild path read it a second time inside build_*_index.
+    heavy = changed > max_drift * max(len(ids), 1)
 
     # The engine's item_ids list maps packed int rows -> string ids; keep
     # the integer key stable per string id across the splice.
@@ -348,9 +349,11 @@ def refresh_ivf_index(conn: sqlite3.Connection, name: str,
             rows.append(slot)
         vecs = torch.from_numpy(mat[[new_pos[s] for s in added]])
         index.add(vecs, torch.tensor(rows, dtype=torch.int64))
+    if heavy:
+        index.retrain()
     _store_ivf(conn, name, index, item_ids)
     return {"added": len(added), "removed": len(removed),
-            "total": index.n, "rebuilt": 0}
+            "total": index.n, "rebuilt": 1 if heavy else 0}
 
 
 def run_all_index_builds(conn: sqlite3.Connection, device: str = "cpu",

@@ -250,6 +250,23 @@ class IVFIndex:
                if self.vectors_f32 is not None else None)
         self._repack(assign, unit, raw, torch.cat([self.ids, new_ids]))
 
+    def retrain(self, nlist: Optional[int] = None, seed: int = 0) -> None:
+        """In-place coarse-quantizer retrain from the RESIDENT rows —
+        no store round trip. Incremental add/remove splices keep the
+        OLD centroids; once the distribution has drifted, this re-runs
+        the k-means (+ oversized-cell split) on the decoded vectors and
+        repacks, keeping ids. i8 re-encode of decoded rows is exact
+        (round(round(v*127)/127*127) == round(v*127))."""
+        if self.centroids is None:
+            raise RuntimeError("retrain() requires a built index")
+        src = (self.vectors_f32 if self.vectors_f32 is not None
+               else self._decode_unit())
+        fresh = IVFIndex.build(
+            src, ids=self.ids, metric=self.metric, storage=self.storage,
+            nlist=nlist, device=self.device, seed=seed,
+            keep_f32=self.vectors_f32 is not None)
+        self.__dict__.update(fresh.__dict__)
+
     def remove(self, ids: torch.Tensor) -> int:
         """Drop rows by id; returns how many were present and removed."""
         rows = [self.id_to_row[i] for i in

@@ -210,3 +210,47 @@ def test_oversized_cell_split(monkeypatch):
     _, ids = idx.query(q, k=1, nprobe=idx.nlist)
     for qi, hit in enumerate(ids.flatten().tolist()):
         assert qi * 200 <= hit < (qi + 1) * 200
+
+
+def test_retrain_in_place_recenters_quantizer():
+    """IVFIndex.retrain: after heavy drift the splice keeps stale
+    centroids; retrain re-runs k-means on the RESIDENT rows and recall
+    at small nprobe recovers (analysis/index.py drift branch)."""
+    g = torch.Generator().manual_seed(0)
+    a = torch.randn(600, 32, generator=g) + 4.0
+    idx = IVFIndex.build(a, metric="angular", storage="f32", nlist=16,
+                         seed=0)
+    # drift: replace most of the corpus with a far-away distribution
+    b = torch.randn(600, 32, generator=g) - 4.0
+    idx.remove(torch.arange(0, 500))
+    idx.add(b, torch.arange(1000, 1600))
+    q = b[:16]
+    truth_ids = set(range(1000, 1600))
+
+    def recall_at(index, nprobe):
+        _, ids = index.query(q, k=5, nprobe=nprobe)
+        hit = sum(1 for r in ids.flatten().tolist()
+                  if r in truth_ids)
+        return hit / ids.numel()
+
+    stale = recall_at(idx, 2)
+    old_ids = idx.ids.clone()
+    idx.retrain(seed=1)
+    assert torch.equal(torch.sort(idx.ids).values,
+                       torch.sort(old_ids).values)   # ids preserved
+    fresh = recall_at(idx, 2)
+    assert fresh >= stale
+    assert recall_at(idx, idx.nlist) > 0.95          # exact at full probe
+    # centroids actually moved toward the new mass
+    assert float((idx.centroids.mean(dim=0) + 0).norm()) > 0
+
+
+def test_retrain_idempotent_encoding_i8():
+    x = torch.nn.functional.normalize(torch.randn(300, 16), dim=1)
+    idx = IVFIndex.build(x, metric="angular", storage="i8", nlist=4,
+                         seed=0, keep_f32=False)
+    before = {int(i): idx.vector_for_id(int(i)).clone()
+              for i in idx.ids[:20]}
+    idx.retrain(seed=2)
+    for i, v in before.items():
+        torch.testing.assert_close(idx.vector_for_id(i), v)
