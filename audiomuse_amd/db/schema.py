@@ -125,6 +125,17 @@ CREATE TABLE IF NOT EXISTS playlist_name_history (
     name TEXT NOT NULL,
     created_at REAL DEFAULT ((julianday('now') - 2440587.5) * 86400.0)
 );
+CREATE TABLE IF NOT EXISTS migration_session (
+    id INTEGER PRIMARY KEY AUTOINCREMENT,
+    status TEXT DEFAULT 'open',
+    server_type TEXT NOT NULL,
+    server_config TEXT DEFAULT '{}',
+    source_server_id TEXT DEFAULT 'default',
+    decisions TEXT DEFAULT '{}',
+    report BLOB,
+    target_meta BLOB,
+    created_at REAL DEFAULT ((julianday('now') - 2440587.5) * 86400.0)
+);
 CREATE TABLE IF NOT EXISTS cron (
     id INTEGER PRIMARY KEY AUTOINCREMENT,
     name TEXT, schedule TEXT NOT NULL, task_type TEXT NOT NULL,

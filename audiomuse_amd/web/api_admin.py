@@ -204,7 +204,8 @@ def save_radio():
 # -- provider migration wizard -----------------------------------------------
 # (reference: app_provider_migration.py:2828 — probe, library select,
 # path-format detection, match preview, transactional rewrite, restart
-# handshake; condensed into 4 endpoints + one queued task)
+# handshake). Two flows: the one-shot probe/preview/start wizard below,
+# and the per-album review sessions (migration_session.py) after it.
 
 @bp.post("/api/migration/probe")
 @require_auth
@@ -266,6 +267,148 @@ def migration_start():
         "min_match_ratio": float(body.get("min_match_ratio", 0.5)),
     }, queue="high")
     return jsonify({"task_id": tid}), 202
+
+
+# -- per-album review sessions (reference: migration_session +
+# matched-albums / match-album / skip-album / dry-run routes,
+# app_provider_migration.py:678-2502) --------------------------------------
+
+@bp.post("/api/migration/session/start")
+@require_auth
+def migration_session_start():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    try:
+        out = ms.create_session(
+            _state().conn(), body.get("server_type", "synthetic"),
+            body.get("server_config", {}),
+            body.get("source_server_id", "default"))
+    except Exception as exc:  # noqa: BLE001 — wizard shows the reason
+        return jsonify({"error": str(exc)}), 502
+    return (jsonify(out), 502) if "error" in out else (jsonify(out), 201)
+
+
+@bp.get("/api/migration/session/<int:sid>")
+@require_auth
+def migration_session_get(sid: int):
+    from audiomuse_amd.analysis import migration_session as ms
+
+    out = ms.get_session(_state().conn(), sid)
+    if out is None:
+        return jsonify({"error": "unknown session"}), 404
+    return jsonify(out)
+
+
+@bp.delete("/api/migration/session/<int:sid>")
+@require_auth
+def migration_session_discard(sid: int):
+    from audiomuse_amd.analysis import migration_session as ms
+
+    ok = ms.discard_session(_state().conn(), sid)
+    return jsonify({"discarded": bool(ok)}), 200 if ok else 409
+
+
+@bp.post("/api/migration/dry-run")
+@require_auth
+def migration_dry_run():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    out = ms.run_dry_run(_state().conn(), int(body.get("session_id", 0)))
+    if out is None:
+        return jsonify({"error": "unknown or closed session"}), 404
+    return jsonify(out)
+
+
+@bp.get("/api/migration/dry-run-report/<int:sid>")
+@require_auth
+def migration_dry_run_report(sid: int):
+    from audiomuse_amd.analysis import migration_session as ms
+
+    out = ms.dry_run_report(_state().conn(), sid)
+    if out is None:
+        return jsonify({"error": "no report for session"}), 404
+    return jsonify(out)
+
+
+@bp.get("/api/migration/matched-albums/<int:sid>")
+@require_auth
+def migration_matched_albums(sid: int):
+    from audiomuse_amd.analysis import migration_session as ms
+
+    out = ms.matched_albums(_state().conn(), sid)
+    if out is None:
+        return jsonify({"error": "no report for session"}), 404
+    return jsonify(out)
+
+
+@bp.post("/api/migration/match-album")
+@require_auth
+def migration_match_album():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    out = ms.set_decision(
+        _state().conn(), int(body.get("session_id", 0)),
+        body.get("album", ""), "map",
+        target_album=body.get("target_album"))
+    if out is None:
+        return jsonify({"error": "unknown or closed session"}), 404
+    return jsonify({"decisions": out})
+
+
+@bp.post("/api/migration/skip-album")
+@require_auth
+def migration_skip_album():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    action = "auto" if body.get("undo") else "skip"
+    out = ms.set_decision(_state().conn(), int(body.get("session_id", 0)),
+                          body.get("album", ""), action)
+    if out is None:
+        return jsonify({"error": "unknown or closed session"}), 404
+    return jsonify({"decisions": out})
+
+
+@bp.post("/api/migration/search-albums")
+@require_auth
+def migration_search_albums():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    return jsonify(ms.search_albums(
+        _state().conn(), int(body.get("session_id", 0)),
+        body.get("q", "")))
+
+
+@bp.post("/api/migration/finalize-dry-run")
+@require_auth
+def migration_finalize():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    out = ms.finalize(_state().conn(), int(body.get("session_id", 0)))
+    if out is None:
+        return jsonify({"error": "session has no current dry run"}), 409
+    return jsonify(out)
+
+
+@bp.post("/api/migration/execute")
+@require_auth
+def migration_execute():
+    from audiomuse_amd.analysis import migration_session as ms
+
+    body = request.get_json(force=True, silent=True) or {}
+    out = ms.execute_session(
+        _state().conn(), int(body.get("session_id", 0)),
+        body.get("target_server_id", "migrated"),
+        remove_source=bool(body.get("remove_source", False)),
+        min_match_ratio=float(body.get("min_match_ratio", 0.5)))
+    if out is None:
+        return jsonify({"error": "session is not finalized"}), 409
+    return jsonify(out)
 
 
 @bp.get("/api/migration/status/<task_id>")

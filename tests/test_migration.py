@@ -200,3 +200,108 @@ def test_wizard_api_flow(db):
     r = client.get(f"/api/migration/status/{tid}")
     assert r.status_code == 200 and r.json["status"] == "SUCCESS"
     assert r.json["result"]["stage"] == "done"
+
+
+# -- per-album review sessions (analysis/migration_session.py) --------------
+
+def test_session_lifecycle_with_decisions(db):
+    """start -> dry-run -> matched-albums -> skip + manual map ->
+    re-dry-run -> finalize -> execute; decisions change the report."""
+    from audiomuse_amd.analysis import migration_session as ms
+
+    conn, _ = db
+    _seed_source(conn)
+    out = ms.create_session(conn, "synthetic", DST_CFG)
+    sid = out["session_id"]
+    assert out["target_tracks"] == 12 and out["libraries"]
+
+    rep = ms.run_dry_run(conn, sid)
+    assert rep["match_ratio"] > 0.9
+    albums = ms.matched_albums(conn, sid)
+    assert albums and all(a["complete"] for a in albums)
+    first_album = albums[0]["album"]
+
+    # skip one album -> its matches disappear from the next dry run
+    ms.set_decision(conn, sid, first_album, "skip")
+    rep2 = ms.run_dry_run(conn, sid)
+    assert rep2["matched"] < rep["matched"]
+    albums2 = ms.matched_albums(conn, sid)
+    skipped = [a for a in albums2 if a["album"] == first_album][0]
+    assert skipped["decision"] == "skip" and skipped["matched"] == 0
+
+    # manual map: re-attach the album by explicit target choice
+    hits = ms.search_albums(conn, sid, first_album[:4])
+    assert any(h["album"] == first_album for h in hits)
+    ms.set_decision(conn, sid, first_album, "map",
+                    target_album=first_album)
+    rep3 = ms.run_dry_run(conn, sid)
+    assert rep3["matched"] == rep["matched"]
+    assert rep3["tiers"].get("manual", 0) > 0
+
+    full = ms.dry_run_report(conn, sid)
+    assert any(m["tier"] == "manual" for m in full["matches"])
+
+    fin = ms.finalize(conn, sid)
+    assert fin["status"] == "finalized"
+    res = ms.execute_session(conn, sid, "migrated")
+    assert res["applied"] and res["written"] == rep["matched"]
+    n = conn.execute("SELECT COUNT(*) AS n FROM track_server_map "
+                     "WHERE server_id='migrated'").fetchone()["n"]
+    assert n == rep["matched"]
+    assert ms.get_session(conn, sid)["status"] == "executed"
+
+
+def test_session_gates(db):
+    from audiomuse_amd.analysis import migration_session as ms
+
+    conn, _ = db
+    _seed_source(conn)
+    sid = ms.create_session(conn, "synthetic", DST_CFG)["session_id"]
+    # execute before finalize refuses
+    assert ms.execute_session(conn, sid, "m2") is None
+    # finalize before dry run refuses
+    assert ms.finalize(conn, sid) is None
+    # a decision after dry-run reopens the session (stale report gate)
+    ms.run_dry_run(conn, sid)
+    ms.set_decision(conn, sid, "whatever", "skip")
+    assert ms.get_session(conn, sid)["status"] == "open"
+    assert ms.finalize(conn, sid) is None
+    # discard
+    assert ms.discard_session(conn, sid)
+    assert ms.run_dry_run(conn, sid) is None
+
+
+def test_session_endpoints(db, monkeypatch):
+    from audiomuse_amd.web.app import create_app
+
+    conn, url = db
+    _seed_source(conn)
+    app = create_app(url, auth_disabled=True)
+    app.testing = True
+    with app.test_client() as client:
+        r = client.post("/api/migration/session/start",
+                        json={"server_type": "synthetic",
+                              "server_config": DST_CFG})
+        assert r.status_code == 201
+        sid = r.json["session_id"]
+        assert client.post("/api/migration/dry-run",
+                           json={"session_id": sid}).status_code == 200
+        r = client.get(f"/api/migration/matched-albums/{sid}")
+        assert r.status_code == 200 and len(r.json) >= 3
+        album = r.json[0]["album"]
+        assert client.post("/api/migration/skip-album",
+                           json={"session_id": sid,
+                                 "album": album}).status_code == 200
+        assert client.post("/api/migration/search-albums",
+                           json={"session_id": sid, "q": ""}).json
+        client.post("/api/migration/dry-run", json={"session_id": sid})
+        assert client.post("/api/migration/finalize-dry-run",
+                           json={"session_id": sid}).status_code == 200
+        r = client.post("/api/migration/execute",
+                        json={"session_id": sid,
+                              "target_server_id": "migrated"})
+        assert r.status_code == 200 and r.json["applied"]
+        r = client.get(f"/api/migration/session/{sid}")
+        assert r.json["status"] == "executed"
+        assert client.delete(
+            f"/api/migration/session/{sid}").status_code == 409
