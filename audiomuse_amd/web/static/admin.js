@@ -65,7 +65,15 @@ const Admin = {
               onclick="Admin.mgApply()">3 · Apply migration</button>
             <label style="margin:0"><input type="checkbox"
               id="mg-remove"> remove source mappings</label></div>
-          <pre id="mg-out"></pre></section>
+          <pre id="mg-out"></pre>
+          <h3>Album-by-album review (session)</h3>
+          <div class="row">
+            <button onclick="Admin.msStart()">Open session</button>
+            <button onclick="Admin.msDryRun()">Dry run</button>
+            <button class="primary" id="ms-exec" disabled
+              onclick="Admin.msExecute()">Finalize + execute</button>
+            <span id="ms-info" class="muted"></span></div>
+          <table id="ms-albums"></table></section>
 
         <section class="wide"><h2>Plugins</h2>
           <table id="ad-plugins"></table>
@@ -179,6 +187,60 @@ const Admin = {
         setTimeout(poll, 1500);
     };
     poll();
+  },
+
+  msId: 0,
+
+  async msStart() {
+    const out = await AM.api("/api/migration/session/start",
+      { json: Object.assign(this.mgConfig(), {
+          source_server_id: document.getElementById("mg-src").value }) })
+      .catch(e => ({ error: e.message }));
+    if (out.error) { AM.status(out.error, "err"); return; }
+    this.msId = out.session_id;
+    document.getElementById("ms-info").textContent =
+      `session ${out.session_id}: ${out.target_tracks} target tracks`;
+  },
+
+  async msDryRun() {
+    if (!this.msId) { AM.status("open a session first", "warn"); return; }
+    const rep = await AM.api("/api/migration/dry-run",
+      { json: { session_id: this.msId } });
+    document.getElementById("ms-info").textContent =
+      `matched ${rep.matched}/${rep.total} ` +
+      `(${Math.round(rep.match_ratio * 100)}%)`;
+    document.getElementById("ms-exec").disabled = !(rep.match_ratio >= 0.5);
+    const albums = await AM.api(
+      `/api/migration/matched-albums/${this.msId}`);
+    document.getElementById("ms-albums").innerHTML =
+      `<tr><th>album</th><th>matched</th><th>decision</th><th></th></tr>` +
+      albums.map(a => `
+        <tr><td>${AM.esc(a.album)}</td>
+        <td>${a.matched}/${a.total}${a.complete ? "" : " ⚠"}</td>
+        <td>${AM.esc(a.decision)}</td>
+        <td><button onclick="Admin.msSkip('${AM.esc(a.album)}',
+          ${a.decision === "skip"})">
+          ${a.decision === "skip" ? "unskip" : "skip"}</button>
+        </td></tr>`).join("");
+  },
+
+  async msSkip(album, undo) {
+    await AM.api("/api/migration/skip-album",
+      { json: { session_id: this.msId, album, undo } });
+    this.msDryRun();
+  },
+
+  async msExecute() {
+    const fin = await AM.api("/api/migration/finalize-dry-run",
+      { json: { session_id: this.msId } }).catch(e => ({ error: e.message }));
+    if (fin.error) { AM.status(fin.error, "err"); return; }
+    const out = await AM.api("/api/migration/execute",
+      { json: { session_id: this.msId,
+                target_server_id: document.getElementById("mg-dst").value,
+                remove_source:
+                  document.getElementById("mg-remove").checked } });
+    AM.status(out.applied ? `migrated ${out.written} mappings`
+                          : out.reason, out.applied ? "ok" : "err");
   },
 
   async uploadPlugin() {

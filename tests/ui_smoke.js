@@ -189,6 +189,15 @@ async function main() {
   if (!E("ad-browse").innerHTML.includes("Song"))
     throw new Error("catalogue browser empty");
 
+  // album-by-album review session: open -> dry run -> album table
+  await Admin.msStart();
+  if (!E("ms-info").textContent.includes("target tracks"))
+    throw new Error("migration session did not open");
+  await Admin.msDryRun();
+  if (!E("ms-albums").innerHTML.includes("decision") &&
+      !E("ms-albums").innerHTML.includes("auto"))
+    throw new Error("matched-albums table empty");
+
   console.log("UI_SMOKE_OK views=" + Object.keys(AM.views).length);
 }
 
