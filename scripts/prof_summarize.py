@@ -31,6 +31,13 @@ def summarize(label: str, pattern: str, limit: int = 9) -> None:
 
 
 if __name__ == "__main__":
+    limit = 9
+    args = []
     for arg in sys.argv[1:]:
+        if arg.startswith("--limit="):
+            limit = int(arg.split("=", 1)[1])
+        else:
+            args.append(arg)
+    for arg in args:
         label, pattern = arg.split("=", 1)
-        summarize(label, pattern)
+        summarize(label, pattern, limit=limit)
