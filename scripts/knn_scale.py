@@ -60,4 +60,4 @@ def main(n=10_000_000, d=512):
 
 
 if __name__ == "__main__":
-    main()
+    main(n=int(sys.argv[1]) if len(sys.argv) > 1 else 10_000_000)
