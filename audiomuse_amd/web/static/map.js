@@ -6,6 +6,7 @@
 const MusicMap = {
   gl: null, prog: null, buf: null, n: 0,
   points: [],                 // [{item_id, x, y}]
+  centroids: [],              // [{mood, x, y, count}] (song map overlay)
   view: { cx: 0, cy: 0, scale: 1 },
   grid: null, cell: 0,        // uniform hover grid
 
@@ -20,7 +21,9 @@ const MusicMap = {
           <button class="primary" onclick="MusicMap.load()">Load</button>
           <span id="map-info" class="muted"></span>
         </div>
-        <canvas class="map" id="map-canvas"></canvas>
+        <div style="position:relative">
+          <canvas class="map" id="map-canvas"></canvas>
+          <div id="map-labels"></div></div>
         <p class="muted">drag to pan · wheel to zoom · hover for the id ·
           click a point for similar songs (opens Library)</p>
         <ul class="list" id="map-similar"></ul>
@@ -74,6 +77,8 @@ const MusicMap = {
     try {
       const pts = await AM.api(`/api/map?kind=${kind}&percent=${pct}`);
       this.points = pts;
+      this.centroids = kind === "song"
+        ? await AM.api("/api/mood_centroids").catch(() => []) : [];
       document.getElementById("map-info").textContent =
         `${pts.length} points`;
       this.upload();
@@ -152,6 +157,22 @@ const MusicMap = {
     gl.uniform1f(gl.getUniformLocation(this.prog, "ps"),
                  ps * (window.devicePixelRatio || 1));
     gl.drawArrays(gl.POINTS, 0, this.n);
+    this.drawLabels();
+  },
+
+  /* mood-centroid labels ride an HTML overlay (no GL text): positions
+     recompute with the same view transform on every draw */
+  drawLabels() {
+    const box = document.getElementById("map-labels");
+    if (!box || !this.gl) return;
+    const w = this.gl.canvas.clientWidth, h = 560;
+    box.innerHTML = this.centroids.map(c => {
+      const nx = (c.x - this.view.cx) * this.view.scale;
+      const ny = (c.y - this.view.cy) * this.view.scale;
+      if (nx < -0.98 || nx > 0.98 || ny < -0.98 || ny > 0.98) return "";
+      return `<span class="map-label" style="left:${(nx + 1) / 2 * w}px;` +
+        `top:${(1 - ny) / 2 * h}px">${AM.esc(c.mood)}</span>`;
+    }).join("");
   },
 
   canvasToData(ev, canvas) {
