@@ -49,10 +49,12 @@ def _seed_source(conn, server_id="default"):
 
 
 @pytest.fixture
-def db(tmp_sqlite_url):
-    conn = connect(tmp_sqlite_url)
+def db(tmp_db_url):
+    # both backends: the session table round-trips zlib BLOBs and the
+    # decisions JSON through sqlite AND the PG wire driver
+    conn = connect(tmp_db_url)
     init_db(conn)
-    yield conn, tmp_sqlite_url
+    yield conn, tmp_db_url
     conn.close()
 
 
