@@ -80,8 +80,8 @@ def test_skeleton_has_no_track_items():
     assert mood["items"] and all(c["leaf"] for c in mood["items"])
 
 
-def test_persist_and_lazy_warm_unload(tmp_sqlite_url):
-    conn = connect(tmp_sqlite_url)
+def test_persist_and_lazy_warm_unload(tmp_db_url):
+    conn = connect(tmp_db_url)
     init_db(conn)
     tree, _ = _build()
     persist_tree(conn, tree)
@@ -113,13 +113,13 @@ def test_persist_and_lazy_warm_unload(tmp_sqlite_url):
     conn.close()
 
 
-def test_stale_version_discarded(tmp_sqlite_url):
+def test_stale_version_discarded(tmp_db_url):
     import json
     import zlib
 
     from audiomuse_amd.db.store import store_index_blob
 
-    conn = connect(tmp_sqlite_url)
+    conn = connect(tmp_db_url)
     init_db(conn)
     old = {"version": 1, "nodes": {"root": {}}, "flat_ids": {},
            "track_count": 5}

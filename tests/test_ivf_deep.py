@@ -113,7 +113,7 @@ def test_query_dim_mismatch_raises():
 
 # -- persistence --------------------------------------------------------------
 
-def test_blob_store_round_trip_with_segmentation(tmp_sqlite_url, monkeypatch):
+def test_blob_store_round_trip_with_segmentation(tmp_db_url, monkeypatch):
     """Index persistence through the segmented blob store with a tiny
     part size: multiple ivf_cell rows, exact reload (reference:
     index_build_helpers store_segmented_blob :399)."""
@@ -123,7 +123,7 @@ def test_blob_store_round_trip_with_segmentation(tmp_sqlite_url, monkeypatch):
     from audiomuse_amd.db.schema import init_db
 
     monkeypatch.setattr(C, "IVF_MAX_PART_SIZE_MB", 1)
-    conn = connect(tmp_sqlite_url)
+    conn = connect(tmp_db_url)
     init_db(conn)
     x = _data(3000, 64, seed=7)          # > 1 MiB encoded
     idx = IVFIndex.build(x, metric="angular", storage="f32", seed=3)
@@ -142,14 +142,14 @@ def test_blob_store_round_trip_with_segmentation(tmp_sqlite_url, monkeypatch):
     conn.close()
 
 
-def test_truncated_blob_refuses_to_load(tmp_sqlite_url, monkeypatch):
+def test_truncated_blob_refuses_to_load(tmp_db_url, monkeypatch):
     from audiomuse_amd import config as C
     from audiomuse_amd.analysis.index import _store_ivf, load_ivf_engine
     from audiomuse_amd.db import connect, write_txn
     from audiomuse_amd.db.schema import init_db
 
     monkeypatch.setattr(C, "IVF_MAX_PART_SIZE_MB", 1)
-    conn = connect(tmp_sqlite_url)
+    conn = connect(tmp_db_url)
     init_db(conn)
     idx = IVFIndex.build(_data(3000, 64), storage="f32", seed=1)
     _store_ivf(conn, "broken", idx, [str(i) for i in range(3000)])
