@@ -1,0 +1,117 @@
+"""HTTP serving load: concurrent clients against the real Flask app
+(threaded werkzeug, the deployment default) on a seeded catalogue —
+/api/similar_tracks + /api/search_tracks + /api/track mixed traffic.
+Reports throughput and latency percentiles end to end (HTTP + engine +
+GPU scan)."""
+
+import json
+import random
+import sys
+import tempfile
+import threading
+import time
+import urllib.request
+
+import numpy as np
+
+sys.path.insert(0, __file__.rsplit("/", 2)[0])
+
+from audiomuse_amd.analysis.index import run_all_index_builds  # noqa: E402
+from audiomuse_amd.db import connect  # noqa: E402
+from audiomuse_amd.db.schema import init_db  # noqa: E402
+from audiomuse_amd.db.store import (save_clap_embedding,  # noqa: E402
+                                    save_track_analysis_and_embedding)
+
+
+def seed(conn, n=20000):
+    rng = np.random.default_rng(0)
+    ids = []
+    for i in range(n):
+        iid = f"fp_4{'%050x' % i}"
+        ids.append(iid)
+        save_track_analysis_and_embedding(
+            conn, iid, title=f"Song {i}", author=f"Artist {i % 500}",
+            album=f"Album {i % 2000}", tempo=100.0, energy=0.5, key="C",
+            scale="major", duration=200.0, mood_vector={"rock": 0.5},
+            other_features={},
+            embedding=rng.standard_normal(200).astype(np.float32))
+        save_clap_embedding(conn, iid,
+                            rng.standard_normal(512).astype(np.float32))
+    conn.commit()
+    return ids
+
+
+def main(n_clients=16, seconds=20.0, n_tracks=20000):
+    import torch
+
+    from audiomuse_amd.web.app import create_app
+
+    td = tempfile.mkdtemp()
+    url = f"sqlite:///{td}/load.db"
+    conn = connect(url)
+    init_db(conn)
+    ids = seed(conn, n_tracks)
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    run_all_index_builds(conn, device=dev)
+    app = create_app(url, auth_disabled=True)
+    import logging
+    logging.getLogger("werkzeug").setLevel(logging.ERROR)
+    from werkzeug.serving import make_server
+
+    srv = make_server("127.0.0.1", 0, app, threaded=True)
+    port = srv.server_port
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+
+    base = f"http://127.0.0.1:{port}"
+    urllib.request.urlopen(f"{base}/api/similar_tracks?item_id={ids[0]}&n=10")
+
+    lats, errors = [], [0]
+    lock = threading.Lock()
+    stop_at = time.perf_counter() + seconds
+
+    def worker(wid):
+        rng = random.Random(wid)
+        my = []
+        while time.perf_counter() < stop_at:
+            r = rng.random()
+            if r < 0.6:
+                u = (f"{base}/api/similar_tracks?item_id="
+                     f"{rng.choice(ids)}&n=10")
+            elif r < 0.8:
+                u = f"{base}/api/search_tracks?q=song {rng.randrange(999)}"
+            else:
+                u = f"{base}/api/track?item_id={rng.choice(ids)}"
+            t0 = time.perf_counter()
+            try:
+                with urllib.request.urlopen(u, timeout=10) as resp:
+                    resp.read()
+            except Exception:
+                with lock:
+                    errors[0] += 1
+                continue
+            my.append((time.perf_counter() - t0) * 1000)
+        with lock:
+            lats.extend(my)
+
+    threads = [threading.Thread(target=worker, args=(i,))
+               for i in range(n_clients)]
+    t0 = time.perf_counter()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    wall = time.perf_counter() - t0
+    srv.shutdown()
+    lats.sort()
+    q = lambda p: lats[min(int(len(lats) * p), len(lats) - 1)]  # noqa: E731
+    print(json.dumps({
+        "clients": n_clients, "seconds": round(wall, 1),
+        "requests": len(lats), "errors": errors[0],
+        "qps": round(len(lats) / wall, 1),
+        "p50_ms": round(q(0.5), 1), "p90_ms": round(q(0.9), 1),
+        "p99_ms": round(q(0.99), 1),
+        "catalogue": n_tracks, "mix": "60% similar / 20% search / 20% track"}))
+
+
+if __name__ == "__main__":
+    main(n_clients=int(sys.argv[1]) if len(sys.argv) > 1 else 16)
