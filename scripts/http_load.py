@@ -65,48 +65,60 @@ def main(n_clients=16, seconds=20.0, n_tracks=20000):
     base = f"http://127.0.0.1:{port}"
     urllib.request.urlopen(f"{base}/api/similar_tracks?item_id={ids[0]}&n=10")
 
-    lats, errors = [], [0]
-    lock = threading.Lock()
-    stop_at = time.perf_counter() + seconds
+    # clients run as SEPARATE PROCESSES: an in-process thread pool
+    # shares the GIL with the (threaded-werkzeug) server and measures
+    # its own contention instead of the server
+    import subprocess
 
-    def worker(wid):
-        rng = random.Random(wid)
-        my = []
-        while time.perf_counter() < stop_at:
-            r = rng.random()
-            if r < 0.6:
-                u = (f"{base}/api/similar_tracks?item_id="
-                     f"{rng.choice(ids)}&n=10")
-            elif r < 0.8:
-                u = f"{base}/api/search_tracks?q=song {rng.randrange(999)}"
-            else:
-                u = f"{base}/api/track?item_id={rng.choice(ids)}"
-            t0 = time.perf_counter()
-            try:
-                with urllib.request.urlopen(u, timeout=10) as resp:
-                    resp.read()
-            except Exception:
-                with lock:
-                    errors[0] += 1
-                continue
-            my.append((time.perf_counter() - t0) * 1000)
-        with lock:
-            lats.extend(my)
-
-    threads = [threading.Thread(target=worker, args=(i,))
-               for i in range(n_clients)]
+    client_src = r"""
+import json, random, sys, time, urllib.request, urllib.error
+base, seed_n, n_ids, seconds = (sys.argv[1], int(sys.argv[2]),
+                                int(sys.argv[3]), float(sys.argv[4]))
+rng = random.Random(seed_n)
+lats, errs = [], {}
+stop_at = time.perf_counter() + seconds
+while time.perf_counter() < stop_at:
+    i = rng.randrange(n_ids)
+    iid = "fp_4" + ("%050x" % i)
+    r = rng.random()
+    if r < 0.6:
+        u = base + "/api/similar_tracks?item_id=" + iid + "&n=10"
+    elif r < 0.8:
+        u = base + "/api/search_tracks?q=song%20" + str(rng.randrange(999))
+    else:
+        u = base + "/api/track?item_id=" + iid
     t0 = time.perf_counter()
-    for t in threads:
-        t.start()
-    for t in threads:
-        t.join()
+    try:
+        with urllib.request.urlopen(u, timeout=20) as resp:
+            resp.read()
+    except urllib.error.HTTPError as e:
+        errs[str(e.code)] = errs.get(str(e.code), 0) + 1
+        continue
+    except Exception as e:
+        errs[type(e).__name__] = errs.get(type(e).__name__, 0) + 1
+        continue
+    lats.append((time.perf_counter() - t0) * 1000)
+print(json.dumps({"lats": lats, "errs": errs}))
+"""
+    procs = [subprocess.Popen(
+        [sys.executable, "-c", client_src, base, str(i), str(len(ids)),
+         str(seconds)], stdout=subprocess.PIPE)
+        for i in range(n_clients)]
+    t0 = time.perf_counter()
+    lats, errors = [], {}
+    for p in procs:
+        out, _ = p.communicate(timeout=seconds + 60)
+        d = json.loads(out)
+        lats.extend(d["lats"])
+        for k, v in d["errs"].items():
+            errors[k] = errors.get(k, 0) + v
     wall = time.perf_counter() - t0
     srv.shutdown()
     lats.sort()
     q = lambda p: lats[min(int(len(lats) * p), len(lats) - 1)]  # noqa: E731
     print(json.dumps({
         "clients": n_clients, "seconds": round(wall, 1),
-        "requests": len(lats), "errors": errors[0],
+        "requests": len(lats), "errors": errors,
         "qps": round(len(lats) / wall, 1),
         "p50_ms": round(q(0.5), 1), "p90_ms": round(q(0.9), 1),
         "p99_ms": round(q(0.99), 1),
