@@ -121,9 +121,18 @@ class ProxyPrefixMiddleware:
         return self.wsgi_app(environ, start_response)
 
 
-def create_app(db_url: Optional[str] = None, device: str = "cpu",
+def create_app(db_url: Optional[str] = None, device: Optional[str] = None,
                auth_disabled: Optional[bool] = None) -> Flask:
     import os as _os
+
+    if device is None:
+        # serve from the GPU when one is present: the index engines
+        # deserialize onto this device and every /api/similar_tracks
+        # scan runs there (a CPU default on an MI355X host measured
+        # ~80x slower under load — profiles/r2_http_load*.log)
+        import torch as _torch
+
+        device = "cuda" if _torch.cuda.is_available() else "cpu"
 
     app = Flask("audiomuse_amd",
                 static_folder=_os.path.join(_os.path.dirname(
