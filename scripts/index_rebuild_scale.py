@@ -23,30 +23,49 @@ from audiomuse_amd.db.store import (save_clap_embedding,  # noqa: E402
                                     save_track_analysis_and_embedding)
 
 
-def seed(conn, n_tracks: int, n_artists: int, seed_: int = 0) -> None:
+def seed(conn, n_tracks: int, n_artists: int, seed_: int = 0,
+         chunk: int = 50_000) -> None:
+    """Bulk seeding via executemany chunks (the row-at-a-time store
+    helpers are the ingest path's shape; at 1M synthetic rows they are
+    pure overhead)."""
+    import json as _json
+
     rng = np.random.default_rng(seed_)
     artist_centers = rng.standard_normal((n_artists, 200)).astype(
         np.float32) * 2
     t0 = time.perf_counter()
-    for i in range(n_tracks):
-        a = i % n_artists
-        iid = f"fp_4{'%050x' % i}"
-        emb = (artist_centers[a]
-               + rng.standard_normal(200).astype(np.float32) * 0.4)
-        save_track_analysis_and_embedding(
-            conn, iid, title=f"T{i}", author=f"Artist {a}",
-            album=f"Album {i // 8}", tempo=float(80 + i % 100),
-            energy=float((i % 10) / 10), key="C", scale="major",
-            duration=200.0,
-            mood_vector={"rock": float(i % 3) / 2, "jazz": 0.3},
-            other_features={"happy": 0.5}, embedding=emb)
-        save_clap_embedding(
-            conn, iid, rng.standard_normal(512).astype(np.float32))
-        if i % 2 == 0:
-            save_lyrics_embedding(
-                conn, iid, rng.standard_normal(768).astype(np.float32),
-                axis_scores={"love": 0.4, "loss": 0.2})
-    conn.commit()
+    mood = _json.dumps({"rock": 0.5, "jazz": 0.3})
+    other = _json.dumps({"happy": 0.5})
+    axis = _json.dumps({"love": 0.4, "loss": 0.2})
+    for c0 in range(0, n_tracks, chunk):
+        c1 = min(c0 + chunk, n_tracks)
+        idx = np.arange(c0, c1)
+        embs = (artist_centers[idx % n_artists]
+                + rng.standard_normal((len(idx), 200)).astype(
+                    np.float32) * 0.4)
+        claps = rng.standard_normal((len(idx), 512)).astype(np.float32)
+        lyr = rng.standard_normal((len(idx), 768)).astype(np.float32)
+        ids = [f"fp_4{'%050x' % i}" for i in idx]
+        conn.executemany(
+            """INSERT INTO score (item_id, title, author, album, tempo,
+                   key, scale, mood_vector, other_features, energy,
+                   duration) VALUES (?,?,?,?,?,?,?,?,?,?,?)""",
+            [(ids[j], f"T{i}", f"Artist {i % n_artists}",
+              f"Album {i // 8}", float(80 + i % 100), "C", "major",
+              mood, other, 0.5, 200.0) for j, i in enumerate(idx)])
+        conn.executemany(
+            "INSERT INTO embedding (item_id, embedding) VALUES (?,?)",
+            [(ids[j], embs[j].tobytes()) for j in range(len(idx))])
+        conn.executemany(
+            "INSERT INTO clap_embedding (item_id, embedding) VALUES (?,?)",
+            [(ids[j], claps[j].tobytes()) for j in range(len(idx))])
+        conn.executemany(
+            """INSERT INTO lyrics_embedding (item_id, embedding,
+                   axis_scores, lyrics_text, language, instrumental)
+               VALUES (?,?,?,?,?,0)""",
+            [(ids[j], lyr[j].tobytes(), axis, "la la", "en")
+             for j in range(len(idx)) if idx[j] % 2 == 0])
+        conn.commit()
     print(f"seeded {n_tracks} tracks / {n_artists} artists "
           f"in {time.perf_counter() - t0:.1f}s")
 
