@@ -43,7 +43,7 @@ def _store_ivf(conn: sqlite3.Connection, name: str, index: IVFIndex,
                item_ids: List[str]) -> None:
     payload = {"index": index.serialize(), "item_ids": item_ids}
     buf = io.BytesIO()
-    torch.save(payload, buf)
+    torch.save(payload, buf, pickle_protocol=4)
     store_index_blob(conn, name, buf.getvalue(),
                      meta={"n": len(item_ids), "dim": index.dim,
                            "metric": index.metric, "storage": index.storage})
@@ -146,7 +146,8 @@ def build_semgrove_index(conn: sqlite3.Connection, device: str = "cpu") -> int:
     # persist whitening stats for query-time merging
     stats = io.BytesIO()
     torch.save({"lyr_mean": merger.lyr_mean, "lyr_std": merger.lyr_std,
-                "aud_mean": merger.aud_mean, "aud_std": merger.aud_std}, stats)
+                "aud_mean": merger.aud_mean, "aud_std": merger.aud_std},
+               stats, pickle_protocol=4)
     store_index_blob(conn, SEMGROVE_INDEX + "_stats", stats.getvalue())
     return len(ids)
 
@@ -171,7 +172,7 @@ def build_artist_index(conn: sqlite3.Connection, seed: int = 0) -> int:
                       "n_tracks": m.n_tracks}
                for name, m in sim.models.items()}
     buf = io.BytesIO()
-    torch.save(payload, buf)
+    torch.save(payload, buf, pickle_protocol=4)
     store_index_blob(conn, ARTIST_INDEX, buf.getvalue(),
                      meta={"n_artists": len(payload)})
     return len(payload)
@@ -242,7 +243,8 @@ def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
         coords = proj.cpu().numpy()
     buf = io.BytesIO()
     torch.save({"item_ids": ids,
-                "coords": torch.from_numpy(coords.astype(np.float32))}, buf)
+                "coords": torch.from_numpy(coords.astype(np.float32))},
+               buf, pickle_protocol=4)
     store_index_blob(conn, SONG_MAP, buf.getvalue(), meta={"n": len(ids)})
     return len(ids)
 
@@ -279,7 +281,8 @@ def build_artist_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
     buf = io.BytesIO()
     torch.save({"item_ids": names,
                 "coords": torch.from_numpy(coords.astype(np.float32)),
-                "n_tracks": [len(per_artist[a]) for a in names]}, buf)
+                "n_tracks": [len(per_artist[a]) for a in names]},
+               buf, pickle_protocol=4)
     store_index_blob(conn, ARTIST_MAP, buf.getvalue(), meta={"n": len(names)})
     return len(names)
 
