@@ -41,12 +41,22 @@ ARTIST_MAP = "artist_map_projection"
 
 def _store_ivf(conn: sqlite3.Connection, name: str, index: IVFIndex,
                item_ids: List[str]) -> None:
-    payload = {"index": index.serialize(), "item_ids": item_ids}
+    # the serialized index rides as a uint8 TENSOR: tensors serialize
+    # through torch storages, while a raw bytes field would be pickled
+    # as one string and overflow pickle's 4 GiB string cap (hit by the
+    # semgrove build at 10^6 tracks)
+    raw = torch.frombuffer(bytearray(index.serialize()), dtype=torch.uint8)
+    payload = {"index": raw, "item_ids": item_ids}
     buf = io.BytesIO()
-    torch.save(payload, buf, pickle_protocol=4)
+    torch.save(payload, buf)
     store_index_blob(conn, name, buf.getvalue(),
                      meta={"n": len(item_ids), "dim": index.dim,
                            "metric": index.metric, "storage": index.storage})
+
+
+def _index_bytes(payload) -> bytes:
+    raw = payload["index"]
+    return raw.numpy().tobytes() if torch.is_tensor(raw) else raw
 
 
 def load_ivf_engine(conn: sqlite3.Connection, name: str,
@@ -58,7 +68,7 @@ def load_ivf_engine(conn: sqlite3.Connection, name: str,
     blob, _meta = got
     payload = torch.load(io.BytesIO(blob), map_location="cpu",
                          weights_only=True)
-    index = IVFIndex.deserialize(payload["index"], device=device)
+    index = IVFIndex.deserialize(_index_bytes(payload), device=device)
     return SimilarityEngine(index, payload["item_ids"], meta_fn=meta_fn)
 
 
@@ -146,8 +156,7 @@ def build_semgrove_index(conn: sqlite3.Connection, device: str = "cpu") -> int:
     # persist whitening stats for query-time merging
     stats = io.BytesIO()
     torch.save({"lyr_mean": merger.lyr_mean, "lyr_std": merger.lyr_std,
-                "aud_mean": merger.aud_mean, "aud_std": merger.aud_std},
-               stats, pickle_protocol=4)
+                "aud_mean": merger.aud_mean, "aud_std": merger.aud_std}, stats)
     store_index_blob(conn, SEMGROVE_INDEX + "_stats", stats.getvalue())
     return len(ids)
 
@@ -172,7 +181,7 @@ def build_artist_index(conn: sqlite3.Connection, seed: int = 0) -> int:
                       "n_tracks": m.n_tracks}
                for name, m in sim.models.items()}
     buf = io.BytesIO()
-    torch.save(payload, buf, pickle_protocol=4)
+    torch.save(payload, buf)
     store_index_blob(conn, ARTIST_INDEX, buf.getvalue(),
                      meta={"n_artists": len(payload)})
     return len(payload)
@@ -243,8 +252,7 @@ def build_song_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
         coords = proj.cpu().numpy()
     buf = io.BytesIO()
     torch.save({"item_ids": ids,
-                "coords": torch.from_numpy(coords.astype(np.float32))},
-               buf, pickle_protocol=4)
+                "coords": torch.from_numpy(coords.astype(np.float32))}, buf)
     store_index_blob(conn, SONG_MAP, buf.getvalue(), meta={"n": len(ids)})
     return len(ids)
 
@@ -281,8 +289,7 @@ def build_artist_map(conn: sqlite3.Connection, device: str = "cpu") -> int:
     buf = io.BytesIO()
     torch.save({"item_ids": names,
                 "coords": torch.from_numpy(coords.astype(np.float32)),
-                "n_tracks": [len(per_artist[a]) for a in names]},
-               buf, pickle_protocol=4)
+                "n_tracks": [len(per_artist[a]) for a in names]}, buf)
     store_index_blob(conn, ARTIST_MAP, buf.getvalue(), meta={"n": len(names)})
     return len(names)
 
@@ -317,7 +324,7 @@ def refresh_ivf_index(conn: sqlite3.Connection, name: str,
 
     payload = torch.load(io.BytesIO(got[0]), map_location="cpu",
                          weights_only=True)
-    index = IVFIndex.deserialize(payload["index"], device=device)
+    index = IVFIndex.deserialize(_index_bytes(payload), device=device)
     old_ids: List[str] = payload["item_ids"]
     old_pos = {s: i for i, s in enumerate(old_ids)}
     new_pos = {s: i for i, s in enumerate(ids)}

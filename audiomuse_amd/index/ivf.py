@@ -486,7 +486,7 @@ class IVFIndex:
             "vectors_f32": None if self.vectors_f32 is None else self.vectors_f32.cpu(),
         }
         buf = io.BytesIO()
-        torch.save(state, buf, pickle_protocol=4)
+        torch.save(state, buf)
         return buf.getvalue()
 
     @classmethod
