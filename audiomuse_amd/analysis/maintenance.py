@@ -33,8 +33,11 @@ _NOISE = re.compile(r"\s*[\(\[].*?[\)\]]\s*|\s*(feat\.|ft\.)\s.*$|[^\w\s]",
 
 def normalize_title(s: str) -> str:
     """Noise-normalized matching key (remaster tags, feat. credits,
-    punctuation stripped)."""
-    return re.sub(r"\s+", " ", _NOISE.sub(" ", (s or "").lower())).strip()
+    punctuation stripped). casefold(), not lower(): unicode caseless
+    matching must be stable through case round trips (µ -> Μ -> μ) —
+    found by the property test."""
+    return re.sub(r"\s+", " ",
+                  _NOISE.sub(" ", (s or "").casefold())).strip()
 
 
 def align_server_tracks(conn: sqlite3.Connection, server_id: str,
