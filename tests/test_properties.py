@@ -125,3 +125,47 @@ def test_rewrite_path_round_trip(src, dst, tail):
             "from_separator": "/", "to_separator": "/"}
     p = src + "/" + "/".join(tail)
     assert rewrite_path(rewrite_path(p, rule), back) == p
+
+
+@FAST
+@given(st.integers(0, 2 ** 31 - 1), st.integers(1, 200))
+def test_ivf_query_invariants_random_shapes(seed, n):
+    """For ANY corpus shape: distances ascend, ids are valid or the -1
+    pad, self-query at full probe finds itself first (angular)."""
+    from audiomuse_amd.index.ivf import IVFIndex
+
+    g = torch.Generator().manual_seed(seed)
+    d = 8 + (seed % 5) * 8
+    x = torch.randn(n, d, generator=g)
+    idx = IVFIndex.build(x, metric="angular", storage="f32", seed=1)
+    k = min(5, n)
+    dist, ids = idx.query(x[:1], k=k, nprobe=idx.nlist)
+    row_d, row_i = dist.reshape(-1), ids.reshape(-1)
+    valid = row_i >= 0
+    assert int(row_i[0]) == 0                       # self first
+    assert float(row_d[0]) < 1e-4
+    vd = row_d[valid]
+    assert bool((vd[:-1] <= vd[1:] + 1e-6).all())   # ascending
+    assert int(row_i[valid].max()) < n
+
+
+@FAST
+@given(st.integers(0, 2 ** 31 - 1), st.integers(40, 90),
+       st.integers(0, 12))
+def test_chromaprint_bit_match_shift_invariance(seed, n_frames, shift):
+    """A fingerprint matches a time-shifted copy of itself (alignment
+    search; overlap stays above CHROMAPRINT_MIN_OVERLAP words) and the
+    ratio stays in [0, 1]."""
+    import zlib
+
+    from audiomuse_amd.engines.chromaprint import bit_match_ratio
+
+    rng = np.random.default_rng(seed)
+    raw = rng.integers(0, 2 ** 32 - 1, size=n_frames, dtype=np.uint32)
+    a = zlib.compress(raw.tobytes())          # the compute() wire format
+    b = zlib.compress(np.roll(raw, shift).tobytes())
+    r = bit_match_ratio(a, b, max_offset=16)
+    assert 0.0 <= r <= 1.0
+    if n_frames - shift >= 20:
+        assert r > 0.95, (shift, r)
+    assert bit_match_ratio(a, a) == pytest.approx(1.0)
