@@ -23,6 +23,10 @@ def main() -> None:
     web.add_argument("--port", type=int, default=8000)
     web.add_argument("--db", default=None)
     web.add_argument("--no-auth", action="store_true")
+    web.add_argument("--procs", type=int, default=1,
+                    help="SO_REUSEPORT web processes sharing the port")
+    web.add_argument("--reuse-port", action="store_true",
+                    help="bind with SO_REUSEPORT (set by --procs children)")
 
     wk = sub.add_parser("worker")
     wk.add_argument("--db", default=None)
@@ -49,10 +53,14 @@ def main() -> None:
     args, rest = ap.parse_known_args()
 
     if args.cmd == "web":
-        from audiomuse_amd.web.app import create_app
+        from audiomuse_amd.web.serve import serve, serve_procs
 
-        app = create_app(args.db, auth_disabled=args.no_auth)
-        app.run(host=args.host, port=args.port)
+        if args.procs > 1:
+            serve_procs(args.db, args.host, args.port, args.procs,
+                        auth_disabled=args.no_auth)
+        else:
+            serve(args.db, args.host, args.port,
+                  auth_disabled=args.no_auth, reuse_port=args.reuse_port)
     elif args.cmd == "worker":
         from audiomuse_amd.taskqueue.worker import Worker
 

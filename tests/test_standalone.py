@@ -105,3 +105,56 @@ def test_standalone_worker_subprocess_drains_queue(tmp_sqlite_url):
     finally:
         sup.shutdown()
     conn.close()
+
+
+def test_reuseport_web_procs_share_one_port(tmp_sqlite_url):
+    """Two SO_REUSEPORT web processes bind the SAME port and both serve
+    (web/serve.py — the web tier's scale-out path; one process's qps
+    ceiling is measured in profiles/r2_http_load16c.log)."""
+    import json
+    import subprocess
+    import sys
+    import time
+    import urllib.request
+
+    from audiomuse_amd.db import connect
+    from audiomuse_amd.db.schema import init_db
+
+    conn = connect(tmp_sqlite_url)
+    init_db(conn)
+    conn.close()
+    import socket as s_mod
+    if not hasattr(s_mod, "SO_REUSEPORT"):
+        import pytest as _pytest
+        _pytest.skip("no SO_REUSEPORT")
+    # pick a free port
+    probe = s_mod.socket()
+    probe.bind(("127.0.0.1", 0))
+    port = probe.getsockname()[1]
+    probe.close()
+    cmd = [sys.executable, "-m", "audiomuse_amd", "web",
+           "--host", "127.0.0.1", "--port", str(port), "--reuse-port",
+           "--no-auth", "--db", tmp_sqlite_url]
+    procs = [subprocess.Popen(cmd) for _ in range(2)]
+    try:
+        deadline = time.time() + 30
+        body = None
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/health", timeout=2) as r:
+                    body = json.loads(r.read())
+                break
+            except Exception:
+                time.sleep(0.3)
+        assert body and body["status"] == "ok"
+        assert all(p.poll() is None for p in procs)  # both still serving
+        for _ in range(6):                           # a few more round trips
+            with urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/health", timeout=2) as r:
+                assert json.loads(r.read())["status"] == "ok"
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            p.wait(timeout=10)
