@@ -41,7 +41,7 @@ def seed(conn, n=20000):
     return ids
 
 
-def main(n_clients=16, seconds=20.0, n_tracks=20000):
+def main(n_clients=16, seconds=20.0, n_tracks=20000, server_procs=1):
     import torch
 
     from audiomuse_amd.web.app import create_app
@@ -53,17 +53,43 @@ def main(n_clients=16, seconds=20.0, n_tracks=20000):
     ids = seed(conn, n_tracks)
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     run_all_index_builds(conn, device=dev)
-    app = create_app(url, auth_disabled=True)
     import logging
     logging.getLogger("werkzeug").setLevel(logging.ERROR)
-    from werkzeug.serving import make_server
+    import subprocess as _sp
 
-    srv = make_server("127.0.0.1", 0, app, threaded=True)
-    port = srv.server_port
-    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    server_children = []
+    if server_procs > 1:
+        # N SO_REUSEPORT processes share the port (web/serve.py)
+        import socket as s_mod
+        probe = s_mod.socket()
+        probe.bind(("127.0.0.1", 0))
+        port = probe.getsockname()[1]
+        probe.close()
+        cmd = [sys.executable, "-m", "audiomuse_amd", "web",
+               "--host", "127.0.0.1", "--port", str(port), "--reuse-port",
+               "--no-auth", "--db", url]
+        server_children = [_sp.Popen(cmd) for _ in range(server_procs)]
+        srv = None
+    else:
+        from werkzeug.serving import make_server
+
+        app = create_app(url, auth_disabled=True)
+        srv = make_server("127.0.0.1", 0, app, threaded=True)
+        port = srv.server_port
+        threading.Thread(target=srv.serve_forever, daemon=True).start()
 
     base = f"http://127.0.0.1:{port}"
-    urllib.request.urlopen(f"{base}/api/similar_tracks?item_id={ids[0]}&n=10")
+    deadline = time.perf_counter() + 120
+    while True:
+        try:
+            urllib.request.urlopen(
+                f"{base}/api/similar_tracks?item_id={ids[0]}&n=10",
+                timeout=5)
+            break
+        except Exception:
+            if time.perf_counter() > deadline:
+                raise
+            time.sleep(0.5)
 
     # clients run as SEPARATE PROCESSES: an in-process thread pool
     # shares the GIL with the (threaded-werkzeug) server and measures
@@ -113,7 +139,10 @@ print(json.dumps({"lats": lats, "errs": errs}))
         for k, v in d["errs"].items():
             errors[k] = errors.get(k, 0) + v
     wall = time.perf_counter() - t0
-    srv.shutdown()
+    if srv is not None:
+        srv.shutdown()
+    for c in server_children:
+        c.terminate()
     lats.sort()
     q = lambda p: lats[min(int(len(lats) * p), len(lats) - 1)]  # noqa: E731
     print(json.dumps({
@@ -122,8 +151,10 @@ print(json.dumps({"lats": lats, "errs": errs}))
         "qps": round(len(lats) / wall, 1),
         "p50_ms": round(q(0.5), 1), "p90_ms": round(q(0.9), 1),
         "p99_ms": round(q(0.99), 1),
+        "server_procs": server_procs,
         "catalogue": n_tracks, "mix": "60% similar / 20% search / 20% track"}))
 
 
 if __name__ == "__main__":
-    main(n_clients=int(sys.argv[1]) if len(sys.argv) > 1 else 16)
+    main(n_clients=int(sys.argv[1]) if len(sys.argv) > 1 else 16,
+         server_procs=int(sys.argv[2]) if len(sys.argv) > 2 else 1)
