@@ -11,7 +11,6 @@ proxy needed on Linux (SO_REUSEPORT).
 
 from __future__ import annotations
 
-import os
 import socket
 import sys
 from typing import Optional
@@ -54,14 +53,13 @@ def serve_procs(db_url: Optional[str], host: str, port: int,
         return []
     import subprocess
 
-    env = dict(os.environ)
     cmd = [sys.executable, "-m", "audiomuse_amd", "web",
            "--host", host, "--port", str(port), "--reuse-port"]
     if db_url:
         cmd += ["--db", db_url]
     if auth_disabled:
         cmd += ["--no-auth"]
-    children = [subprocess.Popen(cmd, env=env) for _ in range(procs)]
+    children = [subprocess.Popen(cmd) for _ in range(procs)]
     try:
         for c in children:
             c.wait()
